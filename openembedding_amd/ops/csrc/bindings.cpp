@@ -1,0 +1,185 @@
+// Torch bindings for the openembedding_amd CDNA4 kernels (embops.hip).
+// Tensor-level API consumed by ops/dispatch.py and core/variable_gpu.py.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>  // ROCm torch masquerades HIP as CUDA here
+#include <c10/cuda/CUDAGuard.h>
+
+#include <cstdint>
+#include <tuple>
+#include <vector>
+
+// torch instantiates data_ptr<int64_t> (long on LP64); the kernel TU uses
+// long long — same width, C linkage, so the pointer casts below are safe.
+typedef unsigned long long u64;
+typedef int64_t i64;
+typedef struct ihipStream_t* hipStream_t_;
+
+extern "C" {
+void emb_unique(const i64*, long, u64*, int*, long, int*, unsigned char*,
+                i64*, i64*, int*, hipStream_t_);
+void emb_ht_lookup(u64*, int*, long, const i64*, long, int*, i64*, i64*,
+                   unsigned char*, int, hipStream_t_);
+void emb_ht_rehash(const u64*, const int*, long, u64*, int*, long,
+                   hipStream_t_);
+void emb_array_touch(unsigned char*, const i64*, long, unsigned char*, int*,
+                     hipStream_t_);
+void emb_gather_init(float*, float*, long, long, const i64*,
+                     const unsigned char*, const i64*, long, float*, int,
+                     float, float, float, u64, const float*, hipStream_t_);
+void emb_reduce_by_inverse(const i64*, const float*, long, long, float*, u64*,
+                           long, hipStream_t_);
+void emb_apply_optimizer(int, float*, float*, long, long, const i64*, long,
+                         const float*, const u64*, const float*, hipStream_t_);
+}
+
+namespace {
+
+hipStream_t_ cur_stream() {
+    return reinterpret_cast<hipStream_t_>(
+        at::cuda::getCurrentCUDAStream().stream());
+}
+
+long next_pow2(long x) {
+    long p = 16;
+    while (p < x) p <<= 1;
+    return p;
+}
+
+#define CHECK_GPU(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+#define CHECK_CONT(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+std::tuple<torch::Tensor, torch::Tensor> unique_inverse(torch::Tensor keys) {
+    CHECK_GPU(keys); CHECK_CONT(keys);
+    TORCH_CHECK(keys.dtype() == torch::kInt64);
+    const c10::cuda::CUDAGuard guard(keys.device());
+    long n = keys.numel();
+    auto opts_i64 = keys.options();
+    auto opts_i32 = keys.options().dtype(torch::kInt32);
+    auto opts_u8 = keys.options().dtype(torch::kUInt8);
+    long cap = next_pow2(2 * n);
+    auto tk = torch::empty({cap}, opts_i64);
+    auto tv = torch::empty({cap}, opts_i32);
+    auto slot_of = torch::empty({n}, opts_i32);
+    auto is_first = torch::empty({n}, opts_u8);
+    auto uk = torch::empty({n}, opts_i64);
+    auto inverse = torch::empty({n}, opts_i64);
+    auto counter = torch::zeros({1}, opts_i32);
+    emb_unique(keys.data_ptr<i64>(), n, (u64*)tk.data_ptr<i64>(),
+               tv.data_ptr<int>(), cap, slot_of.data_ptr<int>(),
+               is_first.data_ptr<uint8_t>(), uk.data_ptr<i64>(),
+               inverse.data_ptr<i64>(), counter.data_ptr<int>(), cur_stream());
+    long u = counter.item<int>();  // host sync (bounded-buffer mode avoids it)
+    return {uk.narrow(0, 0, u), inverse};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> ht_lookup(
+    torch::Tensor tk, torch::Tensor tv, torch::Tensor keys,
+    torch::Tensor nrows, torch::Tensor slot_keys, bool insert) {
+    CHECK_GPU(keys); CHECK_CONT(keys);
+    const c10::cuda::CUDAGuard guard(keys.device());
+    long n = keys.numel();
+    auto slots = torch::empty({n}, keys.options());
+    auto new_mask = torch::zeros({n}, keys.options().dtype(torch::kUInt8));
+    if (n)
+        emb_ht_lookup((u64*)tk.data_ptr<i64>(), tv.data_ptr<int>(),
+                      tk.numel(), keys.data_ptr<i64>(), n,
+                      nrows.data_ptr<int>(), slot_keys.data_ptr<i64>(),
+                      slots.data_ptr<i64>(), new_mask.data_ptr<uint8_t>(),
+                      insert ? 1 : 0, cur_stream());
+    return {slots, new_mask};
+}
+
+void ht_rehash(torch::Tensor tk_old, torch::Tensor tv_old, torch::Tensor tk_new,
+               torch::Tensor tv_new) {
+    const c10::cuda::CUDAGuard guard(tk_old.device());
+    emb_ht_rehash((u64*)tk_old.data_ptr<i64>(), tv_old.data_ptr<int>(),
+                  tk_old.numel(), (u64*)tk_new.data_ptr<i64>(),
+                  tv_new.data_ptr<int>(), tk_new.numel(), cur_stream());
+}
+
+std::tuple<torch::Tensor, torch::Tensor> array_touch(torch::Tensor valid,
+                                                     torch::Tensor slots) {
+    const c10::cuda::CUDAGuard guard(valid.device());
+    long n = slots.numel();
+    auto new_mask = torch::zeros({n}, slots.options().dtype(torch::kUInt8));
+    auto n_new = torch::zeros({1}, slots.options().dtype(torch::kInt32));
+    if (n)
+        emb_array_touch(valid.data_ptr<uint8_t>(), slots.data_ptr<i64>(), n,
+                        new_mask.data_ptr<uint8_t>(), n_new.data_ptr<int>(),
+                        cur_stream());
+    return {new_mask, n_new};
+}
+
+torch::Tensor gather_init(torch::Tensor weights, torch::Tensor state,
+                          torch::Tensor slots, torch::Tensor new_mask,
+                          torch::Tensor keys, int64_t init_cat, double p0,
+                          double p1, double p2, int64_t seed,
+                          torch::Tensor state_init_row, bool want_out) {
+    CHECK_GPU(weights); CHECK_CONT(weights);
+    const c10::cuda::CUDAGuard guard(weights.device());
+    long n = keys.numel();
+    long dim = weights.size(1);
+    long sd = state.numel() ? state.size(1) : 0;
+    torch::Tensor out;
+    float* out_ptr = nullptr;
+    if (want_out) {
+        out = torch::empty({n, dim}, weights.options());
+        out_ptr = out.data_ptr<float>();
+    } else {
+        out = torch::empty({0}, weights.options());
+    }
+    emb_gather_init(weights.data_ptr<float>(),
+                    sd ? state.data_ptr<float>() : nullptr, dim, sd,
+                    slots.data_ptr<i64>(),
+                    new_mask.numel() ? new_mask.data_ptr<uint8_t>() : nullptr,
+                    keys.data_ptr<i64>(), n, out_ptr, (int)init_cat,
+                    (float)p0, (float)p1, (float)p2, (u64)seed,
+                    sd ? state_init_row.data_ptr<float>() : nullptr,
+                    cur_stream());
+    return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> reduce_by_inverse(
+    torch::Tensor inverse, torch::Tensor grads, int64_t u) {
+    CHECK_GPU(grads); CHECK_CONT(grads);
+    TORCH_CHECK(grads.dtype() == torch::kFloat32,
+                "reduce_by_inverse: grads must be float32");
+    const c10::cuda::CUDAGuard guard(grads.device());
+    long n = grads.size(0);
+    long dim = grads.size(1);
+    auto ugrads = torch::empty({u, dim}, grads.options());
+    auto counts = torch::empty({u}, grads.options().dtype(torch::kInt64));
+    emb_reduce_by_inverse(inverse.data_ptr<i64>(), grads.data_ptr<float>(), n,
+                          dim, ugrads.data_ptr<float>(),
+                          (u64*)counts.data_ptr<i64>(), u, cur_stream());
+    return {ugrads, counts};
+}
+
+void apply_optimizer(int64_t opt, torch::Tensor weights, torch::Tensor state,
+                     torch::Tensor slots, torch::Tensor grads,
+                     torch::Tensor counts, std::vector<double> cfg) {
+    CHECK_GPU(weights); CHECK_CONT(weights); CHECK_CONT(grads);
+    const c10::cuda::CUDAGuard guard(weights.device());
+    long n = slots.numel();
+    long dim = weights.size(1);
+    long sd = state.numel() ? state.size(1) : 0;
+    float c[7] = {0, 0, 0, 0, 0, 0, 0};
+    for (size_t i = 0; i < cfg.size() && i < 7; ++i) c[i] = (float)cfg[i];
+    emb_apply_optimizer((int)opt, weights.data_ptr<float>(),
+                        sd ? state.data_ptr<float>() : nullptr, dim, sd,
+                        slots.data_ptr<i64>(), n, grads.data_ptr<float>(),
+                        (const u64*)counts.data_ptr<i64>(), c, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("unique_inverse", &unique_inverse, "hash-based unique+inverse");
+    m.def("ht_lookup", &ht_lookup, "hash table lookup/insert");
+    m.def("ht_rehash", &ht_rehash, "hash table rehash into larger table");
+    m.def("array_touch", &array_touch, "array-table valid-bitmap touch");
+    m.def("gather_init", &gather_init, "row gather with fused lazy init");
+    m.def("reduce_by_inverse", &reduce_by_inverse, "grad reduce-by-key");
+    m.def("apply_optimizer", &apply_optimizer, "fused sparse optimizer step");
+}

@@ -1,0 +1,540 @@
+// openembedding_amd CDNA4 (gfx950 / MI355X) kernels.
+//
+// GPU-resident rebuild of the reference's CPU parameter-server primitives
+// (see SURVEY.md §2.8 for the mapping):
+//   unique+inverse        <- client dedup, EmbeddingPullOperator.cpp:67-79
+//   hash lookup/insert    <- EasyHashMap row lookup, EmbeddingTable.h:72-98
+//   gather + lazy init    <- pull_weights miss path,
+//                            EmbeddingOptimizerVariable.h:242-266
+//   reduce-by-inverse     <- client grad pre-agg + MpscGradientReducer
+//   fused sparse optimizers (9) <- EmbeddingOptimizer.h:49-390, applied once
+//                            per unique key per batch with occurrence counts
+//
+// Design notes (MI355X):
+//   - wave = 64 lanes; rows are handled by power-of-2 lane groups (G=16 for
+//     dim<=32, G=64 above) so a group's row read is one coalesced segment;
+//   - all tables live in HBM; probing is 1 load/probe on a splitmix64 hash
+//     (load factor kept <= 0.5 by the host);
+//   - lazy row init is a deterministic function of (seed, key, col, attempt)
+//     via splitmix64 — bit-identical to the torch oracle in core/rng.py;
+//   - compile with -ffp-contract=off so optimizer/init math matches the
+//     torch float32 oracle op-for-op (no silent fma contraction).
+//
+// No CUDA-compat paths: this file is HIP-for-gfx950 only.
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define DEV static __device__ __forceinline__
+
+typedef unsigned long long u64;
+typedef long long i64;
+
+static constexpr u64 EMPTY = ~0ull;
+static constexpr int BLOCK = 256;
+
+static inline int cdiv(long a, long b) { return (int)((a + b - 1) / b); }
+static inline int grid1d(long n) {
+    long g = (n + BLOCK - 1) / BLOCK;
+    return (int)(g < 1 ? 1 : g);
+}
+
+// ------------------------------------------------------------------ rng
+
+DEV u64 splitmix64(u64 z) {
+    z += 0x9E3779B97F4A7C15ull;
+    z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+    z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+    return z ^ (z >> 31);
+}
+// NOTE: core/rng.py applies the same constants with the "+C1" folded into the
+// first line; both sides compute splitmix64(x) identically.
+
+DEV u64 uniform_bits(u64 seed, u64 key, u64 col, u64 attempt) {
+    u64 x = key * 0x100000ull + col + (attempt << 40);
+    return splitmix64(splitmix64(x) ^ seed);
+}
+
+DEV float uniform01(u64 seed, u64 key, u64 col, u64 attempt) {
+    return (float)(uniform_bits(seed, key, col, attempt) >> 40)
+           * (1.0f / 16777216.0f);
+}
+
+// init categories
+enum { INIT_CONSTANT = 0, INIT_UNIFORM = 1, INIT_NORMAL = 2 };
+
+DEV float normal01(u64 seed, u64 key, u64 col, u64 attempt) {
+    float u1 = uniform01(seed, key, col, 2 * attempt);
+    float u2 = uniform01(seed, key, col, 2 * attempt + 1);
+    float r = sqrtf(-2.0f * logf(1.0f - u1));
+    return r * cosf(2.0f * 3.14159265358979323846f * u2);
+}
+
+DEV float init_value(int cat, float p0, float p1, float p2,
+                     u64 seed, u64 key, u64 col) {
+    // constant: p0=value; uniform: p0=minval p1=maxval;
+    // normal: p0=mean p1=stddev p2=truncated (one-sided rejection, reference
+    // EmbeddingInitializer.h:76-81 resamples only while (w-mean)/stddev > trunc)
+    if (cat == INIT_CONSTANT) return p0;
+    if (cat == INIT_UNIFORM) {
+        float u = uniform01(seed, key, col, 0);
+        return p0 + u * (p1 - p0);
+    }
+    float z = normal01(seed, key, col, 0);
+    if (p2 > 0.1f) {
+        for (int attempt = 1; attempt < 16 && z > p2; ++attempt)
+            z = normal01(seed, key, col, attempt);
+    }
+    return p0 + p1 * z;
+}
+
+// ------------------------------------------------------- unique + inverse
+// Batch-local dedup via a scratch open-addressed table (3 passes, no spin).
+
+__global__ void k_unique_insert(const i64* __restrict__ keys, long n,
+                                u64* __restrict__ tk, long mask,
+                                int* __restrict__ slot_of,
+                                unsigned char* __restrict__ is_first) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    u64 k = (u64)keys[i];
+    u64 h = splitmix64(k) & (u64)mask;
+    for (;;) {
+        u64 cur = tk[h];
+        if (cur == k) break;
+        if (cur == EMPTY) {
+            u64 prev = atomicCAS(&tk[h], EMPTY, k);
+            if (prev == EMPTY) { is_first[i] = 1; break; }
+            if (prev == k) break;
+        }
+        h = (h + 1) & (u64)mask;
+    }
+    slot_of[i] = (int)h;
+}
+
+__global__ void k_unique_assign(const i64* __restrict__ keys, long n,
+                                const int* __restrict__ slot_of,
+                                const unsigned char* __restrict__ is_first,
+                                int* __restrict__ tv,
+                                i64* __restrict__ unique_keys,
+                                int* __restrict__ counter) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    if (!is_first[i]) return;
+    int uid = atomicAdd(counter, 1);
+    tv[slot_of[i]] = uid;
+    unique_keys[uid] = keys[i];
+}
+
+__global__ void k_unique_inverse(long n, const int* __restrict__ slot_of,
+                                 const int* __restrict__ tv,
+                                 i64* __restrict__ inverse) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    inverse[i] = (i64)tv[slot_of[i]];
+}
+
+// --------------------------------------------------- persistent hash table
+// keys arriving here are UNIQUE within one call (callers dedup first), so a
+// CAS loser on slot h can only be a different key -> keep probing; the value
+// written by an insert in an earlier kernel is visible (kernel-boundary
+// ordering), no spin needed.
+
+__global__ void k_ht_lookup(u64* __restrict__ tk, int* __restrict__ tv,
+                            long mask, const i64* __restrict__ keys, long n,
+                            int* __restrict__ nrows,
+                            i64* __restrict__ slot_keys,
+                            i64* __restrict__ slots,
+                            unsigned char* __restrict__ new_mask,
+                            int insert) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    u64 k = (u64)keys[i];
+    u64 h = splitmix64(k) & (u64)mask;
+    i64 slot = -1;
+    unsigned char is_new = 0;
+    for (;;) {
+        u64 cur = tk[h];
+        if (cur == k) { slot = (i64)tv[h]; break; }
+        if (cur == EMPTY) {
+            if (!insert) { slot = -1; break; }
+            u64 prev = atomicCAS(&tk[h], EMPTY, k);
+            if (prev == EMPTY) {
+                int s = atomicAdd(nrows, 1);
+                tv[h] = s;
+                slot_keys[s] = (i64)k;
+                slot = s;
+                is_new = 1;
+                break;
+            }
+            if (prev == k) { slot = (i64)tv[h]; break; }
+        }
+        h = (h + 1) & (u64)mask;
+    }
+    slots[i] = slot;
+    if (new_mask) new_mask[i] = is_new;
+}
+
+__global__ void k_ht_rehash(const u64* __restrict__ tk_old,
+                            const int* __restrict__ tv_old, long cap_old,
+                            u64* __restrict__ tk_new, int* __restrict__ tv_new,
+                            long mask_new) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= cap_old) return;
+    u64 k = tk_old[i];
+    if (k == EMPTY) return;
+    u64 h = splitmix64(k) & (u64)mask_new;
+    for (;;) {
+        u64 prev = atomicCAS(&tk_new[h], EMPTY, k);
+        if (prev == EMPTY) { tv_new[h] = tv_old[i]; return; }
+        h = (h + 1) & (u64)mask_new;
+    }
+}
+
+// ------------------------------------------------------ array-table touch
+// slots unique within a call; marks rows live and reports which were new.
+
+__global__ void k_array_touch(unsigned char* __restrict__ valid,
+                              const i64* __restrict__ slots, long n,
+                              unsigned char* __restrict__ new_mask,
+                              int* __restrict__ n_new) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    i64 s = slots[i];
+    unsigned char was = valid[s];
+    valid[s] = 1;
+    new_mask[i] = !was;
+    if (!was) atomicAdd(n_new, 1);
+}
+
+// -------------------------------------------------- gather + lazy row init
+// One G-lane group per row; new rows are initialized in the miss path
+// (weights from the deterministic initializer, optimizer state copied from
+// the host-prepared state_init_row) and simultaneously returned.
+
+template <int G>
+__global__ void k_gather_init(float* __restrict__ weights,
+                              float* __restrict__ state,
+                              long dim, long sd,
+                              const i64* __restrict__ slots,
+                              const unsigned char* __restrict__ new_mask,
+                              const i64* __restrict__ keys, long n,
+                              float* __restrict__ out,
+                              int init_cat, float p0, float p1, float p2,
+                              u64 seed,
+                              const float* __restrict__ state_init_row) {
+    long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) / G;
+    int lane = threadIdx.x % G;
+    if (g >= n) return;
+    i64 slot = slots[g];
+    if (slot < 0) {  // read-only miss: zeros out
+        if (out)
+            for (long j = lane; j < dim; j += G) out[g * dim + j] = 0.0f;
+        return;
+    }
+    float* wrow = weights + (u64)slot * dim;
+    if (new_mask && new_mask[g]) {
+        u64 key = (u64)keys[g];
+        for (long j = lane; j < dim; j += G) {
+            float v = init_value(init_cat, p0, p1, p2, seed, key, (u64)j);
+            wrow[j] = v;
+            if (out) out[g * dim + j] = v;
+        }
+        if (sd > 0) {
+            float* srow = state + (u64)slot * sd;
+            for (long j = lane; j < sd; j += G) srow[j] = state_init_row[j];
+        }
+    } else {
+        if (out)
+            for (long j = lane; j < dim; j += G)
+                out[g * dim + j] = wrow[j];
+    }
+}
+
+// -------------------------------------------------------- reduce-by-inverse
+
+__global__ void k_reduce_grads(const i64* __restrict__ inverse,
+                               const float* __restrict__ grads,
+                               long n, long dim,
+                               float* __restrict__ ugrads) {
+    long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (e >= n * dim) return;
+    long i = e / dim, j = e % dim;
+    atomicAdd(&ugrads[inverse[i] * dim + j], grads[e]);
+}
+
+__global__ void k_reduce_counts(const i64* __restrict__ inverse, long n,
+                                u64* __restrict__ counts) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    atomicAdd(&counts[inverse[i]], 1ull);
+}
+
+// ------------------------------------------------------------- optimizers
+// State layouts documented in core/optimizers.py; formulas are the
+// reference's (EmbeddingOptimizer.h) applied per unique row. Scalar state
+// (adam/adamax beta powers, test flip) is updated by lane 0 of the row's
+// group and broadcast with __shfl.
+
+enum {
+    OPT_DEFAULT = 0, OPT_ADADELTA = 1, OPT_ADAGRAD = 2, OPT_ADAM = 3,
+    OPT_ADAMAX = 4, OPT_FTRL = 5, OPT_RMSPROP = 6, OPT_SGD = 7, OPT_TEST = 8
+};
+
+template <int G, int OPT>
+__global__ void k_apply_opt(float* __restrict__ weights,
+                            float* __restrict__ state,
+                            long dim, long sd,
+                            const i64* __restrict__ slots, long n,
+                            const float* __restrict__ grads,
+                            const u64* __restrict__ counts,
+                            float c0, float c1, float c2, float c3,
+                            float c4, float c5, float c6) {
+    long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) / G;
+    int lane = threadIdx.x % G;
+    if (g >= n) return;
+    i64 slot = slots[g];
+    float* w = weights + (u64)slot * dim;
+    float* s = state + (u64)slot * sd;
+    const float* gr = grads + (u64)g * dim;
+
+    if (OPT == OPT_DEFAULT) {
+        const float lr = c0;
+        if (lr != 0.0f)
+            for (long j = lane; j < dim; j += G) w[j] -= lr * gr[j];
+    } else if (OPT == OPT_ADADELTA) {
+        const float lr = c0, rho = c1, eps = c2;
+        float* accum = s;
+        float* au = s + dim;
+        for (long j = lane; j < dim; j += G) {
+            float gj = gr[j];
+            float a = accum[j] * rho + gj * gj * (1.0f - rho);
+            accum[j] = a;
+            float upd = gj * sqrtf(au[j] + eps) / sqrtf(a + eps);
+            au[j] = au[j] * rho + upd * upd * (1.0f - rho);
+            w[j] -= lr * upd;
+        }
+    } else if (OPT == OPT_ADAGRAD) {
+        const float lr = c0, eps = c2;
+        float* accum = s;
+        for (long j = lane; j < dim; j += G) {
+            float gj = gr[j];
+            float a = accum[j] + gj * gj;
+            accum[j] = a;
+            w[j] -= lr * gj / (sqrtf(a) + eps);
+        }
+    } else if (OPT == OPT_ADAM) {
+        const float lr = c0, b1 = c1, b2 = c2, eps = c3;
+        float* m = s;
+        float* v = s + dim;
+        float lr_t = 0.0f;
+        if (lane == 0) {
+            float b1t = s[2 * dim] * b1;
+            float b2t = s[2 * dim + 1] * b2;
+            s[2 * dim] = b1t;
+            s[2 * dim + 1] = b2t;
+            lr_t = lr * sqrtf(1.0f - b2t) / (1.0f - b1t);
+        }
+        lr_t = __shfl(lr_t, 0, G);
+        for (long j = lane; j < dim; j += G) {
+            float gj = gr[j];
+            float mj = m[j] * b1 + gj * (1.0f - b1);
+            float vj = v[j] * b2 + gj * gj * (1.0f - b2);
+            m[j] = mj; v[j] = vj;
+            w[j] -= lr_t * mj / (sqrtf(vj) + eps);
+        }
+    } else if (OPT == OPT_ADAMAX) {
+        const float lr = c0, b1 = c1, b2 = c2, eps = c3;
+        float* m = s;
+        float* v = s + dim;
+        float lr_t = 0.0f;
+        if (lane == 0) {
+            float b1t = s[2 * dim] * b1;
+            s[2 * dim] = b1t;
+            lr_t = lr / (1.0f - b1t);
+        }
+        lr_t = __shfl(lr_t, 0, G);
+        for (long j = lane; j < dim; j += G) {
+            float gj = gr[j];
+            float mj = m[j] * b1 + gj * (1.0f - b1);
+            float vj = fmaxf(fabsf(gj), v[j] * b2);
+            m[j] = mj; v[j] = vj;
+            w[j] -= lr_t * mj / (vj + eps);
+        }
+    } else if (OPT == OPT_FTRL) {
+        const float lr = c0, l1 = c1, l2 = c2, l2s = c3, lrp = c4, beta = c5;
+        float* accum = s;
+        float* linear = s + dim;
+        const float adj_l2 = l2 + beta / lr / 2.0f;
+        for (long j = lane; j < dim; j += G) {
+            float gj = gr[j];
+            float wj = w[j];
+            float gg = gj + 2.0f * l2s * wj;
+            float a_old = accum[j];
+            float a_new = a_old + gj * gj;
+            float sigma, quad;
+            if (lrp == -0.5f) {
+                sigma = (sqrtf(a_new) - sqrtf(a_old)) / lr;
+                quad = sqrtf(a_new) / lr + 2.0f * adj_l2;
+            } else {
+                float p = -lrp;
+                sigma = (powf(a_new, p) - powf(a_old, p)) / lr;
+                quad = powf(a_new, p) / lr + 2.0f * adj_l2;
+            }
+            float lin = linear[j] + gg - sigma * wj;
+            linear[j] = lin;
+            accum[j] = a_new;
+            float l1a = fminf(fmaxf(lin, -l1), l1);
+            w[j] = (l1a - lin) / quad;
+        }
+    } else if (OPT == OPT_RMSPROP) {
+        const float lr = c0, rho = c1, mom = c2, eps = c3;
+        float* accum = s;
+        float* moment = s + dim;
+        for (long j = lane; j < dim; j += G) {
+            float gj = gr[j];
+            float a = accum[j] * rho + gj * gj * (1.0f - rho);
+            accum[j] = a;
+            float mo = moment[j] * mom + lr * gj / sqrtf(a + eps);
+            moment[j] = mo;
+            w[j] -= mo;
+        }
+    } else if (OPT == OPT_SGD) {
+        const float lr = c0, mom = c1;
+        const bool nesterov = c2 != 0.0f;
+        float* moment = s;
+        for (long j = lane; j < dim; j += G) {
+            float gj = gr[j];
+            float mo = moment[j] * mom + lr * gj;
+            moment[j] = mo;
+            w[j] -= nesterov ? (mo * mom + lr * gj) : mo;
+        }
+    } else if (OPT == OPT_TEST) {
+        const float lr = c0, flip = c1;
+        float s0 = 0.0f;
+        if (lane == 0) {
+            s0 = flip - s[0];
+            s[0] = s0;
+        }
+        s0 = __shfl(s0, 0, G);
+        float cnt = (float)counts[g];
+        for (long j = lane; j < dim; j += G)
+            w[j] += lr * gr[j] / cnt + s0;
+    }
+}
+
+// ============================================================== launchers
+
+extern "C" {
+
+void emb_unique(const i64* keys, long n, u64* tk, int* tv, long cap,
+                int* slot_of, unsigned char* is_first, i64* unique_keys,
+                i64* inverse, int* counter, hipStream_t stream) {
+    hipMemsetAsync(tk, 0xFF, cap * sizeof(u64), stream);
+    hipMemsetAsync(counter, 0, sizeof(int), stream);
+    hipMemsetAsync(is_first, 0, n, stream);
+    long mask = cap - 1;
+    k_unique_insert<<<grid1d(n), BLOCK, 0, stream>>>(keys, n, tk, mask,
+                                                     slot_of, is_first);
+    k_unique_assign<<<grid1d(n), BLOCK, 0, stream>>>(keys, n, slot_of,
+                                                     is_first, tv,
+                                                     unique_keys, counter);
+    k_unique_inverse<<<grid1d(n), BLOCK, 0, stream>>>(n, slot_of, tv, inverse);
+}
+
+void emb_ht_lookup(u64* tk, int* tv, long cap, const i64* keys, long n,
+                   int* nrows, i64* slot_keys, i64* slots,
+                   unsigned char* new_mask, int insert, hipStream_t stream) {
+    k_ht_lookup<<<grid1d(n), BLOCK, 0, stream>>>(tk, tv, cap - 1, keys, n,
+                                                 nrows, slot_keys, slots,
+                                                 new_mask, insert);
+}
+
+void emb_ht_rehash(const u64* tk_old, const int* tv_old, long cap_old,
+                   u64* tk_new, int* tv_new, long cap_new,
+                   hipStream_t stream) {
+    hipMemsetAsync(tk_new, 0xFF, cap_new * sizeof(u64), stream);
+    k_ht_rehash<<<grid1d(cap_old), BLOCK, 0, stream>>>(
+        tk_old, tv_old, cap_old, tk_new, tv_new, cap_new - 1);
+}
+
+void emb_array_touch(unsigned char* valid, const i64* slots, long n,
+                     unsigned char* new_mask, int* n_new, hipStream_t stream) {
+    k_array_touch<<<grid1d(n), BLOCK, 0, stream>>>(valid, slots, n, new_mask,
+                                                   n_new);
+}
+
+void emb_gather_init(float* weights, float* state, long dim, long sd,
+                     const i64* slots, const unsigned char* new_mask,
+                     const i64* keys, long n, float* out, int init_cat,
+                     float p0, float p1, float p2, u64 seed,
+                     const float* state_init_row, hipStream_t stream) {
+    if (n == 0) return;
+    if (dim <= 32) {
+        const int G = 16;
+        long threads = n * G;
+        k_gather_init<G><<<grid1d(threads), BLOCK, 0, stream>>>(
+            weights, state, dim, sd, slots, new_mask, keys, n, out,
+            init_cat, p0, p1, p2, seed, state_init_row);
+    } else {
+        const int G = 64;
+        long threads = n * G;
+        k_gather_init<G><<<grid1d(threads), BLOCK, 0, stream>>>(
+            weights, state, dim, sd, slots, new_mask, keys, n, out,
+            init_cat, p0, p1, p2, seed, state_init_row);
+    }
+}
+
+void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
+                           long dim, float* ugrads, u64* counts, long u,
+                           hipStream_t stream) {
+    hipMemsetAsync(ugrads, 0, (size_t)u * dim * sizeof(float), stream);
+    hipMemsetAsync(counts, 0, (size_t)u * sizeof(u64), stream);
+    if (n == 0) return;
+    k_reduce_grads<<<grid1d(n * dim), BLOCK, 0, stream>>>(inverse, grads, n,
+                                                          dim, ugrads);
+    k_reduce_counts<<<grid1d(n), BLOCK, 0, stream>>>(inverse, n, counts);
+}
+
+#define LAUNCH_OPT(G, OPT)                                                  \
+    k_apply_opt<G, OPT><<<grid1d(n * G), BLOCK, 0, stream>>>(               \
+        weights, state, dim, sd, slots, n, grads, counts, c[0], c[1], c[2], \
+        c[3], c[4], c[5], c[6])
+
+void emb_apply_optimizer(int opt, float* weights, float* state, long dim,
+                         long sd, const i64* slots, long n,
+                         const float* grads, const u64* counts,
+                         const float* c, hipStream_t stream) {
+    if (n == 0) return;
+    if (dim <= 32) {
+        const int G = 16;
+        switch (opt) {
+            case OPT_DEFAULT: LAUNCH_OPT(16, OPT_DEFAULT); break;
+            case OPT_ADADELTA: LAUNCH_OPT(16, OPT_ADADELTA); break;
+            case OPT_ADAGRAD: LAUNCH_OPT(16, OPT_ADAGRAD); break;
+            case OPT_ADAM: LAUNCH_OPT(16, OPT_ADAM); break;
+            case OPT_ADAMAX: LAUNCH_OPT(16, OPT_ADAMAX); break;
+            case OPT_FTRL: LAUNCH_OPT(16, OPT_FTRL); break;
+            case OPT_RMSPROP: LAUNCH_OPT(16, OPT_RMSPROP); break;
+            case OPT_SGD: LAUNCH_OPT(16, OPT_SGD); break;
+            case OPT_TEST: LAUNCH_OPT(16, OPT_TEST); break;
+        }
+        (void)G;
+    } else {
+        const int G = 64;
+        switch (opt) {
+            case OPT_DEFAULT: LAUNCH_OPT(64, OPT_DEFAULT); break;
+            case OPT_ADADELTA: LAUNCH_OPT(64, OPT_ADADELTA); break;
+            case OPT_ADAGRAD: LAUNCH_OPT(64, OPT_ADAGRAD); break;
+            case OPT_ADAM: LAUNCH_OPT(64, OPT_ADAM); break;
+            case OPT_ADAMAX: LAUNCH_OPT(64, OPT_ADAMAX); break;
+            case OPT_FTRL: LAUNCH_OPT(64, OPT_FTRL); break;
+            case OPT_RMSPROP: LAUNCH_OPT(64, OPT_RMSPROP); break;
+            case OPT_SGD: LAUNCH_OPT(64, OPT_SGD); break;
+            case OPT_TEST: LAUNCH_OPT(64, OPT_TEST); break;
+        }
+        (void)G;
+    }
+}
+
+}  // extern "C"

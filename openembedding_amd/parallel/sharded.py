@@ -1,0 +1,165 @@
+"""ShardedVariable: the distributed pull/push engine.
+
+This file collapses the reference's whole RPC data path —
+EmbeddingPullOperator / EmbeddingPushOperator / EmbeddingStoreOperator +
+client handles (reference openembedding/server/*.cpp, client/
+EmbeddingVariableHandle.cpp) — into RCCL all-to-all over xGMI:
+
+  pull  = unique+bucketize -> all_to_all_v(keys) -> owner gather (+lazy init)
+          -> all_to_all_v(rows) -> scatter to duplicated positions
+  push  = reduce-by-key locally (the reference's client-side pre-aggregation,
+          EmbeddingPushOperator.cpp:39-58 — dedup before the wire is the key
+          bandwidth win) -> all_to_all_v(grads,counts) -> owner merge
+  commit= local fused optimizer apply, no communication
+          (reference EmbeddingStoreOperator.cpp:23-81; ordering that the
+          reference built from batch-id pending queues comes free from
+          stream ordering of collectives).
+
+Owner of global key k is rank ``k % world_size`` (reference shard routing,
+EmbeddingPullOperator.cpp:74-78, kept for checkpoint compatibility).
+"""
+
+from __future__ import annotations
+
+import dataclasses
+from typing import List, Optional
+
+import torch
+
+from ..core.variable import VariableShard
+from ..ops import dispatch as ops
+from . import comm
+
+
+@dataclasses.dataclass
+class PullHandle:
+    """Routing state captured by pull, reused by the matching push
+    (the reference kept it as saved block offsets in the pull handler,
+    EmbeddingPullOperator.cpp:67-79,229-249)."""
+
+    shape: torch.Size
+    unique: torch.Tensor            # [u] unique keys, local order
+    inverse: torch.Tensor           # [n] position -> unique id
+    order: Optional[torch.Tensor] = None       # [u] perm grouping unique by owner
+    send_splits: Optional[List[int]] = None    # keys sent to each rank
+    recv_splits: Optional[List[int]] = None    # keys received from each rank
+    owner_unique: Optional[torch.Tensor] = None  # [u2] deduped keys owned here
+    owner_inverse: Optional[torch.Tensor] = None  # recv pos -> owner_unique id
+
+
+class ShardedVariable:
+    """User-facing variable handle; wraps the local shard + collectives."""
+
+    def __init__(self, shard: VariableShard, storage=None):
+        self.shard = shard
+        self.storage = storage
+        self.world_size = shard.shard_num
+        self.rank = shard.shard_id
+        # perf counters (reference pull_indices/pull_unique accumulators,
+        # EmbeddingPullOperator.cpp:208-247 — they diagnose all-to-all sizing)
+        self.stat_pull_indices = 0
+        self.stat_pull_unique = 0
+
+    # -------------------------------------------------------------- properties
+
+    @property
+    def meta(self):
+        return self.shard.meta
+
+    @property
+    def variable_id(self):
+        return self.shard.meta.variable_id
+
+    @property
+    def embedding_dim(self):
+        return self.shard.dim
+
+    def set_initializer(self, category: str, **cfg):
+        self.shard.set_initializer(category, **cfg)
+
+    def set_optimizer(self, category: str, **cfg):
+        self.shard.set_optimizer(category, **cfg)
+
+    # ------------------------------------------------------------------- pull
+
+    def pull(self, indices: torch.Tensor, readonly: bool = False):
+        """indices: int64 tensor of any shape -> (weights [*shape, dim], handle).
+
+        Collective when world_size > 1: every rank must call it the same
+        number of times per variable per step."""
+        flat = indices.reshape(-1).to(torch.int64)
+        n = flat.numel()
+        unique, inverse = ops.unique_inverse(flat)
+        self.stat_pull_indices += n
+        self.stat_pull_unique += unique.numel()
+        h = PullHandle(shape=indices.shape, unique=unique, inverse=inverse)
+        if self.world_size == 1:
+            rows_u = (self.shard.pull_readonly(unique) if readonly
+                      else self.shard.pull(unique))
+        else:
+            rows_u = self._pull_remote(h, readonly)
+        out = rows_u.index_select(0, inverse)
+        out = out.view(*h.shape, self.shard.dim)
+        return out, h
+
+    def _pull_remote(self, h: PullHandle, readonly: bool) -> torch.Tensor:
+        world = self.world_size
+        owner = h.unique % world
+        order = torch.argsort(owner, stable=True)
+        send_keys = h.unique.index_select(0, order)
+        send_counts = torch.bincount(owner, minlength=world)
+        recv_counts = comm.all_to_all_lengths(send_counts)
+        send_splits = send_counts.tolist()
+        recv_splits = recv_counts.tolist()
+        recv_keys = comm.all_to_all_v(send_keys, send_splits, recv_splits)
+        # owner side: dedup across ranks, gather (+lazy init), fan back out
+        uk2, inv2 = ops.unique_inverse(recv_keys)
+        rows_u2 = (self.shard.pull_readonly(uk2) if readonly
+                   else self.shard.pull(uk2))
+        rows_back = rows_u2.index_select(0, inv2)
+        rows_sorted = comm.all_to_all_v(rows_back, recv_splits, send_splits)
+        rows_u = torch.empty_like(rows_sorted)
+        rows_u.index_copy_(0, order, rows_sorted)
+        h.order = order
+        h.send_splits = send_splits
+        h.recv_splits = recv_splits
+        h.owner_unique = uk2
+        h.owner_inverse = inv2
+        return rows_u
+
+    # ------------------------------------------------------------------- push
+
+    def push(self, h: PullHandle, grads: torch.Tensor) -> None:
+        """grads: [*shape, dim] gradient of the pulled weights."""
+        dim = self.shard.dim
+        g = grads.reshape(-1, dim)
+        u = h.unique.numel()
+        ugrads, counts = ops.reduce_by_inverse(h.inverse, g, u)
+        if self.world_size == 1:
+            self.shard.push(h.unique, ugrads, counts)
+            return
+        send_g = ugrads.index_select(0, h.order)
+        send_c = counts.index_select(0, h.order)
+        recv_g = comm.all_to_all_v(send_g, h.send_splits, h.recv_splits)
+        recv_c = comm.all_to_all_v(send_c, h.send_splits, h.recv_splits)
+        u2 = h.owner_unique.numel()
+        g2, _ = ops.reduce_by_inverse(h.owner_inverse, recv_g, u2)
+        c2 = torch.zeros(u2, dtype=recv_c.dtype, device=recv_c.device)
+        c2.index_add_(0, h.owner_inverse, recv_c)
+        self.shard.push(h.owner_unique, g2, c2)
+
+    # ----------------------------------------------------------------- commit
+
+    def update_weights(self) -> None:
+        self.shard.update_weights()
+
+    # ------------------------------------------------------- whole-table pulls
+
+    def pull_dense(self, start: int, count: int) -> torch.Tensor:
+        """Materialize rows [start, start+count) of a bounded-vocab variable on
+        every rank (export path; reference save_as_original_model bulk pull,
+        exb.py:506-547). Collective."""
+        keys = torch.arange(start, start + count, dtype=torch.int64,
+                            device=self.shard.device)
+        out, _ = self.pull(keys, readonly=True)
+        return out
