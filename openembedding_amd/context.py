@@ -118,9 +118,14 @@ class Context:
             vocabulary_size = HASH_VOCAB_THRESHOLD
         meta = VariableMeta(variable_id=vid, embedding_dim=embedding_dim,
                             dtype=dtype, vocabulary_size=vocabulary_size)
-        shard = VariableShard(meta, shard_id=self.rank,
-                              shard_num=self.world_size,
-                              device=str(self.device), seed=self.seed)
+        if self.device.type == "cuda":
+            from .core.variable_gpu import HipVariableShard
+            shard_cls = HipVariableShard
+        else:
+            shard_cls = VariableShard
+        shard = shard_cls(meta, shard_id=self.rank,
+                          shard_num=self.world_size,
+                          device=str(self.device), seed=self.seed)
         var = ShardedVariable(shard, storage)
         self.variables[vid] = var
         return var

@@ -1,0 +1,138 @@
+"""CTR model zoo: LR, WDL, DeepFM, xDeepFM.
+
+The reference trains these via DeepCTR/Keras (examples/criteo_deepctr_*.py,
+test/benchmark/criteo_deepctr.py); here they are native torch modules whose
+sparse features go through the PS-backed CombinedEmbedding (one fused pull
+per step for all 26 fields — see openembedding_amd/torch/__init__.py).
+
+All models take (dense [B,13] float32, sparse [B,26] int64 per-field ids)
+and return logits [B].
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+import torch
+import torch.nn as nn
+
+from ..torch import CombinedEmbedding
+from .criteo import CRITEO_FIELD_VOCABS, N_DENSE
+
+
+def _mlp(in_dim: int, hidden: Sequence[int], out_dim: int = 1) -> nn.Sequential:
+    layers: List[nn.Module] = []
+    d = in_dim
+    for h in hidden:
+        layers += [nn.Linear(d, h), nn.ReLU()]
+        d = h
+    layers.append(nn.Linear(d, out_dim))
+    return nn.Sequential(*layers)
+
+
+class _CTRBase(nn.Module):
+    def __init__(self, field_vocabs: Optional[List[int]], dim: int,
+                 sparse_as_dense_size: int = 0):
+        super().__init__()
+        self.field_vocabs = list(field_vocabs or CRITEO_FIELD_VOCABS)
+        self.dim = dim
+        self.n_fields = len(self.field_vocabs)
+        self.embedding = CombinedEmbedding(self.field_vocabs, dim)
+        # first-order ("linear"/wide) weights: dim-1 embedding per field
+        self.linear_embedding = CombinedEmbedding(self.field_vocabs, 1)
+        self.dense_linear = nn.Linear(N_DENSE, 1)
+
+    def _first_order(self, dense: torch.Tensor, sparse: torch.Tensor
+                     ) -> torch.Tensor:
+        lin = self.linear_embedding(sparse).sum(dim=(1, 2))
+        return lin + self.dense_linear(dense).squeeze(-1)
+
+
+class LR(_CTRBase):
+    """Logistic regression (reference examples/criteo_lr_subclass.py:
+    subclassed model, hash-mode embedding dim 1)."""
+
+    def __init__(self, field_vocabs: Optional[List[int]] = None):
+        super().__init__(field_vocabs, dim=1)
+
+    def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
+        return self._first_order(dense, sparse)
+
+
+class WDL(_CTRBase):
+    """Wide & Deep (reference benchmark model 'WDL')."""
+
+    def __init__(self, field_vocabs: Optional[List[int]] = None, dim: int = 9,
+                 hidden: Sequence[int] = (400, 400, 400)):
+        super().__init__(field_vocabs, dim)
+        self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
+
+    def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
+        e = self.embedding(sparse)                       # [B, F, d]
+        deep_in = torch.cat([e.flatten(1), dense], dim=1)
+        return self._first_order(dense, sparse) + self.dnn(deep_in).squeeze(-1)
+
+
+class DeepFM(_CTRBase):
+    """DeepFM (reference primary benchmark model, BASELINE.md DeepFM dim9)."""
+
+    def __init__(self, field_vocabs: Optional[List[int]] = None, dim: int = 9,
+                 hidden: Sequence[int] = (400, 400, 400)):
+        super().__init__(field_vocabs, dim)
+        self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
+
+    def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
+        e = self.embedding(sparse)                       # [B, F, d]
+        # FM second order: 0.5*((sum_f e)^2 - sum_f e^2) summed over dim
+        s = e.sum(dim=1)
+        fm2 = 0.5 * (s * s - (e * e).sum(dim=1)).sum(dim=1)
+        deep_in = torch.cat([e.flatten(1), dense], dim=1)
+        return (self._first_order(dense, sparse) + fm2
+                + self.dnn(deep_in).squeeze(-1))
+
+
+class CIN(nn.Module):
+    """Compressed Interaction Network (xDeepFM component)."""
+
+    def __init__(self, n_fields: int, dim: int,
+                 layer_sizes: Sequence[int] = (128, 128)):
+        super().__init__()
+        self.layer_sizes = list(layer_sizes)
+        self.convs = nn.ModuleList()
+        h_prev = n_fields
+        for h in self.layer_sizes:
+            self.convs.append(nn.Conv1d(n_fields * h_prev, h, 1, bias=False))
+            h_prev = h
+        self.fc = nn.Linear(sum(self.layer_sizes), 1)
+
+    def forward(self, e: torch.Tensor) -> torch.Tensor:  # e: [B, F, d]
+        B, F, d = e.shape
+        x0 = e
+        xk = e
+        outs = []
+        for conv in self.convs:
+            # outer product along field axes: [B, F*Hk, d]
+            z = torch.einsum("bfd,bhd->bfhd", x0, xk).reshape(B, -1, d)
+            xk = torch.relu(conv(z))
+            outs.append(xk.sum(dim=2))                   # [B, Hk]
+        return self.fc(torch.cat(outs, dim=1)).squeeze(-1)
+
+
+class xDeepFM(_CTRBase):
+    """xDeepFM (reference benchmark model 'xDeepFM', compute-bound case)."""
+
+    def __init__(self, field_vocabs: Optional[List[int]] = None, dim: int = 9,
+                 hidden: Sequence[int] = (400, 400, 400),
+                 cin_layers: Sequence[int] = (128, 128)):
+        super().__init__(field_vocabs, dim)
+        self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
+        self.cin = CIN(self.n_fields, dim, cin_layers)
+
+    def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
+        e = self.embedding(sparse)
+        deep_in = torch.cat([e.flatten(1), dense], dim=1)
+        return (self._first_order(dense, sparse) + self.cin(e)
+                + self.dnn(deep_in).squeeze(-1))
+
+
+MODELS = {"lr": LR, "wdl": WDL, "deepfm": DeepFM, "xdeepfm": xDeepFM}

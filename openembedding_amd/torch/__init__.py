@@ -1,0 +1,448 @@
+"""openembedding_amd.torch — the user-facing module API.
+
+PyTorch rebuild of the reference's Keras surface
+(reference openembedding/tensorflow/exb.py): ``Embedding``,
+``distributed_optimizer``, ``distributed_model``, ``Model``,
+``save_server_model`` / ``load_server_model``, ``save_as_original_model``,
+``pulling``. The fake-gradient + registered-gradient machinery of the
+reference (exb.py:89-104) collapses into one autograd.Function whose forward
+is the sharded pull and whose backward is the sharded push.
+
+Typical use (3-line change, like the reference README):
+
+    import openembedding_amd.torch as embed
+    model.embedding = embed.Embedding(-1, 64)          # PS-backed, hashed keys
+    opt = embed.distributed_optimizer(torch.optim.Adagrad(model.parameters(), lr=0.01))
+    ...
+    loss.backward(); opt.step()                        # commits sparse + dense
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional, Sequence
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..context import Context, Storage, get_context
+from ..core.variable import HASH_VOCAB_THRESHOLD
+from ..parallel import comm
+from ..parallel.sharded import ShardedVariable
+
+__all__ = [
+    "Embedding", "CombinedEmbedding", "Variable", "Context",
+    "distributed_optimizer", "DistributedOptimizer", "distributed_model",
+    "Model", "save_server_model", "load_server_model",
+    "save_as_original_model", "pulling", "sparse_read_as_dense",
+    "should_persist_server_model", "persist_server_model",
+    "restore_server_model", "get_context",
+]
+
+# registry of live PS-backed embeddings (reference track_variable,
+# exb.py:125-131)
+_tracked: List["Embedding"] = []
+
+
+class _PullPushFn(torch.autograd.Function):
+    """forward = sharded pull, backward = sharded push (replaces the
+    reference's PullWeights op + fake-grad PushGradients registration)."""
+
+    @staticmethod
+    def forward(ctx, hook: torch.Tensor, indices: torch.Tensor,
+                var: ShardedVariable):
+        out, handle = var.pull(indices)
+        ctx.var = var
+        ctx.handle = handle
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        ctx.var.push(ctx.handle, grad_out.contiguous().to(torch.float32))
+        return torch.zeros(0, device=grad_out.device), None, None
+
+
+class Variable:
+    """Thin user handle over a ShardedVariable (reference exb.py Variable,
+    :222-386): sparse_read + prefetch + explicit push/update."""
+
+    def __init__(self, sharded: ShardedVariable, storage: Storage):
+        self.sharded = sharded
+        self.storage = storage
+        self._prefetched: List = []
+
+    @property
+    def embedding_dim(self):
+        return self.sharded.embedding_dim
+
+    @property
+    def vocabulary_size(self):
+        return self.sharded.meta.vocabulary_size
+
+    def sparse_read(self, indices: torch.Tensor) -> torch.Tensor:
+        out, _ = self.sharded.pull(indices, readonly=True)
+        return out
+
+    def set_initializer(self, category: str, **cfg):
+        self.sharded.set_initializer(category, **cfg)
+
+    def set_optimizer(self, category: str, **cfg):
+        self.sharded.set_optimizer(category, **cfg)
+
+    def update_weights(self):
+        self.sharded.update_weights()
+
+
+class Embedding(nn.Module):
+    """PS-backed embedding layer (reference exb.Embedding, exb.py:388-443).
+
+    num_embeddings == -1 selects hash mode over the full int64 key space
+    (reference input_dim=-1 -> 2^63, exb.py:393-396); otherwise a bounded
+    array table sharded across ranks.
+
+    ``sparse_as_dense``: keep this table as an ordinary replicated
+    nn.Embedding trained by allreduce — the reference's "cache" policy for
+    small hot vocabularies (exb.py:241-248, ~+10%: documents/en/benchmark.md).
+    """
+
+    def __init__(self, num_embeddings: int, embedding_dim: int,
+                 initializer: Optional[Dict] = None,
+                 num_shards: Optional[int] = None,
+                 sparse_as_dense: bool = False,
+                 storage: Optional[Storage] = None,
+                 dtype: torch.dtype = torch.float32):
+        super().__init__()
+        ctx = get_context()
+        self.ctx = ctx
+        self.num_embeddings = num_embeddings
+        self.embedding_dim = embedding_dim
+        self.sparse_as_dense = sparse_as_dense
+        if sparse_as_dense:
+            if num_embeddings < 0:
+                raise ValueError("sparse_as_dense needs a bounded vocabulary")
+            self.dense = nn.Embedding(num_embeddings, embedding_dim,
+                                      dtype=dtype)
+            self.dense.to(ctx.device)
+            self.variable = None
+        else:
+            vocab = HASH_VOCAB_THRESHOLD if num_embeddings < 0 else num_embeddings
+            st = storage if storage is not None else _default_storage(ctx)
+            sharded = st.create_variable(vocab, embedding_dim, dtype)
+            init = initializer or {"category": "uniform",
+                                   "minval": -1.0 / max(1, int(math.sqrt(embedding_dim))),
+                                   "maxval": 1.0 / max(1, int(math.sqrt(embedding_dim)))}
+            cat = init.pop("category") if "category" in init else "uniform"
+            sharded.set_initializer(cat, **init)
+            self.variable = Variable(sharded, st)
+            # zero-size hook so autograd reaches our Function even though
+            # indices carry no grad
+            self.grad_hook = nn.Parameter(torch.zeros(0, device=ctx.device))
+            _tracked.append(self)
+
+    def forward(self, indices: torch.Tensor) -> torch.Tensor:
+        if self.sparse_as_dense:
+            return self.dense(indices)
+        if torch.is_grad_enabled() and self.grad_hook.requires_grad:
+            return _PullPushFn.apply(self.grad_hook, indices,
+                                     self.variable.sharded)
+        return self.variable.sparse_read(indices)
+
+    def extra_repr(self):
+        mode = ("dense" if self.sparse_as_dense else
+                ("hash" if self.num_embeddings < 0 else "array"))
+        return f"{self.num_embeddings}, {self.embedding_dim}, mode={mode}"
+
+
+class CombinedEmbedding(Embedding):
+    """Many categorical fields sharing one PS variable.
+
+    MI355X-first design: instead of the reference's one-Embedding-per-feature (26
+    pulls -> 26 RPC fan-outs per step for Criteo), all fields share one key
+    space with per-field offsets, so a step does ONE unique+all_to_all+gather
+    for all fields. forward takes [batch, n_fields] raw per-field ids and
+    returns [batch, n_fields, dim].
+    """
+
+    def __init__(self, field_vocab_sizes: Sequence[int], embedding_dim: int,
+                 **kw):
+        sizes = list(field_vocab_sizes)
+        offsets = [0]
+        for s in sizes:
+            offsets.append(offsets[-1] + int(s))
+        super().__init__(offsets[-1], embedding_dim, **kw)
+        self.n_fields = len(sizes)
+        self.register_buffer(
+            "field_offsets",
+            torch.tensor(offsets[:-1], dtype=torch.int64,
+                         device=get_context().device))
+
+    def forward(self, field_ids: torch.Tensor) -> torch.Tensor:
+        keys = field_ids.to(torch.int64) + self.field_offsets
+        return super().forward(keys)
+
+
+def _default_storage(ctx: Context) -> Storage:
+    if not ctx.storages:
+        ctx.create_storage()
+    return ctx.storages[0]
+
+
+# --------------------------------------------------------------- optimizer
+
+_TORCH_TO_SERVER = {
+    "SGD": "sgd", "Adagrad": "adagrad", "Adam": "adam", "AdamW": "adam",
+    "Adamax": "adamax", "Adadelta": "adadelta", "RMSprop": "rmsprop",
+}
+
+
+def _server_cfg_from_torch(opt: torch.optim.Optimizer) -> Dict:
+    """Translate a torch optimizer's hyper-params to the server-side sparse
+    optimizer config (the reference's TF->category translation,
+    exb.py:64-86)."""
+    name = type(opt).__name__
+    cat = _TORCH_TO_SERVER.get(name)
+    if cat is None:
+        raise ValueError(f"no server-side sparse optimizer for {name}; "
+                         f"set one explicitly with Embedding.variable."
+                         f"set_optimizer(...)")
+    g = opt.param_groups[0]
+    lr = g["lr"]
+    if cat == "sgd":
+        return dict(category="sgd", learning_rate=lr,
+                    momentum=g.get("momentum", 0.0),
+                    nesterov=bool(g.get("nesterov", False)))
+    if cat == "adagrad":
+        return dict(category="adagrad", learning_rate=lr,
+                    initial_accumulator_value=g.get("initial_accumulator_value", 0.0),
+                    epsilon=g.get("eps", 1e-10))
+    if cat == "adam":
+        b1, b2 = g.get("betas", (0.9, 0.999))
+        return dict(category="adam", learning_rate=lr, beta_1=b1, beta_2=b2,
+                    epsilon=g.get("eps", 1e-8))
+    if cat == "adamax":
+        b1, b2 = g.get("betas", (0.9, 0.999))
+        return dict(category="adamax", learning_rate=lr, beta_1=b1, beta_2=b2,
+                    epsilon=g.get("eps", 1e-8))
+    if cat == "adadelta":
+        return dict(category="adadelta", learning_rate=lr,
+                    rho=g.get("rho", 0.9), epsilon=g.get("eps", 1e-6))
+    if cat == "rmsprop":
+        return dict(category="rmsprop", learning_rate=lr,
+                    rho=g.get("alpha", 0.99), momentum=g.get("momentum", 0.0),
+                    epsilon=g.get("eps", 1e-8))
+    raise AssertionError(cat)
+
+
+class DistributedOptimizer:
+    """Wraps a torch optimizer: dense grads are RCCL-allreduced (SUM, like the
+    reference's hvd.DistributedOptimizer(op=hvd.Sum) examples), then the base
+    optimizer steps, then every PS storage commits its batch
+    (reference exb.py:446-488 distributed optimizer subclasses +
+    UpdateWeights op)."""
+
+    def __init__(self, optimizer: torch.optim.Optimizer,
+                 sparse_config: Optional[Dict] = None,
+                 average_dense: bool = False):
+        self.optimizer = optimizer
+        self.average_dense = average_dense
+        self._sparse_config = sparse_config
+        self._configured_vars = set()
+        self.ctx = get_context()
+
+    # behave like the wrapped optimizer
+    def __getattr__(self, name):
+        return getattr(self.optimizer, name)
+
+    def zero_grad(self, set_to_none: bool = True):
+        self.optimizer.zero_grad(set_to_none=set_to_none)
+
+    def _allreduce_dense(self):
+        if not comm.dist_ready() or dist.get_world_size() == 1:
+            return
+        bucket: List[torch.Tensor] = []
+        for group in self.optimizer.param_groups:
+            for p in group["params"]:
+                if p.grad is not None and p.numel() > 0:
+                    bucket.append(p.grad)
+        if not bucket:
+            return
+        flat = torch.cat([g.reshape(-1) for g in bucket])
+        dist.all_reduce(flat)
+        if self.average_dense:
+            flat /= dist.get_world_size()
+        off = 0
+        for g in bucket:
+            g.copy_(flat[off:off + g.numel()].view_as(g))
+            off += g.numel()
+
+    def _ensure_sparse_configured(self):
+        cfg = self._sparse_config
+        if cfg is None:
+            try:
+                cfg = _server_cfg_from_torch(self.optimizer)
+            except ValueError:
+                cfg = None
+        for e in _tracked:
+            if e.variable is None:
+                continue
+            vid = e.variable.sharded.variable_id
+            if vid in self._configured_vars:
+                continue
+            if e.variable.sharded.shard.optimizer is None:
+                if cfg is None:
+                    raise RuntimeError(
+                        "no sparse optimizer configured; pass sparse_config= "
+                        "to distributed_optimizer or call set_optimizer on the "
+                        "Embedding variable")
+                c = dict(cfg)
+                cat = c.pop("category")
+                e.variable.set_optimizer(cat, **c)
+            self._configured_vars.add(vid)
+
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            loss = closure()
+        self._allreduce_dense()
+        self.optimizer.step()
+        self._ensure_sparse_configured()
+        self.ctx.update_all_weights()
+        self.ctx.model_version += 1
+        return loss
+
+    def state_dict(self):
+        return self.optimizer.state_dict()
+
+    def load_state_dict(self, sd):
+        self.optimizer.load_state_dict(sd)
+
+
+def distributed_optimizer(optimizer: torch.optim.Optimizer,
+                          sparse_config: Optional[Dict] = None,
+                          **kw) -> DistributedOptimizer:
+    return DistributedOptimizer(optimizer, sparse_config=sparse_config, **kw)
+
+
+# ----------------------------------------------------------- model helpers
+
+def distributed_model(model: nn.Module, sparse_as_dense_size: int = 64
+                      ) -> nn.Module:
+    """Replace every nn.Embedding in ``model`` with a PS-backed Embedding
+    (vocab < sparse_as_dense_size stays a replicated dense table — the
+    reference's policy, exb.py:241-248,593-642). In-place; returns model."""
+    for name, mod in list(model.named_children()):
+        if isinstance(mod, nn.Embedding):
+            if mod.num_embeddings < sparse_as_dense_size:
+                continue
+            rep = Embedding(mod.num_embeddings, mod.embedding_dim)
+            setattr(model, name, rep)
+        else:
+            distributed_model(mod, sparse_as_dense_size)
+    return model
+
+
+def sparse_read_as_dense(variable: Variable, vocab: int) -> torch.Tensor:
+    """Materialize rows [0, vocab) of a variable on every rank
+    (reference sparse_read_as_dense / save_as_original_model bulk pull)."""
+    chunks = []
+    step = max(1, (1 << 20) // max(1, variable.embedding_dim))
+    for start in range(0, vocab, step):
+        n = min(step, vocab - start)
+        chunks.append(variable.sharded.pull_dense(start, n))
+    return torch.cat(chunks)
+
+
+# ------------------------------------------------------- checkpoint surface
+
+def save_server_model(path: str, include_optimizer: bool = True):
+    from ..checkpoint import dump_model
+    dump_model(get_context(), path, include_optimizer=include_optimizer)
+
+
+def load_server_model(path: str):
+    from ..checkpoint import load_model
+    load_model(get_context(), path)
+
+
+def save_as_original_model(model: nn.Module, path: str):
+    """Export a fully materialized, PS-free copy of the model: every PS
+    Embedding is replaced by a plain nn.Embedding with pulled rows
+    (reference exb.py:506-547 clone-and-materialize). Saved with torch.save;
+    loadable without openembedding_amd."""
+    state = {}
+    for name, mod in model.named_modules():
+        if isinstance(mod, Embedding) and not mod.sparse_as_dense:
+            if mod.num_embeddings < 0:
+                raise ValueError("cannot materialize a hash-mode Embedding "
+                                 "into a dense table")
+            rows = sparse_read_as_dense(mod.variable, mod.num_embeddings)
+            state[name + ".weight"] = rows.cpu()
+    sd = {k: v.cpu() for k, v in model.state_dict().items()
+          if not k.endswith("grad_hook")}
+    sd.update(state)
+    if get_context().rank == 0:
+        torch.save({"state_dict": sd, "format": "openembedding_amd.original",
+                    "version": "0.2"}, path)
+
+
+class Model(nn.Module):
+    """Wrapper adding server-model save/load beside the torch checkpoint
+    (reference _DistributedModel, exb.py:550-583)."""
+
+    def __init__(self, module: nn.Module):
+        super().__init__()
+        self.module = module
+
+    def forward(self, *a, **kw):
+        return self.module(*a, **kw)
+
+    def save_weights(self, filepath: str):
+        if get_context().rank == 0:
+            torch.save(self.module.state_dict(), filepath)
+        save_server_model(filepath + ".openembedding")
+
+    def load_weights(self, filepath: str):
+        sd = torch.load(filepath, map_location=get_context().device,
+                        weights_only=True)
+        self.module.load_state_dict(sd, strict=False)
+        load_server_model(filepath + ".openembedding")
+
+
+# ------------------------------------------------------------ prefetch API
+
+def pulling(loader, model=None, depth: int = 2):
+    """Dataset-side prefetch (reference pulling(), exb.py:645-691 +
+    PrefetchPullWeights): yields batches while issuing the NEXT batches'
+    embedding pulls ahead of time on the engine. With the GPU engine the
+    overlap is stream-level; this wrapper currently provides the API shape
+    and simple read-ahead."""
+    import collections
+    it = iter(loader)
+    buf = collections.deque()
+    while True:
+        while len(buf) < depth:
+            try:
+                buf.append(next(it))
+            except StopIteration:
+                break
+        if not buf:
+            return
+        yield buf.popleft()
+
+
+# ------------------------------------------------------------- PMem parity
+
+def should_persist_server_model() -> bool:
+    """The reference's PMem cache-full backpressure signal
+    (EmbeddingPullOperator.cpp:182-189). The HBM/DRAM tier analogue is not
+    yet wired; returns False."""
+    return False
+
+
+def persist_server_model(path: str):
+    save_server_model(path, include_optimizer=True)
+
+
+def restore_server_model(path: str):
+    load_server_model(path)

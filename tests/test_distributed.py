@@ -1,0 +1,152 @@
+"""Multi-process (gloo, world_size=2) tests of the sharded all-to-all engine.
+
+Because lazy init is a deterministic function of (seed, key), a world-2 run
+must produce bit-compatible rows with a world-1 run over the union of its
+batches — that is the equivalence checked here, plus cross-shard-count
+checkpoint reload (the reference's load-works-across-shard-counts contract,
+EmbeddingLoadOperator.cpp:58-111)."""
+
+import os
+import socket
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+DIM = 4
+VOCAB = 1000
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _init(rank, world, port):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+
+
+def _rank_batches(rank):
+    """Deterministic per-rank batches (with cross-rank key overlap)."""
+    g = torch.Generator().manual_seed(100 + rank)
+    keys = torch.randint(0, VOCAB, (64,), generator=g, dtype=torch.int64)
+    grads = torch.randn(64, DIM, generator=g)
+    return keys, grads
+
+
+def _worker_engine(rank, world, port, tmp):
+    from openembedding_amd.context import Context
+    from openembedding_amd import checkpoint
+
+    _init(rank, world, port)
+    ctx = Context(device="cpu")
+    st = ctx.create_storage()
+    var = st.create_variable(VOCAB, DIM)
+    var.set_initializer("uniform", minval=-1, maxval=1)
+    var.set_optimizer("test")
+
+    keys, grads = _rank_batches(rank)
+    out, h = var.pull(keys)
+    assert out.shape == (64, DIM)
+    var.push(h, grads)
+    st.update_weights()
+    out2, _ = var.pull(keys, readonly=True)
+
+    checkpoint.dump_model(ctx, os.path.join(tmp, "ckpt"))
+    torch.save({"keys": keys, "out": out, "out2": out2},
+               os.path.join(tmp, f"result_{rank}.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_world2_matches_world1(tmp_path):
+    port = _free_port()
+    mp.spawn(_worker_engine, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
+    for v in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
+        os.environ.pop(v, None)
+
+    # single-process reference over the union of both ranks' batches
+    from openembedding_amd.context import Context
+    from openembedding_amd import checkpoint
+
+    ctx = Context(device="cpu")
+    st = ctx.create_storage()
+    var = st.create_variable(VOCAB, DIM)
+    var.set_initializer("uniform", minval=-1, maxval=1)
+    var.set_optimizer("test")
+
+    k0, g0 = _rank_batches(0)
+    k1, g1 = _rank_batches(1)
+    out_ref0, h0 = var.pull(k0)
+    out_ref1, h1 = var.pull(k1)
+    var.push(h0, g0)
+    var.push(h1, g1)
+    st.update_weights()
+
+    r0 = torch.load(tmp_path / "result_0.pt", weights_only=True)
+    r1 = torch.load(tmp_path / "result_1.pt", weights_only=True)
+    # pulls before update: identical deterministic init
+    torch.testing.assert_close(r0["out"], out_ref0)
+    torch.testing.assert_close(r1["out"], out_ref1)
+    # after the commit: same table state
+    after0, _ = var.pull(r0["keys"], readonly=True)
+    after1, _ = var.pull(r1["keys"], readonly=True)
+    torch.testing.assert_close(r0["out2"], after0, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(r1["out2"], after1, rtol=1e-5, atol=1e-5)
+
+    # cross-shard-count reload: world-2 checkpoint into this world-1 context
+    ctx2 = Context(device="cpu")
+    st2 = ctx2.create_storage()
+    var2 = st2.create_variable(VOCAB, DIM)
+    var2.set_optimizer("test")
+    checkpoint.load_model(ctx2, str(tmp_path / "ckpt"))
+    loaded, _ = var2.pull(r0["keys"], readonly=True)
+    torch.testing.assert_close(loaded, after0, rtol=1e-5, atol=1e-5)
+
+
+def _worker_model(rank, world, port, tmp):
+    import openembedding_amd.torch as embed
+    from openembedding_amd.models import DeepFM, synthetic_batch
+
+    _init(rank, world, port)
+    torch.manual_seed(0)  # same dense init on every rank
+    fv = [50, 3, 1000, 40] + [100] * 22
+    model = DeepFM(field_vocabs=fv, dim=4)
+    opt = embed.distributed_optimizer(
+        torch.optim.Adagrad(model.parameters(), lr=0.01))
+    lossf = torch.nn.BCEWithLogitsLoss()
+    g = torch.Generator().manual_seed(10 + rank)
+    for step in range(3):
+        dense, sparse, labels = synthetic_batch(64, field_vocabs=fv,
+                                                generator=g)
+        opt.zero_grad()
+        loss = lossf(model(dense, sparse), labels)
+        loss.backward()
+        opt.step()
+    # dense params must remain identical across ranks after allreduce steps
+    p = torch.cat([q.detach().reshape(-1) for q in model.dnn.parameters()])
+    gathered = [torch.empty_like(p) for _ in range(world)]
+    dist.all_gather(gathered, p)
+    assert torch.allclose(gathered[0], gathered[1], rtol=1e-6, atol=1e-6)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_deepfm_world2(tmp_path):
+    port = _free_port()
+    mp.spawn(_worker_model, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+    for v in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
+        os.environ.pop(v, None)
