@@ -1,0 +1,136 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: DeepFM on Criteo-shaped synthetic data.
+
+Measures the BASELINE.json metric — samples/sec for DeepFM (embedding dim 9,
+Adagrad, batch 4096 per GPU, Criteo Kaggle field cardinalities) — on
+1..8 MI355X GPUs, one rank per GPU over RCCL.
+
+    python bench.py --gpus N --steps K --warmup W
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N ...
+
+Rank 0 prints one JSON line. Baseline: the reference's published DeepFM dim9
+throughput on 8x Tesla T4: 293/458/727/935 k samples/s at
+1/2/4/8 GPUs.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+# reference OE DeepFM dim9 samples/s at 1/2/4/8 GPUs
+BASELINE_SAMPLES_PER_SEC = {1: 293_000.0, 2: 458_000.0, 4: 727_000.0,
+                            8: 935_000.0}
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=4096, help="per-GPU batch")
+    p.add_argument("--model", default="deepfm",
+                   choices=["deepfm", "wdl", "xdeepfm", "lr"])
+    p.add_argument("--dim", type=int, default=9)
+    p.add_argument("--data-pool", type=int, default=8,
+                   help="pre-generated synthetic batches, rotated")
+    args = p.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+
+    import openembedding_amd.torch as embed
+    from openembedding_amd.models import MODELS, synthetic_batch
+
+    ctx = embed.get_context()
+    device = ctx.device
+    on_gpu = device.type == "cuda"
+    if on_gpu:
+        from openembedding_amd.ops import require_hip
+        require_hip()  # fail loudly if the native extension is missing
+
+    torch.manual_seed(1234)  # identical dense init on all ranks
+    kw = {} if args.model == "lr" else {"dim": args.dim}
+    model = MODELS[args.model](**kw).to(device)
+    opt = embed.distributed_optimizer(
+        torch.optim.Adagrad(model.parameters(), lr=0.005))
+    lossf = torch.nn.BCEWithLogitsLoss()
+
+    gen = torch.Generator(device="cpu").manual_seed(4242 + rank)
+    pool = []
+    for _ in range(args.data_pool):
+        dense, sparse, labels = synthetic_batch(args.batch, generator=gen)
+        pool.append((dense.to(device), sparse.to(device), labels.to(device)))
+
+    def step(i):
+        dense, sparse, labels = pool[i % len(pool)]
+        opt.zero_grad()
+        loss = lossf(model(dense, sparse), labels)
+        loss.backward()
+        opt.step()
+        return loss
+
+    for i in range(args.warmup):
+        step(i)
+
+    if dist.is_initialized():
+        dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max across ranks = whole-job time
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if dist.is_initialized()
+                     and dist.get_backend() == "nccl" else "cpu")
+    if dist.is_initialized():
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    n_gpus = world if world > 1 else args.gpus
+    ms_per_step = elapsed / args.steps * 1000.0
+    samples_per_sec = args.batch * n_gpus * args.steps / elapsed
+    baseline = BASELINE_SAMPLES_PER_SEC.get(n_gpus)
+    result = {
+        "metric": "samples/sec DeepFM Criteo",
+        "value": samples_per_sec,
+        "unit": "samples/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": ms_per_step,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": (samples_per_sec / baseline) if (
+            baseline and args.model == "deepfm" and args.dim == 9
+            and args.batch == 4096) else None,
+        "dtype": "fp32",
+        "data": "synthetic Criteo-shaped (random ids, Criteo-Kaggle "
+                "cardinalities, random labels), random-init weights",
+        "config": {
+            "model": f"{args.model}-dim{args.dim}",
+            "global_batch": args.batch * n_gpus,
+            "fields": 26,
+            "dense_features": 13,
+            "optimizer": "adagrad",
+            "parallelism": (f"dense-dp{n_gpus} + embedding sharded "
+                            f"all_to_all" if n_gpus > 1 else "single-gpu"),
+        },
+    }
+    if rank == 0:
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()
