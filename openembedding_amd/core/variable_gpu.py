@@ -83,6 +83,7 @@ class HipVariableShard(VariableShard):
             raise NotImplementedError("GPU shards are float32 (reference default); "
                                       "float64 runs on the CPU backend")
         super().__init__(meta, shard_id, shard_num, device, seed)
+        self._pending_bounded = []
         dev = self.device
         if meta.use_hash_table:
             self._cap = self.INITIAL_TABLE_CAP
@@ -132,32 +133,34 @@ class HipVariableShard(VariableShard):
 
     # -------------------------------------------------------------- lookups
 
-    def _lookup_or_insert(self, keys: torch.Tensor):
+    def _lookup_or_insert(self, keys: torch.Tensor, u_dev=None):
         n = keys.numel()
         if self.meta.use_hash_table:
             self._ensure_rows(self._nrows_upper + n)
             self._maybe_rehash()
             slots, new_mask = self.ext.ht_lookup(self.tk, self.tv, keys,
                                                  self.nrows_dev, self.slot_keys,
-                                                 True)
+                                                 True, u_dev)
             self._nrows_upper += n
         else:
-            slots = torch.div(keys, self.shard_num, rounding_mode="floor")
-            new_mask, _ = self.ext.array_touch(self.valid_u8, slots)
+            slots, new_mask = self.ext.array_touch(self.valid_u8, keys,
+                                                   self.shard_num, u_dev)
         return slots, new_mask
 
     def _lookup_readonly(self, keys: torch.Tensor) -> torch.Tensor:
         if self.meta.use_hash_table:
             slots, _ = self.ext.ht_lookup(self.tk, self.tv, keys,
-                                          self.nrows_dev, self.slot_keys, False)
+                                          self.nrows_dev, self.slot_keys,
+                                          False, None)
             return slots
         slots = torch.div(keys, self.shard_num, rounding_mode="floor")
-        hit = self.valid_u8[slots] != 0
+        hit = self.valid_u8[slots.clamp(0, self._array_cap - 1)] != 0
         return torch.where(hit, slots, torch.full_like(slots, -1))
 
     # ------------------------------------------------------------- hot paths
 
-    def _gather(self, keys, slots, new_mask, want_out: bool) -> torch.Tensor:
+    def _gather(self, keys, slots, new_mask, want_out: bool,
+                inverse=None, u_dev=None) -> torch.Tensor:
         cat, p0, p1, p2 = _init_params(self.initializer)
         sir = self._state_init_row
         if sir is None:
@@ -167,50 +170,79 @@ class HipVariableShard(VariableShard):
                                     new_mask if new_mask is not None
                                     else torch.empty(0, dtype=torch.uint8,
                                                      device=self.device),
-                                    keys, cat, p0, p1, p2, self.seed,
-                                    sir.reshape(-1), want_out)
+                                    keys, inverse, cat, p0, p1, p2, self.seed,
+                                    sir.reshape(-1), want_out, u_dev)
 
     def pull(self, keys: torch.Tensor) -> torch.Tensor:
         slots, new_mask = self._lookup_or_insert(keys)
         return self._gather(keys, slots, new_mask, True)
 
+    def pull_bounded(self, keys_buf: torch.Tensor, u_dev: torch.Tensor,
+                     inverse: torch.Tensor):
+        """Sync-free pull: keys_buf is an n-sized unique buffer with the live
+        count in u_dev (device int32); the gather iterates the full inverse
+        and fuses the duplicate scatter. Returns (out [n_elems, dim], slots)."""
+        slots, new_mask = self._lookup_or_insert(keys_buf, u_dev)
+        out = self._gather(keys_buf, slots, new_mask, True,
+                           inverse=inverse, u_dev=u_dev)
+        return out, slots
+
+    def push_slots(self, keys_buf, u_dev, slots, grads, counts) -> None:
+        """Queue a bounded pre-reduced block whose table slots are already
+        known (saved from the matching pull)."""
+        self._pending_bounded.append((keys_buf, u_dev, slots, grads, counts))
+
     def pull_readonly(self, keys: torch.Tensor) -> torch.Tensor:
         slots = self._lookup_readonly(keys)
         empty = torch.empty(0, dtype=torch.uint8, device=self.device)
         return self.ext.gather_init(self.weights, self.state, slots, empty,
-                                    keys, 0, 0.0, 0.0, 0.0, self.seed,
+                                    keys, None, 0, 0.0, 0.0, 0.0, self.seed,
                                     torch.empty(0, dtype=self.dtype,
-                                                device=self.device), True)
+                                                device=self.device), True,
+                                    None)
 
     def update_weights(self) -> None:
-        if not self._pending:
+        if not self._pending and not self._pending_bounded:
             return
         if self.optimizer is None:
             raise RuntimeError("update_weights called before set_optimizer")
-        if len(self._pending) == 1:
-            keys, grads, counts = self._pending[0]
-        else:
-            keys0 = torch.cat([b[0] for b in self._pending])
-            grads0 = torch.cat([b[1] for b in self._pending])
-            counts0 = torch.cat([b[2] for b in self._pending])
-            keys, inv = self.ext.unique_inverse(keys0)
-            grads, _ = self.ext.reduce_by_inverse(inv, grads0, keys.numel())
-            counts = torch.zeros(keys.numel(), dtype=counts0.dtype,
-                                 device=self.device)
-            counts.index_add_(0, inv, counts0)
-        self._pending = []
-        slots, new_mask = self._lookup_or_insert(keys)
-        # init rows that were pushed without a prior pull
-        self._gather(keys, slots, new_mask, False)
-        self.ext.apply_optimizer(_OPT_IDS[self.optimizer.category],
-                                 self.weights, self.state, slots,
-                                 grads.contiguous(), counts,
-                                 _opt_cfg_vector(self.optimizer))
-        if self.meta.use_hash_table:
+        opt_id = _OPT_IDS[self.optimizer.category]
+        cfg = _opt_cfg_vector(self.optimizer)
+        # bounded blocks: slots known from pull, zero-sync apply
+        for keys_buf, u_dev, slots, grads, counts in self._pending_bounded:
+            self.ext.apply_optimizer(opt_id, self.weights, self.state, slots,
+                                     grads.contiguous(), counts, cfg, u_dev)
+        self._pending_bounded = []
+        if self._pending:
+            if len(self._pending) == 1:
+                keys, grads, counts = self._pending[0]
+            else:
+                keys0 = torch.cat([b[0] for b in self._pending])
+                grads0 = torch.cat([b[1] for b in self._pending])
+                counts0 = torch.cat([b[2] for b in self._pending])
+                keys, inv = self.ext.unique_inverse(keys0)
+                grads, _ = self.ext.reduce_by_inverse(inv, grads0, keys.numel())
+                counts = torch.zeros(keys.numel(), dtype=counts0.dtype,
+                                     device=self.device)
+                counts.index_add_(0, inv, counts0)
+            self._pending = []
+            slots, new_mask = self._lookup_or_insert(keys)
+            # init rows that were pushed without a prior pull
+            self._gather(keys, slots, new_mask, False)
+            self.ext.apply_optimizer(opt_id, self.weights, self.state, slots,
+                                     grads.contiguous(), counts, cfg, None)
+        if self.meta.use_hash_table and not self._in_graph_capture():
             # once per committed batch: tighten the row-count bound (one small
             # D2H read; the only sync of the commit path)
             self._nrows_exact = int(self.nrows_dev.item())
             self._nrows_upper = self._nrows_exact
+
+    @staticmethod
+    def _in_graph_capture() -> bool:
+        try:
+            return torch.cuda.is_current_stream_capturing()
+        except Exception:
+            return False
 
     # ----------------------------------------------------------- checkpoint
 

@@ -31,21 +31,32 @@ def _mlp(in_dim: int, hidden: Sequence[int], out_dim: int = 1) -> nn.Sequential:
 
 
 class _CTRBase(nn.Module):
+    """Shared CTR skeleton.
+
+    MI355X-first layout: the per-key first-order ("linear"/wide) weight is
+    stored as an EXTRA COLUMN of the same PS variable (rows are dim+1 wide)
+    instead of the reference's separate dim-1 Embedding — one fused
+    unique/all_to_all/gather per step instead of two, with identical
+    per-element optimizer math (all sparse optimizers are element-wise; the
+    per-row scalar states advance on the same touch pattern either way)."""
+
     def __init__(self, field_vocabs: Optional[List[int]], dim: int,
                  sparse_as_dense_size: int = 0):
         super().__init__()
         self.field_vocabs = list(field_vocabs or CRITEO_FIELD_VOCABS)
         self.dim = dim
         self.n_fields = len(self.field_vocabs)
-        self.embedding = CombinedEmbedding(self.field_vocabs, dim)
-        # first-order ("linear"/wide) weights: dim-1 embedding per field
-        self.linear_embedding = CombinedEmbedding(self.field_vocabs, 1)
+        self.embedding = CombinedEmbedding(self.field_vocabs, dim + 1)
         self.dense_linear = nn.Linear(N_DENSE, 1)
 
-    def _first_order(self, dense: torch.Tensor, sparse: torch.Tensor
+    def _embed(self, sparse: torch.Tensor):
+        """-> (e [B,F,dim], linear_w [B,F])."""
+        e_all = self.embedding(sparse)
+        return e_all[..., :self.dim], e_all[..., self.dim]
+
+    def _first_order(self, dense: torch.Tensor, linear_w: torch.Tensor
                      ) -> torch.Tensor:
-        lin = self.linear_embedding(sparse).sum(dim=(1, 2))
-        return lin + self.dense_linear(dense).squeeze(-1)
+        return linear_w.sum(dim=1) + self.dense_linear(dense).squeeze(-1)
 
 
 class LR(_CTRBase):
@@ -53,10 +64,11 @@ class LR(_CTRBase):
     subclassed model, hash-mode embedding dim 1)."""
 
     def __init__(self, field_vocabs: Optional[List[int]] = None):
-        super().__init__(field_vocabs, dim=1)
+        super().__init__(field_vocabs, dim=0)
 
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
-        return self._first_order(dense, sparse)
+        _, lin = self._embed(sparse)
+        return self._first_order(dense, lin)
 
 
 class WDL(_CTRBase):
@@ -68,9 +80,9 @@ class WDL(_CTRBase):
         self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
 
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
-        e = self.embedding(sparse)                       # [B, F, d]
+        e, lin = self._embed(sparse)                     # [B, F, d], [B, F]
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
-        return self._first_order(dense, sparse) + self.dnn(deep_in).squeeze(-1)
+        return self._first_order(dense, lin) + self.dnn(deep_in).squeeze(-1)
 
 
 class DeepFM(_CTRBase):
@@ -82,12 +94,12 @@ class DeepFM(_CTRBase):
         self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
 
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
-        e = self.embedding(sparse)                       # [B, F, d]
+        e, lin = self._embed(sparse)                     # [B, F, d], [B, F]
         # FM second order: 0.5*((sum_f e)^2 - sum_f e^2) summed over dim
         s = e.sum(dim=1)
         fm2 = 0.5 * (s * s - (e * e).sum(dim=1)).sum(dim=1)
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
-        return (self._first_order(dense, sparse) + fm2
+        return (self._first_order(dense, lin) + fm2
                 + self.dnn(deep_in).squeeze(-1))
 
 
@@ -129,9 +141,9 @@ class xDeepFM(_CTRBase):
         self.cin = CIN(self.n_fields, dim, cin_layers)
 
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
-        e = self.embedding(sparse)
+        e, lin = self._embed(sparse)
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
-        return (self._first_order(dense, sparse) + self.cin(e)
+        return (self._first_order(dense, lin) + self.cin(e)
                 + self.dnn(deep_in).squeeze(-1))
 
 

@@ -1,5 +1,13 @@
 // Torch bindings for the openembedding_amd CDNA4 kernels (embops.hip).
 // Tensor-level API consumed by ops/dispatch.py and core/variable_gpu.py.
+//
+// Two calling modes:
+//   exact   — unique_inverse() syncs once to return a tightly-sized unique
+//             tensor (needed for the multi-GPU all_to_all splits anyway);
+//   bounded — unique_bounded() returns an n-sized buffer + device count and
+//             every downstream kernel takes the count as a device pointer,
+//             so a whole train step runs with ZERO host syncs (and is
+//             hipGraph-capturable).
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>  // ROCm torch masquerades HIP as CUDA here
@@ -19,21 +27,25 @@ extern "C" {
 void emb_unique(const i64*, long, u64*, int*, long, int*, unsigned char*,
                 i64*, i64*, int*, hipStream_t_);
 void emb_ht_lookup(u64*, int*, long, const i64*, long, int*, i64*, i64*,
-                   unsigned char*, int, hipStream_t_);
+                   unsigned char*, int, const int*, hipStream_t_);
 void emb_ht_rehash(const u64*, const int*, long, u64*, int*, long,
                    hipStream_t_);
-void emb_array_touch(unsigned char*, const i64*, long, unsigned char*, int*,
-                     hipStream_t_);
+void emb_array_touch(unsigned char*, const i64*, long, long, long, i64*,
+                     unsigned char*, const int*, hipStream_t_);
 void emb_gather_init(float*, float*, long, long, const i64*,
-                     const unsigned char*, const i64*, long, float*, int,
-                     float, float, float, u64, const float*, hipStream_t_);
+                     const unsigned char*, const i64*, long, const i64*,
+                     float*, int, float, float, float, u64, const float*,
+                     const int*, hipStream_t_);
 void emb_reduce_by_inverse(const i64*, const float*, long, long, float*, u64*,
                            long, hipStream_t_);
 void emb_apply_optimizer(int, float*, float*, long, long, const i64*, long,
-                         const float*, const u64*, const float*, hipStream_t_);
+                         const float*, const u64*, const float*, const int*,
+                         hipStream_t_);
 }
 
 namespace {
+
+using OptTensor = c10::optional<torch::Tensor>;
 
 hipStream_t_ cur_stream() {
     return reinterpret_cast<hipStream_t_>(
@@ -46,10 +58,19 @@ long next_pow2(long x) {
     return p;
 }
 
+const int* u_ptr(const OptTensor& u_dev) {
+    if (!u_dev.has_value()) return nullptr;
+    TORCH_CHECK(u_dev->dtype() == torch::kInt32, "u_dev must be int32");
+    return u_dev->data_ptr<int>();
+}
+
 #define CHECK_GPU(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
 #define CHECK_CONT(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
 
-std::tuple<torch::Tensor, torch::Tensor> unique_inverse(torch::Tensor keys) {
+// ---- unique ----------------------------------------------------------
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> unique_bounded(
+    torch::Tensor keys) {
     CHECK_GPU(keys); CHECK_CONT(keys);
     TORCH_CHECK(keys.dtype() == torch::kInt64);
     const c10::cuda::CUDAGuard guard(keys.device());
@@ -62,20 +83,28 @@ std::tuple<torch::Tensor, torch::Tensor> unique_inverse(torch::Tensor keys) {
     auto tv = torch::empty({cap}, opts_i32);
     auto slot_of = torch::empty({n}, opts_i32);
     auto is_first = torch::empty({n}, opts_u8);
-    auto uk = torch::empty({n}, opts_i64);
+    auto uk = torch::zeros({n}, opts_i64);
     auto inverse = torch::empty({n}, opts_i64);
-    auto counter = torch::zeros({1}, opts_i32);
+    auto counter = torch::empty({1}, opts_i32);
     emb_unique(keys.data_ptr<i64>(), n, (u64*)tk.data_ptr<i64>(),
                tv.data_ptr<int>(), cap, slot_of.data_ptr<int>(),
                is_first.data_ptr<uint8_t>(), uk.data_ptr<i64>(),
                inverse.data_ptr<i64>(), counter.data_ptr<int>(), cur_stream());
-    long u = counter.item<int>();  // host sync (bounded-buffer mode avoids it)
-    return {uk.narrow(0, 0, u), inverse};
+    return {uk, inverse, counter};
 }
+
+std::tuple<torch::Tensor, torch::Tensor> unique_inverse(torch::Tensor keys) {
+    auto r = unique_bounded(keys);
+    long u = std::get<2>(r).item<int>();  // host sync
+    return {std::get<0>(r).narrow(0, 0, u), std::get<1>(r)};
+}
+
+// ---- persistent hash table -------------------------------------------
 
 std::tuple<torch::Tensor, torch::Tensor> ht_lookup(
     torch::Tensor tk, torch::Tensor tv, torch::Tensor keys,
-    torch::Tensor nrows, torch::Tensor slot_keys, bool insert) {
+    torch::Tensor nrows, torch::Tensor slot_keys, bool insert,
+    OptTensor u_dev) {
     CHECK_GPU(keys); CHECK_CONT(keys);
     const c10::cuda::CUDAGuard guard(keys.device());
     long n = keys.numel();
@@ -86,7 +115,7 @@ std::tuple<torch::Tensor, torch::Tensor> ht_lookup(
                       tk.numel(), keys.data_ptr<i64>(), n,
                       nrows.data_ptr<int>(), slot_keys.data_ptr<i64>(),
                       slots.data_ptr<i64>(), new_mask.data_ptr<uint8_t>(),
-                      insert ? 1 : 0, cur_stream());
+                      insert ? 1 : 0, u_ptr(u_dev), cur_stream());
     return {slots, new_mask};
 }
 
@@ -98,27 +127,34 @@ void ht_rehash(torch::Tensor tk_old, torch::Tensor tv_old, torch::Tensor tk_new,
                   tv_new.data_ptr<int>(), tk_new.numel(), cur_stream());
 }
 
-std::tuple<torch::Tensor, torch::Tensor> array_touch(torch::Tensor valid,
-                                                     torch::Tensor slots) {
+// ---- array table ------------------------------------------------------
+
+std::tuple<torch::Tensor, torch::Tensor> array_touch(
+    torch::Tensor valid, torch::Tensor keys, int64_t shard_num,
+    OptTensor u_dev) {
     const c10::cuda::CUDAGuard guard(valid.device());
-    long n = slots.numel();
-    auto new_mask = torch::zeros({n}, slots.options().dtype(torch::kUInt8));
-    auto n_new = torch::zeros({1}, slots.options().dtype(torch::kInt32));
+    long n = keys.numel();
+    auto slots = torch::empty({n}, keys.options());
+    auto new_mask = torch::zeros({n}, keys.options().dtype(torch::kUInt8));
     if (n)
-        emb_array_touch(valid.data_ptr<uint8_t>(), slots.data_ptr<i64>(), n,
-                        new_mask.data_ptr<uint8_t>(), n_new.data_ptr<int>(),
+        emb_array_touch(valid.data_ptr<uint8_t>(), keys.data_ptr<i64>(), n,
+                        shard_num, valid.numel(), slots.data_ptr<i64>(),
+                        new_mask.data_ptr<uint8_t>(), u_ptr(u_dev),
                         cur_stream());
-    return {new_mask, n_new};
+    return {slots, new_mask};
 }
+
+// ---- gather + lazy init ----------------------------------------------
 
 torch::Tensor gather_init(torch::Tensor weights, torch::Tensor state,
                           torch::Tensor slots, torch::Tensor new_mask,
-                          torch::Tensor keys, int64_t init_cat, double p0,
-                          double p1, double p2, int64_t seed,
-                          torch::Tensor state_init_row, bool want_out) {
+                          torch::Tensor keys, OptTensor inverse,
+                          int64_t init_cat, double p0, double p1, double p2,
+                          int64_t seed, torch::Tensor state_init_row,
+                          bool want_out, OptTensor u_dev) {
     CHECK_GPU(weights); CHECK_CONT(weights);
     const c10::cuda::CUDAGuard guard(weights.device());
-    long n = keys.numel();
+    long n = inverse.has_value() ? inverse->numel() : keys.numel();
     long dim = weights.size(1);
     long sd = state.numel() ? state.size(1) : 0;
     torch::Tensor out;
@@ -133,12 +169,16 @@ torch::Tensor gather_init(torch::Tensor weights, torch::Tensor state,
                     sd ? state.data_ptr<float>() : nullptr, dim, sd,
                     slots.data_ptr<i64>(),
                     new_mask.numel() ? new_mask.data_ptr<uint8_t>() : nullptr,
-                    keys.data_ptr<i64>(), n, out_ptr, (int)init_cat,
+                    keys.data_ptr<i64>(), n,
+                    inverse.has_value() ? inverse->data_ptr<i64>() : nullptr,
+                    out_ptr, (int)init_cat,
                     (float)p0, (float)p1, (float)p2, (u64)seed,
                     sd ? state_init_row.data_ptr<float>() : nullptr,
-                    cur_stream());
+                    u_ptr(u_dev), cur_stream());
     return out;
 }
+
+// ---- reduce-by-key ----------------------------------------------------
 
 std::tuple<torch::Tensor, torch::Tensor> reduce_by_inverse(
     torch::Tensor inverse, torch::Tensor grads, int64_t u) {
@@ -156,9 +196,12 @@ std::tuple<torch::Tensor, torch::Tensor> reduce_by_inverse(
     return {ugrads, counts};
 }
 
+// ---- fused optimizers --------------------------------------------------
+
 void apply_optimizer(int64_t opt, torch::Tensor weights, torch::Tensor state,
                      torch::Tensor slots, torch::Tensor grads,
-                     torch::Tensor counts, std::vector<double> cfg) {
+                     torch::Tensor counts, std::vector<double> cfg,
+                     OptTensor u_dev) {
     CHECK_GPU(weights); CHECK_CONT(weights); CHECK_CONT(grads);
     const c10::cuda::CUDAGuard guard(weights.device());
     long n = slots.numel();
@@ -169,17 +212,23 @@ void apply_optimizer(int64_t opt, torch::Tensor weights, torch::Tensor state,
     emb_apply_optimizer((int)opt, weights.data_ptr<float>(),
                         sd ? state.data_ptr<float>() : nullptr, dim, sd,
                         slots.data_ptr<i64>(), n, grads.data_ptr<float>(),
-                        (const u64*)counts.data_ptr<i64>(), c, cur_stream());
+                        (const u64*)counts.data_ptr<i64>(), c, u_ptr(u_dev),
+                        cur_stream());
 }
 
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("unique_inverse", &unique_inverse, "hash-based unique+inverse");
+    m.def("unique_bounded", &unique_bounded,
+          "unique+inverse without host sync (n-sized buffer + device count)");
     m.def("ht_lookup", &ht_lookup, "hash table lookup/insert");
     m.def("ht_rehash", &ht_rehash, "hash table rehash into larger table");
-    m.def("array_touch", &array_touch, "array-table valid-bitmap touch");
-    m.def("gather_init", &gather_init, "row gather with fused lazy init");
-    m.def("reduce_by_inverse", &reduce_by_inverse, "grad reduce-by-key");
+    m.def("array_touch", &array_touch,
+          "array-table slot compute + valid-bitmap touch");
+    m.def("gather_init", &gather_init,
+          "row gather with fused lazy init (+ optional duplicate scatter)");
+    m.def("reduce_by_inverse", &reduce_by_inverse,
+          "grad reduce-by-key with counts (LDS-aggregated)");
     m.def("apply_optimizer", &apply_optimizer, "fused sparse optimizer step");
 }

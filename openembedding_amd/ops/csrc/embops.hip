@@ -90,26 +90,58 @@ DEV float init_value(int cat, float p0, float p1, float p2,
 
 // ------------------------------------------------------- unique + inverse
 // Batch-local dedup via a scratch open-addressed table (3 passes, no spin).
+//
+// Contention design: CTR batches contain extremely hot keys (a vocab-3 field
+// contributes batch_size copies of 3 keys), so a naive per-element CAS
+// serializes thousands of atomics on one cache line (measured 24us for a
+// 106k batch). An LDS pre-filter dedups per block first: one block-winner
+// probes the global table and publishes the slot through LDS; hot-key global
+// CAS count drops from O(batch) to O(blocks).
+
+#define ULDS 1024  // LDS pre-filter entries (8 KiB keys + 4 KiB slots)
 
 __global__ void k_unique_insert(const i64* __restrict__ keys, long n,
                                 u64* __restrict__ tk, long mask,
                                 int* __restrict__ slot_of,
                                 unsigned char* __restrict__ is_first) {
+    __shared__ u64 lkeys[ULDS];
+    __shared__ int lslot[ULDS];
+    for (int t = threadIdx.x; t < ULDS; t += blockDim.x) lkeys[t] = EMPTY;
+    __syncthreads();
+
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n) return;
-    u64 k = (u64)keys[i];
-    u64 h = splitmix64(k) & (u64)mask;
-    for (;;) {
-        u64 cur = tk[h];
-        if (cur == k) break;
-        if (cur == EMPTY) {
-            u64 prev = atomicCAS(&tk[h], EMPTY, k);
-            if (prev == EMPTY) { is_first[i] = 1; break; }
-            if (prev == k) break;
+    u64 k = 0;
+    int lh = -1;           // LDS slot of this element's key
+    bool lds_winner = false, lds_ok = false;
+    if (i < n) {
+        k = (u64)keys[i];
+        u64 hh = splitmix64(k);
+        lh = (int)(hh & (ULDS - 1));
+        for (int probes = 0; probes < 64; ++probes) {
+            u64 cur = atomicCAS(&lkeys[lh], EMPTY, k);
+            if (cur == EMPTY) { lds_winner = true; lds_ok = true; break; }
+            if (cur == k) { lds_ok = true; break; }
+            lh = (lh + 1) & (ULDS - 1);
         }
-        h = (h + 1) & (u64)mask;
+        if (lds_winner || !lds_ok) {
+            // block winner (or LDS overflow): probe the global table
+            u64 h = splitmix64(k) & (u64)mask;
+            for (;;) {
+                u64 cur = tk[h];
+                if (cur == k) break;
+                if (cur == EMPTY) {
+                    u64 prev = atomicCAS(&tk[h], EMPTY, k);
+                    if (prev == EMPTY) { is_first[i] = 1; break; }
+                    if (prev == k) break;
+                }
+                h = (h + 1) & (u64)mask;
+            }
+            slot_of[i] = (int)h;
+            if (lds_winner) lslot[lh] = (int)h;
+        }
     }
-    slot_of[i] = (int)h;
+    __syncthreads();
+    if (i < n && lds_ok && !lds_winner) slot_of[i] = lslot[lh];
 }
 
 __global__ void k_unique_assign(const i64* __restrict__ keys, long n,
@@ -119,11 +151,22 @@ __global__ void k_unique_assign(const i64* __restrict__ keys, long n,
                                 i64* __restrict__ unique_keys,
                                 int* __restrict__ counter) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n) return;
-    if (!is_first[i]) return;
-    int uid = atomicAdd(counter, 1);
-    tv[slot_of[i]] = uid;
-    unique_keys[uid] = keys[i];
+    // wave-aggregated counter: one atomic per wave, not per first-occurrence
+    bool first = (i < n) && is_first[i];
+    u64 ballot = __ballot(first);
+    if (!ballot) return;
+    int lane = threadIdx.x & 63;
+    int leader = __ffsll((unsigned long long)ballot) - 1;
+    int base = 0;
+    if (lane == leader)
+        base = atomicAdd(counter, __popcll(ballot));
+    base = __shfl(base, leader);
+    if (first) {
+        int rank = __popcll(ballot & ((1ull << lane) - 1ull));
+        int uid = base + rank;
+        tv[slot_of[i]] = uid;
+        unique_keys[uid] = keys[i];
+    }
 }
 
 __global__ void k_unique_inverse(long n, const int* __restrict__ slot_of,
@@ -146,9 +189,10 @@ __global__ void k_ht_lookup(u64* __restrict__ tk, int* __restrict__ tv,
                             i64* __restrict__ slot_keys,
                             i64* __restrict__ slots,
                             unsigned char* __restrict__ new_mask,
-                            int insert) {
+                            int insert, const int* __restrict__ u_dev) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
+    if (u_dev && i >= *u_dev) return;
     u64 k = (u64)keys[i];
     u64 h = splitmix64(k) & (u64)mask;
     i64 slot = -1;
@@ -192,19 +236,26 @@ __global__ void k_ht_rehash(const u64* __restrict__ tk_old,
 }
 
 // ------------------------------------------------------ array-table touch
-// slots unique within a call; marks rows live and reports which were new.
+// keys unique within a call; computes local slots (key / shard_num, the
+// reference's layout EmbeddingShardFile.h:23-25), marks rows live and
+// reports which were new.
 
 __global__ void k_array_touch(unsigned char* __restrict__ valid,
-                              const i64* __restrict__ slots, long n,
+                              const i64* __restrict__ keys, long n,
+                              long shard_num, long cap,
+                              i64* __restrict__ slots,
                               unsigned char* __restrict__ new_mask,
-                              int* __restrict__ n_new) {
+                              const int* __restrict__ u_dev) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    i64 s = slots[i];
+    if (u_dev && i >= *u_dev) { slots[i] = -1; return; }
+    i64 s = keys[i] / shard_num;
+    if (s < 0 || s >= cap) s = -1;  // guarded; host validates separately
+    slots[i] = s;
+    if (s < 0) { new_mask[i] = 0; return; }
     unsigned char was = valid[s];
     valid[s] = 1;
     new_mask[i] = !was;
-    if (!was) atomicAdd(n_new, 1);
 }
 
 // -------------------------------------------------- gather + lazy row init
@@ -212,6 +263,11 @@ __global__ void k_array_touch(unsigned char* __restrict__ valid,
 // (weights from the deterministic initializer, optimizer state copied from
 // the host-prepared state_init_row) and simultaneously returned.
 
+// When ``inverse`` is given the kernel iterates the FULL (duplicated)
+// element list and fuses the scatter-to-duplicates (the reference's client
+// response scatter, EmbeddingPullOperator.cpp:229-249) into the gather: all
+// duplicates of a new key recompute the same deterministic init value, so
+// the racy-looking table writes are benign (same bytes).
 template <int G>
 __global__ void k_gather_init(float* __restrict__ weights,
                               float* __restrict__ state,
@@ -219,22 +275,26 @@ __global__ void k_gather_init(float* __restrict__ weights,
                               const i64* __restrict__ slots,
                               const unsigned char* __restrict__ new_mask,
                               const i64* __restrict__ keys, long n,
+                              const i64* __restrict__ inverse,
                               float* __restrict__ out,
                               int init_cat, float p0, float p1, float p2,
                               u64 seed,
-                              const float* __restrict__ state_init_row) {
+                              const float* __restrict__ state_init_row,
+                              const int* __restrict__ u_dev) {
     long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) / G;
     int lane = threadIdx.x % G;
     if (g >= n) return;
-    i64 slot = slots[g];
+    long uid = inverse ? inverse[g] : g;
+    if (!inverse && u_dev && g >= *u_dev) return;
+    i64 slot = slots[uid];
     if (slot < 0) {  // read-only miss: zeros out
         if (out)
             for (long j = lane; j < dim; j += G) out[g * dim + j] = 0.0f;
         return;
     }
     float* wrow = weights + (u64)slot * dim;
-    if (new_mask && new_mask[g]) {
-        u64 key = (u64)keys[g];
+    if (new_mask && new_mask[uid]) {
+        u64 key = (u64)keys[uid];
         for (long j = lane; j < dim; j += G) {
             float v = init_value(init_cat, p0, p1, p2, seed, key, (u64)j);
             wrow[j] = v;
@@ -252,6 +312,62 @@ __global__ void k_gather_init(float* __restrict__ weights,
 }
 
 // -------------------------------------------------------- reduce-by-inverse
+// Hot rows (small-vocab fields) make direct global atomics serialize; for
+// dim <= 32 gradients+counts are pre-aggregated in an LDS hash per block
+// (one global atomic per distinct uid per block), fused with the counts.
+
+#define RED_ITERS 8
+
+template <int H>
+__global__ void k_reduce_lds(const i64* __restrict__ inverse,
+                             const float* __restrict__ grads,
+                             long n, long dim,
+                             float* __restrict__ ugrads,
+                             u64* __restrict__ counts) {
+    extern __shared__ char smem[];
+    int* luid = (int*)smem;
+    int* lcnt = (int*)(smem + H * 4);
+    float* lacc = (float*)(smem + H * 8);
+    for (int t = threadIdx.x; t < H; t += blockDim.x) {
+        luid[t] = -1;
+        lcnt[t] = 0;
+    }
+    for (long t = threadIdx.x; t < (long)H * dim; t += blockDim.x)
+        lacc[t] = 0.0f;
+    __syncthreads();
+    long base = (long)blockIdx.x * blockDim.x * RED_ITERS;
+    for (int it = 0; it < RED_ITERS; ++it) {
+        long i = base + (long)it * blockDim.x + threadIdx.x;
+        if (i >= n) break;
+        int uid = (int)inverse[i];
+        int h = (int)(((unsigned)uid * 2654435761u) & (H - 1));
+        bool got = false;
+        for (int probes = 0; probes < H; ++probes) {
+            int cur = atomicCAS(&luid[h], -1, uid);
+            if (cur == -1 || cur == uid) { got = true; break; }
+            h = (h + 1) & (H - 1);
+        }
+        const float* g = grads + (u64)i * dim;
+        if (got) {
+            atomicAdd(&lcnt[h], 1);
+            float* acc = lacc + (u64)h * dim;
+            for (long j = 0; j < dim; ++j) atomicAdd(&acc[j], g[j]);
+        } else {  // LDS table full: fall through to global atomics
+            atomicAdd(&counts[uid], 1ull);
+            float* ug = ugrads + (u64)uid * dim;
+            for (long j = 0; j < dim; ++j) atomicAdd(&ug[j], g[j]);
+        }
+    }
+    __syncthreads();
+    for (int h = threadIdx.x; h < H; h += blockDim.x) {
+        int uid = luid[h];
+        if (uid < 0) continue;
+        atomicAdd(&counts[uid], (u64)lcnt[h]);
+        float* ug = ugrads + (u64)uid * dim;
+        const float* acc = lacc + (u64)h * dim;
+        for (long j = 0; j < dim; ++j) atomicAdd(&ug[j], acc[j]);
+    }
+}
 
 __global__ void k_reduce_grads(const i64* __restrict__ inverse,
                                const float* __restrict__ grads,
@@ -289,11 +405,14 @@ __global__ void k_apply_opt(float* __restrict__ weights,
                             const float* __restrict__ grads,
                             const u64* __restrict__ counts,
                             float c0, float c1, float c2, float c3,
-                            float c4, float c5, float c6) {
+                            float c4, float c5, float c6,
+                            const int* __restrict__ u_dev) {
     long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) / G;
     int lane = threadIdx.x % G;
     if (g >= n) return;
+    if (u_dev && g >= *u_dev) return;
     i64 slot = slots[g];
+    if (slot < 0) return;
     float* w = weights + (u64)slot * dim;
     float* s = state + (u64)slot * sd;
     const float* gr = grads + (u64)g * dim;
@@ -430,9 +549,9 @@ extern "C" {
 void emb_unique(const i64* keys, long n, u64* tk, int* tv, long cap,
                 int* slot_of, unsigned char* is_first, i64* unique_keys,
                 i64* inverse, int* counter, hipStream_t stream) {
-    hipMemsetAsync(tk, 0xFF, cap * sizeof(u64), stream);
-    hipMemsetAsync(counter, 0, sizeof(int), stream);
-    hipMemsetAsync(is_first, 0, n, stream);
+    (void)hipMemsetAsync(tk, 0xFF, cap * sizeof(u64), stream);
+    (void)hipMemsetAsync(counter, 0, sizeof(int), stream);
+    (void)hipMemsetAsync(is_first, 0, n, stream);
     long mask = cap - 1;
     k_unique_insert<<<grid1d(n), BLOCK, 0, stream>>>(keys, n, tk, mask,
                                                      slot_of, is_first);
@@ -444,67 +563,85 @@ void emb_unique(const i64* keys, long n, u64* tk, int* tv, long cap,
 
 void emb_ht_lookup(u64* tk, int* tv, long cap, const i64* keys, long n,
                    int* nrows, i64* slot_keys, i64* slots,
-                   unsigned char* new_mask, int insert, hipStream_t stream) {
+                   unsigned char* new_mask, int insert, const int* u_dev,
+                   hipStream_t stream) {
     k_ht_lookup<<<grid1d(n), BLOCK, 0, stream>>>(tk, tv, cap - 1, keys, n,
                                                  nrows, slot_keys, slots,
-                                                 new_mask, insert);
+                                                 new_mask, insert, u_dev);
 }
 
 void emb_ht_rehash(const u64* tk_old, const int* tv_old, long cap_old,
                    u64* tk_new, int* tv_new, long cap_new,
                    hipStream_t stream) {
-    hipMemsetAsync(tk_new, 0xFF, cap_new * sizeof(u64), stream);
+    (void)hipMemsetAsync(tk_new, 0xFF, cap_new * sizeof(u64), stream);
     k_ht_rehash<<<grid1d(cap_old), BLOCK, 0, stream>>>(
         tk_old, tv_old, cap_old, tk_new, tv_new, cap_new - 1);
 }
 
-void emb_array_touch(unsigned char* valid, const i64* slots, long n,
-                     unsigned char* new_mask, int* n_new, hipStream_t stream) {
-    k_array_touch<<<grid1d(n), BLOCK, 0, stream>>>(valid, slots, n, new_mask,
-                                                   n_new);
+void emb_array_touch(unsigned char* valid, const i64* keys, long n,
+                     long shard_num, long cap, i64* slots,
+                     unsigned char* new_mask, const int* u_dev,
+                     hipStream_t stream) {
+    k_array_touch<<<grid1d(n), BLOCK, 0, stream>>>(valid, keys, n, shard_num,
+                                                   cap, slots, new_mask,
+                                                   u_dev);
 }
 
 void emb_gather_init(float* weights, float* state, long dim, long sd,
                      const i64* slots, const unsigned char* new_mask,
-                     const i64* keys, long n, float* out, int init_cat,
-                     float p0, float p1, float p2, u64 seed,
-                     const float* state_init_row, hipStream_t stream) {
+                     const i64* keys, long n, const i64* inverse, float* out,
+                     int init_cat, float p0, float p1, float p2, u64 seed,
+                     const float* state_init_row, const int* u_dev,
+                     hipStream_t stream) {
     if (n == 0) return;
     if (dim <= 32) {
-        const int G = 16;
-        long threads = n * G;
-        k_gather_init<G><<<grid1d(threads), BLOCK, 0, stream>>>(
-            weights, state, dim, sd, slots, new_mask, keys, n, out,
-            init_cat, p0, p1, p2, seed, state_init_row);
+        k_gather_init<16><<<grid1d(n * 16), BLOCK, 0, stream>>>(
+            weights, state, dim, sd, slots, new_mask, keys, n, inverse, out,
+            init_cat, p0, p1, p2, seed, state_init_row, u_dev);
     } else {
-        const int G = 64;
-        long threads = n * G;
-        k_gather_init<G><<<grid1d(threads), BLOCK, 0, stream>>>(
-            weights, state, dim, sd, slots, new_mask, keys, n, out,
-            init_cat, p0, p1, p2, seed, state_init_row);
+        k_gather_init<64><<<grid1d(n * 64), BLOCK, 0, stream>>>(
+            weights, state, dim, sd, slots, new_mask, keys, n, inverse, out,
+            init_cat, p0, p1, p2, seed, state_init_row, u_dev);
     }
 }
 
 void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
                            long dim, float* ugrads, u64* counts, long u,
                            hipStream_t stream) {
-    hipMemsetAsync(ugrads, 0, (size_t)u * dim * sizeof(float), stream);
-    hipMemsetAsync(counts, 0, (size_t)u * sizeof(u64), stream);
+    hipError_t e1 = hipMemsetAsync(ugrads, 0, (size_t)u * dim * sizeof(float),
+                                   stream);
+    hipError_t e2 = hipMemsetAsync(counts, 0, (size_t)u * sizeof(u64), stream);
+    (void)e1; (void)e2;
     if (n == 0) return;
-    k_reduce_grads<<<grid1d(n * dim), BLOCK, 0, stream>>>(inverse, grads, n,
-                                                          dim, ugrads);
-    k_reduce_counts<<<grid1d(n), BLOCK, 0, stream>>>(inverse, n, counts);
+    if (dim <= 16) {
+        const int H = 512;
+        size_t smem = H * 8 + (size_t)H * dim * 4;
+        int grid = cdiv(n, (long)BLOCK * RED_ITERS);
+        k_reduce_lds<H><<<grid, BLOCK, smem, stream>>>(inverse, grads, n, dim,
+                                                       ugrads, counts);
+    } else if (dim <= 64) {
+        const int H = 128;
+        size_t smem = H * 8 + (size_t)H * dim * 4;
+        int grid = cdiv(n, (long)BLOCK * RED_ITERS);
+        k_reduce_lds<H><<<grid, BLOCK, smem, stream>>>(inverse, grads, n, dim,
+                                                       ugrads, counts);
+    } else {
+        k_reduce_grads<<<grid1d(n * dim), BLOCK, 0, stream>>>(inverse, grads,
+                                                              n, dim, ugrads);
+        k_reduce_counts<<<grid1d(n), BLOCK, 0, stream>>>(inverse, n, counts);
+    }
 }
 
 #define LAUNCH_OPT(G, OPT)                                                  \
     k_apply_opt<G, OPT><<<grid1d(n * G), BLOCK, 0, stream>>>(               \
         weights, state, dim, sd, slots, n, grads, counts, c[0], c[1], c[2], \
-        c[3], c[4], c[5], c[6])
+        c[3], c[4], c[5], c[6], u_dev)
 
 void emb_apply_optimizer(int opt, float* weights, float* state, long dim,
                          long sd, const i64* slots, long n,
                          const float* grads, const u64* counts,
-                         const float* c, hipStream_t stream) {
+                         const float* c, const int* u_dev,
+                         hipStream_t stream) {
     if (n == 0) return;
     if (dim <= 32) {
         const int G = 16;

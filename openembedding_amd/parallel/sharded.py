@@ -45,6 +45,10 @@ class PullHandle:
     recv_splits: Optional[List[int]] = None    # keys received from each rank
     owner_unique: Optional[torch.Tensor] = None  # [u2] deduped keys owned here
     owner_inverse: Optional[torch.Tensor] = None  # recv pos -> owner_unique id
+    # bounded (sync-free) single-GPU fast path:
+    bounded: bool = False
+    u_dev: Optional[torch.Tensor] = None       # device int32 live-count
+    slots: Optional[torch.Tensor] = None       # table slots saved from pull
 
 
 class ShardedVariable:
@@ -89,8 +93,11 @@ class ShardedVariable:
         number of times per variable per step."""
         flat = indices.reshape(-1).to(torch.int64)
         n = flat.numel()
-        unique, inverse = ops.unique_inverse(flat)
         self.stat_pull_indices += n
+        if (self.world_size == 1 and not readonly
+                and hasattr(self.shard, "pull_bounded")):
+            return self._pull_local_bounded(indices, flat)
+        unique, inverse = ops.unique_inverse(flat)
         self.stat_pull_unique += unique.numel()
         h = PullHandle(shape=indices.shape, unique=unique, inverse=inverse)
         if self.world_size == 1:
@@ -101,6 +108,16 @@ class ShardedVariable:
         out = rows_u.index_select(0, inverse)
         out = out.view(*h.shape, self.shard.dim)
         return out, h
+
+    def _pull_local_bounded(self, indices: torch.Tensor, flat: torch.Tensor):
+        """Single-GPU sync-free pull: bounded unique buffer + fused
+        gather/init/scatter; zero host round-trips (hipGraph-capturable)."""
+        ext = self.shard.ext
+        uk_buf, inverse, u_dev = ext.unique_bounded(flat)
+        out, slots = self.shard.pull_bounded(uk_buf, u_dev, inverse)
+        h = PullHandle(shape=indices.shape, unique=uk_buf, inverse=inverse,
+                       bounded=True, u_dev=u_dev, slots=slots)
+        return out.view(*indices.shape, self.shard.dim), h
 
     def _pull_remote(self, h: PullHandle, readonly: bool) -> torch.Tensor:
         world = self.world_size
@@ -135,6 +152,9 @@ class ShardedVariable:
         g = grads.reshape(-1, dim)
         u = h.unique.numel()
         ugrads, counts = ops.reduce_by_inverse(h.inverse, g, u)
+        if h.bounded:
+            self.shard.push_slots(h.unique, h.u_dev, h.slots, ugrads, counts)
+            return
         if self.world_size == 1:
             self.shard.push(h.unique, ugrads, counts)
             return
