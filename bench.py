@@ -75,7 +75,10 @@ def main():
 
     def run_step(dense, sparse, labels):
         opt.zero_grad(set_to_none=False)
-        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=amp):
+        # cache_enabled=False is required for hipGraph capture under autocast
+        # (the autocast weight-cast cache is not capture-safe)
+        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=amp,
+                            cache_enabled=False):
             out = model(dense, sparse)
         loss = lossf(out.float(), labels)
         loss.backward()

@@ -316,8 +316,11 @@ __global__ void k_gather_init(float* __restrict__ weights,
 // dim <= 32 gradients+counts are pre-aggregated in an LDS hash per block
 // (one global atomic per distinct uid per block), fused with the counts.
 
-#define RED_ITERS 8
-
+// One element per thread; the LDS hash only absorbs the duplicates a block
+// actually sees (hot small-vocab fields), everything else flushes exactly
+// once — so vs direct global atomics the extra cost is one LDS CAS + dim LDS
+// adds per element, and the win is that a hot key takes O(blocks) instead of
+// O(batch) serialized global atomics.
 template <int H>
 __global__ void k_reduce_lds(const i64* __restrict__ inverse,
                              const float* __restrict__ grads,
@@ -335,10 +338,8 @@ __global__ void k_reduce_lds(const i64* __restrict__ inverse,
     for (long t = threadIdx.x; t < (long)H * dim; t += blockDim.x)
         lacc[t] = 0.0f;
     __syncthreads();
-    long base = (long)blockIdx.x * blockDim.x * RED_ITERS;
-    for (int it = 0; it < RED_ITERS; ++it) {
-        long i = base + (long)it * blockDim.x + threadIdx.x;
-        if (i >= n) break;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) {
         int uid = (int)inverse[i];
         int h = (int)(((unsigned)uid * 2654435761u) & (H - 1));
         bool got = false;
@@ -616,13 +617,13 @@ void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
     if (dim <= 16) {
         const int H = 512;
         size_t smem = H * 8 + (size_t)H * dim * 4;
-        int grid = cdiv(n, (long)BLOCK * RED_ITERS);
+        int grid = cdiv(n, BLOCK);
         k_reduce_lds<H><<<grid, BLOCK, smem, stream>>>(inverse, grads, n, dim,
                                                        ugrads, counts);
     } else if (dim <= 64) {
-        const int H = 128;
+        const int H = 256;  // 256*(8+dim*4) <= 68 KiB LDS -> 2 blocks/CU
         size_t smem = H * 8 + (size_t)H * dim * 4;
-        int grid = cdiv(n, (long)BLOCK * RED_ITERS);
+        int grid = cdiv(n, BLOCK);
         k_reduce_lds<H><<<grid, BLOCK, smem, stream>>>(inverse, grads, n, dim,
                                                        ugrads, counts);
     } else {
