@@ -95,6 +95,9 @@ def main():
         # capturable and replays with fresh data copied into static buffers.
         static = tuple(t.clone() for t in pool[0])
         try:
+            for _ in range(3):  # eager warmup (lazy state, GEMM algo select)
+                run_step(*static)
+            torch.cuda.synchronize()
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):
