@@ -26,8 +26,10 @@ import torch
 import torch.distributed as dist
 
 from . import flags
+from .config import EnvConfig
 from .core.variable import VariableMeta, VariableShard, HASH_VOCAB_THRESHOLD
 from .parallel.sharded import ShardedVariable
+from .utils.metrics import Reporter
 
 _context: Optional["Context"] = None
 
@@ -100,6 +102,12 @@ class Context:
             self.model_uuid = uuid.uuid4().hex
         self.model_version = 0
         self._t0 = time.time()
+        # typed config tree from flags.config (reference EnvConfig via
+        # embed.flags.config YAML, openembedding/__init__.py:8-41)
+        self.config = EnvConfig.parse(flags.config)
+        self._reporter = Reporter(self.config.server.report_interval,
+                                  rank=self.rank)
+        self._reporter.start()
 
     # ------------------------------------------------------------- factories
 
@@ -149,6 +157,7 @@ class Context:
 
     def finalize(self) -> None:
         global _context
+        self._reporter.stop()
         if self._owns_pg and _dist_ready():
             dist.destroy_process_group()
         if _context is self:

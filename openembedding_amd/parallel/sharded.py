@@ -28,6 +28,7 @@ import torch
 
 from ..core.variable import VariableShard
 from ..ops import dispatch as ops
+from ..utils.metrics import REGISTRY, stage_timer
 from . import comm
 
 
@@ -99,6 +100,10 @@ class ShardedVariable:
             return self._pull_local_bounded(indices, flat)
         unique, inverse = ops.unique_inverse(flat)
         self.stat_pull_unique += unique.numel()
+        # reference pull_indices/pull_unique accumulators
+        # (EmbeddingPullOperator.cpp:208-247): dedup rate sizes the wire
+        REGISTRY.add("pull_indices", n)
+        REGISTRY.add("pull_unique", unique.numel())
         h = PullHandle(shape=indices.shape, unique=unique, inverse=inverse)
         if self.world_size == 1:
             rows_u = (self.shard.pull_readonly(unique) if readonly
@@ -120,6 +125,10 @@ class ShardedVariable:
         return out.view(*indices.shape, self.shard.dim), h
 
     def _pull_remote(self, h: PullHandle, readonly: bool) -> torch.Tensor:
+        with stage_timer("pull", "remote"):
+            return self._pull_remote_impl(h, readonly)
+
+    def _pull_remote_impl(self, h: PullHandle, readonly: bool) -> torch.Tensor:
         world = self.world_size
         owner = h.unique % world
         order = torch.argsort(owner, stable=True)
