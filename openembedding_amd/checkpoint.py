@@ -131,12 +131,55 @@ def _apply_config(var, hdr: dict) -> None:
             var.shard.set_optimizer(cat, **c)
 
 
+def read_meta(uri: str) -> dict:
+    with open(os.path.join(uri, "model_meta")) as f:
+        return json.load(f)
+
+
+def iter_blocks(uri: str, meta: Optional[dict] = None):
+    """Stream every shard file of a dump: yields (header_dict, keys <i8 [n],
+    weights <f4 [n,dim], states <f4 [n,sd] or None) per block. The streaming
+    analogue of the reference's EmbeddingLoadOperator generate_push_items
+    (EmbeddingLoadOperator.cpp:58-111)."""
+    meta = meta or read_meta(uri)
+    for st_ord in range(meta["num_storages"]):
+        d = os.path.join(uri, str(st_ord))
+        for fname in sorted(os.listdir(d)):
+            if not fname.startswith("model_"):
+                continue
+            with open(os.path.join(d, fname), "rb") as f:
+                while True:
+                    magic = f.read(8)
+                    if not magic:
+                        break
+                    if magic != MAGIC:
+                        raise RuntimeError(f"bad shard file {fname}")
+                    (hlen,) = struct.unpack("<q", f.read(8))
+                    hdr = json.loads(f.read(hlen))
+                    dim = hdr["embedding_dim"]
+                    sd = hdr["state_dim"]
+                    while True:
+                        (blk,) = struct.unpack("<q", f.read(8))
+                        if blk <= 0:
+                            if blk == 0:
+                                (end,) = struct.unpack("<q", f.read(8))
+                                assert end == -1
+                            break
+                        keys = np.frombuffer(f.read(blk * 8), dtype="<i8")
+                        w = np.frombuffer(f.read(blk * dim * 4),
+                                          dtype="<f4").reshape(blk, dim)
+                        s = None
+                        if sd:
+                            s = np.frombuffer(f.read(blk * sd * 4),
+                                              dtype="<f4").reshape(blk, sd)
+                        yield hdr, keys, w, s
+
+
 def load_model(ctx, uri: str) -> None:
     """Collective: clears all variables, streams every shard file and keeps
     the keys this rank owns (key % world == rank), re-sharding like the
     reference's load-through-init-push (EmbeddingLoadOperator.cpp:58-111)."""
-    with open(os.path.join(uri, "model_meta")) as f:
-        meta = json.load(f)
+    meta = read_meta(uri)
     by_id = {}
     for st in ctx.storages:
         for var in st.variables:
@@ -152,47 +195,21 @@ def load_model(ctx, uri: str) -> None:
         v.shard.clear()
     rank, world = ctx.rank, ctx.world_size
     configured = set()
-    for st_ord in range(meta["num_storages"]):
-        d = os.path.join(uri, str(st_ord))
-        for fname in sorted(os.listdir(d)):
-            if not fname.startswith("model_"):
-                continue
-            with open(os.path.join(d, fname), "rb") as f:
-                while True:
-                    magic = f.read(8)
-                    if not magic:
-                        break
-                    if magic != MAGIC:
-                        raise RuntimeError(f"bad shard file {fname}")
-                    (hlen,) = struct.unpack("<q", f.read(8))
-                    hdr = json.loads(f.read(hlen))
-                    var = by_id.get(hdr["variable_id"])
-                    dim = hdr["embedding_dim"]
-                    sd = hdr["state_dim"]
-                    if var is not None and hdr["variable_id"] not in configured:
-                        configured.add(hdr["variable_id"])
-                        _apply_config(var, hdr)
-                    while True:
-                        (blk,) = struct.unpack("<q", f.read(8))
-                        if blk <= 0:
-                            if blk == 0:
-                                (end,) = struct.unpack("<q", f.read(8))
-                                assert end == -1
-                            break
-                        keys = np.frombuffer(f.read(blk * 8), dtype="<i8")
-                        w = np.frombuffer(f.read(blk * dim * 4),
-                                          dtype="<f4").reshape(blk, dim)
-                        s = None
-                        if sd:
-                            s = np.frombuffer(f.read(blk * sd * 4),
-                                              dtype="<f4").reshape(blk, sd)
-                        mine = (keys % world) == rank
-                        if var is None or not mine.any():
-                            continue
-                        kt = torch.from_numpy(keys[mine].copy()).to(ctx.device)
-                        wt = torch.from_numpy(w[mine].copy()).to(ctx.device)
-                        st_t: Optional[torch.Tensor] = None
-                        if s is not None and sd == var.shard.state_dim:
-                            st_t = torch.from_numpy(s[mine].copy()).to(ctx.device)
-                        var.shard.import_rows(kt, wt, st_t)
+    for hdr, keys, w, s in iter_blocks(uri, meta):
+        var = by_id.get(hdr["variable_id"])
+        if var is None:
+            continue
+        if hdr["variable_id"] not in configured:
+            configured.add(hdr["variable_id"])
+            _apply_config(var, hdr)
+        sd = hdr["state_dim"]
+        mine = (keys % world) == rank
+        if not mine.any():
+            continue
+        kt = torch.from_numpy(keys[mine].copy()).to(ctx.device)
+        wt = torch.from_numpy(w[mine].copy()).to(ctx.device)
+        st_t: Optional[torch.Tensor] = None
+        if s is not None and sd == var.shard.state_dim:
+            st_t = torch.from_numpy(s[mine].copy()).to(ctx.device)
+        var.shard.import_rows(kt, wt, st_t)
     ctx.barrier()
