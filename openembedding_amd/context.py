@@ -126,14 +126,25 @@ class Context:
             vocabulary_size = HASH_VOCAB_THRESHOLD
         meta = VariableMeta(variable_id=vid, embedding_dim=embedding_dim,
                             dtype=dtype, vocabulary_size=vocabulary_size)
+        cache_mb = self.config.server.cache_size_mb
         if self.device.type == "cuda":
             from .core.variable_gpu import HipVariableShard
             shard_cls = HipVariableShard
+            # GPU capacity tier (HBM cache over host DRAM) lands with the
+            # HIP-backed tier; 288 GB HBM per MI355X holds any single-node
+            # sharded table the reference's benchmarks use.
+            kw = {}
+        elif cache_mb > 0 and meta.use_hash_table:
+            from .core.tiered import TieredVariableShard
+            row_bytes = 4 * (embedding_dim + 64)  # dim + worst-case state
+            shard_cls = TieredVariableShard
+            kw = {"cache_rows": max(1024, (cache_mb << 20) // row_bytes)}
         else:
             shard_cls = VariableShard
+            kw = {}
         shard = shard_cls(meta, shard_id=self.rank,
                           shard_num=self.world_size,
-                          device=str(self.device), seed=self.seed)
+                          device=str(self.device), seed=self.seed, **kw)
         var = ShardedVariable(shard, storage)
         self.variables[vid] = var
         return var

@@ -434,14 +434,28 @@ def pulling(loader, model=None, depth: int = 2):
 # ------------------------------------------------------------- PMem parity
 
 def should_persist_server_model() -> bool:
-    """The reference's PMem cache-full backpressure signal
-    (EmbeddingPullOperator.cpp:182-189). The HBM/DRAM tier analogue is not
-    yet wired; returns False."""
-    return False
+    """The cache-full backpressure signal of the capacity tier (reference
+    should_persist, PmemEmbeddingOptimizerVariable.h:84-86 propagated
+    through EmbeddingPullOperator.cpp:182-189): True when any tiered
+    variable's device cache filled since the last persist."""
+    ctx = get_context()
+    return any(getattr(v.shard, "should_persist", lambda: False)()
+               for v in ctx.variables.values())
 
 
 def persist_server_model(path: str):
+    """Lightweight checkpoint of the capacity tier (reference
+    exb_persist_model): flush dirty cache rows to the host tier, dump, and
+    commit the watermark. For untired variables this equals
+    save_server_model."""
+    ctx = get_context()
+    for v in ctx.variables.values():
+        if hasattr(v.shard, "persist"):
+            v.shard.persist()
     save_server_model(path, include_optimizer=True)
+    for v in ctx.variables.values():
+        if hasattr(v.shard, "checkpoint_committed"):
+            v.shard.checkpoint_committed()
 
 
 def restore_server_model(path: str):

@@ -51,8 +51,51 @@ def fwd_bwd():
     return loss
 
 
+emb = model.embedding
+keys = sparse + emb.field_offsets
+svar = emb.variable.sharded
+
+
+def pull_only():
+    out, h = svar.pull(keys)
+    return out.sum()
+
+
+def pull_push_update():
+    out, h = svar.pull(keys)
+    svar.push(h, torch.ones_like(out))
+    svar.update_weights()
+    return out.sum()
+
+
+static_emb = torch.randn(4096, 26, 10, device=DEV)
+
+
+def mlp_only():
+    opt.zero_grad(set_to_none=False)
+    e = static_emb[..., :9]
+    lin = static_emb[..., 9]
+    s = e.sum(dim=1)
+    fm2 = 0.5 * (s * s - (e * e).sum(dim=1)).sum(dim=1)
+    deep_in = torch.cat([e.flatten(1), dense], dim=1)
+    with torch.autocast("cuda", dtype=torch.bfloat16, enabled=AMP,
+                        cache_enabled=False):
+        out = model.dnn(deep_in).squeeze(-1)
+    logit = (lin.sum(dim=1) + model.dense_linear(dense).squeeze(-1)
+             + fm2 + out.float())
+    loss = lossf(logit, labels)
+    loss.backward()
+    opt.optimizer.step()
+    return loss
+
+
 stage = sys.argv[1] if len(sys.argv) > 1 else "full"
-fn = {"full": full_step, "fwd": fwd_only, "fwdbwd": fwd_bwd}[stage]
+fn = {"full": full_step, "fwd": fwd_only, "fwdbwd": fwd_bwd,
+      "pull": pull_only, "pullpush": pull_push_update,
+      "mlp": mlp_only}[stage.replace("fresh", "")]
+FRESH = "fresh" in stage
+if stage.startswith(("pull", "pullpush")):
+    svar.set_optimizer("adagrad", learning_rate=0.005)
 
 log(f"stage={stage}: eager warmups")
 for _ in range(3):
@@ -71,13 +114,29 @@ graph = torch.cuda.CUDAGraph()
 with torch.cuda.graph(graph):
     fn()
 log("captured; replaying x5")
-for _ in range(5):
+
+
+def refresh(i):
+    """Copy a fresh batch into the static tensors (what bench.py does)."""
+    if not FRESH:
+        return
+    d2, s2, l2 = synthetic_batch(4096, generator=torch.Generator().manual_seed(100 + i))
+    dense.copy_(d2.to(DEV))
+    sparse.copy_(s2.to(DEV))
+    labels.copy_(l2.to(DEV))
+    log(f"  fresh batch {i} copied")
+
+
+for i in range(5):
+    refresh(i)
     graph.replay()
-torch.cuda.synchronize()
+    torch.cuda.synchronize()
+    log(f"  replay {i} ok")
 log("replay ok")
 import time
 t0 = time.perf_counter()
-for _ in range(100):
+for i in range(100):
+    refresh(100 + i)
     graph.replay()
 torch.cuda.synchronize()
 dt = (time.perf_counter() - t0) / 100
