@@ -39,6 +39,32 @@ static inline int grid1d(long n) {
     return (int)(g < 1 ? 1 : g);
 }
 
+// ---------------------------------------------------------------- fills
+// Graph-capture-safe buffer resets. hipMemsetAsync on a capturing stream is
+// NOT reliably captured into the graph on this stack (observed: scratch
+// tables were cleared at capture time only, overflowed after two replays
+// and the probe loops spun) — plain fill kernels capture like any launch.
+
+template <typename T>
+__global__ void k_fill(T* __restrict__ p, long n, T v) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) p[i] = v;
+}
+
+static inline void fill_u64(u64* p, long n, u64 v, hipStream_t s) {
+    if (n) k_fill<u64><<<(int)((n + 255) / 256), 256, 0, s>>>(p, n, v);
+}
+static inline void fill_i32(int* p, long n, int v, hipStream_t s) {
+    if (n) k_fill<int><<<(int)((n + 255) / 256), 256, 0, s>>>(p, n, v);
+}
+static inline void fill_u8(unsigned char* p, long n, unsigned char v,
+                           hipStream_t s) {
+    if (n) k_fill<unsigned char><<<(int)((n + 255) / 256), 256, 0, s>>>(p, n, v);
+}
+static inline void fill_f32(float* p, long n, float v, hipStream_t s) {
+    if (n) k_fill<float><<<(int)((n + 255) / 256), 256, 0, s>>>(p, n, v);
+}
+
 // ------------------------------------------------------------------ rng
 
 DEV u64 splitmix64(u64 z) {
@@ -124,20 +150,30 @@ __global__ void k_unique_insert(const i64* __restrict__ keys, long n,
             lh = (lh + 1) & (ULDS - 1);
         }
         if (lds_winner || !lds_ok) {
-            // block winner (or LDS overflow): probe the global table
+            // block winner (or LDS overflow): probe the global table.
+            // Probes are BOUNDED by the table size: an overfull table (can
+            // only happen on a host sizing bug) must never hang the GPU —
+            // the overflow element becomes its own unique (slot_of
+            // encoding: -1 = overflow winner, assigned directly in the
+            // assign kernel; <=-2 = duplicate pointing at winner element).
             u64 h = splitmix64(k) & (u64)mask;
-            for (;;) {
+            long found = -1;
+            for (long p = 0; p <= mask; ++p) {
                 u64 cur = tk[h];
-                if (cur == k) break;
+                if (cur == k) { found = (long)h; break; }
                 if (cur == EMPTY) {
                     u64 prev = atomicCAS(&tk[h], EMPTY, k);
-                    if (prev == EMPTY) { is_first[i] = 1; break; }
-                    if (prev == k) break;
+                    if (prev == EMPTY) {
+                        is_first[i] = 1; found = (long)h; break;
+                    }
+                    if (prev == k) { found = (long)h; break; }
                 }
                 h = (h + 1) & (u64)mask;
             }
-            slot_of[i] = (int)h;
-            if (lds_winner) lslot[lh] = (int)h;
+            if (found < 0) is_first[i] = 1;  // overflow: own unique
+            slot_of[i] = found >= 0 ? (int)found : -1;
+            if (lds_winner)
+                lslot[lh] = found >= 0 ? (int)found : (int)(-2 - i);
         }
     }
     __syncthreads();
@@ -149,7 +185,8 @@ __global__ void k_unique_assign(const i64* __restrict__ keys, long n,
                                 const unsigned char* __restrict__ is_first,
                                 int* __restrict__ tv,
                                 i64* __restrict__ unique_keys,
-                                int* __restrict__ counter) {
+                                int* __restrict__ counter,
+                                i64* __restrict__ inverse) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     // wave-aggregated counter: one atomic per wave, not per first-occurrence
     bool first = (i < n) && is_first[i];
@@ -164,8 +201,10 @@ __global__ void k_unique_assign(const i64* __restrict__ keys, long n,
     if (first) {
         int rank = __popcll(ballot & ((1ull << lane) - 1ull));
         int uid = base + rank;
-        tv[slot_of[i]] = uid;
         unique_keys[uid] = keys[i];
+        int s = slot_of[i];
+        if (s >= 0) tv[s] = uid;
+        else inverse[i] = uid;  // overflow winner: direct assignment
     }
 }
 
@@ -174,7 +213,10 @@ __global__ void k_unique_inverse(long n, const int* __restrict__ slot_of,
                                  i64* __restrict__ inverse) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    inverse[i] = (i64)tv[slot_of[i]];
+    int s = slot_of[i];
+    if (s >= 0) { inverse[i] = (i64)tv[s]; return; }
+    if (s == -1) return;           // overflow winner: set by assign kernel
+    inverse[i] = inverse[-2 - s];  // overflow duplicate: winner's uid
 }
 
 // --------------------------------------------------- persistent hash table
@@ -197,7 +239,9 @@ __global__ void k_ht_lookup(u64* __restrict__ tk, int* __restrict__ tv,
     u64 h = splitmix64(k) & (u64)mask;
     i64 slot = -1;
     unsigned char is_new = 0;
-    for (;;) {
+    // bounded probe: a full table (host sizing bug) yields slot -1 (miss)
+    // instead of hanging the GPU
+    for (long p = 0; p <= mask; ++p) {
         u64 cur = tk[h];
         if (cur == k) { slot = (i64)tv[h]; break; }
         if (cur == EMPTY) {
@@ -228,7 +272,7 @@ __global__ void k_ht_rehash(const u64* __restrict__ tk_old,
     u64 k = tk_old[i];
     if (k == EMPTY) return;
     u64 h = splitmix64(k) & (u64)mask_new;
-    for (;;) {
+    for (long p = 0; p <= mask_new; ++p) {  // bounded (new table is larger)
         u64 prev = atomicCAS(&tk_new[h], EMPTY, k);
         if (prev == EMPTY) { tv_new[h] = tv_old[i]; return; }
         h = (h + 1) & (u64)mask_new;
@@ -550,15 +594,19 @@ extern "C" {
 void emb_unique(const i64* keys, long n, u64* tk, int* tv, long cap,
                 int* slot_of, unsigned char* is_first, i64* unique_keys,
                 i64* inverse, int* counter, hipStream_t stream) {
-    (void)hipMemsetAsync(tk, 0xFF, cap * sizeof(u64), stream);
-    (void)hipMemsetAsync(counter, 0, sizeof(int), stream);
-    (void)hipMemsetAsync(is_first, 0, n, stream);
+    // fill kernels, not hipMemsetAsync: memsets issued here were NOT
+    // replayed inside hipGraph captures (scratch kept stale keys across
+    // replays until the probe loops hung); kernels always capture
+    fill_u64(tk, cap, EMPTY, stream);
+    fill_i32(counter, 1, 0, stream);
+    fill_u8(is_first, n, 0, stream);
     long mask = cap - 1;
     k_unique_insert<<<grid1d(n), BLOCK, 0, stream>>>(keys, n, tk, mask,
                                                      slot_of, is_first);
     k_unique_assign<<<grid1d(n), BLOCK, 0, stream>>>(keys, n, slot_of,
                                                      is_first, tv,
-                                                     unique_keys, counter);
+                                                     unique_keys, counter,
+                                                     inverse);
     k_unique_inverse<<<grid1d(n), BLOCK, 0, stream>>>(n, slot_of, tv, inverse);
 }
 
@@ -574,7 +622,7 @@ void emb_ht_lookup(u64* tk, int* tv, long cap, const i64* keys, long n,
 void emb_ht_rehash(const u64* tk_old, const int* tv_old, long cap_old,
                    u64* tk_new, int* tv_new, long cap_new,
                    hipStream_t stream) {
-    (void)hipMemsetAsync(tk_new, 0xFF, cap_new * sizeof(u64), stream);
+    fill_u64(tk_new, cap_new, EMPTY, stream);
     k_ht_rehash<<<grid1d(cap_old), BLOCK, 0, stream>>>(
         tk_old, tv_old, cap_old, tk_new, tv_new, cap_new - 1);
 }
@@ -609,10 +657,8 @@ void emb_gather_init(float* weights, float* state, long dim, long sd,
 void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
                            long dim, float* ugrads, u64* counts, long u,
                            hipStream_t stream) {
-    hipError_t e1 = hipMemsetAsync(ugrads, 0, (size_t)u * dim * sizeof(float),
-                                   stream);
-    hipError_t e2 = hipMemsetAsync(counts, 0, (size_t)u * sizeof(u64), stream);
-    (void)e1; (void)e2;
+    fill_f32(ugrads, u * dim, 0.0f, stream);
+    fill_u64(counts, u, 0ull, stream);
     if (n == 0) return;
     if (dim <= 16) {
         const int H = 512;
