@@ -50,6 +50,11 @@ class PullHandle:
     bounded: bool = False
     u_dev: Optional[torch.Tensor] = None       # device int32 live-count
     slots: Optional[torch.Tensor] = None       # table slots saved from pull
+    # bounded owner-side state of the remote path (GPU multi-rank): the
+    # owner's dedup/gather runs sync-free and its table slots are reused by
+    # the matching push
+    owner_u_dev: Optional[torch.Tensor] = None
+    owner_slots: Optional[torch.Tensor] = None
 
 
 class ShardedVariable:
@@ -138,19 +143,30 @@ class ShardedVariable:
         send_splits = send_counts.tolist()
         recv_splits = recv_counts.tolist()
         recv_keys = comm.all_to_all_v(send_keys, send_splits, recv_splits)
-        # owner side: dedup across ranks, gather (+lazy init), fan back out
-        uk2, inv2 = ops.unique_inverse(recv_keys)
-        rows_u2 = (self.shard.pull_readonly(uk2) if readonly
-                   else self.shard.pull(uk2))
-        rows_back = rows_u2.index_select(0, inv2)
+        # owner side: dedup across ranks, gather (+lazy init), fan back out.
+        # On the GPU engine this runs the bounded sync-free path (fused
+        # dedup+gather+duplicate-scatter, slots kept for the push).
+        if (not readonly and hasattr(self.shard, "pull_bounded")
+                and recv_keys.numel()):
+            uk2_buf, inv2, u2_dev = self.shard.ext.unique_bounded(recv_keys)
+            rows_back, slots2 = self.shard.pull_bounded(uk2_buf, u2_dev, inv2)
+            h.owner_unique = uk2_buf
+            h.owner_inverse = inv2
+            h.owner_u_dev = u2_dev
+            h.owner_slots = slots2
+        else:
+            uk2, inv2 = ops.unique_inverse(recv_keys)
+            rows_u2 = (self.shard.pull_readonly(uk2) if readonly
+                       else self.shard.pull(uk2))
+            rows_back = rows_u2.index_select(0, inv2)
+            h.owner_unique = uk2
+            h.owner_inverse = inv2
         rows_sorted = comm.all_to_all_v(rows_back, recv_splits, send_splits)
         rows_u = torch.empty_like(rows_sorted)
         rows_u.index_copy_(0, order, rows_sorted)
         h.order = order
         h.send_splits = send_splits
         h.recv_splits = recv_splits
-        h.owner_unique = uk2
-        h.owner_inverse = inv2
         return rows_u
 
     # ------------------------------------------------------------------- push
@@ -175,7 +191,12 @@ class ShardedVariable:
         g2, _ = ops.reduce_by_inverse(h.owner_inverse, recv_g, u2)
         c2 = torch.zeros(u2, dtype=recv_c.dtype, device=recv_c.device)
         c2.index_add_(0, h.owner_inverse, recv_c)
-        self.shard.push(h.owner_unique, g2, c2)
+        if h.owner_slots is not None:
+            # bounded owner path: slots saved from pull, sync-free apply
+            self.shard.push_slots(h.owner_unique, h.owner_u_dev,
+                                  h.owner_slots, g2, c2)
+        else:
+            self.shard.push(h.owner_unique, g2, c2)
 
     # ----------------------------------------------------------------- commit
 
