@@ -183,14 +183,18 @@ class ShardedVariable:
         if self.world_size == 1:
             self.shard.push(h.unique, ugrads, counts)
             return
-        send_g = ugrads.index_select(0, h.order)
-        send_c = counts.index_select(0, h.order)
-        recv_g = comm.all_to_all_v(send_g, h.send_splits, h.recv_splits)
-        recv_c = comm.all_to_all_v(send_c, h.send_splits, h.recv_splits)
+        # one fused payload [u, dim+1] = grads ‖ counts-as-f32: a single
+        # all_to_all instead of two (counts <= batch size, exact in fp32),
+        # and the owner reduces grads+counts in ONE reduce-by-key pass
+        payload = torch.cat([ugrads, counts.to(ugrads.dtype).unsqueeze(1)],
+                            dim=1)
+        send_p = payload.index_select(0, h.order)
+        recv_p = comm.all_to_all_v(send_p, h.send_splits, h.recv_splits)
         u2 = h.owner_unique.numel()
-        g2, _ = ops.reduce_by_inverse(h.owner_inverse, recv_g, u2)
-        c2 = torch.zeros(u2, dtype=recv_c.dtype, device=recv_c.device)
-        c2.index_add_(0, h.owner_inverse, recv_c)
+        g2c, _ = ops.reduce_by_inverse(h.owner_inverse,
+                                       recv_p.contiguous(), u2)
+        g2 = g2c[:, :dim].contiguous()
+        c2 = g2c[:, dim].round().to(torch.int64)
         if h.owner_slots is not None:
             # bounded owner path: slots saved from pull, sync-free apply
             self.shard.push_slots(h.owner_unique, h.owner_u_dev,
