@@ -108,6 +108,15 @@ class Context:
         self._reporter = Reporter(self.config.server.report_interval,
                                   rank=self.rank)
         self._reporter.start()
+        self._prefetch_stream = None
+
+    @property
+    def prefetch_stream(self):
+        """Side HIP stream for ahead-of-time pulls (reference dataset-thread
+        prefetch; here overlap is stream-level)."""
+        if self._prefetch_stream is None:
+            self._prefetch_stream = torch.cuda.Stream()
+        return self._prefetch_stream
 
     # ------------------------------------------------------------- factories
 

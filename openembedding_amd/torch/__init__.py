@@ -51,8 +51,11 @@ class _PullPushFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, hook: torch.Tensor, indices: torch.Tensor,
-                var: ShardedVariable):
-        out, handle = var.pull(indices)
+                var: ShardedVariable, pre=None):
+        if pre is None:
+            out, handle = var.pull(indices)
+        else:  # prefetched (out, handle) from Variable.prefetch
+            out, handle = pre
         ctx.var = var
         ctx.handle = handle
         return out
@@ -60,7 +63,7 @@ class _PullPushFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_out: torch.Tensor):
         ctx.var.push(ctx.handle, grad_out.contiguous().to(torch.float32))
-        return torch.zeros(0, device=grad_out.device), None, None
+        return torch.zeros(0, device=grad_out.device), None, None, None
 
 
 class Variable:
@@ -70,7 +73,45 @@ class Variable:
     def __init__(self, sharded: ShardedVariable, storage: Storage):
         self.sharded = sharded
         self.storage = storage
-        self._prefetched: List = []
+        self._prefetched: List = []  # FIFO of (match_id, out, handle, event)
+
+    def prefetch(self, indices: torch.Tensor, match_id=None) -> None:
+        """Issue the pull for a FUTURE batch now, on the prefetch stream
+        (reference PrefetchPullWeights issued from the dataset thread,
+        exb_ops.cpp:109-205). ``match_id`` identifies the batch at consume
+        time (default: the identity of the indices tensor). Collective —
+        every rank must prefetch the same variables in the same order."""
+        ctx = get_context()
+        if match_id is None:
+            match_id = id(indices)
+        if ctx.device.type == "cuda":
+            stream = ctx.prefetch_stream
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                out, handle = self.sharded.pull(indices)
+                ev = torch.cuda.Event()
+                ev.record(stream)
+        else:
+            out, handle = self.sharded.pull(indices)
+            ev = None
+        self._prefetched.append((match_id, out, handle, ev))
+
+    def _take_prefetched(self, indices: torch.Tensor):
+        """Pop the prefetched entry for these indices, or None. FIFO with
+        identity check (the reference validated with a sampled-index hash,
+        Prefetch.h:34-44; object identity is exact here because pulling()
+        passes the same tensors through)."""
+        want = id(indices)
+        for pos, (match_id, out, handle, ev) in enumerate(self._prefetched):
+            if match_id == want:
+                # drop skipped older entries along with the match
+                del self._prefetched[:pos + 1]
+                if ev is not None:
+                    torch.cuda.current_stream().wait_event(ev)
+                return out, handle
+        if len(self._prefetched) > 8:  # never-consumed entries: bound memory
+            self._prefetched.pop(0)
+        return None
 
     @property
     def embedding_dim(self):
@@ -141,12 +182,23 @@ class Embedding(nn.Module):
             _tracked.append(self)
 
     def forward(self, indices: torch.Tensor) -> torch.Tensor:
+        return self._forward_keys(indices, indices)
+
+    def _forward_keys(self, keys: torch.Tensor,
+                      match_ref: torch.Tensor) -> torch.Tensor:
         if self.sparse_as_dense:
-            return self.dense(indices)
+            return self.dense(keys)
         if torch.is_grad_enabled() and self.grad_hook.requires_grad:
-            return _PullPushFn.apply(self.grad_hook, indices,
-                                     self.variable.sharded)
-        return self.variable.sparse_read(indices)
+            pre = self.variable._take_prefetched(match_ref)
+            return _PullPushFn.apply(self.grad_hook, keys,
+                                     self.variable.sharded, pre)
+        return self.variable.sparse_read(keys)
+
+    def prefetch(self, indices: torch.Tensor) -> None:
+        """Issue this batch's pull ahead of time (reference prefetch,
+        exb.py:331-340)."""
+        if not self.sparse_as_dense:
+            self.variable.prefetch(indices)
 
     def extra_repr(self):
         mode = ("dense" if self.sparse_as_dense else
@@ -179,7 +231,15 @@ class CombinedEmbedding(Embedding):
 
     def forward(self, field_ids: torch.Tensor) -> torch.Tensor:
         keys = field_ids.to(torch.int64) + self.field_offsets
-        return super().forward(keys)
+        # prefetch matching is by the ORIGINAL field_ids tensor (keys is a
+        # fresh tensor every call)
+        return self._forward_keys(keys, field_ids)
+
+    def prefetch(self, field_ids: torch.Tensor) -> None:
+        if self.sparse_as_dense:
+            return
+        keys = field_ids.to(torch.int64) + self.field_offsets
+        self.variable.prefetch(keys, match_id=id(field_ids))
 
 
 def _default_storage(ctx: Context) -> Storage:
@@ -411,24 +471,60 @@ class Model(nn.Module):
 
 # ------------------------------------------------------------ prefetch API
 
-def pulling(loader, model=None, depth: int = 2):
-    """Dataset-side prefetch (reference pulling(), exb.py:645-691 +
-    PrefetchPullWeights): yields batches while issuing the NEXT batches'
-    embedding pulls ahead of time on the engine. With the GPU engine the
-    overlap is stream-level; this wrapper currently provides the API shape
-    and simple read-ahead."""
+def pulling(loader, model=None, depth: int = 2, sparse_index: int = 1,
+            getter=None):
+    """Dataset-side prefetch pipeline (reference pulling(), exb.py:645-691 +
+    PrefetchPullWeights): while batch t trains, the embedding pulls of
+    batches t+1..t+depth are already issued on the prefetch stream, so the
+    unique/all_to_all/gather work overlaps the MLP compute.
+
+    Batch -> indices mapping: ``getter(batch) -> {embedding_module: indices}``
+    or by default every tracked PS Embedding of ``model`` gets
+    ``batch[sparse_index]`` (the (dense, sparse, labels) convention of the
+    model zoo; the reference matched model input layers by name,
+    exb.py:656-682).
+
+    The prefetched result is matched by tensor identity at forward time, so
+    the SAME batch objects yielded here must flow into the model."""
     import collections
+
+    if getter is None:
+        embs = ([m for m in model.modules()
+                 if isinstance(m, Embedding) and not m.sparse_as_dense]
+                if model is not None else [])
+
+        def getter(batch):
+            return {e: batch[sparse_index] for e in embs}
+
     it = iter(loader)
     buf = collections.deque()
-    while True:
-        while len(buf) < depth:
+
+    def fill():
+        while len(buf) < depth + 1:
             try:
                 buf.append(next(it))
             except StopIteration:
-                break
-        if not buf:
-            return
+                return
+
+    def issue(b):
+        for emb, idx in getter(b).items():
+            emb.prefetch(idx)
+
+    # Ordering guarantee (the reference built it from the server-side
+    # batch-id pending queue, EmbeddingPullOperator.cpp:125-141): a batch's
+    # pull must see the PREVIOUS batch's commit. Issuing the prefetch for
+    # batch t+1 only after batch t was trained puts it after the commit in
+    # stream order — exactly one batch of pipelining, the steady state the
+    # reference's deferral also converges to. ``depth`` only sizes the
+    # host-side loader read-ahead.
+    fill()
+    if buf:
+        issue(buf[0])  # first batch: nothing committed yet
+    while buf:
         yield buf.popleft()
+        fill()
+        if buf:
+            issue(buf[0])  # after the just-trained batch's commit
 
 
 # ------------------------------------------------------------- PMem parity
