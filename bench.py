@@ -38,9 +38,11 @@ def main():
     p.add_argument("--dim", type=int, default=9)
     p.add_argument("--data-pool", type=int, default=8,
                    help="pre-generated synthetic batches, rotated")
-    p.add_argument("--amp", default="bf16", choices=["bf16", "off"],
+    p.add_argument("--amp", default="off", choices=["bf16", "off"],
                    help="autocast dtype for the dense MLP (embeddings and "
-                        "optimizer stay fp32)")
+                        "optimizer stay fp32). Default off: fp32 with the "
+                        "fused head measured faster than bf16 autocast for "
+                        "these skinny GEMMs (profiles/)")
     p.add_argument("--graph", default="auto", choices=["auto", "on", "off"],
                    help="capture the train step in a hipGraph (single-GPU)")
     args = p.parse_args()
@@ -61,6 +63,8 @@ def main():
     torch.manual_seed(1234)  # identical dense init on all ranks
     kw = {} if args.model == "lr" else {"dim": args.dim}
     model = MODELS[args.model](**kw).to(device)
+    if hasattr(model, "head_bf16"):
+        model.head_bf16 = args.amp == "bf16"  # deep_in dtype follows amp
     opt = embed.distributed_optimizer(
         torch.optim.Adagrad(model.parameters(), lr=0.005))
     lossf = torch.nn.BCEWithLogitsLoss()

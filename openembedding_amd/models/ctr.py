@@ -16,8 +16,38 @@ from typing import List, Optional, Sequence
 import torch
 import torch.nn as nn
 
+from ..ops import hip_available
 from ..torch import CombinedEmbedding
 from .criteo import CRITEO_FIELD_VOCABS, N_DENSE
+
+
+class _FusedCTRHeadFn(torch.autograd.Function):
+    """Fused interaction head (ops/csrc/ctrhead.hip): one kernel assembles
+    deep_in (cast fused) and computes first-order + FM + dense-linear
+    partial logits; one kernel does the whole backward. Replaces ~20 small
+    elementwise/reduce launches per step (profiles/)."""
+
+    @staticmethod
+    def forward(ctx, e_all, dense, w, b, use_fm, out_bf16):
+        from ..ops import require_hip
+        ext = require_hip()
+        wf = w.reshape(-1).contiguous()
+        deep_in, partial = ext.ctr_head_fwd(
+            e_all.contiguous(), dense.contiguous(), wf,
+            b.reshape(-1).contiguous(), use_fm, out_bf16)
+        ctx.save_for_backward(e_all, dense, wf)
+        ctx.use_fm = use_fm
+        return deep_in, partial
+
+    @staticmethod
+    def backward(ctx, d_deep_in, d_partial):
+        from ..ops import require_hip
+        ext = require_hip()
+        e_all, dense, wf = ctx.saved_tensors
+        de_all, d_dense, dw, db = ext.ctr_head_bwd(
+            e_all, dense, wf, d_deep_in.contiguous(),
+            d_partial.contiguous(), ctx.use_fm)
+        return de_all, d_dense, dw.view(1, -1), db, None, None
 
 
 def _mlp(in_dim: int, hidden: Sequence[int], out_dim: int = 1) -> nn.Sequential:
@@ -48,6 +78,10 @@ class _CTRBase(nn.Module):
         self.n_fields = len(self.field_vocabs)
         self.embedding = CombinedEmbedding(self.field_vocabs, dim + 1)
         self.dense_linear = nn.Linear(N_DENSE, 1)
+        # fused interaction head on GPU (ctrhead.hip); deep_in dtype follows
+        # head_bf16 (fp32 measured faster than bf16 autocast on MI355X for
+        # these skinny GEMMs — gpurun_out/bench_fp32graph.log)
+        self.head_bf16 = False
 
     def _embed(self, sparse: torch.Tensor):
         """-> (e [B,F,dim], linear_w [B,F])."""
@@ -57,6 +91,18 @@ class _CTRBase(nn.Module):
     def _first_order(self, dense: torch.Tensor, linear_w: torch.Tensor
                      ) -> torch.Tensor:
         return linear_w.sum(dim=1) + self.dense_linear(dense).squeeze(-1)
+
+    def _use_fused_head(self, t: torch.Tensor) -> bool:
+        return t.is_cuda and self.dim >= 1 and hip_available()
+
+    def _fused_head(self, dense: torch.Tensor, sparse: torch.Tensor,
+                    use_fm: bool):
+        """-> (e_all [B,F,dim+1], deep_in [B,F*dim+ND], partial [B])."""
+        e_all = self.embedding(sparse)
+        deep_in, partial = _FusedCTRHeadFn.apply(
+            e_all, dense, self.dense_linear.weight, self.dense_linear.bias,
+            use_fm, self.head_bf16)
+        return e_all, deep_in, partial
 
 
 class LR(_CTRBase):
@@ -80,6 +126,9 @@ class WDL(_CTRBase):
         self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
 
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
+        if self._use_fused_head(dense):
+            _, deep_in, partial = self._fused_head(dense, sparse, use_fm=False)
+            return partial + self.dnn(deep_in).squeeze(-1)
         e, lin = self._embed(sparse)                     # [B, F, d], [B, F]
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
         return self._first_order(dense, lin) + self.dnn(deep_in).squeeze(-1)
@@ -94,6 +143,9 @@ class DeepFM(_CTRBase):
         self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
 
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
+        if self._use_fused_head(dense):
+            _, deep_in, partial = self._fused_head(dense, sparse, use_fm=True)
+            return partial + self.dnn(deep_in).squeeze(-1)
         e, lin = self._embed(sparse)                     # [B, F, d], [B, F]
         # FM second order: 0.5*((sum_f e)^2 - sum_f e^2) summed over dim
         s = e.sum(dim=1)
@@ -141,6 +193,11 @@ class xDeepFM(_CTRBase):
         self.cin = CIN(self.n_fields, dim, cin_layers)
 
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
+        if self._use_fused_head(dense):
+            e_all, deep_in, partial = self._fused_head(dense, sparse,
+                                                       use_fm=False)
+            e = e_all[..., :self.dim]
+            return partial + self.cin(e) + self.dnn(deep_in).squeeze(-1)
         e, lin = self._embed(sparse)
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
         return (self._first_order(dense, lin) + self.cin(e)

@@ -41,6 +41,11 @@ void emb_reduce_by_inverse(const i64*, const float*, long, long, float*, u64*,
 void emb_apply_optimizer(int, float*, float*, long, long, const i64*, long,
                          const float*, const u64*, const float*, const int*,
                          hipStream_t_);
+void emb_ctr_head_fwd(const float*, const float*, const float*, const float*, long,
+                      long, long, long, void*, float*, int, int, hipStream_t_);
+void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
+                      const float*, long, long, long, long, float*, float*,
+                      float*, float*, int, int, hipStream_t_);
 }
 
 namespace {
@@ -216,6 +221,49 @@ void apply_optimizer(int64_t opt, torch::Tensor weights, torch::Tensor state,
                         cur_stream());
 }
 
+// ---- fused CTR interaction head ---------------------------------------
+
+std::tuple<torch::Tensor, torch::Tensor> ctr_head_fwd(
+    torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
+    torch::Tensor bias, bool use_fm, bool out_bf16) {
+    CHECK_GPU(e_all); CHECK_CONT(e_all); CHECK_CONT(dense); CHECK_CONT(w);
+    const c10::cuda::CUDAGuard guard(e_all.device());
+    long B = e_all.size(0), F = e_all.size(1), D1 = e_all.size(2);
+    long dim = D1 - 1, nd = dense.size(1);
+    TORCH_CHECK(D1 <= 64, "ctr_head: dim+1 must be <= 64");
+    auto out_opts = e_all.options().dtype(out_bf16 ? torch::kBFloat16
+                                                   : torch::kFloat32);
+    auto deep_in = torch::empty({B, F * dim + nd}, out_opts);
+    auto partial = torch::empty({B}, e_all.options());
+    CHECK_CONT(bias);
+    emb_ctr_head_fwd(e_all.data_ptr<float>(), dense.data_ptr<float>(),
+                     w.data_ptr<float>(), bias.data_ptr<float>(), B, F, dim, nd,
+                     deep_in.data_ptr(), partial.data_ptr<float>(),
+                     use_fm ? 1 : 0, out_bf16 ? 1 : 0, cur_stream());
+    return {deep_in, partial};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+ctr_head_bwd(torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
+             torch::Tensor d_deep_in, torch::Tensor d_partial, bool use_fm) {
+    CHECK_GPU(e_all); CHECK_CONT(e_all); CHECK_CONT(d_deep_in);
+    const c10::cuda::CUDAGuard guard(e_all.device());
+    long B = e_all.size(0), F = e_all.size(1), D1 = e_all.size(2);
+    long dim = D1 - 1, nd = dense.size(1);
+    bool out_bf16 = d_deep_in.dtype() == torch::kBFloat16;
+    auto de_all = torch::empty_like(e_all);
+    auto d_dense = torch::empty_like(dense);
+    auto dw = torch::zeros_like(w);
+    auto db = torch::zeros({1}, w.options());
+    emb_ctr_head_bwd(e_all.data_ptr<float>(), dense.data_ptr<float>(),
+                     w.data_ptr<float>(), d_deep_in.data_ptr(),
+                     d_partial.data_ptr<float>(), B, F, dim, nd,
+                     de_all.data_ptr<float>(), d_dense.data_ptr<float>(),
+                     dw.data_ptr<float>(), db.data_ptr<float>(),
+                     use_fm ? 1 : 0, out_bf16 ? 1 : 0, cur_stream());
+    return {de_all, d_dense, dw, db};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -231,4 +279,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("reduce_by_inverse", &reduce_by_inverse,
           "grad reduce-by-key with counts (LDS-aggregated)");
     m.def("apply_optimizer", &apply_optimizer, "fused sparse optimizer step");
+    m.def("ctr_head_fwd", &ctr_head_fwd,
+          "fused CTR head fwd: deep_in assembly (+cast) + FM + first-order "
+          "+ dense linear");
+    m.def("ctr_head_bwd", &ctr_head_bwd, "fused CTR head backward");
 }

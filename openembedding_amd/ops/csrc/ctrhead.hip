@@ -1,0 +1,166 @@
+// Fused CTR interaction head (gfx950 / MI355X).
+//
+// The DeepFM/WDL "glue" between the embedding gather and the MLP GEMMs was
+// ~320 us/step of small elementwise/reduce kernels + ~110 us of autocast
+// casts (profiles/bench_deepfm_1gpu_kernels.md). These two kernels fuse, in
+// one pass over the gathered rows:
+//
+//   forward:  e_all [B, F, D1] (D1 = dim+1: dim embedding cols + the merged
+//             first-order/"wide" column — see models/ctr.py) and dense
+//             [B, ND] ->
+//               deep_in  bf16 [B, F*dim + ND]   (cast fused into the write)
+//               partial  f32  [B] = sum_f e[.,dim]            (first order)
+//                              + 0.5*sum_d((sum_f e)^2 - sum_f e^2)  (FM)
+//                              + dense @ w + b             (dense linear)
+//   backward: d_deep_in bf16, d_partial f32 ->
+//               de_all f32 (FM + first-order + deep contributions),
+//               d_dense f32, dw/db accumulated with atomics.
+//
+// One 64-lane wave per sample; lane c covers column c of the row (D1 <= 64
+// enforced host-side), so the per-field row load is one coalesced segment.
+// Reference math: DeepFM second order 0.5*((sum e)^2 - sum e^2) — the same
+// formula the TF/DeepCTR models of the reference benchmark compute
+// (test/benchmark/criteo_deepctr.py DeepFM).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+typedef long long i64;
+typedef unsigned long long u64;
+typedef __hip_bfloat16 bf16;
+
+// lanes c < dim: embedding columns; lane dim: wide column; F fields.
+// use_fm: 0 for WDL/LR-style heads.
+template <typename OutT>
+__global__ void k_ctr_head_fwd(const float* __restrict__ e_all,
+                               const float* __restrict__ dense,
+                               const float* __restrict__ w,  // [nd]
+                               const float* __restrict__ bias,  // [1]
+                               long B, long F, long dim, long nd,
+                               OutT* __restrict__ deep_in,
+                               float* __restrict__ partial,
+                               int use_fm) {
+    const long D1 = dim + 1;
+    const int lane = threadIdx.x & 63;
+    const long b = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    if (b >= B) return;
+    const float* e = e_all + b * F * D1;
+    OutT* di = deep_in + b * (F * dim + nd);
+
+    float s = 0.0f, sq = 0.0f;   // per-column running sums (lane c)
+    const bool emb_col = lane < dim;
+    const bool wide_col = lane == dim;
+    for (long f = 0; f < F; ++f) {
+        float v = (lane < D1) ? e[f * D1 + lane] : 0.0f;
+        if (emb_col) {
+            s += v;
+            sq += v * v;
+            di[f * dim + lane] = (OutT)v;
+        } else if (wide_col) {
+            s += v;                       // first-order sum
+        }
+    }
+    // dense tail: cast + dot
+    float dsum = 0.0f;
+    for (long j = lane; j < nd; j += 64) {
+        float v = dense[b * nd + j];
+        di[F * dim + j] = (OutT)v;
+        dsum += v * w[j];
+    }
+    // wave reduction: fm2 over lanes<dim, lin from lane dim, dsum over all
+    float fm = emb_col && use_fm ? (s * s - sq) : 0.0f;
+    float lin = wide_col ? s : 0.0f;
+    float acc = 0.5f * fm + lin + dsum;
+    for (int off = 32; off; off >>= 1)
+        acc += __shfl_down(acc, off, 64);
+    if (lane == 0) partial[b] = acc + bias[0];
+}
+
+// backward. sums s[d] are recomputed from e_all (cheaper than saving them).
+template <typename OutT>
+__global__ void k_ctr_head_bwd(const float* __restrict__ e_all,
+                               const float* __restrict__ dense,
+                               const float* __restrict__ w,
+                               const OutT* __restrict__ d_deep_in,
+                               const float* __restrict__ d_partial,
+                               long B, long F, long dim, long nd,
+                               float* __restrict__ de_all,
+                               float* __restrict__ d_dense,
+                               float* __restrict__ dw,   // [nd] atomic
+                               float* __restrict__ db,   // [1] atomic
+                               int use_fm) {
+    const long D1 = dim + 1;
+    const int lane = threadIdx.x & 63;
+    const long b = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    if (b >= B) return;
+    const float* e = e_all + b * F * D1;
+    const OutT* ddi = d_deep_in + b * (F * dim + nd);
+    float* de = de_all + b * F * D1;
+    const float gp = d_partial[b];
+
+    const bool emb_col = lane < dim;
+    float s = 0.0f;
+    if (use_fm && emb_col)
+        for (long f = 0; f < F; ++f) s += e[f * D1 + lane];
+
+    for (long f = 0; f < F; ++f) {
+        if (emb_col) {
+            float g = (float)ddi[f * dim + lane];
+            if (use_fm) g += gp * (s - e[f * D1 + lane]);
+            de[f * D1 + lane] = g;
+        } else if (lane == dim) {
+            de[f * D1 + lane] = gp;          // wide column
+        } else if (lane < D1) {
+            de[f * D1 + lane] = 0.0f;
+        }
+    }
+    float dbl = 0.0f;
+    for (long j = lane; j < nd; j += 64) {
+        d_dense[b * nd + j] = (float)ddi[F * dim + j] + gp * w[j];
+        atomicAdd(&dw[j], gp * dense[b * nd + j]);
+    }
+    if (lane == 0) {
+        dbl = gp;
+        atomicAdd(db, dbl);
+    }
+}
+
+extern "C" {
+
+void emb_ctr_head_fwd(const float* e_all, const float* dense, const float* w,
+                      const float* bias, long B, long F, long dim, long nd,
+                      void* deep_in, float* partial, int use_fm,
+                      int out_bf16, hipStream_t stream) {
+    if (B == 0) return;
+    int block = 256;                    // 4 waves per block
+    long grid = (B * 64 + block - 1) / block;
+    if (out_bf16)
+        k_ctr_head_fwd<bf16><<<(int)grid, block, 0, stream>>>(
+            e_all, dense, w, bias, B, F, dim, nd, (bf16*)deep_in, partial,
+            use_fm);
+    else
+        k_ctr_head_fwd<float><<<(int)grid, block, 0, stream>>>(
+            e_all, dense, w, bias, B, F, dim, nd, (float*)deep_in, partial,
+            use_fm);
+}
+
+void emb_ctr_head_bwd(const float* e_all, const float* dense, const float* w,
+                      const void* d_deep_in, const float* d_partial,
+                      long B, long F, long dim, long nd,
+                      float* de_all, float* d_dense, float* dw, float* db,
+                      int use_fm, int out_bf16, hipStream_t stream) {
+    if (B == 0) return;
+    int block = 256;
+    long grid = (B * 64 + block - 1) / block;
+    if (out_bf16)
+        k_ctr_head_bwd<bf16><<<(int)grid, block, 0, stream>>>(
+            e_all, dense, w, (const bf16*)d_deep_in, d_partial, B, F, dim, nd,
+            de_all, d_dense, dw, db, use_fm);
+    else
+        k_ctr_head_bwd<float><<<(int)grid, block, 0, stream>>>(
+            e_all, dense, w, (const float*)d_deep_in, d_partial, B, F, dim, nd,
+            de_all, d_dense, dw, db, use_fm);
+}
+
+}  // extern "C"

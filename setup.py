@@ -20,6 +20,7 @@ ext = CUDAExtension(
     sources=[
         "openembedding_amd/ops/csrc/bindings.cpp",
         "openembedding_amd/ops/csrc/embops.hip",
+        "openembedding_amd/ops/csrc/ctrhead.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3"],
