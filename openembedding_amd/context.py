@@ -137,12 +137,15 @@ class Context:
                             dtype=dtype, vocabulary_size=vocabulary_size)
         cache_mb = self.config.server.cache_size_mb
         if self.device.type == "cuda":
-            from .core.variable_gpu import HipVariableShard
-            shard_cls = HipVariableShard
-            # GPU capacity tier (HBM cache over host DRAM) lands with the
-            # HIP-backed tier; 288 GB HBM per MI355X holds any single-node
-            # sharded table the reference's benchmarks use.
-            kw = {}
+            if cache_mb > 0 and meta.use_hash_table:
+                from .core.tiered_gpu import HipTieredVariableShard
+                row_bytes = 4 * (embedding_dim + 64)
+                shard_cls = HipTieredVariableShard
+                kw = {"cache_rows": max(1024, (cache_mb << 20) // row_bytes)}
+            else:
+                from .core.variable_gpu import HipVariableShard
+                shard_cls = HipVariableShard
+                kw = {}
         elif cache_mb > 0 and meta.use_hash_table:
             from .core.tiered import TieredVariableShard
             row_bytes = 4 * (embedding_dim + 64)  # dim + worst-case state

@@ -101,7 +101,7 @@ class ShardedVariable:
         n = flat.numel()
         self.stat_pull_indices += n
         if (self.world_size == 1 and not readonly
-                and hasattr(self.shard, "pull_bounded")):
+                and getattr(self.shard, "pull_bounded", None) is not None):
             return self._pull_local_bounded(indices, flat)
         unique, inverse = ops.unique_inverse(flat)
         self.stat_pull_unique += unique.numel()
@@ -146,7 +146,8 @@ class ShardedVariable:
         # owner side: dedup across ranks, gather (+lazy init), fan back out.
         # On the GPU engine this runs the bounded sync-free path (fused
         # dedup+gather+duplicate-scatter, slots kept for the push).
-        if (not readonly and hasattr(self.shard, "pull_bounded")
+        if (not readonly
+                and getattr(self.shard, "pull_bounded", None) is not None
                 and recv_keys.numel()):
             uk2_buf, inv2, u2_dev = self.shard.ext.unique_bounded(recv_keys)
             rows_back, slots2 = self.shard.pull_bounded(uk2_buf, u2_dev, inv2)
