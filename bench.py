@@ -66,7 +66,8 @@ def main():
     if hasattr(model, "head_bf16"):
         model.head_bf16 = args.amp == "bf16"  # deep_in dtype follows amp
     opt = embed.distributed_optimizer(
-        torch.optim.Adagrad(model.parameters(), lr=0.005))
+        torch.optim.Adagrad(model.parameters(), lr=0.005),
+        flatten_dense=True)
     lossf = torch.nn.BCEWithLogitsLoss()
 
     gen = torch.Generator(device="cpu").manual_seed(4242 + rank)

@@ -32,10 +32,11 @@ class _FusedCTRHeadFn(torch.autograd.Function):
         from ..ops import require_hip
         ext = require_hip()
         wf = w.reshape(-1).contiguous()
-        deep_in, partial = ext.ctr_head_fwd(
-            e_all.contiguous(), dense.contiguous(), wf,
-            b.reshape(-1).contiguous(), use_fm, out_bf16)
-        ctx.save_for_backward(e_all, dense, wf)
+        e_all = e_all.contiguous()
+        dense = dense.contiguous()
+        deep_in, partial, s = ext.ctr_head_fwd(
+            e_all, dense, wf, b.reshape(-1).contiguous(), use_fm, out_bf16)
+        ctx.save_for_backward(e_all, dense, wf, s)
         ctx.use_fm = use_fm
         return deep_in, partial
 
@@ -43,10 +44,10 @@ class _FusedCTRHeadFn(torch.autograd.Function):
     def backward(ctx, d_deep_in, d_partial):
         from ..ops import require_hip
         ext = require_hip()
-        e_all, dense, wf = ctx.saved_tensors
+        e_all, dense, wf, s = ctx.saved_tensors
         de_all, d_dense, dw, db = ext.ctr_head_bwd(
             e_all, dense, wf, d_deep_in.contiguous(),
-            d_partial.contiguous(), ctx.use_fm)
+            d_partial.contiguous(), s, ctx.use_fm)
         return de_all, d_dense, dw.view(1, -1), db, None, None
 
 

@@ -42,10 +42,11 @@ void emb_apply_optimizer(int, float*, float*, long, long, const i64*, long,
                          const float*, const u64*, const float*, const int*,
                          hipStream_t_);
 void emb_ctr_head_fwd(const float*, const float*, const float*, const float*, long,
-                      long, long, long, void*, float*, int, int, hipStream_t_);
+                      long, long, long, void*, float*, float*, int, int,
+                      hipStream_t_);
 void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
-                      const float*, long, long, long, long, float*, float*,
-                      float*, float*, int, int, hipStream_t_);
+                      const float*, const float*, long, long, long, long,
+                      float*, float*, float*, float*, int, int, hipStream_t_);
 }
 
 namespace {
@@ -88,7 +89,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> unique_bounded(
     auto tv = torch::empty({cap}, opts_i32);
     auto slot_of = torch::empty({n}, opts_i32);
     auto is_first = torch::empty({n}, opts_u8);
-    auto uk = torch::zeros({n}, opts_i64);
+    auto uk = torch::empty({n}, opts_i64);  // tail unread
     auto inverse = torch::empty({n}, opts_i64);
     auto counter = torch::empty({1}, opts_i32);
     emb_unique(keys.data_ptr<i64>(), n, (u64*)tk.data_ptr<i64>(),
@@ -114,7 +115,7 @@ std::tuple<torch::Tensor, torch::Tensor> ht_lookup(
     const c10::cuda::CUDAGuard guard(keys.device());
     long n = keys.numel();
     auto slots = torch::empty({n}, keys.options());
-    auto new_mask = torch::zeros({n}, keys.options().dtype(torch::kUInt8));
+    auto new_mask = torch::empty({n}, keys.options().dtype(torch::kUInt8));
     if (n)
         emb_ht_lookup((u64*)tk.data_ptr<i64>(), tv.data_ptr<int>(),
                       tk.numel(), keys.data_ptr<i64>(), n,
@@ -140,7 +141,7 @@ std::tuple<torch::Tensor, torch::Tensor> array_touch(
     const c10::cuda::CUDAGuard guard(valid.device());
     long n = keys.numel();
     auto slots = torch::empty({n}, keys.options());
-    auto new_mask = torch::zeros({n}, keys.options().dtype(torch::kUInt8));
+    auto new_mask = torch::empty({n}, keys.options().dtype(torch::kUInt8));
     if (n)
         emb_array_touch(valid.data_ptr<uint8_t>(), keys.data_ptr<i64>(), n,
                         shard_num, valid.numel(), slots.data_ptr<i64>(),
@@ -223,7 +224,7 @@ void apply_optimizer(int64_t opt, torch::Tensor weights, torch::Tensor state,
 
 // ---- fused CTR interaction head ---------------------------------------
 
-std::tuple<torch::Tensor, torch::Tensor> ctr_head_fwd(
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ctr_head_fwd(
     torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
     torch::Tensor bias, bool use_fm, bool out_bf16) {
     CHECK_GPU(e_all); CHECK_CONT(e_all); CHECK_CONT(dense); CHECK_CONT(w);
@@ -235,17 +236,20 @@ std::tuple<torch::Tensor, torch::Tensor> ctr_head_fwd(
                                                    : torch::kFloat32);
     auto deep_in = torch::empty({B, F * dim + nd}, out_opts);
     auto partial = torch::empty({B}, e_all.options());
+    auto s_out = torch::empty({B, dim}, e_all.options());
     CHECK_CONT(bias);
     emb_ctr_head_fwd(e_all.data_ptr<float>(), dense.data_ptr<float>(),
                      w.data_ptr<float>(), bias.data_ptr<float>(), B, F, dim, nd,
                      deep_in.data_ptr(), partial.data_ptr<float>(),
+                     s_out.data_ptr<float>(),
                      use_fm ? 1 : 0, out_bf16 ? 1 : 0, cur_stream());
-    return {deep_in, partial};
+    return {deep_in, partial, s_out};
 }
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 ctr_head_bwd(torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
-             torch::Tensor d_deep_in, torch::Tensor d_partial, bool use_fm) {
+             torch::Tensor d_deep_in, torch::Tensor d_partial,
+             torch::Tensor s_in, bool use_fm) {
     CHECK_GPU(e_all); CHECK_CONT(e_all); CHECK_CONT(d_deep_in);
     const c10::cuda::CUDAGuard guard(e_all.device());
     long B = e_all.size(0), F = e_all.size(1), D1 = e_all.size(2);
@@ -257,7 +261,8 @@ ctr_head_bwd(torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
     auto db = torch::zeros({1}, w.options());
     emb_ctr_head_bwd(e_all.data_ptr<float>(), dense.data_ptr<float>(),
                      w.data_ptr<float>(), d_deep_in.data_ptr(),
-                     d_partial.data_ptr<float>(), B, F, dim, nd,
+                     d_partial.data_ptr<float>(),
+                     s_in.data_ptr<float>(), B, F, dim, nd,
                      de_all.data_ptr<float>(), d_dense.data_ptr<float>(),
                      dw.data_ptr<float>(), db.data_ptr<float>(),
                      use_fm ? 1 : 0, out_bf16 ? 1 : 0, cur_stream());

@@ -294,31 +294,102 @@ def _server_cfg_from_torch(opt: torch.optim.Optimizer) -> Dict:
     raise AssertionError(cat)
 
 
+class _FlatDenseAdagrad:
+    """Dense params re-based onto ONE flat buffer: zero_grad is one fill,
+    the distributed allreduce is one collective on one tensor, and the
+    Adagrad step is 3 flat kernels (torch's multi-tensor apply was ~60us +
+    ~60us of per-param grad fills per step on the DeepFM profile).
+    Numerically identical to torch.optim.Adagrad with default lr_decay=0,
+    weight_decay=0."""
+
+    def __init__(self, optimizer: torch.optim.Optimizer):
+        g = optimizer.param_groups[0]
+        if (type(optimizer).__name__ != "Adagrad"
+                or len(optimizer.param_groups) != 1
+                or g.get("lr_decay", 0) or g.get("weight_decay", 0)):
+            raise ValueError("flat dense path supports plain Adagrad "
+                             "(single group, no lr_decay/weight_decay)")
+        self.lr = g["lr"]
+        self.eps = g.get("eps", 1e-10)
+        init_acc = g.get("initial_accumulator_value", 0.0)
+        params = [p for p in g["params"]
+                  if p.requires_grad and p.numel() > 0]
+        if not params:
+            raise ValueError("no dense params")
+        dev, dt = params[0].device, params[0].dtype
+        if any(p.device != dev or p.dtype != dt for p in params):
+            raise ValueError("flat dense path needs one device/dtype")
+        total = sum(p.numel() for p in params)
+        self.flat = torch.empty(total, device=dev, dtype=dt)
+        self.flat_grad = torch.zeros(total, device=dev, dtype=dt)
+        off = 0
+        for p in params:
+            n = p.numel()
+            self.flat[off:off + n] = p.data.reshape(-1)
+            p.data = self.flat[off:off + n].view_as(p)
+            p.grad = self.flat_grad[off:off + n].view_as(p)
+            off += n
+        self.params = params
+        self.accum = torch.full_like(self.flat, float(init_acc))
+        self._std = torch.empty_like(self.flat)
+
+    def zero_grad(self):
+        self.flat_grad.zero_()
+
+    def step(self):
+        self.accum.addcmul_(self.flat_grad, self.flat_grad)
+        torch.sqrt(self.accum, out=self._std)
+        self._std.add_(self.eps)
+        self.flat.addcdiv_(self.flat_grad, self._std, value=-self.lr)
+
+    def state_dict(self):
+        return {"flat": self.flat, "accum": self.accum, "lr": self.lr}
+
+    def load_state_dict(self, sd):
+        self.flat.copy_(sd["flat"])
+        self.accum.copy_(sd["accum"])
+
+
 class DistributedOptimizer:
     """Wraps a torch optimizer: dense grads are RCCL-allreduced (SUM, like the
     reference's hvd.DistributedOptimizer(op=hvd.Sum) examples), then the base
     optimizer steps, then every PS storage commits its batch
     (reference exb.py:446-488 distributed optimizer subclasses +
-    UpdateWeights op)."""
+    UpdateWeights op).
+
+    ``flatten_dense=True`` re-bases the dense params onto one flat buffer
+    (single-fill zero_grad, single-tensor allreduce, 3-kernel Adagrad)."""
 
     def __init__(self, optimizer: torch.optim.Optimizer,
                  sparse_config: Optional[Dict] = None,
-                 average_dense: bool = False):
+                 average_dense: bool = False,
+                 flatten_dense: bool = False):
         self.optimizer = optimizer
         self.average_dense = average_dense
         self._sparse_config = sparse_config
         self._configured_vars = set()
         self.ctx = get_context()
+        self._flat: Optional[_FlatDenseAdagrad] = None
+        if flatten_dense:
+            self._flat = _FlatDenseAdagrad(optimizer)
 
     # behave like the wrapped optimizer
     def __getattr__(self, name):
         return getattr(self.optimizer, name)
 
     def zero_grad(self, set_to_none: bool = True):
+        if self._flat is not None:
+            self._flat.zero_grad()
+            return
         self.optimizer.zero_grad(set_to_none=set_to_none)
 
     def _allreduce_dense(self):
         if not comm.dist_ready() or dist.get_world_size() == 1:
+            return
+        if self._flat is not None:
+            dist.all_reduce(self._flat.flat_grad)
+            if self.average_dense:
+                self._flat.flat_grad /= dist.get_world_size()
             return
         bucket: List[torch.Tensor] = []
         for group in self.optimizer.param_groups:
@@ -365,16 +436,24 @@ class DistributedOptimizer:
         if closure is not None:
             loss = closure()
         self._allreduce_dense()
-        self.optimizer.step()
+        if self._flat is not None:
+            self._flat.step()
+        else:
+            self.optimizer.step()
         self._ensure_sparse_configured()
         self.ctx.update_all_weights()
         self.ctx.model_version += 1
         return loss
 
     def state_dict(self):
+        if self._flat is not None:
+            return {"flat_dense": self._flat.state_dict()}
         return self.optimizer.state_dict()
 
     def load_state_dict(self, sd):
+        if self._flat is not None and "flat_dense" in sd:
+            self._flat.load_state_dict(sd["flat_dense"])
+            return
         self.optimizer.load_state_dict(sd)
 
 
