@@ -232,6 +232,7 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ctr_head_fwd(
     long B = e_all.size(0), F = e_all.size(1), D1 = e_all.size(2);
     long dim = D1 - 1, nd = dense.size(1);
     TORCH_CHECK(D1 <= 64, "ctr_head: dim+1 must be <= 64");
+    TORCH_CHECK(nd <= 32, "ctr_head: dense features must be <= 32");
     auto out_opts = e_all.options().dtype(out_bf16 ? torch::kBFloat16
                                                    : torch::kFloat32);
     auto deep_in = torch::empty({B, F * dim + nd}, out_opts);
