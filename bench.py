@@ -19,6 +19,11 @@ import json
 import os
 import time
 
+# hipBLASLt algo autotuning during the (untimed) warmup: measured +8% on the
+# DeepFM step (gpurun_out/bench7_tunable.log vs bench7.log). Must be set
+# before torch initializes its blas handles.
+os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+
 import torch
 import torch.distributed as dist
 
@@ -38,11 +43,13 @@ def main():
     p.add_argument("--dim", type=int, default=9)
     p.add_argument("--data-pool", type=int, default=8,
                    help="pre-generated synthetic batches, rotated")
-    p.add_argument("--amp", default="off", choices=["bf16", "off"],
-                   help="autocast dtype for the dense MLP (embeddings and "
-                        "optimizer stay fp32). Default off: fp32 with the "
-                        "fused head measured faster than bf16 autocast for "
-                        "these skinny GEMMs (profiles/)")
+    p.add_argument("--amp", default="off",
+                   choices=["bf16", "off", "native"],
+                   help="dense-MLP precision: off = fp32; bf16 = autocast "
+                        "(per-step weight casts); native = bf16-resident "
+                        "MLP weights with fp32 master in the flat optimizer "
+                        "(MFMA rate, no cast kernels). Embeddings, FM math, "
+                        "loss and all optimizer state stay fp32.")
     p.add_argument("--graph", default="auto", choices=["auto", "on", "off"],
                    help="capture the train step in a hipGraph (single-GPU)")
     args = p.parse_args()
@@ -65,6 +72,9 @@ def main():
     model = MODELS[args.model](**kw).to(device)
     if hasattr(model, "head_bf16"):
         model.head_bf16 = args.amp == "bf16"  # deep_in dtype follows amp
+    if args.amp == "native" and on_gpu:
+        from openembedding_amd.models.ctr import convert_mlp_bf16
+        convert_mlp_bf16(model)
     opt = embed.distributed_optimizer(
         torch.optim.Adagrad(model.parameters(), lr=0.005),
         flatten_dense=True)
@@ -170,7 +180,8 @@ def main():
         "vs_baseline": (samples_per_sec / baseline) if (
             baseline and args.model == "deepfm" and args.dim == 9
             and args.batch == 4096) else None,
-        "dtype": "bf16" if amp else "fp32",
+        "dtype": ("bf16" if (amp or (args.amp == "native" and on_gpu))
+                  else "fp32"),
         "data": "synthetic Criteo-shaped (random ids, Criteo-Kaggle "
                 "cardinalities, random labels), random-init weights",
         "config": {
@@ -179,9 +190,12 @@ def main():
             "fields": 26,
             "dense_features": 13,
             "optimizer": "adagrad",
-            "precision_note": ("dense MLP bf16 autocast (MFMA); embeddings, "
-                               "FM reductions in fp32; sparse+dense optimizer "
-                               "state fp32" if amp else "all fp32"),
+            "precision_note": (
+                "dense MLP bf16 autocast (MFMA); embeddings, FM reductions "
+                "fp32; optimizer state fp32" if amp else
+                "dense MLP weights bf16-resident (MFMA) with fp32 master + "
+                "fp32 accumulators; embeddings, FM math, loss fp32"
+                if (args.amp == "native" and on_gpu) else "all fp32"),
             "graph": graph is not None,
             "parallelism": (f"dense-dp{n_gpus} + embedding sharded "
                             f"all_to_all" if n_gpus > 1 else "single-gpu"),

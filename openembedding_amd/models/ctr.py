@@ -205,4 +205,16 @@ class xDeepFM(_CTRBase):
                 + self.dnn(deep_in).squeeze(-1))
 
 
+def convert_mlp_bf16(model: _CTRBase) -> _CTRBase:
+    """Native-bf16 MLP: the dnn's weights LIVE in bf16 (straight into MFMA
+    GEMMs, no per-step autocast casts; the flat optimizer keeps an fp32
+    master) and the fused head emits bf16 deep_in. dense_linear stays fp32
+    (it is consumed by the fused head kernel); embeddings, FM math and the
+    loss stay fp32."""
+    if hasattr(model, "dnn"):
+        model.dnn.to(torch.bfloat16)
+    model.head_bf16 = True
+    return model
+
+
 MODELS = {"lr": LR, "wdl": WDL, "deepfm": DeepFM, "xdeepfm": xDeepFM}

@@ -47,6 +47,10 @@ void emb_ctr_head_fwd(const float*, const float*, const float*, const float*, lo
 void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
                       const float*, const float*, long, long, long, long,
                       float*, float*, float*, float*, int, int, hipStream_t_);
+void emb_flat_adagrad_f32(float*, float*, const float*, long, float, float,
+                          hipStream_t_);
+void emb_flat_adagrad_bf16(float*, float*, const void*, void*, long, float,
+                           float, hipStream_t_);
 }
 
 namespace {
@@ -270,6 +274,28 @@ ctr_head_bwd(torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
     return {de_all, d_dense, dw, db};
 }
 
+// ---- flat dense Adagrad -----------------------------------------------
+
+void flat_adagrad(torch::Tensor param, torch::Tensor accum,
+                  torch::Tensor grad, c10::optional<torch::Tensor> master,
+                  double lr, double eps) {
+    CHECK_GPU(param); CHECK_CONT(param); CHECK_CONT(accum); CHECK_CONT(grad);
+    const c10::cuda::CUDAGuard guard(param.device());
+    long n = param.numel();
+    if (param.dtype() == torch::kFloat32) {
+        emb_flat_adagrad_f32(param.data_ptr<float>(), accum.data_ptr<float>(),
+                             grad.data_ptr<float>(), n, (float)lr, (float)eps,
+                             cur_stream());
+    } else {
+        TORCH_CHECK(param.dtype() == torch::kBFloat16 && master.has_value(),
+                    "bf16 flat_adagrad needs a float32 master tensor");
+        emb_flat_adagrad_bf16(master->data_ptr<float>(),
+                              accum.data_ptr<float>(), grad.data_ptr(),
+                              param.data_ptr(), n, (float)lr, (float)eps,
+                              cur_stream());
+    }
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -289,4 +315,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused CTR head fwd: deep_in assembly (+cast) + FM + first-order "
           "+ dense linear");
     m.def("ctr_head_bwd", &ctr_head_bwd, "fused CTR head backward");
+    m.def("flat_adagrad", &flat_adagrad,
+          "fused flat-buffer Adagrad (f32, or bf16 weights + f32 master)");
 }

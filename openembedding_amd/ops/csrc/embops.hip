@@ -597,6 +597,43 @@ __global__ void k_apply_opt(float* __restrict__ weights,
     }
 }
 
+// ---------------------------------------------------- flat dense optimizer
+// One fused elementwise pass for the flattened dense-parameter buffer:
+// bf16 form keeps an fp32 master + fp32 accumulator and writes the bf16
+// working weights (native-bf16 MLP: the MFMA GEMMs read bf16 weights that
+// are never re-cast per step); f32 form is plain Adagrad in one kernel
+// (replaces 3 torch elementwise launches).
+
+#include <hip/hip_bf16.h>
+typedef __hip_bfloat16 oebf16;
+
+__global__ void k_flat_adagrad_f32(float* __restrict__ p,
+                                   float* __restrict__ accum,
+                                   const float* __restrict__ g,
+                                   long n, float lr, float eps) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    float gi = g[i];
+    float a = accum[i] + gi * gi;
+    accum[i] = a;
+    p[i] -= lr * gi / (sqrtf(a) + eps);
+}
+
+__global__ void k_flat_adagrad_bf16(float* __restrict__ master,
+                                    float* __restrict__ accum,
+                                    const oebf16* __restrict__ g,
+                                    oebf16* __restrict__ p,
+                                    long n, float lr, float eps) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    float gi = (float)g[i];
+    float a = accum[i] + gi * gi;
+    accum[i] = a;
+    float m = master[i] - lr * gi / (sqrtf(a) + eps);
+    master[i] = m;
+    p[i] = (oebf16)m;
+}
+
 // ============================================================== launchers
 
 extern "C" {
@@ -733,6 +770,19 @@ void emb_apply_optimizer(int opt, float* weights, float* state, long dim,
         }
         (void)G;
     }
+}
+
+void emb_flat_adagrad_f32(float* p, float* accum, const float* g, long n,
+                          float lr, float eps, hipStream_t stream) {
+    if (n) k_flat_adagrad_f32<<<grid1d(n), BLOCK, 0, stream>>>(p, accum, g, n,
+                                                               lr, eps);
+}
+
+void emb_flat_adagrad_bf16(float* master, float* accum, const void* g,
+                           void* p, long n, float lr, float eps,
+                           hipStream_t stream) {
+    if (n) k_flat_adagrad_bf16<<<grid1d(n), BLOCK, 0, stream>>>(
+        master, accum, (const oebf16*)g, (oebf16*)p, n, lr, eps);
 }
 
 }  // extern "C"

@@ -316,38 +316,72 @@ class _FlatDenseAdagrad:
                   if p.requires_grad and p.numel() > 0]
         if not params:
             raise ValueError("no dense params")
-        dev, dt = params[0].device, params[0].dtype
-        if any(p.device != dev or p.dtype != dt for p in params):
-            raise ValueError("flat dense path needs one device/dtype")
-        total = sum(p.numel() for p in params)
-        self.flat = torch.empty(total, device=dev, dtype=dt)
-        self.flat_grad = torch.zeros(total, device=dev, dtype=dt)
-        off = 0
-        for p in params:
-            n = p.numel()
-            self.flat[off:off + n] = p.data.reshape(-1)
-            p.data = self.flat[off:off + n].view_as(p)
-            p.grad = self.flat_grad[off:off + n].view_as(p)
-            off += n
+        dev = params[0].device
+        if any(p.device != dev for p in params):
+            raise ValueError("flat dense path needs one device")
+        # one flat group per dtype: bf16 params get an fp32 master (the
+        # working bf16 weights feed MFMA GEMMs with no per-step casts;
+        # accumulator/update math is fp32)
+        self.groups = []
+        for dt in (torch.float32, torch.bfloat16):
+            ps = [p for p in params if p.dtype == dt]
+            if not ps:
+                continue
+            total = sum(p.numel() for p in ps)
+            flat = torch.empty(total, device=dev, dtype=dt)
+            flat_grad = torch.zeros(total, device=dev, dtype=dt)
+            off = 0
+            for p in ps:
+                n = p.numel()
+                flat[off:off + n] = p.data.reshape(-1)
+                p.data = flat[off:off + n].view_as(p)
+                p.grad = flat_grad[off:off + n].view_as(p)
+                off += n
+            grp = {
+                "dtype": dt,
+                "flat": flat,
+                "flat_grad": flat_grad,
+                "accum": torch.full((total,), float(init_acc),
+                                    device=dev, dtype=torch.float32),
+                "master": (flat.to(torch.float32)
+                           if dt == torch.bfloat16 else None),
+            }
+            self.groups.append(grp)
         self.params = params
-        self.accum = torch.full_like(self.flat, float(init_acc))
-        self._std = torch.empty_like(self.flat)
+        self._ext = None
+        if dev.type == "cuda":
+            from ..ops import require_hip
+            self._ext = require_hip()
 
     def zero_grad(self):
-        self.flat_grad.zero_()
+        for g in self.groups:
+            g["flat_grad"].zero_()
 
     def step(self):
-        self.accum.addcmul_(self.flat_grad, self.flat_grad)
-        torch.sqrt(self.accum, out=self._std)
-        self._std.add_(self.eps)
-        self.flat.addcdiv_(self.flat_grad, self._std, value=-self.lr)
+        for g in self.groups:
+            if self._ext is not None:
+                self._ext.flat_adagrad(g["flat"], g["accum"], g["flat_grad"],
+                                       g["master"], self.lr, self.eps)
+                continue
+            grad = g["flat_grad"].to(torch.float32)
+            g["accum"].addcmul_(grad, grad)
+            std = g["accum"].sqrt().add_(self.eps)
+            if g["master"] is not None:
+                g["master"].addcdiv_(grad, std, value=-self.lr)
+                g["flat"].copy_(g["master"])
+            else:
+                g["flat"].addcdiv_(grad, std, value=-self.lr)
 
     def state_dict(self):
-        return {"flat": self.flat, "accum": self.accum, "lr": self.lr}
+        return {"groups": [{k: v for k, v in g.items() if k != "flat_grad"}
+                           for g in self.groups], "lr": self.lr}
 
     def load_state_dict(self, sd):
-        self.flat.copy_(sd["flat"])
-        self.accum.copy_(sd["accum"])
+        for g, s in zip(self.groups, sd["groups"]):
+            g["flat"].copy_(s["flat"])
+            g["accum"].copy_(s["accum"])
+            if g["master"] is not None and s.get("master") is not None:
+                g["master"].copy_(s["master"])
 
 
 class DistributedOptimizer:
