@@ -421,9 +421,10 @@ class DistributedOptimizer:
         if not comm.dist_ready() or dist.get_world_size() == 1:
             return
         if self._flat is not None:
-            dist.all_reduce(self._flat.flat_grad)
-            if self.average_dense:
-                self._flat.flat_grad /= dist.get_world_size()
+            for g in self._flat.groups:
+                dist.all_reduce(g["flat_grad"])
+                if self.average_dense:
+                    g["flat_grad"] /= dist.get_world_size()
             return
         bucket: List[torch.Tensor] = []
         for group in self.optimizer.param_groups:
