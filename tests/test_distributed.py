@@ -116,7 +116,7 @@ def test_world2_matches_world1(tmp_path):
     torch.testing.assert_close(loaded, after0, rtol=1e-5, atol=1e-5)
 
 
-def _worker_model(rank, world, port, tmp):
+def _worker_model(rank, world, port, tmp, flatten=False):
     import openembedding_amd.torch as embed
     from openembedding_amd.models import DeepFM, synthetic_batch
 
@@ -125,7 +125,8 @@ def _worker_model(rank, world, port, tmp):
     fv = [50, 3, 1000, 40] + [100] * 22
     model = DeepFM(field_vocabs=fv, dim=4)
     opt = embed.distributed_optimizer(
-        torch.optim.Adagrad(model.parameters(), lr=0.01))
+        torch.optim.Adagrad(model.parameters(), lr=0.01),
+        flatten_dense=flatten)
     lossf = torch.nn.BCEWithLogitsLoss()
     g = torch.Generator().manual_seed(10 + rank)
     for step in range(3):
@@ -148,5 +149,15 @@ def _worker_model(rank, world, port, tmp):
 def test_deepfm_world2(tmp_path):
     port = _free_port()
     mp.spawn(_worker_model, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+    for v in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
+        os.environ.pop(v, None)
+
+
+@pytest.mark.timeout(300)
+def test_deepfm_world2_flat_dense(tmp_path):
+    """The bench path: flat dense buffer + allreduce on the flat grads."""
+    port = _free_port()
+    mp.spawn(_worker_model, args=(2, port, str(tmp_path), True), nprocs=2,
+             join=True)
     for v in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
         os.environ.pop(v, None)
