@@ -52,7 +52,17 @@ def main():
                         "loss and all optimizer state stay fp32.")
     p.add_argument("--graph", default="auto", choices=["auto", "on", "off"],
                    help="capture the train step in a hipGraph (single-GPU)")
+    p.add_argument("--hash", action="store_true",
+                   help="store embeddings in the open-addressed hash table "
+                        "(lazy rows) instead of the array table")
+    p.add_argument("--cache-mb", type=int, default=0,
+                   help="capacity tier: device row-cache budget in MB, cold "
+                        "rows spill to host DRAM (implies --hash)")
     args = p.parse_args()
+    if args.cache_mb:
+        args.hash = True
+        import openembedding_amd as oe
+        oe.flags.config = f"server:\n  cache_size: {args.cache_mb}\n"
 
     world = int(os.environ.get("WORLD_SIZE", 1))
     rank = int(os.environ.get("RANK", 0))
@@ -69,6 +79,8 @@ def main():
 
     torch.manual_seed(1234)  # identical dense init on all ranks
     kw = {} if args.model == "lr" else {"dim": args.dim}
+    if args.hash and args.model != "lr":
+        kw["hash_mode"] = True
     model = MODELS[args.model](**kw).to(device)
     if hasattr(model, "head_bf16"):
         model.head_bf16 = args.amp == "bf16"  # deep_in dtype follows amp
@@ -197,6 +209,8 @@ def main():
                 "fp32 accumulators; embeddings, FM math, loss fp32"
                 if (args.amp == "native" and on_gpu) else "all fp32"),
             "graph": graph is not None,
+            "table": "hash" if args.hash else "array",
+            "cache_mb": args.cache_mb,
             "parallelism": (f"dense-dp{n_gpus} + embedding sharded "
                             f"all_to_all" if n_gpus > 1 else "single-gpu"),
         },

@@ -72,12 +72,13 @@ class _CTRBase(nn.Module):
     per-row scalar states advance on the same touch pattern either way)."""
 
     def __init__(self, field_vocabs: Optional[List[int]], dim: int,
-                 sparse_as_dense_size: int = 0):
+                 sparse_as_dense_size: int = 0, hash_mode: bool = False):
         super().__init__()
         self.field_vocabs = list(field_vocabs or CRITEO_FIELD_VOCABS)
         self.dim = dim
         self.n_fields = len(self.field_vocabs)
-        self.embedding = CombinedEmbedding(self.field_vocabs, dim + 1)
+        self.embedding = CombinedEmbedding(self.field_vocabs, dim + 1,
+                                           hash_mode=hash_mode)
         self.dense_linear = nn.Linear(N_DENSE, 1)
         # fused interaction head on GPU (ctrhead.hip); deep_in dtype follows
         # head_bf16 (fp32 measured faster than bf16 autocast on MI355X for
@@ -122,8 +123,8 @@ class WDL(_CTRBase):
     """Wide & Deep (reference benchmark model 'WDL')."""
 
     def __init__(self, field_vocabs: Optional[List[int]] = None, dim: int = 9,
-                 hidden: Sequence[int] = (400, 400, 400)):
-        super().__init__(field_vocabs, dim)
+                 hidden: Sequence[int] = (400, 400, 400), **kw):
+        super().__init__(field_vocabs, dim, **kw)
         self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
 
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
@@ -139,8 +140,8 @@ class DeepFM(_CTRBase):
     """DeepFM (reference primary benchmark model, BASELINE.md DeepFM dim9)."""
 
     def __init__(self, field_vocabs: Optional[List[int]] = None, dim: int = 9,
-                 hidden: Sequence[int] = (400, 400, 400)):
-        super().__init__(field_vocabs, dim)
+                 hidden: Sequence[int] = (400, 400, 400), **kw):
+        super().__init__(field_vocabs, dim, **kw)
         self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
 
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
@@ -188,8 +189,8 @@ class xDeepFM(_CTRBase):
 
     def __init__(self, field_vocabs: Optional[List[int]] = None, dim: int = 9,
                  hidden: Sequence[int] = (400, 400, 400),
-                 cin_layers: Sequence[int] = (128, 128)):
-        super().__init__(field_vocabs, dim)
+                 cin_layers: Sequence[int] = (128, 128), **kw):
+        super().__init__(field_vocabs, dim, **kw)
         self.dnn = _mlp(self.n_fields * dim + N_DENSE, hidden)
         self.cin = CIN(self.n_fields, dim, cin_layers)
 

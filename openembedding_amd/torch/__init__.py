@@ -217,12 +217,15 @@ class CombinedEmbedding(Embedding):
     """
 
     def __init__(self, field_vocab_sizes: Sequence[int], embedding_dim: int,
-                 **kw):
+                 hash_mode: bool = False, **kw):
         sizes = list(field_vocab_sizes)
         offsets = [0]
         for s in sizes:
             offsets.append(offsets[-1] + int(s))
-        super().__init__(offsets[-1], embedding_dim, **kw)
+        # hash_mode: same offset key space, but stored in the open-addressed
+        # hash table (rows created lazily; enables the capacity tier)
+        super().__init__(-1 if hash_mode else offsets[-1], embedding_dim,
+                         **kw)
         self.n_fields = len(sizes)
         self.register_buffer(
             "field_offsets",
