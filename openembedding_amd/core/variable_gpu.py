@@ -136,6 +136,14 @@ class HipVariableShard(VariableShard):
     def _lookup_or_insert(self, keys: torch.Tensor, u_dev=None):
         n = keys.numel()
         if self.meta.use_hash_table:
+            if self._in_graph_capture():
+                # growth (slab realloc, rehash) needs host control; a
+                # captured insert would write past the frozen capacity on
+                # replay. Raising here makes graph capture fall back to
+                # eager (bench.py catches it). The kernel additionally
+                # bounds the slot counter by the slab capacity.
+                raise RuntimeError(
+                    "hash-table insert path is not hipGraph-capturable")
             self._ensure_rows(self._nrows_upper + n)
             self._maybe_rehash()
             slots, new_mask = self.ext.ht_lookup(self.tk, self.tv, keys,
@@ -232,10 +240,14 @@ class HipVariableShard(VariableShard):
             self.ext.apply_optimizer(opt_id, self.weights, self.state, slots,
                                      grads.contiguous(), counts, cfg, None)
         if self.meta.use_hash_table and not self._in_graph_capture():
-            # once per committed batch: tighten the row-count bound (one small
-            # D2H read; the only sync of the commit path)
-            self._nrows_exact = int(self.nrows_dev.item())
-            self._nrows_upper = self._nrows_exact
+            # every 16 commits: tighten the row-count bound (the D2H read
+            # syncs the stream — a per-step sync costs a pipeline bubble;
+            # between tightenings the bound only inflates by batch size)
+            self._commits = getattr(self, "_commits", 0) + 1
+            if self._commits % 16 == 0 or self._nrows_upper > (
+                    self.weights.shape[0] * 3) // 4:
+                self._nrows_exact = int(self.nrows_dev.item())
+                self._nrows_upper = self._nrows_exact
 
     @staticmethod
     def _in_graph_capture() -> bool:

@@ -95,7 +95,8 @@ class _CTRBase(nn.Module):
         return linear_w.sum(dim=1) + self.dense_linear(dense).squeeze(-1)
 
     def _use_fused_head(self, t: torch.Tensor) -> bool:
-        return t.is_cuda and self.dim >= 1 and hip_available()
+        # kernel limit: one wave lane per row column -> dim+1 <= 64
+        return (t.is_cuda and 1 <= self.dim + 1 <= 64 and hip_available())
 
     def _fused_head(self, dense: torch.Tensor, sparse: torch.Tensor,
                     use_fm: bool):

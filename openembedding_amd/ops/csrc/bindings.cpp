@@ -27,7 +27,7 @@ extern "C" {
 void emb_unique(const i64*, long, u64*, int*, long, int*, unsigned char*,
                 i64*, i64*, int*, hipStream_t_);
 void emb_ht_lookup(u64*, int*, long, const i64*, long, int*, i64*, i64*,
-                   unsigned char*, int, const int*, hipStream_t_);
+                   unsigned char*, int, const int*, long, hipStream_t_);
 void emb_ht_rehash(const u64*, const int*, long, u64*, int*, long,
                    hipStream_t_);
 void emb_array_touch(unsigned char*, const i64*, long, long, long, i64*,
@@ -125,7 +125,8 @@ std::tuple<torch::Tensor, torch::Tensor> ht_lookup(
                       tk.numel(), keys.data_ptr<i64>(), n,
                       nrows.data_ptr<int>(), slot_keys.data_ptr<i64>(),
                       slots.data_ptr<i64>(), new_mask.data_ptr<uint8_t>(),
-                      insert ? 1 : 0, u_ptr(u_dev), cur_stream());
+                      insert ? 1 : 0, u_ptr(u_dev), slot_keys.numel(),
+                      cur_stream());
     return {slots, new_mask};
 }
 

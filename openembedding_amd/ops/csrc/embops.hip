@@ -231,7 +231,8 @@ __global__ void k_ht_lookup(u64* __restrict__ tk, int* __restrict__ tv,
                             i64* __restrict__ slot_keys,
                             i64* __restrict__ slots,
                             unsigned char* __restrict__ new_mask,
-                            int insert, const int* __restrict__ u_dev) {
+                            int insert, const int* __restrict__ u_dev,
+                            long row_cap) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     if (u_dev && i >= *u_dev) return;
@@ -249,6 +250,11 @@ __global__ void k_ht_lookup(u64* __restrict__ tk, int* __restrict__ tv,
             u64 prev = atomicCAS(&tk[h], EMPTY, k);
             if (prev == EMPTY) {
                 int s = atomicAdd(nrows, 1);
+                if (s >= row_cap) {  // slab full (host sizing bug or a
+                    tk[h] = k;       // graph-captured insert): report miss
+                    slot = -1;       // instead of corrupting memory
+                    break;
+                }
                 tv[h] = s;
                 slot_keys[s] = (i64)k;
                 slot = s;
@@ -660,10 +666,11 @@ void emb_unique(const i64* keys, long n, u64* tk, int* tv, long cap,
 void emb_ht_lookup(u64* tk, int* tv, long cap, const i64* keys, long n,
                    int* nrows, i64* slot_keys, i64* slots,
                    unsigned char* new_mask, int insert, const int* u_dev,
-                   hipStream_t stream) {
+                   long row_cap, hipStream_t stream) {
     k_ht_lookup<<<grid1d(n), BLOCK, 0, stream>>>(tk, tv, cap - 1, keys, n,
                                                  nrows, slot_keys, slots,
-                                                 new_mask, insert, u_dev);
+                                                 new_mask, insert, u_dev,
+                                                 row_cap);
 }
 
 void emb_ht_rehash(const u64* tk_old, const int* tv_old, long cap_old,
