@@ -28,14 +28,15 @@ class _FusedCTRHeadFn(torch.autograd.Function):
     elementwise/reduce launches per step (profiles/)."""
 
     @staticmethod
-    def forward(ctx, e_all, dense, w, b, use_fm, out_bf16):
+    def forward(ctx, e_all, dense, w, b, use_fm, out_bf16, pad32=False):
         from ..ops import require_hip
         ext = require_hip()
         wf = w.reshape(-1).contiguous()
         e_all = e_all.contiguous()
         dense = dense.contiguous()
         deep_in, partial, s = ext.ctr_head_fwd(
-            e_all, dense, wf, b.reshape(-1).contiguous(), use_fm, out_bf16)
+            e_all, dense, wf, b.reshape(-1).contiguous(), use_fm, out_bf16,
+            pad32)
         ctx.save_for_backward(e_all, dense, wf, s)
         ctx.use_fm = use_fm
         return deep_in, partial
@@ -48,7 +49,48 @@ class _FusedCTRHeadFn(torch.autograd.Function):
         de_all, d_dense, dw, db = ext.ctr_head_bwd(
             e_all, dense, wf, d_deep_in.contiguous(),
             d_partial.contiguous(), s, ctx.use_fm)
-        return de_all, d_dense, dw.view(1, -1), db, None, None
+        return de_all, d_dense, dw.view(1, -1), db, None, None, None
+
+
+class _FusedMLP3Fn(torch.autograd.Function):
+    """Whole 3-hidden-layer MLP forward in ONE bf16 MFMA kernel
+    (ops/csrc/mlp.hip): activations flow layer-to-layer through LDS,
+    bias+ReLU fused, hidden activations saved for backward. Backward:
+    dgrad chain + wgrads via bf16 torch matmuls (wgrad shapes K=M are the
+    library-friendly ones)."""
+
+    @staticmethod
+    def forward(ctx, x0, w1, b1, w2, b2, w3, b3, w4, b4, w1pad):
+        from ..ops import require_hip
+        ext = require_hip()
+        w1pad[:, :w1.shape[1]].copy_(w1)   # refresh padded copy (16B rows)
+        out, a1, a2, a3 = ext.mlp3_fwd(x0, w1pad, b1, w2, b2, w3, b3,
+                                       w4.reshape(-1).contiguous(), b4)
+        ctx.save_for_backward(x0, w1, w2, w3, w4, a1, a2, a3)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        x0, w1, w2, w3, w4, a1, a2, a3 = ctx.saved_tensors
+        bf = a3.dtype
+        K0 = w1.shape[1]
+        dout = dout.contiguous()
+        d = dout.unsqueeze(1).to(bf)                   # [M, 1]
+        dz3 = (d * w4.reshape(1, -1)) * (a3 > 0)       # [M, H] bf16
+        dz2 = (dz3 @ w3) * (a2 > 0)
+        dz1 = (dz2 @ w2) * (a1 > 0)
+        dx0 = dz1 @ w1                                 # [M, K0]
+        if x0.shape[1] != K0:                          # pad tail grads = 0
+            dx0 = torch.nn.functional.pad(dx0, (0, x0.shape[1] - K0))
+        dw1 = dz1.t() @ x0[:, :K0]
+        dw2 = dz2.t() @ a1
+        dw3 = dz3.t() @ a2
+        dw4 = (d.t() @ a3)                             # [1, H]
+        db1 = dz1.sum(0)
+        db2 = dz2.sum(0)
+        db3 = dz3.sum(0)
+        db4 = d.sum(0)
+        return (dx0, dw1, db1, dw2, db2, dw3, db3, dw4, db4, None)
 
 
 def _mlp(in_dim: int, hidden: Sequence[int], out_dim: int = 1) -> nn.Sequential:
@@ -84,6 +126,10 @@ class _CTRBase(nn.Module):
         # head_bf16 (fp32 measured faster than bf16 autocast on MI355X for
         # these skinny GEMMs — gpurun_out/bench_fp32graph.log)
         self.head_bf16 = False
+        # whole-MLP fused kernel (mlp.hip): enabled by convert_mlp_bf16 when
+        # the dnn matches Linear/ReLU x3 + Linear(H,1), H <= 512 (% 16)
+        self.fused_mlp = False
+        self._w1pad = None
 
     def _embed(self, sparse: torch.Tensor):
         """-> (e [B,F,dim], linear_w [B,F])."""
@@ -95,17 +141,30 @@ class _CTRBase(nn.Module):
         return linear_w.sum(dim=1) + self.dense_linear(dense).squeeze(-1)
 
     def _use_fused_head(self, t: torch.Tensor) -> bool:
-        # kernel limit: one wave lane per row column -> dim+1 <= 64
-        return (t.is_cuda and 1 <= self.dim + 1 <= 64 and hip_available())
+        # kernel limit: each wave lane covers <=2 row columns -> dim+1 <= 128
+        return (t.is_cuda and 1 <= self.dim + 1 <= 128 and hip_available())
 
     def _fused_head(self, dense: torch.Tensor, sparse: torch.Tensor,
                     use_fm: bool):
-        """-> (e_all [B,F,dim+1], deep_in [B,F*dim+ND], partial [B])."""
+        """-> (e_all [B,F,dim+1], deep_in [B,F*dim+ND(+pad)], partial [B])."""
         e_all = self.embedding(sparse)
         deep_in, partial = _FusedCTRHeadFn.apply(
             e_all, dense, self.dense_linear.weight, self.dense_linear.bias,
-            use_fm, self.head_bf16)
+            use_fm, self.head_bf16, self.fused_mlp)
         return e_all, deep_in, partial
+
+    def _dnn_out(self, deep_in: torch.Tensor) -> torch.Tensor:
+        """dnn logits [B]; the fused single-kernel MLP when enabled."""
+        if self.fused_mlp and deep_in.is_cuda:
+            l1, l2, l3, l4 = self.dnn[0], self.dnn[2], self.dnn[4], self.dnn[6]
+            if self._w1pad is None or self._w1pad.shape[1] != deep_in.shape[1]:
+                self._w1pad = torch.zeros(
+                    l1.weight.shape[0], deep_in.shape[1],
+                    dtype=deep_in.dtype, device=deep_in.device)
+            return _FusedMLP3Fn.apply(
+                deep_in, l1.weight, l1.bias, l2.weight, l2.bias,
+                l3.weight, l3.bias, l4.weight, l4.bias, self._w1pad)
+        return self.dnn(deep_in).squeeze(-1)
 
 
 class LR(_CTRBase):
@@ -131,7 +190,7 @@ class WDL(_CTRBase):
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
         if self._use_fused_head(dense):
             _, deep_in, partial = self._fused_head(dense, sparse, use_fm=False)
-            return partial + self.dnn(deep_in).squeeze(-1)
+            return partial + self._dnn_out(deep_in)
         e, lin = self._embed(sparse)                     # [B, F, d], [B, F]
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
         return self._first_order(dense, lin) + self.dnn(deep_in).squeeze(-1)
@@ -148,7 +207,7 @@ class DeepFM(_CTRBase):
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
         if self._use_fused_head(dense):
             _, deep_in, partial = self._fused_head(dense, sparse, use_fm=True)
-            return partial + self.dnn(deep_in).squeeze(-1)
+            return partial + self._dnn_out(deep_in)
         e, lin = self._embed(sparse)                     # [B, F, d], [B, F]
         # FM second order: 0.5*((sum_f e)^2 - sum_f e^2) summed over dim
         s = e.sum(dim=1)
@@ -200,7 +259,7 @@ class xDeepFM(_CTRBase):
             e_all, deep_in, partial = self._fused_head(dense, sparse,
                                                        use_fm=False)
             e = e_all[..., :self.dim]
-            return partial + self.cin(e) + self.dnn(deep_in).squeeze(-1)
+            return partial + self.cin(e) + self._dnn_out(deep_in)
         e, lin = self._embed(sparse)
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
         return (self._first_order(dense, lin) + self.cin(e)
@@ -215,6 +274,15 @@ def convert_mlp_bf16(model: _CTRBase) -> _CTRBase:
     loss stay fp32."""
     if hasattr(model, "dnn"):
         model.dnn.to(torch.bfloat16)
+        seq = model.dnn
+        ok = (len(seq) == 7
+              and all(isinstance(seq[i], nn.Linear) for i in (0, 2, 4, 6))
+              and seq[0].out_features == seq[2].out_features
+              == seq[4].out_features == seq[6].in_features
+              and seq[6].out_features == 1
+              and seq[0].out_features % 16 == 0
+              and seq[0].out_features <= 512)
+        model.fused_mlp = ok
     model.head_bf16 = True
     return model
 

@@ -42,15 +42,20 @@ void emb_apply_optimizer(int, float*, float*, long, long, const i64*, long,
                          const float*, const u64*, const float*, const int*,
                          hipStream_t_);
 void emb_ctr_head_fwd(const float*, const float*, const float*, const float*, long,
-                      long, long, long, void*, float*, float*, int, int,
-                      hipStream_t_);
+                      long, long, long, long, void*, float*, float*, int,
+                      int, hipStream_t_);
 void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
                       const float*, const float*, long, long, long, long,
-                      float*, float*, float*, float*, int, int, hipStream_t_);
+                      long, float*, float*, float*, float*, int, int,
+                      hipStream_t_);
 void emb_flat_adagrad_f32(float*, float*, const float*, long, float, float,
                           hipStream_t_);
 void emb_flat_adagrad_bf16(float*, float*, const void*, void*, long, float,
                            float, hipStream_t_);
+void emb_mlp3_fwd(const void*, long, long, const void*, const void*,
+                  const void*, const void*, const void*, const void*,
+                  const void*, const void*, long, void*, void*, void*,
+                  float*, hipStream_t_);
 }
 
 namespace {
@@ -231,22 +236,24 @@ void apply_optimizer(int64_t opt, torch::Tensor weights, torch::Tensor state,
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ctr_head_fwd(
     torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
-    torch::Tensor bias, bool use_fm, bool out_bf16) {
+    torch::Tensor bias, bool use_fm, bool out_bf16, bool pad32) {
     CHECK_GPU(e_all); CHECK_CONT(e_all); CHECK_CONT(dense); CHECK_CONT(w);
     const c10::cuda::CUDAGuard guard(e_all.device());
     long B = e_all.size(0), F = e_all.size(1), D1 = e_all.size(2);
     long dim = D1 - 1, nd = dense.size(1);
-    TORCH_CHECK(D1 <= 64, "ctr_head: dim+1 must be <= 64");
+    TORCH_CHECK(D1 <= 128, "ctr_head: dim+1 must be <= 128");
     TORCH_CHECK(nd <= 32, "ctr_head: dense features must be <= 32");
     auto out_opts = e_all.options().dtype(out_bf16 ? torch::kBFloat16
                                                    : torch::kFloat32);
-    auto deep_in = torch::empty({B, F * dim + nd}, out_opts);
+    long width = F * dim + nd;
+    long stride = pad32 ? ((width + 31) / 32) * 32 : width;
+    auto deep_in = torch::empty({B, stride}, out_opts);
     auto partial = torch::empty({B}, e_all.options());
     auto s_out = torch::empty({B, dim}, e_all.options());
     CHECK_CONT(bias);
     emb_ctr_head_fwd(e_all.data_ptr<float>(), dense.data_ptr<float>(),
                      w.data_ptr<float>(), bias.data_ptr<float>(), B, F, dim, nd,
-                     deep_in.data_ptr(), partial.data_ptr<float>(),
+                     stride, deep_in.data_ptr(), partial.data_ptr<float>(),
                      s_out.data_ptr<float>(),
                      use_fm ? 1 : 0, out_bf16 ? 1 : 0, cur_stream());
     return {deep_in, partial, s_out};
@@ -260,6 +267,7 @@ ctr_head_bwd(torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
     const c10::cuda::CUDAGuard guard(e_all.device());
     long B = e_all.size(0), F = e_all.size(1), D1 = e_all.size(2);
     long dim = D1 - 1, nd = dense.size(1);
+    long stride = d_deep_in.size(1);  // may be 32-padded
     bool out_bf16 = d_deep_in.dtype() == torch::kBFloat16;
     auto de_all = torch::empty_like(e_all);
     auto d_dense = torch::empty_like(dense);
@@ -268,7 +276,7 @@ ctr_head_bwd(torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
     emb_ctr_head_bwd(e_all.data_ptr<float>(), dense.data_ptr<float>(),
                      w.data_ptr<float>(), d_deep_in.data_ptr(),
                      d_partial.data_ptr<float>(),
-                     s_in.data_ptr<float>(), B, F, dim, nd,
+                     s_in.data_ptr<float>(), B, F, dim, nd, stride,
                      de_all.data_ptr<float>(), d_dense.data_ptr<float>(),
                      dw.data_ptr<float>(), db.data_ptr<float>(),
                      use_fm ? 1 : 0, out_bf16 ? 1 : 0, cur_stream());
@@ -297,6 +305,30 @@ void flat_adagrad(torch::Tensor param, torch::Tensor accum,
     }
 }
 
+// ---- fused 3-layer MLP forward ----------------------------------------
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+mlp3_fwd(torch::Tensor x0, torch::Tensor w1, torch::Tensor b1,
+         torch::Tensor w2, torch::Tensor b2, torch::Tensor w3,
+         torch::Tensor b3, torch::Tensor w4, torch::Tensor b4) {
+    CHECK_GPU(x0); CHECK_CONT(x0);
+    TORCH_CHECK(x0.dtype() == torch::kBFloat16, "mlp3_fwd: x0 must be bf16");
+    const c10::cuda::CUDAGuard guard(x0.device());
+    long M = x0.size(0), K0 = x0.size(1), H = w1.size(0);
+    TORCH_CHECK(H % 16 == 0 && H <= 512, "mlp3_fwd: H must be <=512, x16");
+    TORCH_CHECK(w2.size(0) == H && w3.size(0) == H && w4.size(1) == H);
+    auto a1 = torch::empty({M, H}, x0.options());
+    auto a2 = torch::empty({M, H}, x0.options());
+    auto a3 = torch::empty({M, H}, x0.options());
+    auto out = torch::empty({M}, x0.options().dtype(torch::kFloat32));
+    emb_mlp3_fwd(x0.data_ptr(), M, K0,
+                 w1.data_ptr(), b1.data_ptr(), w2.data_ptr(), b2.data_ptr(),
+                 w3.data_ptr(), b3.data_ptr(), w4.data_ptr(), b4.data_ptr(),
+                 H, a1.data_ptr(), a2.data_ptr(), a3.data_ptr(),
+                 out.data_ptr<float>(), cur_stream());
+    return {out, a1, a2, a3};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -316,6 +348,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused CTR head fwd: deep_in assembly (+cast) + FM + first-order "
           "+ dense linear");
     m.def("ctr_head_bwd", &ctr_head_bwd, "fused CTR head backward");
+    m.def("mlp3_fwd", &mlp3_fwd,
+          "fused 3-hidden-layer MLP forward (bf16 MFMA, bias+ReLU fused)");
     m.def("flat_adagrad", &flat_adagrad,
           "fused flat-buffer Adagrad (f32, or bf16 weights + f32 master)");
 }

@@ -38,6 +38,7 @@ __global__ void k_ctr_head_fwd(const float* __restrict__ e_all,
                                const float* __restrict__ w,  // [nd]
                                const float* __restrict__ bias,  // [1]
                                long B, long F, long dim, long nd,
+                               long out_stride,
                                OutT* __restrict__ deep_in,
                                float* __restrict__ partial,
                                float* __restrict__ s_out,  // [B, dim] for bwd
@@ -47,19 +48,27 @@ __global__ void k_ctr_head_fwd(const float* __restrict__ e_all,
     const long b = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     if (b >= B) return;
     const float* e = e_all + b * F * D1;
-    OutT* di = deep_in + b * (F * dim + nd);
+    OutT* di = deep_in + b * out_stride;
+    // zero the alignment pad tail (the fused MLP reads 16-byte fragments)
+    for (long j = F * dim + nd + lane; j < out_stride; j += 64)
+        di[j] = (OutT)0.0f;
 
-    float s = 0.0f, sq = 0.0f;   // per-column running sums (lane c)
-    const bool emb_col = lane < dim;
-    const bool wide_col = lane == dim;
+    // lane covers columns lane and lane+64 (D1 <= 128 — covers the
+    // reference's dim-64 benchmark config, D1 = 65)
+    float s[2] = {0.0f, 0.0f}, sq[2] = {0.0f, 0.0f};
     for (long f = 0; f < F; ++f) {
-        float v = (lane < D1) ? e[f * D1 + lane] : 0.0f;
-        if (emb_col) {
-            s += v;
-            sq += v * v;
-            di[f * dim + lane] = (OutT)v;
-        } else if (wide_col) {
-            s += v;                       // first-order sum
+        #pragma unroll
+        for (int t = 0; t < 2; ++t) {
+            const long cc = lane + 64 * t;
+            if (cc >= D1) break;
+            float v = e[f * D1 + cc];
+            if (cc < dim) {
+                s[t] += v;
+                sq[t] += v * v;
+                di[f * dim + cc] = (OutT)v;
+            } else {
+                s[t] += v;                // first-order (wide) column
+            }
         }
     }
     // dense tail: cast + dot
@@ -69,10 +78,18 @@ __global__ void k_ctr_head_fwd(const float* __restrict__ e_all,
         di[F * dim + j] = (OutT)v;
         dsum += v * w[j];
     }
-    if (emb_col && s_out) s_out[b * dim + lane] = s;
-    // wave reduction: fm2 over lanes<dim, lin from lane dim, dsum over all
-    float fm = emb_col && use_fm ? (s * s - sq) : 0.0f;
-    float lin = wide_col ? s : 0.0f;
+    float fm = 0.0f, lin = 0.0f;
+    #pragma unroll
+    for (int t = 0; t < 2; ++t) {
+        const long cc = lane + 64 * t;
+        if (cc >= D1) break;
+        if (cc < dim) {
+            if (use_fm) fm += s[t] * s[t] - sq[t];
+            if (s_out) s_out[b * dim + cc] = s[t];
+        } else {
+            lin = s[t];
+        }
+    }
     float acc = 0.5f * fm + lin + dsum;
     for (int off = 32; off; off >>= 1)
         acc += __shfl_down(acc, off, 64);
@@ -89,7 +106,7 @@ __global__ void k_ctr_head_bwd_e(const float* __restrict__ e_all,
                                  const float* __restrict__ d_partial,
                                  const float* __restrict__ s_in, // [B, dim]
                                  long B, long F, long dim,
-                                 long nd,
+                                 long nd, long out_stride,
                                  float* __restrict__ de_all,
                                  int use_fm) {
     const long D1 = dim + 1;
@@ -100,7 +117,7 @@ __global__ void k_ctr_head_bwd_e(const float* __restrict__ e_all,
     const long b = i / (D1 * F);
     const float gp = d_partial[b];
     if (c == dim) { de_all[i] = gp; return; }        // wide column
-    float g = (float)d_deep_in[b * (F * dim + nd) + f * dim + c];
+    float g = (float)d_deep_in[b * out_stride + f * dim + c];
     if (use_fm) g += gp * (s_in[b * dim + c] - e_all[i]);
     de_all[i] = g;
 }
@@ -116,6 +133,7 @@ __global__ void k_ctr_head_bwd_d(const float* __restrict__ dense,
                                  const OutT* __restrict__ d_deep_in,
                                  const float* __restrict__ d_partial,
                                  long B, long F, long dim, long nd,
+                                 long out_stride,
                                  float* __restrict__ d_dense,
                                  float* __restrict__ dw,   // [nd] atomic
                                  float* __restrict__ db) { // [1] atomic
@@ -127,7 +145,7 @@ __global__ void k_ctr_head_bwd_d(const float* __restrict__ dense,
     float accb = 0.0f;
     for (long b = (long)blockIdx.x * blockDim.x + threadIdx.x; b < B;
          b += (long)gridDim.x * blockDim.x) {
-        const OutT* ddi = d_deep_in + b * (F * dim + nd);
+        const OutT* ddi = d_deep_in + b * out_stride;
         const float gp = d_partial[b];
         accb += gp;
         for (long j = 0; j < nd; ++j) {
@@ -149,25 +167,26 @@ extern "C" {
 
 void emb_ctr_head_fwd(const float* e_all, const float* dense, const float* w,
                       const float* bias, long B, long F, long dim, long nd,
-                      void* deep_in, float* partial, float* s_out,
-                      int use_fm, int out_bf16, hipStream_t stream) {
+                      long out_stride, void* deep_in, float* partial,
+                      float* s_out, int use_fm, int out_bf16,
+                      hipStream_t stream) {
     if (B == 0) return;
     int block = 256;                    // 4 waves per block
     long grid = (B * 64 + block - 1) / block;
     if (out_bf16)
         k_ctr_head_fwd<bf16><<<(int)grid, block, 0, stream>>>(
-            e_all, dense, w, bias, B, F, dim, nd, (bf16*)deep_in, partial,
-            s_out, use_fm);
+            e_all, dense, w, bias, B, F, dim, nd, out_stride,
+            (bf16*)deep_in, partial, s_out, use_fm);
     else
         k_ctr_head_fwd<float><<<(int)grid, block, 0, stream>>>(
-            e_all, dense, w, bias, B, F, dim, nd, (float*)deep_in, partial,
-            s_out, use_fm);
+            e_all, dense, w, bias, B, F, dim, nd, out_stride,
+            (float*)deep_in, partial, s_out, use_fm);
 }
 
 void emb_ctr_head_bwd(const float* e_all, const float* dense, const float* w,
                       const void* d_deep_in, const float* d_partial,
                       const float* s_in,
-                      long B, long F, long dim, long nd,
+                      long B, long F, long dim, long nd, long out_stride,
                       float* de_all, float* d_dense, float* dw, float* db,
                       int use_fm, int out_bf16, hipStream_t stream) {
     if (B == 0) return;
@@ -180,17 +199,17 @@ void emb_ctr_head_bwd(const float* e_all, const float* dense, const float* w,
     if (out_bf16) {
         k_ctr_head_bwd_e<bf16><<<(int)grid_e, block, 0, stream>>>(
             e_all, (const bf16*)d_deep_in, d_partial, s_in, B, F, dim, nd,
-            de_all, use_fm);
+            out_stride, de_all, use_fm);
         k_ctr_head_bwd_d<bf16><<<(int)grid_d, block, 0, stream>>>(
             dense, w, (const bf16*)d_deep_in, d_partial, B, F, dim, nd,
-            d_dense, dw, db);
+            out_stride, d_dense, dw, db);
     } else {
         k_ctr_head_bwd_e<float><<<(int)grid_e, block, 0, stream>>>(
             e_all, (const float*)d_deep_in, d_partial, s_in, B, F, dim, nd,
-            de_all, use_fm);
+            out_stride, de_all, use_fm);
         k_ctr_head_bwd_d<float><<<(int)grid_d, block, 0, stream>>>(
             dense, w, (const float*)d_deep_in, d_partial, B, F, dim, nd,
-            d_dense, dw, db);
+            out_stride, d_dense, dw, db);
     }
 }
 

@@ -77,7 +77,6 @@ static __device__ __forceinline__ void layer(
     const int cstart = wave >> 1;               // 0..3
     for (long c = cstart * 16; c < H; c += 64) {
         f32x4 acc = tile_16x16(A, lda, half * 16, W, K, c, K, lane);
-        // NOTE: tile_16x16(A, lda, m0, W, ldw, n0, K, lane)
         const int col = lane & 15;
         float bv = (float)bias[c + col];
         #pragma unroll

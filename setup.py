@@ -21,6 +21,7 @@ ext = CUDAExtension(
         "openembedding_amd/ops/csrc/bindings.cpp",
         "openembedding_amd/ops/csrc/embops.hip",
         "openembedding_amd/ops/csrc/ctrhead.hip",
+        "openembedding_amd/ops/csrc/mlp.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3"],

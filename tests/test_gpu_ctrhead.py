@@ -23,7 +23,7 @@ def _torch_head(e_all, dense, w, b, use_fm):
 
 @pytest.mark.parametrize("use_fm", [True, False])
 @pytest.mark.parametrize("out_bf16", [False, True])
-@pytest.mark.parametrize("dim,nd", [(9, 13), (4, 13), (63, 7)])
+@pytest.mark.parametrize("dim,nd", [(9, 13), (4, 13), (63, 7), (64, 13), (127, 5)])
 def test_head_matches_torch(use_fm, out_bf16, dim, nd):
     from openembedding_amd.models.ctr import _FusedCTRHeadFn
 
