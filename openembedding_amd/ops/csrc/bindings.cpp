@@ -316,7 +316,7 @@ mlp3_fwd(torch::Tensor x0, torch::Tensor w1, torch::Tensor b1,
     const c10::cuda::CUDAGuard guard(x0.device());
     long M = x0.size(0), K0 = x0.size(1), H = w1.size(0);
     TORCH_CHECK(H % 16 == 0 && H <= 512, "mlp3_fwd: H must be <=512, x16");
-    TORCH_CHECK(w2.size(0) == H && w3.size(0) == H && w4.size(1) == H);
+    TORCH_CHECK(w2.size(0) == H && w3.size(0) == H && w4.numel() == H);
     auto a1 = torch::empty({M, H}, x0.options());
     auto a2 = torch::empty({M, H}, x0.options());
     auto a3 = torch::empty({M, H}, x0.options());
