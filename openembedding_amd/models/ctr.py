@@ -60,36 +60,44 @@ class _FusedMLP3Fn(torch.autograd.Function):
     library-friendly ones)."""
 
     @staticmethod
-    def forward(ctx, x0, w1, b1, w2, b2, w3, b3, w4, b4, w1pad):
+    def forward(ctx, x0, w1, b1, w2, b2, w3, b3, w4, b4, bufs):
         from ..ops import require_hip
         ext = require_hip()
-        w1pad[:, :w1.shape[1]].copy_(w1)   # refresh padded copy (16B rows)
-        out, a1, a2, a3 = ext.mlp3_fwd(x0, w1pad, b1, w2, b2, w3, b3,
-                                       w4.reshape(-1).contiguous(), b4)
-        ctx.save_for_backward(x0, w1, w2, w3, w4, a1, a2, a3)
+        # refresh padded weight copies (zero tails allocated once in bufs)
+        bufs["w1p"][:, :w1.shape[1]].copy_(w1)
+        bufs["w2p"][:, :w2.shape[1]].copy_(w2)
+        bufs["w3p"][:, :w3.shape[1]].copy_(w3)
+        w4f = w4.reshape(-1).contiguous()
+        out, a1, a2, a3 = ext.mlp3_fwd(x0, bufs["w1p"], b1, bufs["w2p"], b2,
+                                       bufs["w3p"], b3, w4f, b4)
+        ctx.save_for_backward(x0, w1, w2, w3, w4f, a1, a2, a3)
+        ctx.bufs = bufs
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        x0, w1, w2, w3, w4, a1, a2, a3 = ctx.saved_tensors
-        bf = a3.dtype
+        from ..ops import require_hip
+        ext = require_hip()
+        x0, w1, w2, w3, w4f, a1, a2, a3 = ctx.saved_tensors
+        bufs = ctx.bufs
         K0 = w1.shape[1]
-        dout = dout.contiguous()
-        d = dout.unsqueeze(1).to(bf)                   # [M, 1]
-        dz3 = (d * w4.reshape(1, -1)) * (a3 > 0)       # [M, H] bf16
-        dz2 = (dz3 @ w3) * (a2 > 0)
-        dz1 = (dz2 @ w2) * (a1 > 0)
-        dx0 = dz1 @ w1                                 # [M, K0]
-        if x0.shape[1] != K0:                          # pad tail grads = 0
-            dx0 = torch.nn.functional.pad(dx0, (0, x0.shape[1] - K0))
+        # padded transposed weights for the fused dgrad chain
+        bufs["w3tp"][:, :w3.shape[0]].copy_(w3.t())
+        bufs["w2tp"][:, :w2.shape[0]].copy_(w2.t())
+        bufs["w1tp"][:, :w1.shape[0]].copy_(bufs["w1p"].t())
+        dx0, dz1, dz2, dz3 = ext.mlp3_bwd(
+            dout.contiguous(), a1, a2, a3, w4f,
+            bufs["w3tp"], bufs["w2tp"], bufs["w1tp"])
+        # wgrads stay library GEMMs — their K = batch shapes run well
         dw1 = dz1.t() @ x0[:, :K0]
         dw2 = dz2.t() @ a1
         dw3 = dz3.t() @ a2
-        dw4 = (d.t() @ a3)                             # [1, H]
+        d = dout.unsqueeze(0).to(a3.dtype)             # [1, M]
+        dw4 = d @ a3                                   # [1, H]
         db1 = dz1.sum(0)
         db2 = dz2.sum(0)
         db3 = dz3.sum(0)
-        db4 = d.sum(0)
+        db4 = d.sum(1)
         return (dx0, dw1, db1, dw2, db2, dw3, db3, dw4, db4, None)
 
 
@@ -157,10 +165,15 @@ class _CTRBase(nn.Module):
         """dnn logits [B]; the fused single-kernel MLP when enabled."""
         if self.fused_mlp and deep_in.is_cuda:
             l1, l2, l3, l4 = self.dnn[0], self.dnn[2], self.dnn[4], self.dnn[6]
-            if self._w1pad is None or self._w1pad.shape[1] != deep_in.shape[1]:
-                self._w1pad = torch.zeros(
-                    l1.weight.shape[0], deep_in.shape[1],
-                    dtype=deep_in.dtype, device=deep_in.device)
+            K0p = deep_in.shape[1]
+            if self._w1pad is None or self._w1pad["w1p"].shape[1] != K0p:
+                H = l1.weight.shape[0]
+                Hp = (H + 31) // 32 * 32
+                z = lambda *s: torch.zeros(*s, dtype=deep_in.dtype,  # noqa: E731
+                                           device=deep_in.device)
+                self._w1pad = {"w1p": z(H, K0p), "w2p": z(H, Hp),
+                               "w3p": z(H, Hp), "w3tp": z(H, Hp),
+                               "w2tp": z(H, Hp), "w1tp": z(K0p, Hp)}
             return _FusedMLP3Fn.apply(
                 deep_in, l1.weight, l1.bias, l2.weight, l2.bias,
                 l3.weight, l3.bias, l4.weight, l4.bias, self._w1pad)
@@ -281,7 +294,7 @@ def convert_mlp_bf16(model: _CTRBase) -> _CTRBase:
               == seq[4].out_features == seq[6].in_features
               and seq[6].out_features == 1
               and seq[0].out_features % 16 == 0
-              and seq[0].out_features <= 512)
+              and seq[0].out_features <= 416)
         model.fused_mlp = ok
     model.head_bf16 = True
     return model

@@ -54,8 +54,12 @@ void emb_flat_adagrad_bf16(float*, float*, const void*, void*, long, float,
                            float, hipStream_t_);
 void emb_mlp3_fwd(const void*, long, long, const void*, const void*,
                   const void*, const void*, const void*, const void*,
-                  const void*, const void*, long, void*, void*, void*,
+                  const void*, const void*, long, long, void*, void*, void*,
                   float*, hipStream_t_);
+void emb_mlp3_bwd(const float*, long, long, const void*, const void*,
+                  const void*, const void*, const void*, const void*,
+                  const void*, long, long, void*, void*, void*, void*,
+                  hipStream_t_);
 }
 
 namespace {
@@ -314,19 +318,45 @@ mlp3_fwd(torch::Tensor x0, torch::Tensor w1, torch::Tensor b1,
     CHECK_GPU(x0); CHECK_CONT(x0);
     TORCH_CHECK(x0.dtype() == torch::kBFloat16, "mlp3_fwd: x0 must be bf16");
     const c10::cuda::CUDAGuard guard(x0.device());
-    long M = x0.size(0), K0 = x0.size(1), H = w1.size(0);
-    TORCH_CHECK(H % 16 == 0 && H <= 512, "mlp3_fwd: H must be <=512, x16");
-    TORCH_CHECK(w2.size(0) == H && w3.size(0) == H && w4.numel() == H);
+    long M = x0.size(0), K0p = x0.size(1), H = w1.size(0);
+    long Hp = w2.size(1);
+    TORCH_CHECK(H % 16 == 0 && H <= 416, "mlp3_fwd: H must be <=416, x16");
+    TORCH_CHECK(K0p % 32 == 0 && Hp % 32 == 0,
+                "mlp3_fwd: padded K dims must be x32");
+    TORCH_CHECK(w1.size(1) == K0p && w2.size(0) == H && w3.size(0) == H
+                && w3.size(1) == Hp && w4.numel() == H);
     auto a1 = torch::empty({M, H}, x0.options());
     auto a2 = torch::empty({M, H}, x0.options());
     auto a3 = torch::empty({M, H}, x0.options());
     auto out = torch::empty({M}, x0.options().dtype(torch::kFloat32));
-    emb_mlp3_fwd(x0.data_ptr(), M, K0,
+    emb_mlp3_fwd(x0.data_ptr(), M, K0p,
                  w1.data_ptr(), b1.data_ptr(), w2.data_ptr(), b2.data_ptr(),
                  w3.data_ptr(), b3.data_ptr(), w4.data_ptr(), b4.data_ptr(),
-                 H, a1.data_ptr(), a2.data_ptr(), a3.data_ptr(),
+                 H, Hp, a1.data_ptr(), a2.data_ptr(), a3.data_ptr(),
                  out.data_ptr<float>(), cur_stream());
     return {out, a1, a2, a3};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+mlp3_bwd(torch::Tensor dout, torch::Tensor a1, torch::Tensor a2,
+         torch::Tensor a3, torch::Tensor w4, torch::Tensor w3t,
+         torch::Tensor w2t, torch::Tensor w1t) {
+    CHECK_GPU(dout); CHECK_CONT(dout); CHECK_CONT(w3t); CHECK_CONT(w2t);
+    CHECK_CONT(w1t);
+    const c10::cuda::CUDAGuard guard(dout.device());
+    long M = a1.size(0), H = a1.size(1), K0p = w1t.size(0);
+    long Hp = w3t.size(1);
+    TORCH_CHECK(Hp % 32 == 0 && w2t.size(1) == Hp && w1t.size(1) == Hp);
+    auto dz1 = torch::empty_like(a1);
+    auto dz2 = torch::empty_like(a2);
+    auto dz3 = torch::empty_like(a3);
+    auto dx0 = torch::empty({M, K0p}, a1.options());
+    emb_mlp3_bwd(dout.data_ptr<float>(), M, K0p, a1.data_ptr(),
+                 a2.data_ptr(), a3.data_ptr(), w4.data_ptr(), w3t.data_ptr(),
+                 w2t.data_ptr(), w1t.data_ptr(), H, Hp, dz1.data_ptr(),
+                 dz2.data_ptr(), dz3.data_ptr(), dx0.data_ptr(),
+                 cur_stream());
+    return {dx0, dz1, dz2, dz3};
 }
 
 }  // namespace
@@ -348,6 +378,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused CTR head fwd: deep_in assembly (+cast) + FM + first-order "
           "+ dense linear");
     m.def("ctr_head_bwd", &ctr_head_bwd, "fused CTR head backward");
+    m.def("mlp3_bwd", &mlp3_bwd,
+          "fused dgrad chain backward of the 3-layer MLP");
     m.def("mlp3_fwd", &mlp3_fwd,
           "fused 3-hidden-layer MLP forward (bf16 MFMA, bias+ReLU fused)");
     m.def("flat_adagrad", &flat_adagrad,

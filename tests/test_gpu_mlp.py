@@ -26,6 +26,7 @@ def test_mlp3_fwd_matches_torch(M, K0):
     torch.manual_seed(0)
     H = 400
     K0p = (K0 + 31) // 32 * 32
+    Hp = (H + 31) // 32 * 32
     x0 = torch.zeros(M, K0p, device=DEV, dtype=torch.bfloat16)
     x0[:, :K0] = (torch.randn(M, K0, device=DEV) * 0.5).to(torch.bfloat16)
     ws = [(torch.randn(H, K0p, device=DEV) * 0.05).to(torch.bfloat16),
@@ -36,8 +37,12 @@ def test_mlp3_fwd_matches_torch(M, K0):
     bs = [(torch.randn(H, device=DEV) * 0.1).to(torch.bfloat16)
           for _ in range(3)] + [(torch.randn(1, device=DEV) * 0.1
                                  ).to(torch.bfloat16)]
-    out, a1, a2, a3 = ext.mlp3_fwd(x0, ws[0], bs[0], ws[1], bs[1],
-                                   ws[2], bs[2], ws[3].reshape(-1), bs[3])
+    w2p = torch.zeros(H, Hp, device=DEV, dtype=torch.bfloat16)
+    w2p[:, :H] = ws[1]
+    w3p = torch.zeros(H, Hp, device=DEV, dtype=torch.bfloat16)
+    w3p[:, :H] = ws[2]
+    out, a1, a2, a3 = ext.mlp3_fwd(x0, ws[0], bs[0], w2p, bs[1],
+                                   w3p, bs[2], ws[3].reshape(-1), bs[3])
     ref_out, ref_acts = _torch_mlp(x0, ws, bs)
     # bf16 accumulation differences: compare with bf16-level tolerance
     for got, ref, name in [(a1.float(), ref_acts[0].float(), "a1"),
@@ -62,8 +67,11 @@ def test_fused_mlp_function_grads():
              * (0.05 if len(shape) == 2 else 0.1)).to(torch.bfloat16)
         t.requires_grad_(True)
         params.append(t)
-    w1pad = torch.zeros(H, K0p, device=DEV, dtype=torch.bfloat16)
-    out = _FusedMLP3Fn.apply(x0, *params, w1pad)
+    Hp = (H + 31) // 32 * 32
+    z = lambda *s: torch.zeros(*s, device=DEV, dtype=torch.bfloat16)  # noqa: E731
+    bufs = {"w1p": z(H, K0p), "w2p": z(H, Hp), "w3p": z(H, Hp),
+            "w3tp": z(H, Hp), "w2tp": z(H, Hp), "w1tp": z(K0p, Hp)}
+    out = _FusedMLP3Fn.apply(x0, *params, bufs)
     g = torch.randn(M, device=DEV)
     (out * g).sum().backward()
     got = [t.grad.clone() for t in [x0] + params]
