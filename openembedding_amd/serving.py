@@ -284,6 +284,24 @@ def make_app(controller: Optional[ModelController] = None,
         controller.shutdown_node(node_id)
         return {"shutdown_requested": node_id}
 
+    @app.get("/metrics")
+    def metrics():
+        """Prometheus-style exposition of the process accumulators
+        (reference server metrics exposer, entry/server.cc:7-12,35-36)."""
+        from starlette.responses import PlainTextResponse
+
+        from .utils.metrics import REGISTRY
+        lines = ["# TYPE openembedding_metric gauge"]
+        for name in REGISTRY.names():
+            a = REGISTRY.accumulator(name)
+            safe = name.replace(".", "_").replace("-", "_")
+            lines.append(f'openembedding_metric{{name="{safe}",stat="count"}}'
+                         f' {a.n}')
+            lines.append(f'openembedding_metric{{name="{safe}",stat="sum"}}'
+                         f' {a.total}')
+        lines.append(f'openembedding_models {len(controller.manager.signs())}')
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.post("/models/{sign}/variables/{variable_id}/pull")
     def pull(sign: str, variable_id: int, req: PullReq):
         try:

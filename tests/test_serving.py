@@ -109,3 +109,17 @@ def test_rest_api(dumped_model):
     assert client.delete(f"/models/{sign}").status_code == 200
     assert client.get(f"/models/{sign}").status_code == 404
     assert client.delete("/nodes/0").json()["shutdown_requested"] == 0
+
+
+def test_metrics_endpoint(dumped_model):
+    from fastapi.testclient import TestClient
+    from openembedding_amd.utils.metrics import REGISTRY
+
+    uri, sign, *_ = dumped_model
+    c = ModelController()
+    c.create_model(uri)
+    REGISTRY.add("pull_indices", 7)
+    client = TestClient(make_app(c))
+    body = client.get("/metrics").text
+    assert "openembedding_metric" in body
+    assert "openembedding_models 1" in body
