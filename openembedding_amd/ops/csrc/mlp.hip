@@ -39,13 +39,40 @@ static __device__ __forceinline__ bf16x8 ld_frag(const mbf16* p) {
 }
 
 // 16x16 tile over a padded contraction: A [.., lda] row-major, W [N, ldw]
-// row-major (B = W[n][k]); Kp % 32 == 0. Two-stage software pipeline.
+// row-major (B = W[n][k]); Kp % 32 == 0.
+//
+// M=16 per block means the W stream has NO reuse inside the block — the
+// "decode-GEMV" regime of the perf guide: load straight to VGPRs with a
+// DEEP UNROLL and late counted waits, so all K-steps' loads are in flight
+// before the first MFMA needs its operands (a 2-deep pipeline left ~200
+// cycles of L2 latency exposed per step: 80-90 us/kernel measured).
+template <int KS>
+static __device__ __forceinline__ f32x4 tile16_u(
+        const mbf16* __restrict__ pa, const mbf16* __restrict__ pb) {
+    bf16x8 a[KS], b[KS];
+    #pragma unroll
+    for (int s = 0; s < KS; ++s) {
+        a[s] = ld_frag(pa + 32 * s);
+        b[s] = ld_frag(pb + 32 * s);
+    }
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    #pragma unroll
+    for (int s = 0; s < KS; ++s)
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[s], b[s], acc,
+                                                      0, 0, 0);
+    return acc;
+}
+
 static __device__ __forceinline__ f32x4 tile16(
         const mbf16* A, long lda, const mbf16* W, long ldw, long n0,
         long Kp, int lane) {
     const long koff = (lane >> 4) * 8;
     const mbf16* pa = A + (lane & 15) * lda + koff;
     const mbf16* pb = W + (n0 + (lane & 15)) * ldw + koff;
+    if (Kp == 256) return tile16_u<8>(pa, pb);    // K0 = 247 padded
+    if (Kp == 416) return tile16_u<13>(pa, pb);   // H = 400 padded
+    if (Kp == 128) return tile16_u<4>(pa, pb);
+    // generic fallback: 2-stage pipeline
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     bf16x8 a0 = ld_frag(pa), b0 = ld_frag(pb);
     for (long kb = 32; kb < Kp; kb += 32) {

@@ -43,13 +43,17 @@ def test_mlp3_fwd_matches_torch(M, K0):
     w3p[:, :H] = ws[2]
     out, a1, a2, a3 = ext.mlp3_fwd(x0, ws[0], bs[0], w2p, bs[1],
                                    w3p, bs[2], ws[3].reshape(-1), bs[3])
-    ref_out, ref_acts = _torch_mlp(x0, ws, bs)
-    # bf16 accumulation differences: compare with bf16-level tolerance
-    for got, ref, name in [(a1.float(), ref_acts[0].float(), "a1"),
-                           (a2.float(), ref_acts[1].float(), "a2"),
-                           (a3.float(), ref_acts[2].float(), "a3")]:
-        assert torch.allclose(got, ref, atol=3e-2, rtol=3e-2), name
-    assert torch.allclose(out, ref_out, atol=5e-2, rtol=3e-2)
+    # STAGE-WISE reference: each layer recomputed from the KERNEL's own
+    # previous activation, so bf16 rounding differences don't compound
+    r1 = torch.relu(x0 @ ws[0].t() + bs[0])
+    r2 = torch.relu(a1 @ ws[1].t() + bs[1])
+    r3 = torch.relu(a2 @ ws[2].t() + bs[2])
+    rout = (a3 @ ws[3].reshape(-1) + bs[3]).float()
+    for got, ref, name in [(a1.float(), r1.float(), "a1"),
+                           (a2.float(), r2.float(), "a2"),
+                           (a3.float(), r3.float(), "a3")]:
+        assert torch.allclose(got, ref, atol=2e-2, rtol=2e-2), name
+    assert torch.allclose(out, rout, atol=3e-2, rtol=2e-2)
 
 
 def test_fused_mlp_function_grads():
