@@ -129,6 +129,13 @@ void k_mlp3_fwd(const mbf16* __restrict__ x0, long M, long K0p,
     const int lane = threadIdx.x & 63;
     const long m0 = (long)blockIdx.x * MLP_BM;
     if (m0 >= M) return;
+    // LDS starts with arbitrary bits; the pad columns [H, MLP_LD) are read
+    // by the next stage's fragments against ZERO weight pads — but
+    // 0 * inf-garbage = NaN, so they must be zeroed (found the hard way:
+    // NaN from step ~3 once other kernels had dirtied the LDS)
+    for (int i = threadIdx.x; i < 2 * MLP_BM * MLP_LD; i += blockDim.x)
+        act[0][i] = (mbf16)0.0f;
+    __syncthreads();
 
     mlp_layer(x0 + m0 * K0p, K0p, m0, M, w1, K0p, b1, H, K0p,
               nullptr, 0, act[0], a1, H, wave, lane, true);
@@ -171,6 +178,9 @@ void k_mlp3_bwd(const float* __restrict__ dout, long M, long K0p,
     const int lane = threadIdx.x & 63;
     const long m0 = (long)blockIdx.x * MLP_BM;
     if (m0 >= M) return;
+    for (int i = threadIdx.x; i < 2 * MLP_BM * MLP_LD; i += blockDim.x)
+        dz[0][i] = (mbf16)0.0f;   // see fwd: 0 * LDS-garbage-inf = NaN
+    __syncthreads();
 
     // dz3 = dout ⊗ w4 ⊙ relu'(a3), elementwise; zero the LDS pad columns
     // once (the B-side pads are zero too, but dz tiles are the A side of
