@@ -122,10 +122,10 @@ __global__ void k_ctr_head_bwd_e(const float* __restrict__ e_all,
     de_all[i] = g;
 }
 
-// dense tail: one thread per sample with block-local dw/db accumulation —
-// a naive per-sample atomic scheme serialized B*nd atomics on nd addresses
-// (measured 56 us/step); here it's one LDS pass + nd+1 global atomics per
-// block. ND_MAX bounds the per-thread register accumulator.
+// dense tail, one thread per (sample, feature): coalesced d_dense/dense/
+// d_deep_in traffic, dw/db via LDS block partials then nd+1 global atomics
+// per block (the thread-per-sample version had only B/256 blocks in flight
+// and measured 34.6 us; this shape is ~population-bound).
 #define ND_MAX 32
 template <typename OutT>
 __global__ void k_ctr_head_bwd_d(const float* __restrict__ dense,
@@ -140,22 +140,17 @@ __global__ void k_ctr_head_bwd_d(const float* __restrict__ dense,
     __shared__ float lacc[ND_MAX + 1];
     for (int t = threadIdx.x; t < nd + 1; t += blockDim.x) lacc[t] = 0.0f;
     __syncthreads();
-    float acc[ND_MAX];
-    for (long j = 0; j < nd; ++j) acc[j] = 0.0f;
-    float accb = 0.0f;
-    for (long b = (long)blockIdx.x * blockDim.x + threadIdx.x; b < B;
-         b += (long)gridDim.x * blockDim.x) {
-        const OutT* ddi = d_deep_in + b * out_stride;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < B * nd) {
+        const long b = i / nd;
+        const long j = i % nd;
         const float gp = d_partial[b];
-        accb += gp;
-        for (long j = 0; j < nd; ++j) {
-            float dv = dense[b * nd + j];
-            d_dense[b * nd + j] = (float)ddi[F * dim + j] + gp * w[j];
-            acc[j] += gp * dv;
-        }
+        const float dv = dense[i];
+        d_dense[i] = (float)d_deep_in[b * out_stride + F * dim + j]
+                     + gp * w[j];
+        atomicAdd(&lacc[j], gp * dv);
+        if (j == 0) atomicAdd(&lacc[nd], gp);
     }
-    for (long j = 0; j < nd; ++j) atomicAdd(&lacc[j], acc[j]);
-    atomicAdd(&lacc[nd], accb);
     __syncthreads();
     for (int t = threadIdx.x; t < nd + 1; t += blockDim.x) {
         if (t < nd) atomicAdd(&dw[t], lacc[t]);
@@ -192,10 +187,7 @@ void emb_ctr_head_bwd(const float* e_all, const float* dense, const float* w,
     if (B == 0) return;
     int block = 256;
     long grid_e = (B * F * (dim + 1) + block - 1) / block;
-    // thread-per-sample grid, capped so per-block accumulators cover many
-    // samples before the nd+1 global atomics
-    long grid_d = (B + block - 1) / block;
-    if (grid_d > 512) grid_d = 512;
+    long grid_d = (B * nd + block - 1) / block;
     if (out_bf16) {
         k_ctr_head_bwd_e<bf16><<<(int)grid_e, block, 0, stream>>>(
             e_all, (const bf16*)d_deep_in, d_partial, s_in, B, F, dim, nd,
