@@ -81,11 +81,22 @@ def dump_model(ctx, uri: str, include_optimizer: bool = True) -> None:
         with open(os.path.join(uri, "model_meta"), "w") as f:
             json.dump(meta, f, indent=1)
     ctx.barrier()
+    # files per rank: the reference's server.server_dump_files knob
+    # (EnvConfig.cpp; files named model_{node}_{file_id},
+    # EmbeddingDumpOperator.cpp:28). Rows round-robin across files so big
+    # shards split for parallel HDFS-style upload.
+    n_files = max(1, getattr(ctx.config.server, "server_dump_files", 1)
+                  if getattr(ctx, "config", None) else 1)
     for st in ctx.storages:
-        path = os.path.join(uri, str(st.storage_id), f"model_{rank}_0")
-        with open(path, "wb") as f:
-            for var in st.variables:
-                _dump_variable(f, var, include_optimizer)
+        files = [open(os.path.join(uri, str(st.storage_id),
+                                   f"model_{rank}_{i}"), "wb")
+                 for i in range(n_files)]
+        try:
+            for j, var in enumerate(st.variables):
+                _dump_variable(files[j % n_files], var, include_optimizer)
+        finally:
+            for f in files:
+                f.close()
     ctx.barrier()
 
 

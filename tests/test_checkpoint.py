@@ -98,3 +98,36 @@ def test_save_as_original_model(tmp_path):
     assert w.shape == (50, 4)
     torch.testing.assert_close(w[1:3],
                                e.variable.sparse_read(torch.tensor([1, 2])))
+
+
+def test_multi_file_dump_reload(tmp_path):
+    """server.server_dump_files splits variables across files per rank
+    (reference model_{node}_{file_id} naming); reload must see all rows."""
+    import openembedding_amd as oe
+    from openembedding_amd.context import Context
+    from openembedding_amd import checkpoint
+
+    old = oe.flags.config
+    oe.flags.config = "server:\n  server_dump_files: 3\n"
+    try:
+        ctx = Context(device="cpu")
+        st = ctx.create_storage()
+        vars_ = [st.create_variable(50, 4) for _ in range(4)]
+        for v in vars_:
+            v.set_initializer("uniform", minval=-1, maxval=1)
+            v.set_optimizer("adagrad", learning_rate=0.1)
+            v.shard.pull(torch.arange(0, 20, dtype=torch.int64))
+        uri = str(tmp_path / "d")
+        checkpoint.dump_model(ctx, uri)
+        import os
+        names = sorted(os.listdir(os.path.join(uri, "0")))
+        assert names == ["model_0_0", "model_0_1", "model_0_2"]
+        before = [v.shard.pull_readonly(torch.arange(20)) for v in vars_]
+        for v in vars_:
+            v.shard.clear()
+        checkpoint.load_model(ctx, uri)
+        for v, b in zip(vars_, before):
+            assert torch.equal(v.shard.pull_readonly(torch.arange(20)), b)
+        ctx.finalize()
+    finally:
+        oe.flags.config = old
