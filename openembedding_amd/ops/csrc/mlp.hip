@@ -84,6 +84,9 @@ static __device__ __forceinline__ f32x4 tile16(
     return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc, 0, 0, 0);
 }
 
+#define MLP_WAVES 8   // 8 waves/block = 2 per SIMD: the partner wave hides
+                      // the other's W-stream latency (1 block/CU grid)
+
 // one layer: lds_out/save <- relu(A @ W^T + b) (relu/bias optional)
 static __device__ __forceinline__ void mlp_layer(
         const mbf16* A, long lda, long m0, long M,
@@ -91,7 +94,7 @@ static __device__ __forceinline__ void mlp_layer(
         const mbf16* mask_act, long mask_ld,     // backward relu' source
         mbf16* lds_out, mbf16* save, long save_ld,
         int wave, int lane, bool relu) {
-    for (long c = wave * 16; c < H; c += 64) {
+    for (long c = wave * 16; c < H; c += 16 * MLP_WAVES) {
         f32x4 acc = tile16(A, lda, W, ldw, c, Kp, lane);
         const int col = lane & 15;
         float bv = bias ? (float)bias[c + col] : 0.f;
@@ -111,7 +114,7 @@ static __device__ __forceinline__ void mlp_layer(
     }
 }
 
-extern "C" __global__ __launch_bounds__(256, 2)
+extern "C" __global__ __launch_bounds__(64 * MLP_WAVES, 2)
 void k_mlp3_fwd(const mbf16* __restrict__ x0, long M, long K0p,
                 const mbf16* __restrict__ w1,   // [H, K0p] padded
                 const mbf16* __restrict__ b1,
@@ -147,9 +150,10 @@ void k_mlp3_fwd(const mbf16* __restrict__ x0, long M, long K0p,
               nullptr, 0, act[0], a3, H, wave, lane, true);
     __syncthreads();
 
-    // final Linear(H, 1): VALU dot per row; wave w covers rows w*4..w*4+3
-    for (int r = 0; r < 4; ++r) {
-        long row = wave * 4 + r;
+    // final Linear(H, 1): VALU dot per row
+    const int rows_per_wave = (MLP_BM + MLP_WAVES - 1) / MLP_WAVES;
+    for (int r = 0; r < rows_per_wave; ++r) {
+        long row = (long)wave * rows_per_wave + r;
         long gm = m0 + row;
         if (row >= MLP_BM || gm >= M) continue;
         float s = 0.f;
@@ -162,7 +166,7 @@ void k_mlp3_fwd(const mbf16* __restrict__ x0, long M, long K0p,
     }
 }
 
-extern "C" __global__ __launch_bounds__(256, 2)
+extern "C" __global__ __launch_bounds__(64 * MLP_WAVES, 2)
 void k_mlp3_bwd(const float* __restrict__ dout, long M, long K0p,
                 const mbf16* __restrict__ a1, const mbf16* __restrict__ a2,
                 const mbf16* __restrict__ a3,
@@ -220,7 +224,7 @@ extern "C" void emb_mlp3_fwd(const void* x0, long M, long K0p,
                              hipStream_t stream) {
     if (M == 0) return;
     long grid = (M + MLP_BM - 1) / MLP_BM;
-    k_mlp3_fwd<<<(int)grid, 256, 0, stream>>>(
+    k_mlp3_fwd<<<(int)grid, 64 * MLP_WAVES, 0, stream>>>(
         (const mbf16*)x0, M, K0p, (const mbf16*)w1, (const mbf16*)b1,
         (const mbf16*)w2, (const mbf16*)b2, (const mbf16*)w3,
         (const mbf16*)b3, (const mbf16*)w4, (const mbf16*)b4, H, Hp,
@@ -236,7 +240,7 @@ extern "C" void emb_mlp3_bwd(const float* dout, long M, long K0p,
                              hipStream_t stream) {
     if (M == 0) return;
     long grid = (M + MLP_BM - 1) / MLP_BM;
-    k_mlp3_bwd<<<(int)grid, 256, 0, stream>>>(
+    k_mlp3_bwd<<<(int)grid, 64 * MLP_WAVES, 0, stream>>>(
         dout, M, K0p, (const mbf16*)a1, (const mbf16*)a2, (const mbf16*)a3,
         (const mbf16*)w4, (const mbf16*)w3t, (const mbf16*)w2t,
         (const mbf16*)w1t, H, Hp, (mbf16*)dz1, (mbf16*)dz2, (mbf16*)dz3,

@@ -420,15 +420,22 @@ class DistributedOptimizer:
             return
         self.optimizer.zero_grad(set_to_none=set_to_none)
 
-    def _allreduce_dense(self):
+    def _allreduce_dense(self, async_op: bool = False):
+        """Returns a list of work handles when async_op (flat path only) —
+        the caller overlaps the sparse commit with the wire time (the
+        reference overlapped via update_early_return; here the dense
+        allreduce rides alongside the local sparse optimizer kernels)."""
         if not comm.dist_ready() or dist.get_world_size() == 1:
-            return
+            return []
         if self._flat is not None:
+            handles = []
             for g in self._flat.groups:
-                dist.all_reduce(g["flat_grad"])
-                if self.average_dense:
+                h = dist.all_reduce(g["flat_grad"], async_op=async_op)
+                if async_op:
+                    handles.append((h, g))
+                elif self.average_dense:
                     g["flat_grad"] /= dist.get_world_size()
-            return
+            return handles
         bucket: List[torch.Tensor] = []
         for group in self.optimizer.param_groups:
             for p in group["params"]:
@@ -473,13 +480,22 @@ class DistributedOptimizer:
         loss = None
         if closure is not None:
             loss = closure()
-        self._allreduce_dense()
         if self._flat is not None:
+            # overlap: dense allreduce in flight while the sparse commit's
+            # local optimizer kernels run; wait, then dense step
+            handles = self._allreduce_dense(async_op=True)
+            self._ensure_sparse_configured()
+            self.ctx.update_all_weights()
+            for h, g in handles:
+                h.wait()
+                if self.average_dense:
+                    g["flat_grad"] /= dist.get_world_size()
             self._flat.step()
         else:
+            self._allreduce_dense()
             self.optimizer.step()
-        self._ensure_sparse_configured()
-        self.ctx.update_all_weights()
+            self._ensure_sparse_configured()
+            self.ctx.update_all_weights()
         self.ctx.model_version += 1
         return loss
 
