@@ -49,13 +49,14 @@ def main():
     p.add_argument("--dim", type=int, default=9)
     p.add_argument("--data-pool", type=int, default=8,
                    help="pre-generated synthetic batches, rotated")
-    p.add_argument("--amp", default="off",
+    p.add_argument("--amp", default="native",
                    choices=["bf16", "off", "native"],
                    help="dense-MLP precision: off = fp32; bf16 = autocast "
-                        "(per-step weight casts); native = bf16-resident "
-                        "MLP weights with fp32 master in the flat optimizer "
-                        "(MFMA rate, no cast kernels). Embeddings, FM math, "
-                        "loss and all optimizer state stay fp32.")
+                        "(per-step weight casts); native (default) = "
+                        "bf16-resident MLP weights + the fused single-kernel "
+                        "MLP, fp32 master/accumulators in the flat optimizer "
+                        "(9.25M vs 8.47M fp32 samples/s measured). "
+                        "Embeddings, FM math, loss, optimizer state fp32.")
     p.add_argument("--graph", default="auto", choices=["auto", "on", "off"],
                    help="capture the train step in a hipGraph (single-GPU)")
     p.add_argument("--hash", action="store_true",
