@@ -116,7 +116,7 @@ def test_world2_matches_world1(tmp_path):
     torch.testing.assert_close(loaded, after0, rtol=1e-5, atol=1e-5)
 
 
-def _worker_model(rank, world, port, tmp, flatten=False):
+def _worker_model(rank, world, port, tmp, flatten=False, bf16=False):
     import openembedding_amd.torch as embed
     from openembedding_amd.models import DeepFM, synthetic_batch
 
@@ -124,6 +124,9 @@ def _worker_model(rank, world, port, tmp, flatten=False):
     torch.manual_seed(0)  # same dense init on every rank
     fv = [50, 3, 1000, 40] + [100] * 22
     model = DeepFM(field_vocabs=fv, dim=4)
+    if bf16:
+        from openembedding_amd.models.ctr import convert_mlp_bf16
+        convert_mlp_bf16(model)
     opt = embed.distributed_optimizer(
         torch.optim.Adagrad(model.parameters(), lr=0.01),
         flatten_dense=flatten)
@@ -159,5 +162,16 @@ def test_deepfm_world2_flat_dense(tmp_path):
     port = _free_port()
     mp.spawn(_worker_model, args=(2, port, str(tmp_path), True), nprocs=2,
              join=True)
+    for v in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
+        os.environ.pop(v, None)
+
+
+@pytest.mark.timeout(300)
+def test_deepfm_world2_native_bf16(tmp_path):
+    """The bench DEFAULT at N>1: bf16-resident MLP (mixed-dtype flat groups,
+    fp32 master) + flat allreduce overlapped with the sparse commit."""
+    port = _free_port()
+    mp.spawn(_worker_model, args=(2, port, str(tmp_path), True, True),
+             nprocs=2, join=True)
     for v in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
         os.environ.pop(v, None)
