@@ -163,7 +163,8 @@ class _CTRBase(nn.Module):
 
     def _dnn_out(self, deep_in: torch.Tensor) -> torch.Tensor:
         """dnn logits [B]; the fused single-kernel MLP when enabled."""
-        if self.fused_mlp and deep_in.is_cuda:
+        if (self.fused_mlp and deep_in.is_cuda
+                and deep_in.dtype == torch.bfloat16):
             l1, l2, l3, l4 = self.dnn[0], self.dnn[2], self.dnn[4], self.dnn[6]
             K0p = deep_in.shape[1]
             if self._w1pad is None or self._w1pad["w1p"].shape[1] != K0p:
@@ -177,7 +178,8 @@ class _CTRBase(nn.Module):
             return _FusedMLP3Fn.apply(
                 deep_in, l1.weight, l1.bias, l2.weight, l2.bias,
                 l3.weight, l3.bias, l4.weight, l4.bias, self._w1pad)
-        return self.dnn(deep_in).squeeze(-1)
+        w_dtype = next(self.dnn.parameters()).dtype
+        return self.dnn(deep_in.to(w_dtype)).squeeze(-1).float()
 
 
 class LR(_CTRBase):
@@ -206,7 +208,7 @@ class WDL(_CTRBase):
             return partial + self._dnn_out(deep_in)
         e, lin = self._embed(sparse)                     # [B, F, d], [B, F]
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
-        return self._first_order(dense, lin) + self.dnn(deep_in).squeeze(-1)
+        return self._first_order(dense, lin) + self._dnn_out(deep_in)
 
 
 class DeepFM(_CTRBase):
@@ -227,7 +229,7 @@ class DeepFM(_CTRBase):
         fm2 = 0.5 * (s * s - (e * e).sum(dim=1)).sum(dim=1)
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
         return (self._first_order(dense, lin) + fm2
-                + self.dnn(deep_in).squeeze(-1))
+                + self._dnn_out(deep_in))
 
 
 class CIN(nn.Module):
@@ -276,7 +278,7 @@ class xDeepFM(_CTRBase):
         e, lin = self._embed(sparse)
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
         return (self._first_order(dense, lin) + self.cin(e)
-                + self.dnn(deep_in).squeeze(-1))
+                + self._dnn_out(deep_in))
 
 
 def convert_mlp_bf16(model: _CTRBase) -> _CTRBase:
