@@ -94,9 +94,16 @@ class _FusedMLP3Fn(torch.autograd.Function):
         dw3 = dz3.t() @ a2
         d = dout.unsqueeze(0).to(a3.dtype)             # [1, M]
         dw4 = d @ a3                                   # [1, H]
-        db1 = dz1.sum(0)
-        db2 = dz2.sum(0)
-        db3 = dz3.sum(0)
+        # bias grads as GEMVs: torch's column-sum of row-major bf16 ran
+        # ~16 us each (reduce_kernel); ones@dz is a hipBLASLt GEMV
+        ones = bufs.get("ones")
+        if ones is None or ones.shape[1] != a1.shape[0]:
+            ones = torch.ones(1, a1.shape[0], dtype=a1.dtype,
+                              device=a1.device)
+            bufs["ones"] = ones
+        db1 = (ones @ dz1).reshape(-1)
+        db2 = (ones @ dz2).reshape(-1)
+        db3 = (ones @ dz3).reshape(-1)
         db4 = d.sum(1)
         return (dx0, dw1, db1, dw2, db2, dw3, db3, dw4, db4, None)
 
