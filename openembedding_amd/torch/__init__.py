@@ -20,6 +20,7 @@ Typical use (3-line change, like the reference README):
 from __future__ import annotations
 
 import math
+import weakref
 from typing import Dict, List, Optional, Sequence
 
 import torch
@@ -41,8 +42,23 @@ __all__ = [
 ]
 
 # registry of live PS-backed embeddings (reference track_variable,
-# exb.py:125-131)
-_tracked: List["Embedding"] = []
+# exb.py:125-131); weakrefs so dropped models don't pin their engine state
+_tracked: List = []
+
+
+def _tracked_live():
+    out = []
+    dead = False
+    for r in _tracked:
+        e = r() if isinstance(r, weakref.ref) else r
+        if e is None:
+            dead = True
+        else:
+            out.append(e)
+    if dead:
+        _tracked[:] = [r for r in _tracked
+                       if not (isinstance(r, weakref.ref) and r() is None)]
+    return out
 
 
 class _PullPushFn(torch.autograd.Function):
@@ -179,7 +195,7 @@ class Embedding(nn.Module):
             # zero-size hook so autograd reaches our Function even though
             # indices carry no grad
             self.grad_hook = nn.Parameter(torch.zeros(0, device=ctx.device))
-            _tracked.append(self)
+            _tracked.append(weakref.ref(self))
 
     def forward(self, indices: torch.Tensor) -> torch.Tensor:
         return self._forward_keys(indices, indices)
@@ -459,7 +475,7 @@ class DistributedOptimizer:
                 cfg = _server_cfg_from_torch(self.optimizer)
             except ValueError:
                 cfg = None
-        for e in _tracked:
+        for e in _tracked_live():
             if e.variable is None:
                 continue
             vid = e.variable.sharded.variable_id
