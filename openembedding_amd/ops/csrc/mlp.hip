@@ -87,13 +87,80 @@ static __device__ __forceinline__ f32x4 tile16(
 #define MLP_WAVES 8   // 8 waves/block = 2 per SIMD: the partner wave hides
                       // the other's W-stream latency (1 block/CU grid)
 
-// one layer: lds_out/save <- relu(A @ W^T + b) (relu/bias optional)
+// one layer: lds_out/save <- relu(A @ W^T + b) (relu/bias optional).
+//
+// Pipelining (PMC showed 86% SQ_WAIT_ANY with per-chunk batched loads):
+// the A fragments depend only on (row, k) — identical for every column
+// chunk — so they load ONCE per layer into registers; the B (weight)
+// stream double-buffers across chunks: chunk i+1's 13 fragment loads are
+// in flight while chunk i's MFMAs run.
+template <int KS>
+static __device__ __forceinline__ void mlp_layer_u(
+        const mbf16* A, long lda, long m0, long M,
+        const mbf16* W, long ldw, const mbf16* bias, long H,
+        const mbf16* mask_act, long mask_ld,
+        mbf16* lds_out, mbf16* save, long save_ld,
+        int wave, int lane, bool relu) {
+    const long koff = (lane >> 4) * 8;
+    const mbf16* pa = A + (lane & 15) * lda + koff;
+    bf16x8 a[KS];
+    #pragma unroll
+    for (int s = 0; s < KS; ++s) a[s] = ld_frag(pa + 32 * s);
+
+    const int col = lane & 15;
+    bf16x8 b0[KS], b1[KS];
+    long c = wave * 16;
+    if (c < H) {
+        const mbf16* pb = W + (c + col) * ldw + koff;
+        #pragma unroll
+        for (int s = 0; s < KS; ++s) b0[s] = ld_frag(pb + 32 * s);
+    }
+    for (; c < H; c += 16 * MLP_WAVES) {
+        const long cn = c + 16 * MLP_WAVES;
+        if (cn < H) {
+            const mbf16* pb = W + (cn + col) * ldw + koff;
+            #pragma unroll
+            for (int s = 0; s < KS; ++s) b1[s] = ld_frag(pb + 32 * s);
+        }
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        #pragma unroll
+        for (int s = 0; s < KS; ++s)
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[s], b0[s], acc,
+                                                          0, 0, 0);
+        float bv = bias ? (float)bias[c + col] : 0.f;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            int row = (lane >> 4) * 4 + r;
+            long gm = m0 + row;
+            float v = acc[r] + bv;
+            if (relu) v = v > 0.f ? v : 0.f;
+            if (mask_act && gm < M
+                && !((float)mask_act[gm * mask_ld + c + col] > 0.f))
+                v = 0.f;
+            mbf16 hv = (mbf16)v;
+            if (lds_out) lds_out[row * MLP_LD + c + col] = hv;
+            if (gm < M) save[gm * save_ld + c + col] = hv;
+        }
+        #pragma unroll
+        for (int s = 0; s < KS; ++s) b0[s] = b1[s];
+    }
+}
+
 static __device__ __forceinline__ void mlp_layer(
         const mbf16* A, long lda, long m0, long M,
         const mbf16* W, long ldw, const mbf16* bias, long H, long Kp,
-        const mbf16* mask_act, long mask_ld,     // backward relu' source
+        const mbf16* mask_act, long mask_ld,
         mbf16* lds_out, mbf16* save, long save_ld,
         int wave, int lane, bool relu) {
+    if (Kp == 256)
+        return mlp_layer_u<8>(A, lda, m0, M, W, ldw, bias, H, mask_act,
+                              mask_ld, lds_out, save, save_ld, wave, lane,
+                              relu);
+    if (Kp == 416)
+        return mlp_layer_u<13>(A, lda, m0, M, W, ldw, bias, H, mask_act,
+                               mask_ld, lds_out, save, save_ld, wave, lane,
+                               relu);
+    // generic (unpipelined) fallback for other shapes
     for (long c = wave * 16; c < H; c += 16 * MLP_WAVES) {
         f32x4 acc = tile16(A, lda, W, ldw, c, Kp, lane);
         const int col = lane & 15;
