@@ -125,6 +125,8 @@ DEV float init_value(int cat, float p0, float p1, float p2,
 // CAS count drops from O(batch) to O(blocks).
 
 #define ULDS 1024  // LDS pre-filter entries (8 KiB keys + 4 KiB slots)
+#define UA_ITERS 8  // elements per thread in the uid-assign kernel
+#define UI_ILP 4    // parallel lookups per thread in the inverse kernel
 
 __global__ void k_unique_insert(const i64* __restrict__ keys, long n,
                                 u64* __restrict__ tk, long mask,
@@ -187,36 +189,68 @@ __global__ void k_unique_assign(const i64* __restrict__ keys, long n,
                                 i64* __restrict__ unique_keys,
                                 int* __restrict__ counter,
                                 i64* __restrict__ inverse) {
-    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    // wave-aggregated counter: one atomic per wave, not per first-occurrence
-    bool first = (i < n) && is_first[i];
-    u64 ballot = __ballot(first);
-    if (!ballot) return;
-    int lane = threadIdx.x & 63;
-    int leader = __ffsll((unsigned long long)ballot) - 1;
+    // UA_ITERS elements per thread with ONE returning atomic per wave:
+    // ballots for all iterations are taken first (independent loads in
+    // flight), their popcounts summed, a single atomicAdd reserves the uid
+    // range, then uids are assigned per iteration from the running prefix.
+    // (The one-atomic-per-wave version was 99% SQ_WAIT_ANY parked on the
+    // dependent L2 atomic: 21.6 us/step.)
+    const int lane = threadIdx.x & 63;
+    const long stride = (long)gridDim.x * blockDim.x;
+    long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    u64 ballots[UA_ITERS];
+    int total = 0;
+    #pragma unroll
+    for (int t = 0; t < UA_ITERS; ++t) {
+        long i = i0 + t * stride;
+        bool first = (i < n) && is_first[i];
+        ballots[t] = __ballot(first);
+        total += __popcll(ballots[t]);
+    }
+    if (!total) return;
     int base = 0;
-    if (lane == leader)
-        base = atomicAdd(counter, __popcll(ballot));
-    base = __shfl(base, leader);
-    if (first) {
-        int rank = __popcll(ballot & ((1ull << lane) - 1ull));
-        int uid = base + rank;
-        unique_keys[uid] = keys[i];
-        int s = slot_of[i];
-        if (s >= 0) tv[s] = uid;
-        else inverse[i] = uid;  // overflow winner: direct assignment
+    if (lane == 0) base = atomicAdd(counter, total);
+    base = __shfl(base, 0, 64);
+    #pragma unroll
+    for (int t = 0; t < UA_ITERS; ++t) {
+        u64 b = ballots[t];
+        if (b & (1ull << lane)) {
+            long i = i0 + t * stride;
+            int uid = base + __popcll(b & ((1ull << lane) - 1ull));
+            unique_keys[uid] = keys[i];
+            int s = slot_of[i];
+            if (s >= 0) tv[s] = uid;
+            else inverse[i] = uid;  // overflow winner: direct assignment
+        }
+        base += __popcll(b);
     }
 }
 
 __global__ void k_unique_inverse(long n, const int* __restrict__ slot_of,
                                  const int* __restrict__ tv,
                                  i64* __restrict__ inverse) {
-    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= n) return;
-    int s = slot_of[i];
-    if (s >= 0) { inverse[i] = (i64)tv[s]; return; }
-    if (s == -1) return;           // overflow winner: set by assign kernel
-    inverse[i] = inverse[-2 - s];  // overflow duplicate: winner's uid
+    // UI_ILP independent lookups in flight per thread (2-deep load chain
+    // slot_of -> tv; single-element version was latency-parked)
+    const long stride = (long)gridDim.x * blockDim.x;
+    long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    int s[UI_ILP];
+    #pragma unroll
+    for (int t = 0; t < UI_ILP; ++t) {
+        long i = i0 + t * stride;
+        s[t] = (i < n) ? slot_of[i] : -1;
+    }
+    int v[UI_ILP];
+    #pragma unroll
+    for (int t = 0; t < UI_ILP; ++t)
+        v[t] = (s[t] >= 0) ? tv[s[t]] : 0;
+    #pragma unroll
+    for (int t = 0; t < UI_ILP; ++t) {
+        long i = i0 + t * stride;
+        if (i >= n) continue;
+        if (s[t] >= 0) inverse[i] = (i64)v[t];
+        else if (s[t] <= -2) inverse[i] = inverse[-2 - s[t]];
+        // s[t] == -1: overflow winner, set by the assign kernel
+    }
 }
 
 // --------------------------------------------------- persistent hash table
@@ -656,11 +690,12 @@ void emb_unique(const i64* keys, long n, u64* tk, int* tv, long cap,
     long mask = cap - 1;
     k_unique_insert<<<grid1d(n), BLOCK, 0, stream>>>(keys, n, tk, mask,
                                                      slot_of, is_first);
-    k_unique_assign<<<grid1d(n), BLOCK, 0, stream>>>(keys, n, slot_of,
-                                                     is_first, tv,
-                                                     unique_keys, counter,
-                                                     inverse);
-    k_unique_inverse<<<grid1d(n), BLOCK, 0, stream>>>(n, slot_of, tv, inverse);
+    int ga = grid1d((n + UA_ITERS - 1) / UA_ITERS);
+    int gi = grid1d((n + UI_ILP - 1) / UI_ILP);
+    k_unique_assign<<<ga, BLOCK, 0, stream>>>(keys, n, slot_of, is_first,
+                                              tv, unique_keys, counter,
+                                              inverse);
+    k_unique_inverse<<<gi, BLOCK, 0, stream>>>(n, slot_of, tv, inverse);
 }
 
 void emb_ht_lookup(u64* tk, int* tv, long cap, const i64* keys, long n,
