@@ -425,13 +425,34 @@ __global__ void k_reduce_lds(const i64* __restrict__ inverse,
     __syncthreads();
     const int lane = threadIdx.x % G;
     const int group = threadIdx.x / G;
-    const int n_groups = blockDim.x / G;
-    const long per_group = blockDim.x / n_groups;  // == G elements per group
-    const long base = (long)blockIdx.x * blockDim.x + group * per_group;
-    for (long t = 0; t < per_group; ++t) {
+    const long base = (long)blockIdx.x * blockDim.x + group * G;
+    // latency plan (PMC: 60% parked): preload every element's uid up front
+    // (independent loads) and stage each element's gradient fragment one
+    // iteration ahead of its LDS-accumulate, so a global load is always in
+    // flight while the previous element's LDS atomics run.
+    int uids[G];
+    #pragma unroll
+    for (int t = 0; t < G; ++t) {
+        long e = base + t;
+        uids[t] = (e < n) ? (int)inverse[e] : -1;
+    }
+    float gfrag[2][4];  // up to 4 grad values per lane (dim <= 4*G)
+    const int nj = (int)((dim + G - 1) / G);
+    #pragma unroll
+    for (int j = 0; j < 4; ++j)
+        gfrag[0][j] = (j < nj && base < n && lane + j * G < dim)
+                          ? grads[(u64)base * dim + lane + j * G] : 0.0f;
+    for (int t = 0; t < G; ++t) {
         long e = base + t;
         if (e >= n) break;
-        int uid = (int)inverse[e];
+        long en = base + t + 1;
+        if (en < n) {
+            #pragma unroll
+            for (int j = 0; j < 4; ++j)
+                gfrag[(t + 1) & 1][j] = (j < nj && lane + j * G < dim)
+                    ? grads[(u64)en * dim + lane + j * G] : 0.0f;
+        }
+        int uid = uids[t];
         int h = -1;
         if (lane == 0) {
             int hh = (int)(((unsigned)uid * 2654435761u) & (H - 1));
@@ -444,17 +465,15 @@ __global__ void k_reduce_lds(const i64* __restrict__ inverse,
             else atomicAdd(&counts[uid], 1ull);
         }
         h = __shfl(h, (threadIdx.x & ~(G - 1)) % 64, 64);
-        const float* g = grads + (u64)e * dim;
-        if (h >= 0) {
-            float* acc = lacc + (u64)h * dim;
-            for (long j = lane; j < dim; j += G) atomicAdd(&acc[j], g[j]);
-        } else {  // LDS table full: fall through to global atomics
-            float* ug = ugrads + (u64)uid * dim;
-            for (long j = lane; j < dim; j += G) atomicAdd(&ug[j], g[j]);
-        }
+        float* dstb = (h >= 0) ? (lacc + (u64)h * dim)
+                               : (ugrads + (u64)uid * dim);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j)
+            if (j < nj && lane + j * G < dim)
+                atomicAdd(&dstb[lane + j * G], gfrag[t & 1][j]);
     }
     __syncthreads();
-    for (int h = group; h < H; h += n_groups) {
+    for (int h = group; h < H; h += (int)(blockDim.x / G)) {
         int uid = luid[h];
         if (uid < 0) continue;
         if (lane == 0) atomicAdd(&counts[uid], (u64)lcnt[h]);
