@@ -352,6 +352,11 @@ __global__ void k_array_touch(unsigned char* __restrict__ valid,
 // response scatter, EmbeddingPullOperator.cpp:229-249) into the gather: all
 // duplicates of a new key recompute the same deterministic init value, so
 // the racy-looking table writes are benign (same bytes).
+// GI_ILP elements per G-lane group with the index chain
+// (inverse -> slot -> mask/key) preloaded for all of them before any row
+// copy: the one-element version was 76% SQ_WAIT_ANY (each group parked on
+// its own dependent gather chain).
+#define GI_ILP 4
 template <int G>
 __global__ void k_gather_init(float* __restrict__ weights,
                               float* __restrict__ state,
@@ -365,33 +370,56 @@ __global__ void k_gather_init(float* __restrict__ weights,
                               u64 seed,
                               const float* __restrict__ state_init_row,
                               const int* __restrict__ u_dev) {
-    long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) / G;
-    int lane = threadIdx.x % G;
-    if (g >= n) return;
-    long uid = inverse ? inverse[g] : g;
-    if (!inverse && u_dev && g >= *u_dev) return;
-    i64 slot = slots[uid];
-    if (slot < 0) {  // read-only miss: zeros out
-        if (out)
-            for (long j = lane; j < dim; j += G) out[g * dim + j] = 0.0f;
-        return;
+    const long stride = ((long)gridDim.x * blockDim.x) / G;
+    long g0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) / G;
+    const int lane = threadIdx.x % G;
+    const int live = u_dev ? *u_dev : 0;
+
+    long uid[GI_ILP];
+    i64 slot[GI_ILP];
+    unsigned char nm[GI_ILP];
+    u64 key[GI_ILP];
+    #pragma unroll
+    for (int t = 0; t < GI_ILP; ++t) {
+        long g = g0 + t * stride;
+        bool act = g < n && !(!inverse && u_dev && g >= live);
+        uid[t] = act ? (inverse ? inverse[g] : g) : -1;
     }
-    float* wrow = weights + (u64)slot * dim;
-    if (new_mask && new_mask[uid]) {
-        u64 key = (u64)keys[uid];
-        for (long j = lane; j < dim; j += G) {
-            float v = init_value(init_cat, p0, p1, p2, seed, key, (u64)j);
-            wrow[j] = v;
-            if (out) out[g * dim + j] = v;
+    #pragma unroll
+    for (int t = 0; t < GI_ILP; ++t)
+        slot[t] = (uid[t] >= 0) ? slots[uid[t]] : -1;
+    #pragma unroll
+    for (int t = 0; t < GI_ILP; ++t) {
+        nm[t] = (uid[t] >= 0 && new_mask) ? new_mask[uid[t]] : 0;
+        key[t] = (uid[t] >= 0 && nm[t]) ? (u64)keys[uid[t]] : 0;
+    }
+    #pragma unroll
+    for (int t = 0; t < GI_ILP; ++t) {
+        long g = g0 + t * stride;
+        if (uid[t] < 0)
+            continue;
+        if (slot[t] < 0) {  // read-only miss: zeros out
+            if (out)
+                for (long j = lane; j < dim; j += G) out[g * dim + j] = 0.0f;
+            continue;
         }
-        if (sd > 0) {
-            float* srow = state + (u64)slot * sd;
-            for (long j = lane; j < sd; j += G) srow[j] = state_init_row[j];
-        }
-    } else {
-        if (out)
+        float* wrow = weights + (u64)slot[t] * dim;
+        if (nm[t]) {
+            for (long j = lane; j < dim; j += G) {
+                float v = init_value(init_cat, p0, p1, p2, seed, key[t],
+                                     (u64)j);
+                wrow[j] = v;
+                if (out) out[g * dim + j] = v;
+            }
+            if (sd > 0) {
+                float* srow = state + (u64)slot[t] * sd;
+                for (long j = lane; j < sd; j += G)
+                    srow[j] = state_init_row[j];
+            }
+        } else if (out) {
             for (long j = lane; j < dim; j += G)
                 out[g * dim + j] = wrow[j];
+        }
     }
 }
 
@@ -751,12 +779,15 @@ void emb_gather_init(float* weights, float* state, long dim, long sd,
                      const float* state_init_row, const int* u_dev,
                      hipStream_t stream) {
     if (n == 0) return;
+    long npg = (n + GI_ILP - 1) / GI_ILP;
     if (dim <= 32) {
-        k_gather_init<16><<<grid1d(n * 16), BLOCK, 0, stream>>>(
+        int gg = grid1d(npg * 16);
+        k_gather_init<16><<<gg, BLOCK, 0, stream>>>(
             weights, state, dim, sd, slots, new_mask, keys, n, inverse, out,
             init_cat, p0, p1, p2, seed, state_init_row, u_dev);
     } else {
-        k_gather_init<64><<<grid1d(n * 64), BLOCK, 0, stream>>>(
+        int gg = grid1d(npg * 64);
+        k_gather_init<64><<<gg, BLOCK, 0, stream>>>(
             weights, state, dim, sd, slots, new_mask, keys, n, inverse, out,
             init_cat, p0, p1, p2, seed, state_init_row, u_dev);
     }
