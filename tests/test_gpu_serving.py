@@ -1,0 +1,49 @@
+"""GPU serving: a trained dump loads into HBM-resident read-only tables
+and serves identical rows through the controller + REST."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def test_serve_from_gpu(tmp_path):
+    import openembedding_amd.torch as embed
+    from openembedding_amd.models import DeepFM, synthetic_batch
+    from openembedding_amd.serving import ModelController, make_app
+
+    torch.manual_seed(0)
+    model = DeepFM(dim=9).to(DEV)
+    opt = embed.distributed_optimizer(
+        torch.optim.Adagrad(model.parameters(), lr=0.01))
+    lossf = torch.nn.BCEWithLogitsLoss()
+    for _ in range(3):
+        dense, sparse, labels = synthetic_batch(512, device=DEV)
+        opt.zero_grad()
+        loss = lossf(model(dense, sparse), labels)
+        loss.backward()
+        opt.step()
+    uri = str(tmp_path / "dump")
+    embed.save_server_model(uri)
+    ctx = embed.get_context()
+    sign = f"{ctx.model_uuid}-{ctx.model_version}"
+
+    probe = sparse[:4] + model.embedding.field_offsets
+    live = model.embedding.variable.sparse_read(probe)
+
+    c = ModelController(device=DEV)
+    c.create_model(uri)
+    var = c.manager.find_model_variable(sign, 0)
+    assert var.shard.device.type == "cuda"
+    got = var.pull_weights(probe)
+    assert torch.allclose(got, live, atol=1e-6)
+
+    from fastapi.testclient import TestClient
+    client = TestClient(make_app(c))
+    r = client.post(f"/models/{sign}/variables/0/pull",
+                    json={"indices": probe.cpu().tolist()})
+    assert r.status_code == 200
+    served = torch.tensor(r.json()["weights"], device=DEV)
+    assert torch.allclose(served, live, atol=1e-6)
