@@ -39,6 +39,8 @@ __all__ = [
     "save_as_original_model", "pulling", "sparse_read_as_dense",
     "should_persist_server_model", "persist_server_model",
     "restore_server_model", "get_context",
+    "Adadelta", "Adagrad", "Adam", "Adamax", "Ftrl", "Nadam", "NAdam",
+    "RMSprop", "SGD",
 ]
 
 # registry of live PS-backed embeddings (reference track_variable,
@@ -531,6 +533,52 @@ def distributed_optimizer(optimizer: torch.optim.Optimizer,
                           sparse_config: Optional[Dict] = None,
                           **kw) -> DistributedOptimizer:
     return DistributedOptimizer(optimizer, sparse_config=sparse_config, **kw)
+
+
+def _make_opt_class(torch_name: str):
+    """Pre-wrapped optimizer constructors (the reference exported
+    distributed subclasses of all keras optimizers, exb.py:446-488):
+    ``embed.Adagrad(model.parameters(), lr=0.01)`` ==
+    ``distributed_optimizer(torch.optim.Adagrad(...))``."""
+    base = getattr(torch.optim, torch_name)
+
+    def ctor(params, *args, sparse_config: Optional[Dict] = None,
+             flatten_dense: bool = False, **kw):
+        return distributed_optimizer(base(params, *args, **kw),
+                                     sparse_config=sparse_config,
+                                     flatten_dense=flatten_dense)
+
+    ctor.__name__ = torch_name
+    ctor.__qualname__ = torch_name
+    return ctor
+
+
+Adadelta = _make_opt_class("Adadelta")
+Adagrad = _make_opt_class("Adagrad")
+Adam = _make_opt_class("Adam")
+Adamax = _make_opt_class("Adamax")
+NAdam = Nadam = _make_opt_class("NAdam")
+RMSprop = _make_opt_class("RMSprop")
+SGD = _make_opt_class("SGD")
+
+
+def Ftrl(params, sparse_config: Optional[Dict] = None, **kw):
+    """FTRL exists only as a SERVER-side sparse optimizer (torch has no
+    dense FTRL; the reference's keras Ftrl also only mattered server-side).
+    Dense params train with Adagrad; the sparse side runs the engine's FTRL
+    (core/optimizers.py, exact reference formulas incl. l2_shrinkage and
+    lr_power branches)."""
+    cfg = dict(category="ftrl",
+               learning_rate=kw.pop("learning_rate", kw.pop("lr", 0.05)),
+               l1_regularization_strength=kw.pop(
+                   "l1_regularization_strength", 0.0),
+               l2_regularization_strength=kw.pop(
+                   "l2_regularization_strength", 0.0))
+    if sparse_config:
+        cfg.update(sparse_config)
+    return distributed_optimizer(
+        torch.optim.Adagrad(params, lr=cfg["learning_rate"], **kw),
+        sparse_config=cfg)
 
 
 # ----------------------------------------------------------- model helpers

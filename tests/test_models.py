@@ -75,3 +75,29 @@ def test_sparse_as_dense_cache_path():
     e = embed.Embedding(10, 4, sparse_as_dense=True)
     out = e(torch.tensor([1, 2]))
     assert out.requires_grad and out.shape == (2, 4)
+
+
+def test_optimizer_class_surface():
+    """Reference exb exports optimizer classes (exb.py:446-488):
+    embed.Adagrad(...) pre-wraps distributed_optimizer."""
+    import openembedding_amd.torch as embed
+    from openembedding_amd.models import DeepFM, synthetic_batch
+
+    torch.manual_seed(0)
+    m = DeepFM(dim=4)
+    opt = embed.Adagrad(m.parameters(), lr=0.01)
+    assert isinstance(opt, embed.DistributedOptimizer)
+    dense, sparse, labels = synthetic_batch(32)
+    loss = torch.nn.functional.binary_cross_entropy_with_logits(
+        m(dense, sparse), labels)
+    loss.backward()
+    opt.step()
+
+    m2 = DeepFM(dim=4)
+    o2 = embed.Ftrl(m2.parameters(), learning_rate=0.05,
+                    l1_regularization_strength=0.01)
+    loss = torch.nn.functional.binary_cross_entropy_with_logits(
+        m2(*synthetic_batch(32)[:2]), synthetic_batch(32)[2])
+    loss.backward()
+    o2.step()
+    assert m2.embedding.variable.sharded.shard.optimizer.category == "ftrl"
