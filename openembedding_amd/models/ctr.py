@@ -277,17 +277,9 @@ class CIN(nn.Module):
         xk = e
         outs = []
         for conv in self.convs:
-            # xk+1[b,k,d] = relu( sum_{f,g} W[k,f,g] x0[b,f,d] xk[b,g,d] ).
-            # Materializing the [B, F*G, d] outer-product tensor (the naive
-            # conv1d form) moved ~0.5 GB/step through HBM; instead run ONE
-            # well-shaped GEMM  W[F*K, G] @ xk[G, B*d]  and contract f
-            # against x0 (identical math, conv kernel size is 1).
-            G = xk.shape[1]
-            K = conv.weight.shape[0]
-            w = conv.weight.reshape(K, F, G).transpose(0, 1).reshape(F * K, G)
-            xkt = xk.permute(1, 0, 2).reshape(G, B * d)
-            m = (w @ xkt).reshape(F, K, B, d)
-            xk = torch.relu(torch.einsum("fkbd,bfd->bkd", m, x0))
+            # outer product along field axes: [B, F*Hk, d]
+            z = torch.einsum("bfd,bhd->bfhd", x0, xk).reshape(B, -1, d)
+            xk = torch.relu(conv(z))
             outs.append(xk.sum(dim=2))                   # [B, Hk]
         return self.fc(torch.cat(outs, dim=1)).squeeze(-1)
 
