@@ -150,3 +150,49 @@ def test_optimizer_reconfig_migrates():
                           torch.full((4,), 0.3))
     shard.set_optimizer("adam")
     assert shard.state_dim == 2 * 4 + 2
+
+
+def test_empty_batch_pull_push():
+    """n=0 pulls/pushes must be no-ops (collective paths call them on every
+    rank even when a rank's batch is empty)."""
+    from openembedding_amd.parallel.sharded import ShardedVariable
+
+    sh = VariableShard(VariableMeta(variable_id=50, embedding_dim=4,
+                                    vocabulary_size=100))
+    sh.set_initializer("constant", value=1.0)
+    sh.set_optimizer("adagrad", learning_rate=0.1)
+    v = ShardedVariable(sh)
+    out, h = v.pull(torch.empty(0, dtype=torch.int64))
+    assert out.shape == (0, 4)
+    v.push(h, torch.empty(0, 4))
+    v.update_weights()
+    assert sh.num_rows == 0
+
+
+def test_update_without_push_is_noop():
+    sh = VariableShard(VariableMeta(variable_id=51, embedding_dim=4,
+                                    vocabulary_size=100))
+    sh.set_initializer("constant", value=2.0)
+    sh.set_optimizer("sgd", learning_rate=0.1, momentum=0.0, nesterov=False)
+    before = sh.pull(torch.tensor([3])).clone()
+    sh.update_weights()
+    sh.update_weights()
+    assert torch.equal(sh.pull(torch.tensor([3])), before)
+
+
+def test_single_key_many_duplicates():
+    """One key duplicated across a whole batch: gradient SUM + one
+    optimizer step (the reference's counts semantics)."""
+    sh = VariableShard(VariableMeta(variable_id=52, embedding_dim=2,
+                                    vocabulary_size=10))
+    sh.set_initializer("constant", value=0.0)
+    sh.set_optimizer("default", learning_rate=1.0)
+    keys = torch.zeros(64, dtype=torch.int64)
+    from openembedding_amd.parallel.sharded import ShardedVariable
+    v = ShardedVariable(sh)
+    out, h = v.pull(keys)
+    v.push(h, torch.ones(64, 2))
+    v.update_weights()
+    got = sh.pull_readonly(torch.tensor([0]))
+    # default optimizer: w -= lr * summed_grad = -64
+    assert torch.allclose(got, torch.full((1, 2), -64.0))
