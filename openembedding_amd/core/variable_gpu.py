@@ -133,17 +133,39 @@ class HipVariableShard(VariableShard):
 
     # -------------------------------------------------------------- lookups
 
+    def reserve_rows(self, rows: int) -> None:
+        """Pre-size the row slab and probe table for ``rows`` keys and mark
+        the shard capture-safe: with capacity fixed up front, the insert
+        path has no host-side growth and a hipGraph may capture it (the
+        kernel's capacity guard turns overflow beyond the reservation into
+        dropped rows instead of corruption). CombinedEmbedding in hash mode
+        auto-reserves its key-space size."""
+        rows = int(rows)
+        self._ensure_rows(max(rows, 1024))
+        want_cap = 1 << max(16, (2 * rows - 1).bit_length())
+        if want_cap > self._cap:
+            old_upper = self._nrows_upper
+            self._nrows_upper = want_cap // 2
+            self._maybe_rehash()
+            self._nrows_upper = old_upper
+        self._reserved_rows = rows
+
     def _lookup_or_insert(self, keys: torch.Tensor, u_dev=None):
         n = keys.numel()
         if self.meta.use_hash_table:
             if self._in_graph_capture():
-                # growth (slab realloc, rehash) needs host control; a
-                # captured insert would write past the frozen capacity on
-                # replay. Raising here makes graph capture fall back to
-                # eager (bench.py catches it). The kernel additionally
-                # bounds the slot counter by the slab capacity.
-                raise RuntimeError(
-                    "hash-table insert path is not hipGraph-capturable")
+                if getattr(self, "_reserved_rows", 0) <= 0:
+                    # growth (slab realloc, rehash) needs host control; a
+                    # captured insert would write past the frozen capacity
+                    # on replay. Raising makes graph capture fall back to
+                    # eager (bench.py catches it). The kernel additionally
+                    # bounds the slot counter by the slab capacity.
+                    raise RuntimeError(
+                        "hash-table insert path is not hipGraph-capturable "
+                        "without reserve_rows()")
+                return self.ext.ht_lookup(self.tk, self.tv, keys,
+                                          self.nrows_dev, self.slot_keys,
+                                          True, u_dev)
             self._ensure_rows(self._nrows_upper + n)
             self._maybe_rehash()
             slots, new_mask = self.ext.ht_lookup(self.tk, self.tv, keys,

@@ -90,6 +90,13 @@ class ShardedVariable:
     def set_optimizer(self, category: str, **cfg):
         self.shard.set_optimizer(category, **cfg)
 
+    def reserve_rows(self, total_rows: int) -> None:
+        """Pre-size this shard for its share of ``total_rows`` keys (hash
+        mode; makes the GPU insert path hipGraph-capturable)."""
+        fn = getattr(self.shard, "reserve_rows", None)
+        if fn is not None:
+            fn((int(total_rows) + self.world_size - 1) // self.world_size)
+
     # ------------------------------------------------------------------- pull
 
     def pull(self, indices: torch.Tensor, readonly: bool = False):

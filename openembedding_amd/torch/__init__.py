@@ -244,6 +244,10 @@ class CombinedEmbedding(Embedding):
         # hash table (rows created lazily; enables the capacity tier)
         super().__init__(-1 if hash_mode else offsets[-1], embedding_dim,
                          **kw)
+        if hash_mode and self.variable is not None:
+            # the key space is known (field offsets): pre-size the table so
+            # the GPU insert path stays hipGraph-capturable
+            self.variable.sharded.reserve_rows(offsets[-1])
         self.n_fields = len(sizes)
         self.register_buffer(
             "field_offsets",

@@ -90,3 +90,65 @@ def test_graph_replay_matches_eager():
                            for p in model2.dnn.parameters()])
     assert torch.allclose(got_rows, ref_rows, atol=1e-5, rtol=1e-5)
     assert torch.allclose(got_dense, ref_dense, atol=1e-5, rtol=1e-5)
+
+
+def test_graph_replay_hash_mode_with_reservation():
+    """CombinedEmbedding hash mode auto-reserves its key space, making the
+    insert path capturable; replayed training must equal eager."""
+    import openembedding_amd.context as cm
+    import openembedding_amd.torch as api
+    import openembedding_amd.torch as embed
+    from openembedding_amd.models import DeepFM, synthetic_batch
+
+    lossf = torch.nn.BCEWithLogitsLoss()
+    gen = torch.Generator().manual_seed(5)
+    batches = [tuple(t.to(DEV) for t in synthetic_batch(512, generator=gen))
+               for _ in range(5)]
+
+    def build():
+        if cm._context is not None:
+            cm._context.finalize()
+            cm._context = None
+        api._tracked.clear()
+        torch.manual_seed(0)
+        m = DeepFM(dim=4, hash_mode=True).to(DEV)
+        o = embed.distributed_optimizer(
+            torch.optim.Adagrad(m.parameters(), lr=0.01), flatten_dense=True)
+        return m, o
+
+    def step(m, o, dense, sparse, labels):
+        o.zero_grad(set_to_none=False)
+        loss = lossf(m(dense, sparse), labels)
+        loss.backward()
+        o.step()
+
+    m1, o1 = build()
+    for _ in range(6):
+        step(m1, o1, *batches[0])
+    for b in batches:
+        step(m1, o1, *b)
+    probe = batches[1][1][:8] + m1.embedding.field_offsets
+    ref = m1.embedding.variable.sparse_read(probe).clone()
+
+    m2, o2 = build()
+    static = tuple(t.clone() for t in batches[0])
+    for _ in range(3):
+        step(m2, o2, *static)
+    torch.cuda.synchronize()
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        for _ in range(3):
+            step(m2, o2, *static)
+    torch.cuda.current_stream().wait_stream(side)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        step(m2, o2, *static)   # must capture (reservation active)
+    for b in batches:
+        static[0].copy_(b[0])
+        static[1].copy_(b[1])
+        static[2].copy_(b[2])
+        g.replay()
+    torch.cuda.synchronize()
+    got = m2.embedding.variable.sparse_read(probe)
+    assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5)
