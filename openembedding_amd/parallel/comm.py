@@ -26,6 +26,8 @@ def backend() -> str:
 def all_to_all_lengths(send_counts: torch.Tensor) -> torch.Tensor:
     """Exchange per-peer element counts. send_counts int64 [world] (any device)
     -> recv_counts int64 [world] on the same device."""
+    if not dist_ready():
+        return send_counts.clone()  # world-1 (forced-remote tests)
     world = dist.get_world_size()
     if backend() == "gloo":
         cpu = send_counts.to("cpu")
@@ -43,6 +45,8 @@ def all_to_all_v(inp: torch.Tensor, in_splits: Sequence[int],
                  out_splits: Sequence[int]) -> torch.Tensor:
     """Variable all-to-all along dim 0. inp [sum(in_splits), ...] ->
     [sum(out_splits), ...]. Splits are python ints (host-known)."""
+    if not dist_ready():
+        return inp.clone()          # world-1 (forced-remote tests)
     world = dist.get_world_size()
     rank = dist.get_rank()
     trailing = list(inp.shape[1:])

@@ -99,6 +99,10 @@ class ShardedVariable:
 
     # ------------------------------------------------------------------- pull
 
+    # test hook: run the multi-rank path at world_size 1 (collectives become
+    # identity) so the GPU bucketize/fan-out logic is validatable on one GPU
+    _force_remote = False
+
     def pull(self, indices: torch.Tensor, readonly: bool = False):
         """indices: int64 tensor of any shape -> (weights [*shape, dim], handle).
 
@@ -107,9 +111,13 @@ class ShardedVariable:
         flat = indices.reshape(-1).to(torch.int64)
         n = flat.numel()
         self.stat_pull_indices += n
-        if (self.world_size == 1 and not readonly
+        remote = self.world_size > 1 or self._force_remote
+        if (not remote and not readonly
                 and getattr(self.shard, "pull_bounded", None) is not None):
             return self._pull_local_bounded(indices, flat)
+        if (remote and not readonly
+                and getattr(self.shard, "pull_bounded", None) is not None):
+            return self._pull_remote_bounded(indices, flat)
         unique, inverse = ops.unique_inverse(flat)
         self.stat_pull_unique += unique.numel()
         # reference pull_indices/pull_unique accumulators
@@ -134,6 +142,53 @@ class ShardedVariable:
         out, slots = self.shard.pull_bounded(uk_buf, u_dev, inverse)
         h = PullHandle(shape=indices.shape, unique=uk_buf, inverse=inverse,
                        bounded=True, u_dev=u_dev, slots=slots)
+        return out.view(*indices.shape, self.shard.dim), h
+
+    def _pull_remote_bounded(self, indices: torch.Tensor,
+                             flat: torch.Tensor):
+        """Multi-rank pull on the GPU engine with ONE host sync per step
+        (the combined split-size read): dedup, owner bucketize and the
+        sentinel-last ordering all run device-side off the bounded unique
+        (the exact path paid a second sync for the tight unique count)."""
+        world = self.world_size
+        ext = self.shard.ext
+        dev = flat.device
+        uk_buf, inverse, u_dev = ext.unique_bounded(flat)   # no sync
+        n = uk_buf.numel()
+        valid = torch.arange(n, device=dev) < u_dev.to(torch.int64)
+        owner = torch.where(valid, uk_buf % world,
+                            torch.full_like(uk_buf, world))
+        send_counts = torch.bincount(owner, minlength=world + 1)[:world]
+        recv_counts = comm.all_to_all_lengths(send_counts)
+        both = torch.cat([send_counts, recv_counts]).tolist()  # ONE sync
+        send_splits, recv_splits = both[:world], both[world:]
+        u = int(sum(send_splits))
+        self.stat_pull_unique += u
+        REGISTRY.add("pull_indices", n)
+        REGISTRY.add("pull_unique", u)
+        order = torch.argsort(owner, stable=True)[:u]  # valid, owner-grouped
+        send_keys = uk_buf.index_select(0, order)
+        recv_keys = comm.all_to_all_v(send_keys, send_splits, recv_splits)
+        h = PullHandle(shape=indices.shape, unique=uk_buf, inverse=inverse,
+                       order=order, send_splits=send_splits,
+                       recv_splits=recv_splits, u_dev=u_dev)
+        if recv_keys.numel():
+            uk2_buf, inv2, u2_dev = ext.unique_bounded(recv_keys)
+            rows_back, slots2 = self.shard.pull_bounded(uk2_buf, u2_dev,
+                                                        inv2)
+            h.owner_unique = uk2_buf
+            h.owner_inverse = inv2
+            h.owner_u_dev = u2_dev
+            h.owner_slots = slots2
+        else:
+            h.owner_unique = recv_keys
+            rows_back = torch.empty(0, self.shard.dim, dtype=torch.float32,
+                                    device=dev)
+        rows_sorted = comm.all_to_all_v(rows_back, recv_splits, send_splits)
+        rows_u = torch.zeros(n, self.shard.dim, dtype=rows_sorted.dtype,
+                             device=dev)
+        rows_u.index_copy_(0, order, rows_sorted)
+        out = rows_u.index_select(0, inverse)
         return out.view(*indices.shape, self.shard.dim), h
 
     def _pull_remote(self, h: PullHandle, readonly: bool) -> torch.Tensor:
@@ -188,7 +243,7 @@ class ShardedVariable:
         if h.bounded:
             self.shard.push_slots(h.unique, h.u_dev, h.slots, ugrads, counts)
             return
-        if self.world_size == 1:
+        if h.send_splits is None:       # pure-local (world 1) exact path
             self.shard.push(h.unique, ugrads, counts)
             return
         # one fused payload [u, dim+1] = grads ‖ counts-as-f32: a single
@@ -199,6 +254,8 @@ class ShardedVariable:
         send_p = payload.index_select(0, h.order)
         recv_p = comm.all_to_all_v(send_p, h.send_splits, h.recv_splits)
         u2 = h.owner_unique.numel()
+        if u2 == 0:
+            return  # no keys owned here this step
         g2c, _ = ops.reduce_by_inverse(h.owner_inverse,
                                        recv_p.contiguous(), u2)
         g2 = g2c[:, :dim].contiguous()
