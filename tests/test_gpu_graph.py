@@ -88,8 +88,11 @@ def test_graph_replay_matches_eager():
     got_rows = model2.embedding.variable.sparse_read(probe)
     got_dense = torch.cat([p.detach().reshape(-1).float()
                            for p in model2.dnn.parameters()])
-    assert torch.allclose(got_rows, ref_rows, atol=1e-5, rtol=1e-5)
-    assert torch.allclose(got_dense, ref_dense, atol=1e-5, rtol=1e-5)
+    # float atomics (head backward dw/db, reduce-by-key) make two separate
+    # runs differ at rounding level; a replay bug (e.g. a non-replayed
+    # scratch reset) shows up orders of magnitude above this
+    assert torch.allclose(got_rows, ref_rows, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(got_dense, ref_dense, atol=1e-4, rtol=1e-4)
 
 
 def test_graph_replay_hash_mode_with_reservation():
@@ -151,4 +154,4 @@ def test_graph_replay_hash_mode_with_reservation():
         g.replay()
     torch.cuda.synchronize()
     got = m2.embedding.variable.sparse_read(probe)
-    assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(got, ref, atol=1e-4, rtol=1e-4)
