@@ -171,6 +171,10 @@ class Embedding(nn.Module):
                  sparse_as_dense: bool = False,
                  storage: Optional[Storage] = None,
                  dtype: torch.dtype = torch.float32):
+        # num_shards is accepted for API parity (reference exb.py:276
+        # decoupled shard count from server count); here shards == ranks by
+        # construction and checkpoints reload across any world size, so the
+        # value is advisory only.
         super().__init__()
         ctx = get_context()
         self.ctx = ctx
