@@ -1,5 +1,6 @@
 """Checkpoint dump/load round-trips (reference §3.4 semantics)."""
 
+import pytest
 import torch
 
 import openembedding_amd.torch as embed
@@ -131,3 +132,74 @@ def test_multi_file_dump_reload(tmp_path):
         ctx.finalize()
     finally:
         oe.flags.config = old
+
+
+def test_load_rejects_corrupt_magic(tmp_path):
+    from openembedding_amd.context import Context
+    from openembedding_amd import checkpoint
+
+    ctx = Context(device="cpu")
+    st = ctx.create_storage()
+    v = st.create_variable(50, 4)
+    v.set_initializer("constant", value=1.0)
+    v.set_optimizer("adagrad", learning_rate=0.1)
+    v.shard.pull(torch.arange(5))
+    uri = str(tmp_path / "c")
+    checkpoint.dump_model(ctx, uri)
+    import os
+    f = os.path.join(uri, "0", "model_0_0")
+    blob = bytearray(open(f, "rb").read())
+    blob[0:8] = b"BADMAGIC"
+    open(f, "wb").write(bytes(blob))
+    with pytest.raises(RuntimeError, match="bad shard file"):
+        checkpoint.load_model(ctx, uri)
+    ctx.finalize()
+
+
+def test_load_rejects_dim_mismatch(tmp_path):
+    from openembedding_amd.context import Context
+    from openembedding_amd import checkpoint
+
+    ctx = Context(device="cpu")
+    st = ctx.create_storage()
+    v = st.create_variable(50, 4)
+    v.set_initializer("constant", value=1.0)
+    v.set_optimizer("adagrad", learning_rate=0.1)
+    v.shard.pull(torch.arange(5))
+    uri = str(tmp_path / "d")
+    checkpoint.dump_model(ctx, uri)
+    ctx.finalize()
+
+    import openembedding_amd.context as cm
+    cm._context = None
+    ctx2 = Context(device="cpu")
+    st2 = ctx2.create_storage()
+    st2.create_variable(50, 8)  # same id, wrong dim
+    with pytest.raises(RuntimeError, match="dim mismatch"):
+        checkpoint.load_model(ctx2, uri)
+    ctx2.finalize()
+
+
+def test_load_missing_variable_raises(tmp_path):
+    from openembedding_amd.context import Context
+    from openembedding_amd import checkpoint
+
+    ctx = Context(device="cpu")
+    st = ctx.create_storage()
+    for _ in range(2):
+        v = st.create_variable(50, 4)
+        v.set_initializer("constant", value=1.0)
+        v.set_optimizer("adagrad", learning_rate=0.1)
+        v.shard.pull(torch.arange(3))
+    uri = str(tmp_path / "m")
+    checkpoint.dump_model(ctx, uri)
+    ctx.finalize()
+
+    import openembedding_amd.context as cm
+    cm._context = None
+    ctx2 = Context(device="cpu")
+    st2 = ctx2.create_storage()
+    st2.create_variable(50, 4)  # only one of two
+    with pytest.raises(RuntimeError, match="missing in model"):
+        checkpoint.load_model(ctx2, uri)
+    ctx2.finalize()
