@@ -94,20 +94,13 @@ static __device__ __forceinline__ f32x4 tile16(
 // chunk — so they load ONCE per layer into registers; the B (weight)
 // stream double-buffers across chunks: chunk i+1's 13 fragment loads are
 // in flight while chunk i's MFMAs run.
-// KSN = next layer's K-steps: the wave's FIRST weight chunk of the next
-// layer is prefetched here, before the inter-layer barrier (plain loads
-// stay in flight across __syncthreads), so each layer no longer starts
-// with a cold ~600-cycle W wait.
-template <int KS, int KSN>
+template <int KS>
 static __device__ __forceinline__ void mlp_layer_u(
         const mbf16* A, long lda, long m0, long M,
         const mbf16* W, long ldw, const mbf16* bias, long H,
         const mbf16* mask_act, long mask_ld,
         mbf16* lds_out, mbf16* save, long save_ld,
-        int wave, int lane, bool relu,
-        bf16x8* b_in,                    // preloaded first chunk, or null
-        const mbf16* Wn, long ldwn, long Hn,
-        bf16x8* b_out) {                 // next layer's first chunk out
+        int wave, int lane, bool relu) {
     const long koff = (lane >> 4) * 8;
     const mbf16* pa = A + (lane & 15) * lda + koff;
     bf16x8 a[KS];
@@ -118,28 +111,16 @@ static __device__ __forceinline__ void mlp_layer_u(
     bf16x8 b0[KS], b1[KS];
     long c = wave * 16;
     if (c < H) {
-        if (b_in) {
-            #pragma unroll
-            for (int s = 0; s < KS; ++s) b0[s] = b_in[s];
-        } else {
-            const mbf16* pb = W + (c + col) * ldw + koff;
-            #pragma unroll
-            for (int s = 0; s < KS; ++s) b0[s] = ld_frag(pb + 32 * s);
-        }
+        const mbf16* pb = W + (c + col) * ldw + koff;
+        #pragma unroll
+        for (int s = 0; s < KS; ++s) b0[s] = ld_frag(pb + 32 * s);
     }
-    bool prefetched = false;
     for (; c < H; c += 16 * MLP_WAVES) {
         const long cn = c + 16 * MLP_WAVES;
         if (cn < H) {
             const mbf16* pb = W + (cn + col) * ldw + koff;
             #pragma unroll
             for (int s = 0; s < KS; ++s) b1[s] = ld_frag(pb + 32 * s);
-        } else if (KSN > 0 && Wn && (long)(wave * 16) < Hn) {
-            // last chunk of this layer: start the next layer's first chunk
-            const mbf16* pb = Wn + (wave * 16 + col) * ldwn + koff;
-            #pragma unroll
-            for (int s = 0; s < KSN; ++s) b_out[s] = ld_frag(pb + 32 * s);
-            prefetched = true;
         }
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
         #pragma unroll
@@ -163,11 +144,6 @@ static __device__ __forceinline__ void mlp_layer_u(
         #pragma unroll
         for (int s = 0; s < KS; ++s) b0[s] = b1[s];
     }
-    if (KSN > 0 && Wn && !prefetched && (long)(wave * 16) < Hn) {
-        const mbf16* pb = Wn + (wave * 16 + col) * ldwn + koff;
-        #pragma unroll
-        for (int s = 0; s < KSN; ++s) b_out[s] = ld_frag(pb + 32 * s);
-    }
 }
 
 static __device__ __forceinline__ void mlp_layer(
@@ -177,15 +153,13 @@ static __device__ __forceinline__ void mlp_layer(
         mbf16* lds_out, mbf16* save, long save_ld,
         int wave, int lane, bool relu) {
     if (Kp == 256)
-        return mlp_layer_u<8, 0>(A, lda, m0, M, W, ldw, bias, H, mask_act,
-                                 mask_ld, lds_out, save, save_ld, wave,
-                                 lane, relu, nullptr, nullptr, 0, 0,
-                                 nullptr);
+        return mlp_layer_u<8>(A, lda, m0, M, W, ldw, bias, H, mask_act,
+                              mask_ld, lds_out, save, save_ld, wave, lane,
+                              relu);
     if (Kp == 416)
-        return mlp_layer_u<13, 0>(A, lda, m0, M, W, ldw, bias, H, mask_act,
-                                  mask_ld, lds_out, save, save_ld, wave,
-                                  lane, relu, nullptr, nullptr, 0, 0,
-                                  nullptr);
+        return mlp_layer_u<13>(A, lda, m0, M, W, ldw, bias, H, mask_act,
+                               mask_ld, lds_out, save, save_ld, wave, lane,
+                               relu);
     // generic (unpipelined) fallback for other shapes
     for (long c = wave * 16; c < H; c += 16 * MLP_WAVES) {
         f32x4 acc = tile16(A, lda, W, ldw, c, Kp, lane);
@@ -233,33 +207,15 @@ void k_mlp3_fwd(const mbf16* __restrict__ x0, long M, long K0p,
         act[0][i] = (mbf16)0.0f;
     __syncthreads();
 
-    if (K0p == 256 && Hp == 416) {
-        // chained: each layer prefetches the next layer's first W chunk
-        // before the barrier (cross-barrier loads hide the cold start)
-        bf16x8 bpre[13];
-        mlp_layer_u<8, 13>(x0 + m0 * K0p, K0p, m0, M, w1, K0p, b1, H,
-                           nullptr, 0, act[0], a1, H, wave, lane, true,
-                           nullptr, w2, Hp, H, bpre);
-        __syncthreads();
-        mlp_layer_u<13, 13>(act[0], MLP_LD, m0, M, w2, Hp, b2, H,
-                            nullptr, 0, act[1], a2, H, wave, lane, true,
-                            bpre, w3, Hp, H, bpre);
-        __syncthreads();
-        mlp_layer_u<13, 0>(act[1], MLP_LD, m0, M, w3, Hp, b3, H,
-                           nullptr, 0, act[0], a3, H, wave, lane, true,
-                           bpre, nullptr, 0, 0, nullptr);
-        __syncthreads();
-    } else {
-        mlp_layer(x0 + m0 * K0p, K0p, m0, M, w1, K0p, b1, H, K0p,
-                  nullptr, 0, act[0], a1, H, wave, lane, true);
-        __syncthreads();
-        mlp_layer(act[0], MLP_LD, m0, M, w2, Hp, b2, H, Hp,
-                  nullptr, 0, act[1], a2, H, wave, lane, true);
-        __syncthreads();
-        mlp_layer(act[1], MLP_LD, m0, M, w3, Hp, b3, H, Hp,
-                  nullptr, 0, act[0], a3, H, wave, lane, true);
-        __syncthreads();
-    }
+    mlp_layer(x0 + m0 * K0p, K0p, m0, M, w1, K0p, b1, H, K0p,
+              nullptr, 0, act[0], a1, H, wave, lane, true);
+    __syncthreads();
+    mlp_layer(act[0], MLP_LD, m0, M, w2, Hp, b2, H, Hp,
+              nullptr, 0, act[1], a2, H, wave, lane, true);
+    __syncthreads();
+    mlp_layer(act[1], MLP_LD, m0, M, w3, Hp, b3, H, Hp,
+              nullptr, 0, act[0], a3, H, wave, lane, true);
+    __syncthreads();
 
     // final Linear(H, 1): VALU dot per row
     const int rows_per_wave = (MLP_BM + MLP_WAVES - 1) / MLP_WAVES;
@@ -315,29 +271,14 @@ void k_mlp3_bwd(const float* __restrict__ dout, long M, long K0p,
     // NOTE: dz tiles' pad columns [H, Hp) may hold garbage after a GEMM
     // stage — harmless, because the B side (padded weight copies) is zero
     // there, so pad products vanish.
-    if (Hp == 416) {
-        bf16x8 bpre[13];
-        mlp_layer_u<13, 13>(dz[0], MLP_LD, m0, M, w3t, Hp, nullptr, H,
-                            a2, H, dz[1], dz2, H, wave, lane, false,
-                            nullptr, w2t, Hp, H, bpre);
-        __syncthreads();
-        mlp_layer_u<13, 13>(dz[1], MLP_LD, m0, M, w2t, Hp, nullptr, H,
-                            a1, H, dz[0], dz1, H, wave, lane, false,
-                            bpre, w1t, Hp, K0p, bpre);
-        __syncthreads();
-        mlp_layer_u<13, 0>(dz[0], MLP_LD, m0, M, w1t, Hp, nullptr, K0p,
-                           nullptr, 0, nullptr, dx0, K0p, wave, lane,
-                           false, bpre, nullptr, 0, 0, nullptr);
-    } else {
-        mlp_layer(dz[0], MLP_LD, m0, M, w3t, Hp, nullptr, H, Hp,
-                  a2, H, dz[1], dz2, H, wave, lane, false);
-        __syncthreads();
-        mlp_layer(dz[1], MLP_LD, m0, M, w2t, Hp, nullptr, H, Hp,
-                  a1, H, dz[0], dz1, H, wave, lane, false);
-        __syncthreads();
-        mlp_layer(dz[0], MLP_LD, m0, M, w1t, Hp, nullptr, K0p, Hp,
-                  nullptr, 0, nullptr, dx0, K0p, wave, lane, false);
-    }
+    mlp_layer(dz[0], MLP_LD, m0, M, w3t, Hp, nullptr, H, Hp,
+              a2, H, dz[1], dz2, H, wave, lane, false);
+    __syncthreads();
+    mlp_layer(dz[1], MLP_LD, m0, M, w2t, Hp, nullptr, H, Hp,
+              a1, H, dz[0], dz1, H, wave, lane, false);
+    __syncthreads();
+    mlp_layer(dz[0], MLP_LD, m0, M, w1t, Hp, nullptr, K0p, Hp,
+              nullptr, 0, nullptr, dx0, K0p, wave, lane, false);
 }
 
 extern "C" void emb_mlp3_fwd(const void* x0, long M, long K0p,
