@@ -97,7 +97,9 @@ def main():
     opt = embed.distributed_optimizer(
         torch.optim.Adagrad(model.parameters(), lr=0.005),
         flatten_dense=True)
-    lossf = torch.nn.BCEWithLogitsLoss()
+    # fused 2-kernel BCE on GPU (torch's spends ~5 launches/step in the
+    # captured graph), plain torch BCE elsewhere
+    from openembedding_amd.ops.dispatch import bce_with_logits as lossf
 
     gen = torch.Generator(device="cpu").manual_seed(4242 + rank)
     pool = []

@@ -48,6 +48,9 @@ void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
                       const float*, const float*, long, long, long, long,
                       long, float*, float*, float*, float*, int, int,
                       hipStream_t_);
+void emb_bce_fwd(const float*, const float*, long, float*, hipStream_t_);
+void emb_bce_bwd(const float*, const float*, long, const float*, float*,
+                 hipStream_t_);
 void emb_flat_adagrad_f32(float*, float*, const float*, long, float, float,
                           hipStream_t_);
 void emb_flat_adagrad_bf16(float*, float*, const void*, void*, long, float,
@@ -309,6 +312,32 @@ void flat_adagrad(torch::Tensor param, torch::Tensor accum,
     }
 }
 
+// ---- fused BCE-with-logits --------------------------------------------
+
+torch::Tensor bce_fwd(torch::Tensor logits, torch::Tensor labels) {
+    CHECK_GPU(logits); CHECK_CONT(logits); CHECK_CONT(labels);
+    TORCH_CHECK(logits.dtype() == torch::kFloat32 &&
+                labels.dtype() == torch::kFloat32, "bce_fwd wants fp32");
+    TORCH_CHECK(logits.numel() == labels.numel(), "logits/labels mismatch");
+    const c10::cuda::CUDAGuard guard(logits.device());
+    auto loss = torch::empty({}, logits.options());
+    emb_bce_fwd(logits.data_ptr<float>(), labels.data_ptr<float>(),
+                logits.numel(), loss.data_ptr<float>(), cur_stream());
+    return loss;
+}
+
+torch::Tensor bce_bwd(torch::Tensor logits, torch::Tensor labels,
+                      torch::Tensor grad_out) {
+    CHECK_GPU(logits); CHECK_CONT(logits); CHECK_CONT(labels);
+    TORCH_CHECK(grad_out.numel() == 1, "grad_out must be scalar");
+    const c10::cuda::CUDAGuard guard(logits.device());
+    auto g = torch::empty_like(logits);
+    emb_bce_bwd(logits.data_ptr<float>(), labels.data_ptr<float>(),
+                logits.numel(), grad_out.contiguous().data_ptr<float>(),
+                g.data_ptr<float>(), cur_stream());
+    return g;
+}
+
 // ---- fused 3-layer MLP forward ----------------------------------------
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
@@ -382,6 +411,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused dgrad chain backward of the 3-layer MLP");
     m.def("mlp3_fwd", &mlp3_fwd,
           "fused 3-hidden-layer MLP forward (bf16 MFMA, bias+ReLU fused)");
+    m.def("bce_fwd", &bce_fwd, "fused BCE-with-logits forward (mean)");
+    m.def("bce_bwd", &bce_bwd, "fused BCE-with-logits backward");
     m.def("flat_adagrad", &flat_adagrad,
           "fused flat-buffer Adagrad (f32, or bf16 weights + f32 master)");
 }

@@ -721,6 +721,36 @@ __global__ void k_flat_adagrad_bf16(float* __restrict__ master,
     p[i] = (oebf16)m;
 }
 
+// ---- fused BCE-with-logits (mean) --------------------------------------
+// torch's BCEWithLogitsLoss costs ~5 launches per step in the captured
+// train graph (log_sigmoid, mean reduce, grad fill, sigmoid-sub-scale
+// chain); these two kernels replace them. Same stable formulation as
+// torch: max(z,0) - z*y + log1p(exp(-|z|)).
+
+__global__ void k_bce_fwd(const float* __restrict__ z,
+                          const float* __restrict__ y, long n, float inv_n,
+                          float* __restrict__ loss) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    float v = 0.f;
+    if (i < n) {
+        float zi = z[i];
+        v = (fmaxf(zi, 0.f) - zi * y[i] + log1pf(__expf(-fabsf(zi)))) * inv_n;
+    }
+    for (int off = 32; off; off >>= 1) v += __shfl_down(v, off);
+    if ((threadIdx.x & 63) == 0 && v != 0.f) atomicAdd(loss, v);
+}
+
+__global__ void k_bce_bwd(const float* __restrict__ z,
+                          const float* __restrict__ y, long n, float inv_n,
+                          const float* __restrict__ go,
+                          float* __restrict__ g) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    float zi = z[i];
+    float s = 1.f / (1.f + __expf(-zi));
+    g[i] = (s - y[i]) * inv_n * go[0];
+}
+
 // ============================================================== launchers
 
 extern "C" {
@@ -875,6 +905,19 @@ void emb_flat_adagrad_bf16(float* master, float* accum, const void* g,
                            hipStream_t stream) {
     if (n) k_flat_adagrad_bf16<<<grid1d(n), BLOCK, 0, stream>>>(
         master, accum, (const oebf16*)g, (oebf16*)p, n, lr, eps);
+}
+
+void emb_bce_fwd(const float* z, const float* y, long n, float* loss,
+                 hipStream_t stream) {
+    fill_f32(loss, 1, 0.f, stream);
+    int ga = grid1d(n);
+    if (n) k_bce_fwd<<<ga, BLOCK, 0, stream>>>(z, y, n, 1.f / (float)n, loss);
+}
+
+void emb_bce_bwd(const float* z, const float* y, long n, const float* go,
+                 float* g, hipStream_t stream) {
+    int ga = grid1d(n);
+    if (n) k_bce_bwd<<<ga, BLOCK, 0, stream>>>(z, y, n, 1.f / (float)n, go, g);
 }
 
 }  // extern "C"

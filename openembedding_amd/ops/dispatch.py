@@ -53,3 +53,38 @@ def reduce_by_inverse(inverse: torch.Tensor, grads: torch.Tensor, u: int
     ugrads.index_add_(0, inverse, grads)
     counts = torch.bincount(inverse, minlength=u).to(torch.int64)
     return ugrads, counts
+
+
+# ----------------------------------------------------------------- fused loss
+
+class _FusedBCEFn(torch.autograd.Function):
+    """BCEWithLogitsLoss(mean) in 2 kernels (torch spends ~5 launches per
+    step on it inside the captured train graph: log_sigmoid + mean reduce +
+    grad fill + the sigmoid-sub-scale chain). Same stable formulation, so
+    it matches torch to fp32 atomic-order noise."""
+
+    @staticmethod
+    def forward(ctx, logits, labels, ext):
+        logits = logits.contiguous()
+        labels = labels.contiguous()
+        ctx.save_for_backward(logits, labels)
+        return ext.bce_fwd(logits, labels)
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        logits, labels = ctx.saved_tensors
+        ext = require_hip()
+        g = ext.bce_bwd(logits, labels, grad_out)
+        return g, None, None
+
+
+def bce_with_logits(logits: torch.Tensor, labels: torch.Tensor
+                    ) -> torch.Tensor:
+    """Mean binary-cross-entropy-with-logits; fused HIP kernels on GPU,
+    torch elsewhere. Differentiable w.r.t. logits."""
+    if _use_hip(logits) and logits.dtype == torch.float32:
+        ext = require_hip()
+        if ext is not None:
+            return _FusedBCEFn.apply(logits, labels.float(), ext)
+    return torch.nn.functional.binary_cross_entropy_with_logits(
+        logits, labels.to(logits.dtype))

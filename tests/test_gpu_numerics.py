@@ -224,3 +224,23 @@ def test_remote_bounded_path_matches_local(tmp_path):
     ra = a.shard.pull_readonly(probe)
     rb = b.shard.pull_readonly(probe)
     assert torch.allclose(ra, rb, atol=1e-6)
+
+
+def test_fused_bce_matches_torch():
+    # fwd + bwd of the 2-kernel BCE vs torch.nn.BCEWithLogitsLoss (fp32)
+    from openembedding_amd.ops.dispatch import bce_with_logits
+    g = torch.Generator(device="cpu").manual_seed(7)
+    for n in (1, 63, 4096, 10000):
+        z = (torch.randn(n, generator=g) * 4).to(DEV).requires_grad_(True)
+        y = (torch.rand(n, generator=g) < 0.3).float().to(DEV)
+        z2 = z.detach().clone().requires_grad_(True)
+
+        loss = bce_with_logits(z, y)
+        ref = torch.nn.functional.binary_cross_entropy_with_logits(z2, y)
+        # atomic-order fp32 sum vs torch's tree reduce
+        assert torch.allclose(loss, ref, atol=1e-5, rtol=1e-5), \
+            (n, loss.item(), ref.item())
+
+        (loss * 3.0).backward()
+        (ref * 3.0).backward()
+        assert torch.allclose(z.grad, z2.grad, atol=1e-6, rtol=1e-5)
