@@ -108,10 +108,24 @@ class _FusedMLP3Fn(torch.autograd.Function):
             w2g.addmm_(dz2.t(), a1)
             w3g.addmm_(dz3.t(), a2)
             w4g.addmm_(d, a3)
-            b1g.unsqueeze(0).addmm_(ones, dz1)
-            b2g.unsqueeze(0).addmm_(ones, dz2)
-            b3g.unsqueeze(0).addmm_(ones, dz3)
-            b4g.add_(d.sum())
+            if b1g.dtype == torch.bfloat16 and b1g.is_contiguous() \
+                    and b2g.is_contiguous() and b3g.is_contiguous():
+                # all 4 bias grads in one pass over the dz mirrors (+ a
+                # finisher that folds the fp32 scratch into the bf16 grads
+                # and re-zeros it) — replaces 3 GEMV launches + reduce + add
+                H = a1.shape[1]
+                scratch = bufs.get("bscratch")
+                if scratch is None or scratch.numel() != 3 * H + 1:
+                    scratch = torch.zeros(3 * H + 1, dtype=torch.float32,
+                                          device=a1.device)
+                    bufs["bscratch"] = scratch
+                ext.mlp3_bias_bwd(dout.contiguous(), dz1, dz2, dz3, scratch,
+                                  b1g, b2g, b3g, b4g)
+            else:
+                b1g.unsqueeze(0).addmm_(ones, dz1)
+                b2g.unsqueeze(0).addmm_(ones, dz2)
+                b3g.unsqueeze(0).addmm_(ones, dz3)
+                b4g.add_(d.sum())
             return (dx0, None, None, None, None, None, None, None, None,
                     None)
         # wgrads stay library GEMMs — their K = batch shapes run well

@@ -48,6 +48,9 @@ void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
                       const float*, const float*, long, long, long, long,
                       long, float*, float*, float*, float*, int, int,
                       hipStream_t_);
+void emb_mlp3_bias_bwd(const float*, const void*, const void*, const void*,
+                       long, long, float*, void*, void*, void*, void*,
+                       hipStream_t_);
 void emb_bce_fwd(const float*, const float*, long, float*, hipStream_t_);
 void emb_bce_bwd(const float*, const float*, long, const float*, float*,
                  hipStream_t_);
@@ -312,6 +315,31 @@ void flat_adagrad(torch::Tensor param, torch::Tensor accum,
     }
 }
 
+// ---- fused MLP bias grads ---------------------------------------------
+
+void mlp3_bias_bwd(torch::Tensor dout, torch::Tensor dz1, torch::Tensor dz2,
+                   torch::Tensor dz3, torch::Tensor scratch,
+                   torch::Tensor db1, torch::Tensor db2, torch::Tensor db3,
+                   torch::Tensor db4) {
+    CHECK_GPU(dout); CHECK_CONT(dout); CHECK_CONT(dz1); CHECK_CONT(dz2);
+    CHECK_CONT(dz3); CHECK_CONT(scratch);
+    long M = dz1.size(0), H = dz1.size(1);
+    TORCH_CHECK(dout.numel() == M && scratch.numel() >= 3 * H + 1,
+                "mlp3_bias_bwd shape mismatch");
+    TORCH_CHECK(db1.dtype() == torch::kBFloat16 &&
+                scratch.dtype() == torch::kFloat32,
+                "mlp3_bias_bwd wants bf16 grads + fp32 scratch");
+    TORCH_CHECK(db1.is_contiguous() && db2.is_contiguous() &&
+                db3.is_contiguous() && db4.is_contiguous() &&
+                db1.numel() == H && db2.numel() == H && db3.numel() == H &&
+                db4.numel() == 1, "bias grad layout");
+    const c10::cuda::CUDAGuard guard(dout.device());
+    emb_mlp3_bias_bwd(dout.data_ptr<float>(), dz1.data_ptr(), dz2.data_ptr(),
+                      dz3.data_ptr(), M, H, scratch.data_ptr<float>(),
+                      db1.data_ptr(), db2.data_ptr(), db3.data_ptr(),
+                      db4.data_ptr(), cur_stream());
+}
+
 // ---- fused BCE-with-logits --------------------------------------------
 
 torch::Tensor bce_fwd(torch::Tensor logits, torch::Tensor labels) {
@@ -411,6 +439,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused dgrad chain backward of the 3-layer MLP");
     m.def("mlp3_fwd", &mlp3_fwd,
           "fused 3-hidden-layer MLP forward (bf16 MFMA, bias+ReLU fused)");
+    m.def("mlp3_bias_bwd", &mlp3_bias_bwd,
+          "all MLP bias grads in one pass over the dz mirrors");
     m.def("bce_fwd", &bce_fwd, "fused BCE-with-logits forward (mean)");
     m.def("bce_bwd", &bce_bwd, "fused BCE-with-logits backward");
     m.def("flat_adagrad", &flat_adagrad,
