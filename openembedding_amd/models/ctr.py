@@ -60,7 +60,7 @@ class _FusedMLP3Fn(torch.autograd.Function):
     library-friendly ones)."""
 
     @staticmethod
-    def forward(ctx, x0, w1, b1, w2, b2, w3, b3, w4, b4, bufs):
+    def forward(ctx, x0, partial, w1, b1, w2, b2, w3, b3, w4, b4, bufs):
         from ..ops import require_hip
         ext = require_hip()
         # refresh padded weight copies (zero tails allocated once in bufs)
@@ -68,11 +68,16 @@ class _FusedMLP3Fn(torch.autograd.Function):
         bufs["w2p"][:, :w2.shape[1]].copy_(w2)
         bufs["w3p"][:, :w3.shape[1]].copy_(w3)
         w4f = w4.reshape(-1).contiguous()
+        # partial (the head's first-order+FM+dense logits) folds into the
+        # final-dot epilogue, replacing the `partial + dnn` add kernel
         out, a1, a2, a3 = ext.mlp3_fwd(x0, bufs["w1p"], b1, bufs["w2p"], b2,
-                                       bufs["w3p"], b3, w4f, b4)
+                                       bufs["w3p"], b3, w4f, b4,
+                                       None if partial is None
+                                       else partial.contiguous())
         ctx.save_for_backward(x0, w1, w2, w3, w4f, a1, a2, a3)
         ctx.bufs = bufs
         ctx._params = (w1, b1, w2, b2, w3, b3, w4, b4)
+        ctx._has_partial = partial is not None
         return out
 
     @staticmethod
@@ -89,14 +94,20 @@ class _FusedMLP3Fn(torch.autograd.Function):
         dx0, dz1, dz2, dz3 = ext.mlp3_bwd(
             dout.contiguous(), a1, a2, a3, w4f,
             bufs["w3tp"], bufs["w2tp"], bufs["w1tp"])
-        d = dout.unsqueeze(0).to(a3.dtype)             # [1, M]
-        # bias grads as GEMVs: torch's column-sum of row-major bf16 ran
-        # ~16 us each (reduce_kernel); ones@dz is a hipBLASLt GEMV
-        ones = bufs.get("ones")
-        if ones is None or ones.shape[1] != a1.shape[0]:
-            ones = torch.ones(1, a1.shape[0], dtype=a1.dtype,
-                              device=a1.device)
-            bufs["ones"] = ones
+        # bias grads as GEMVs (fallback paths): torch's column-sum of
+        # row-major bf16 ran ~16 us each (reduce_kernel); ones@dz is a
+        # hipBLASLt GEMV. The default path below does them in the fused
+        # bias kernel instead and never materializes d/ones.
+        def _d():
+            return dout.unsqueeze(0).to(a3.dtype)      # [1, M]
+
+        def _ones():
+            ones = bufs.get("ones")
+            if ones is None or ones.shape[1] != a1.shape[0]:
+                ones = torch.ones(1, a1.shape[0], dtype=a1.dtype,
+                                  device=a1.device)
+                bufs["ones"] = ones
+            return ones
         # With pre-bound .grad views (the flat-optimizer bench path), the
         # wgrads ACCUMULATE in place via beta=1 addmm_ — one fused GEMM per
         # param instead of GEMM + autograd's separate accumulate-add (8 add
@@ -107,28 +118,35 @@ class _FusedMLP3Fn(torch.autograd.Function):
             w1g.addmm_(dz1.t(), x0[:, :K0])
             w2g.addmm_(dz2.t(), a1)
             w3g.addmm_(dz3.t(), a2)
-            w4g.addmm_(d, a3)
+            dpart = dout if ctx._has_partial else None
             if b1g.dtype == torch.bfloat16 and b1g.is_contiguous() \
-                    and b2g.is_contiguous() and b3g.is_contiguous():
-                # all 4 bias grads in one pass over the dz mirrors (+ a
-                # finisher that folds the fp32 scratch into the bf16 grads
-                # and re-zeros it) — replaces 3 GEMV launches + reduce + add
+                    and b2g.is_contiguous() and b3g.is_contiguous() \
+                    and w4g.is_contiguous():
+                # bias grads for all 4 layers + the head wgrad in one pass
+                # over the dz mirrors (+ a finisher that folds the fp32
+                # scratch into the bf16 grads and re-zeros it) — replaces
+                # 4 GEMV launches + a reduce + an add + a dout bf16 cast
                 H = a1.shape[1]
                 scratch = bufs.get("bscratch")
-                if scratch is None or scratch.numel() != 3 * H + 1:
-                    scratch = torch.zeros(3 * H + 1, dtype=torch.float32,
+                if scratch is None or scratch.numel() != 4 * H + 1:
+                    scratch = torch.zeros(4 * H + 1, dtype=torch.float32,
                                           device=a1.device)
                     bufs["bscratch"] = scratch
-                ext.mlp3_bias_bwd(dout.contiguous(), dz1, dz2, dz3, scratch,
-                                  b1g, b2g, b3g, b4g)
+                ext.mlp3_bias_bwd(dout.contiguous(), dz1, dz2, dz3, a3,
+                                  scratch, b1g, b2g, b3g,
+                                  w4g.reshape(-1), b4g)
             else:
+                d, ones = _d(), _ones()
+                w4g.addmm_(d, a3)
                 b1g.unsqueeze(0).addmm_(ones, dz1)
                 b2g.unsqueeze(0).addmm_(ones, dz2)
                 b3g.unsqueeze(0).addmm_(ones, dz3)
                 b4g.add_(d.sum())
-            return (dx0, None, None, None, None, None, None, None, None,
-                    None)
+            return (dx0, dpart, None, None, None, None, None, None, None,
+                    None, None)
         # wgrads stay library GEMMs — their K = batch shapes run well
+        d, ones = _d(), _ones()
+        dpart = dout if ctx._has_partial else None
         dw1 = dz1.t() @ x0[:, :K0]
         dw2 = dz2.t() @ a1
         dw3 = dz3.t() @ a2
@@ -137,7 +155,7 @@ class _FusedMLP3Fn(torch.autograd.Function):
         db2 = (ones @ dz2).reshape(-1)
         db3 = (ones @ dz3).reshape(-1)
         db4 = d.sum(1)
-        return (dx0, dw1, db1, dw2, db2, dw3, db3, dw4, db4, None)
+        return (dx0, dpart, dw1, db1, dw2, db2, dw3, db3, dw4, db4, None)
 
 
 def _mlp(in_dim: int, hidden: Sequence[int], out_dim: int = 1) -> nn.Sequential:
@@ -200,8 +218,10 @@ class _CTRBase(nn.Module):
             use_fm, self.head_bf16, self.fused_mlp)
         return e_all, deep_in, partial
 
-    def _dnn_out(self, deep_in: torch.Tensor) -> torch.Tensor:
-        """dnn logits [B]; the fused single-kernel MLP when enabled."""
+    def _dnn_out(self, deep_in: torch.Tensor,
+                 partial: torch.Tensor = None) -> torch.Tensor:
+        """dnn logits [B] (+ `partial` carried in, when given — the fused
+        kernel folds the add into its final-dot epilogue)."""
         if (self.fused_mlp and deep_in.is_cuda
                 and deep_in.dtype == torch.bfloat16):
             l1, l2, l3, l4 = self.dnn[0], self.dnn[2], self.dnn[4], self.dnn[6]
@@ -215,10 +235,11 @@ class _CTRBase(nn.Module):
                                "w3p": z(H, Hp), "w3tp": z(H, Hp),
                                "w2tp": z(H, Hp), "w1tp": z(K0p, Hp)}
             return _FusedMLP3Fn.apply(
-                deep_in, l1.weight, l1.bias, l2.weight, l2.bias,
+                deep_in, partial, l1.weight, l1.bias, l2.weight, l2.bias,
                 l3.weight, l3.bias, l4.weight, l4.bias, self._w1pad)
         w_dtype = next(self.dnn.parameters()).dtype
-        return self.dnn(deep_in.to(w_dtype)).squeeze(-1).float()
+        out = self.dnn(deep_in.to(w_dtype)).squeeze(-1).float()
+        return out if partial is None else partial + out
 
 
 class LR(_CTRBase):
@@ -244,7 +265,7 @@ class WDL(_CTRBase):
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
         if self._use_fused_head(dense):
             _, deep_in, partial = self._fused_head(dense, sparse, use_fm=False)
-            return partial + self._dnn_out(deep_in)
+            return self._dnn_out(deep_in, partial)
         e, lin = self._embed(sparse)                     # [B, F, d], [B, F]
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
         return self._first_order(dense, lin) + self._dnn_out(deep_in)
@@ -261,7 +282,7 @@ class DeepFM(_CTRBase):
     def forward(self, dense: torch.Tensor, sparse: torch.Tensor) -> torch.Tensor:
         if self._use_fused_head(dense):
             _, deep_in, partial = self._fused_head(dense, sparse, use_fm=True)
-            return partial + self._dnn_out(deep_in)
+            return self._dnn_out(deep_in, partial)
         e, lin = self._embed(sparse)                     # [B, F, d], [B, F]
         # FM second order: 0.5*((sum_f e)^2 - sum_f e^2) summed over dim
         s = e.sum(dim=1)
@@ -313,7 +334,7 @@ class xDeepFM(_CTRBase):
             e_all, deep_in, partial = self._fused_head(dense, sparse,
                                                        use_fm=False)
             e = e_all[..., :self.dim]
-            return partial + self.cin(e) + self._dnn_out(deep_in)
+            return self._dnn_out(deep_in, partial + self.cin(e))
         e, lin = self._embed(sparse)
         deep_in = torch.cat([e.flatten(1), dense], dim=1)
         return (self._first_order(dense, lin) + self.cin(e)

@@ -49,8 +49,8 @@ void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
                       long, float*, float*, float*, float*, int, int,
                       hipStream_t_);
 void emb_mlp3_bias_bwd(const float*, const void*, const void*, const void*,
-                       long, long, float*, void*, void*, void*, void*,
-                       hipStream_t_);
+                       const void*, long, long, float*, void*, void*, void*,
+                       void*, void*, hipStream_t_);
 void emb_bce_fwd(const float*, const float*, long, float*, hipStream_t_);
 void emb_bce_bwd(const float*, const float*, long, const float*, float*,
                  hipStream_t_);
@@ -60,8 +60,8 @@ void emb_flat_adagrad_bf16(float*, float*, const void*, void*, long, float,
                            float, hipStream_t_);
 void emb_mlp3_fwd(const void*, long, long, const void*, const void*,
                   const void*, const void*, const void*, const void*,
-                  const void*, const void*, long, long, void*, void*, void*,
-                  float*, hipStream_t_);
+                  const void*, const void*, const float*, long, long, void*,
+                  void*, void*, float*, hipStream_t_);
 void emb_mlp3_bwd(const float*, long, long, const void*, const void*,
                   const void*, const void*, const void*, const void*,
                   const void*, long, long, void*, void*, void*, void*,
@@ -318,26 +318,28 @@ void flat_adagrad(torch::Tensor param, torch::Tensor accum,
 // ---- fused MLP bias grads ---------------------------------------------
 
 void mlp3_bias_bwd(torch::Tensor dout, torch::Tensor dz1, torch::Tensor dz2,
-                   torch::Tensor dz3, torch::Tensor scratch,
+                   torch::Tensor dz3, torch::Tensor a3, torch::Tensor scratch,
                    torch::Tensor db1, torch::Tensor db2, torch::Tensor db3,
-                   torch::Tensor db4) {
+                   torch::Tensor dw4, torch::Tensor db4) {
     CHECK_GPU(dout); CHECK_CONT(dout); CHECK_CONT(dz1); CHECK_CONT(dz2);
-    CHECK_CONT(dz3); CHECK_CONT(scratch);
+    CHECK_CONT(dz3); CHECK_CONT(a3); CHECK_CONT(scratch);
     long M = dz1.size(0), H = dz1.size(1);
-    TORCH_CHECK(dout.numel() == M && scratch.numel() >= 3 * H + 1,
-                "mlp3_bias_bwd shape mismatch");
+    TORCH_CHECK(dout.numel() == M && scratch.numel() >= 4 * H + 1 &&
+                a3.sizes() == dz1.sizes(), "mlp3_bias_bwd shape mismatch");
     TORCH_CHECK(db1.dtype() == torch::kBFloat16 &&
                 scratch.dtype() == torch::kFloat32,
                 "mlp3_bias_bwd wants bf16 grads + fp32 scratch");
     TORCH_CHECK(db1.is_contiguous() && db2.is_contiguous() &&
-                db3.is_contiguous() && db4.is_contiguous() &&
+                db3.is_contiguous() && dw4.is_contiguous() &&
+                db4.is_contiguous() &&
                 db1.numel() == H && db2.numel() == H && db3.numel() == H &&
-                db4.numel() == 1, "bias grad layout");
+                dw4.numel() == H && db4.numel() == 1, "bias grad layout");
     const c10::cuda::CUDAGuard guard(dout.device());
     emb_mlp3_bias_bwd(dout.data_ptr<float>(), dz1.data_ptr(), dz2.data_ptr(),
-                      dz3.data_ptr(), M, H, scratch.data_ptr<float>(),
+                      dz3.data_ptr(), a3.data_ptr(), M, H,
+                      scratch.data_ptr<float>(),
                       db1.data_ptr(), db2.data_ptr(), db3.data_ptr(),
-                      db4.data_ptr(), cur_stream());
+                      dw4.data_ptr(), db4.data_ptr(), cur_stream());
 }
 
 // ---- fused BCE-with-logits --------------------------------------------
@@ -371,7 +373,8 @@ torch::Tensor bce_bwd(torch::Tensor logits, torch::Tensor labels,
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 mlp3_fwd(torch::Tensor x0, torch::Tensor w1, torch::Tensor b1,
          torch::Tensor w2, torch::Tensor b2, torch::Tensor w3,
-         torch::Tensor b3, torch::Tensor w4, torch::Tensor b4) {
+         torch::Tensor b3, torch::Tensor w4, torch::Tensor b4,
+         c10::optional<torch::Tensor> partial) {
     CHECK_GPU(x0); CHECK_CONT(x0);
     TORCH_CHECK(x0.dtype() == torch::kBFloat16, "mlp3_fwd: x0 must be bf16");
     const c10::cuda::CUDAGuard guard(x0.device());
@@ -386,10 +389,17 @@ mlp3_fwd(torch::Tensor x0, torch::Tensor w1, torch::Tensor b1,
     auto a2 = torch::empty({M, H}, x0.options());
     auto a3 = torch::empty({M, H}, x0.options());
     auto out = torch::empty({M}, x0.options().dtype(torch::kFloat32));
+    const float* pp = nullptr;
+    if (partial.has_value()) {
+        TORCH_CHECK(partial->is_contiguous() && partial->numel() == M &&
+                    partial->dtype() == torch::kFloat32,
+                    "mlp3_fwd: partial must be contiguous fp32 [M]");
+        pp = partial->data_ptr<float>();
+    }
     emb_mlp3_fwd(x0.data_ptr(), M, K0p,
                  w1.data_ptr(), b1.data_ptr(), w2.data_ptr(), b2.data_ptr(),
                  w3.data_ptr(), b3.data_ptr(), w4.data_ptr(), b4.data_ptr(),
-                 H, Hp, a1.data_ptr(), a2.data_ptr(), a3.data_ptr(),
+                 pp, H, Hp, a1.data_ptr(), a2.data_ptr(), a3.data_ptr(),
                  out.data_ptr<float>(), cur_stream());
     return {out, a1, a2, a3};
 }
@@ -440,7 +450,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("mlp3_fwd", &mlp3_fwd,
           "fused 3-hidden-layer MLP forward (bf16 MFMA, bias+ReLU fused)");
     m.def("mlp3_bias_bwd", &mlp3_bias_bwd,
-          "all MLP bias grads in one pass over the dz mirrors");
+          "MLP bias grads + head wgrad in one pass over the dz mirrors");
     m.def("bce_fwd", &bce_fwd, "fused BCE-with-logits forward (mean)");
     m.def("bce_bwd", &bce_bwd, "fused BCE-with-logits backward");
     m.def("flat_adagrad", &flat_adagrad,

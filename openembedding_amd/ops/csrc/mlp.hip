@@ -191,6 +191,7 @@ void k_mlp3_fwd(const mbf16* __restrict__ x0, long M, long K0p,
                 const mbf16* __restrict__ b3,
                 const mbf16* __restrict__ w4,   // [H]
                 const mbf16* __restrict__ b4,
+                const float* __restrict__ partial,  // [M] logit carry-in or null
                 long H, long Hp,
                 mbf16* __restrict__ a1, mbf16* __restrict__ a2,
                 mbf16* __restrict__ a3, float* __restrict__ out) {
@@ -229,7 +230,8 @@ void k_mlp3_fwd(const mbf16* __restrict__ x0, long M, long K0p,
         #pragma unroll
         for (int off = 32; off; off >>= 1)
             s += __shfl_down(s, off, 64);
-        if (lane == 0) out[gm] = s + (float)b4[0];
+        if (lane == 0)
+            out[gm] = s + (float)b4[0] + (partial ? partial[gm] : 0.f);
     }
 }
 
@@ -291,22 +293,26 @@ void k_mlp3_bwd(const float* __restrict__ dout, long M, long K0p,
 extern "C" __global__ void k_mlp3_bias_bwd(
         const float* __restrict__ dout,
         const mbf16* __restrict__ dz1, const mbf16* __restrict__ dz2,
-        const mbf16* __restrict__ dz3, long M, long H, long rows_per_blk,
+        const mbf16* __restrict__ dz3, const mbf16* __restrict__ a3,
+        long M, long H, long rows_per_blk,
         float* __restrict__ scratch) {
     const long r0 = (long)blockIdx.x * rows_per_blk;
     if (r0 >= M) return;
     const long r1 = min(M, r0 + rows_per_blk);
     // column sums: thread t covers columns t, t+blockDim.x, ... of each dz
+    // (+ the head wgrad dw4[c] = sum_r dout[r]*a3[r,c] — same access shape)
     for (long c = threadIdx.x; c < H; c += blockDim.x) {
-        float s1 = 0.f, s2 = 0.f, s3 = 0.f;
+        float s1 = 0.f, s2 = 0.f, s3 = 0.f, sw = 0.f;
         for (long r = r0; r < r1; ++r) {
             s1 += (float)dz1[r * H + c];
             s2 += (float)dz2[r * H + c];
             s3 += (float)dz3[r * H + c];
+            sw += dout[r] * (float)a3[r * H + c];
         }
         atomicAdd(scratch + c, s1);
         atomicAdd(scratch + H + c, s2);
         atomicAdd(scratch + 2 * H + c, s3);
+        atomicAdd(scratch + 3 * H + c, sw);
     }
     // head bias: sum of dout rows, one atomic per wave
     float s4 = 0.f;
@@ -314,38 +320,43 @@ extern "C" __global__ void k_mlp3_bias_bwd(
         s4 += dout[r];
     #pragma unroll
     for (int off = 32; off; off >>= 1) s4 += __shfl_down(s4, off, 64);
-    if ((threadIdx.x & 63) == 0 && s4 != 0.f) atomicAdd(scratch + 3 * H, s4);
+    if ((threadIdx.x & 63) == 0 && s4 != 0.f) atomicAdd(scratch + 4 * H, s4);
 }
 
 extern "C" __global__ void k_mlp3_bias_finish(
         float* __restrict__ scratch, long H,
         mbf16* __restrict__ db1, mbf16* __restrict__ db2,
-        mbf16* __restrict__ db3, mbf16* __restrict__ db4) {
+        mbf16* __restrict__ db3, mbf16* __restrict__ dw4,
+        mbf16* __restrict__ db4) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i > 3 * H) return;
+    if (i > 4 * H) return;
     float v = scratch[i];
     scratch[i] = 0.f;               // self-cleaning for the next step
     mbf16* dst = i < H        ? db1 + i
                : i < 2 * H    ? db2 + (i - H)
                : i < 3 * H    ? db3 + (i - 2 * H)
+               : i < 4 * H    ? dw4 + (i - 3 * H)
                :                db4;
     *dst = (mbf16)((float)*dst + v);    // zero_grad zeroed it; += matches
 }                                       // torch's beta=1 accumulation
 
 extern "C" void emb_mlp3_bias_bwd(const float* dout, const void* dz1,
                                   const void* dz2, const void* dz3,
+                                  const void* a3,
                                   long M, long H, float* scratch,
-                                  void* db1, void* db2, void* db3, void* db4,
+                                  void* db1, void* db2, void* db3,
+                                  void* dw4, void* db4,
                                   hipStream_t stream) {
     if (M == 0) return;
     const long rows_per_blk = 32;
     int ga = (int)((M + rows_per_blk - 1) / rows_per_blk);
     k_mlp3_bias_bwd<<<ga, 256, 0, stream>>>(
         dout, (const mbf16*)dz1, (const mbf16*)dz2, (const mbf16*)dz3,
-        M, H, rows_per_blk, scratch);
-    int gb = (int)((3 * H + 1 + 255) / 256);
+        (const mbf16*)a3, M, H, rows_per_blk, scratch);
+    int gb = (int)((4 * H + 1 + 255) / 256);
     k_mlp3_bias_finish<<<gb, 256, 0, stream>>>(
-        scratch, H, (mbf16*)db1, (mbf16*)db2, (mbf16*)db3, (mbf16*)db4);
+        scratch, H, (mbf16*)db1, (mbf16*)db2, (mbf16*)db3, (mbf16*)dw4,
+        (mbf16*)db4);
 }
 
 extern "C" void emb_mlp3_fwd(const void* x0, long M, long K0p,
@@ -353,6 +364,7 @@ extern "C" void emb_mlp3_fwd(const void* x0, long M, long K0p,
                              const void* w2, const void* b2,
                              const void* w3, const void* b3,
                              const void* w4, const void* b4,
+                             const float* partial,
                              long H, long Hp,
                              void* a1, void* a2, void* a3, float* out,
                              hipStream_t stream) {
@@ -361,7 +373,7 @@ extern "C" void emb_mlp3_fwd(const void* x0, long M, long K0p,
     k_mlp3_fwd<<<(int)grid, 64 * MLP_WAVES, 0, stream>>>(
         (const mbf16*)x0, M, K0p, (const mbf16*)w1, (const mbf16*)b1,
         (const mbf16*)w2, (const mbf16*)b2, (const mbf16*)w3,
-        (const mbf16*)b3, (const mbf16*)w4, (const mbf16*)b4, H, Hp,
+        (const mbf16*)b3, (const mbf16*)w4, (const mbf16*)b4, partial, H, Hp,
         (mbf16*)a1, (mbf16*)a2, (mbf16*)a3, out);
 }
 
