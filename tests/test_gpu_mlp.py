@@ -42,7 +42,8 @@ def test_mlp3_fwd_matches_torch(M, K0):
     w3p = torch.zeros(H, Hp, device=DEV, dtype=torch.bfloat16)
     w3p[:, :H] = ws[2]
     out, a1, a2, a3 = ext.mlp3_fwd(x0, ws[0], bs[0], w2p, bs[1],
-                                   w3p, bs[2], ws[3].reshape(-1), bs[3])
+                                   w3p, bs[2], ws[3].reshape(-1), bs[3],
+                                   None)
     # STAGE-WISE reference: each layer recomputed from the KERNEL's own
     # previous activation, so bf16 rounding differences don't compound
     r1 = torch.relu(x0 @ ws[0].t() + bs[0])
@@ -75,7 +76,7 @@ def test_fused_mlp_function_grads():
     z = lambda *s: torch.zeros(*s, device=DEV, dtype=torch.bfloat16)  # noqa: E731
     bufs = {"w1p": z(H, K0p), "w2p": z(H, Hp), "w3p": z(H, Hp),
             "w3tp": z(H, Hp), "w2tp": z(H, Hp), "w1tp": z(K0p, Hp)}
-    out = _FusedMLP3Fn.apply(x0, *params, bufs)
+    out = _FusedMLP3Fn.apply(x0, None, *params, bufs)
     g = torch.randn(M, device=DEV)
     (out * g).sum().backward()
     got = [t.grad.clone() for t in [x0] + params]
@@ -120,3 +121,33 @@ def test_deepfm_native_fused_mlp_trains():
         losses.append(loss.item())
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]
+
+
+def test_fused_mlp_partial_fold():
+    # partial carried into the final-dot epilogue == explicit add outside
+    from openembedding_amd.models.ctr import _FusedMLP3Fn
+    torch.manual_seed(3)
+    M, K0, H = 512, 256, 400
+    K0p = (K0 + 31) // 32 * 32
+    Hp = (H + 31) // 32 * 32
+    x0 = torch.zeros(M, K0p, device=DEV, dtype=torch.bfloat16)
+    x0[:, :K0] = (torch.randn(M, K0, device=DEV) * 0.1).to(torch.bfloat16)
+    params = []
+    for shape in [(H, K0), (H,), (H, H), (H,), (H, H), (H,), (1, H), (1,)]:
+        t = (torch.randn(*shape, device=DEV)
+             * (0.05 if len(shape) == 2 else 0.1)).to(torch.bfloat16)
+        t.requires_grad_(True)
+        params.append(t)
+    z = lambda *s: torch.zeros(*s, device=DEV, dtype=torch.bfloat16)  # noqa: E731
+    bufs = {"w1p": z(H, K0p), "w2p": z(H, Hp), "w3p": z(H, Hp),
+            "w3tp": z(H, Hp), "w2tp": z(H, Hp), "w1tp": z(K0p, Hp)}
+    partial = torch.randn(M, device=DEV, requires_grad=True)
+
+    out_folded = _FusedMLP3Fn.apply(x0, partial, *params, bufs)
+    base = _FusedMLP3Fn.apply(x0, None, *params, bufs)
+    out_added = partial + base
+    assert torch.allclose(out_folded, out_added, atol=1e-5, rtol=1e-5)
+
+    g = torch.randn(M, device=DEV)
+    (out_folded * g).sum().backward()
+    assert torch.allclose(partial.grad, g, atol=1e-6)
