@@ -89,10 +89,12 @@ def test_graph_replay_matches_eager():
     got_dense = torch.cat([p.detach().reshape(-1).float()
                            for p in model2.dnn.parameters()])
     # float atomics (head backward dw/db, reduce-by-key) make two separate
-    # runs differ at rounding level; a replay bug (e.g. a non-replayed
-    # scratch reset) shows up orders of magnitude above this
-    assert torch.allclose(got_rows, ref_rows, atol=1e-4, rtol=1e-4)
-    assert torch.allclose(got_dense, ref_dense, atol=1e-4, rtol=1e-4)
+    # runs differ at rounding level, and 12 Adagrad steps amplify a 1-ulp
+    # grad flip into ~1e-4-scale weight deltas (observed flaking at 1e-4);
+    # a replay bug (e.g. a non-replayed scratch reset) shows up orders of
+    # magnitude above 1e-3
+    assert torch.allclose(got_rows, ref_rows, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(got_dense, ref_dense, atol=1e-3, rtol=1e-3)
 
 
 def test_graph_replay_hash_mode_with_reservation():
