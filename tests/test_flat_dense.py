@@ -63,3 +63,27 @@ def test_flat_rejects_unsupported():
         embed.distributed_optimizer(
             torch.optim.Adagrad(model.parameters(), lr=0.1, weight_decay=0.1),
             flatten_dense=True)
+
+
+def test_bce_with_logits_cpu_fallback_matches_torch():
+    # the fused-loss dispatch must be a drop-in for BCEWithLogitsLoss off-GPU
+    import torch
+    from openembedding_amd.ops.dispatch import bce_with_logits
+    g = torch.Generator().manual_seed(11)
+    z = (torch.randn(513, generator=g) * 3).requires_grad_(True)
+    y = (torch.rand(513, generator=g) < 0.25).float()
+    z2 = z.detach().clone().requires_grad_(True)
+    loss = bce_with_logits(z, y)
+    ref = torch.nn.functional.binary_cross_entropy_with_logits(z2, y)
+    assert torch.allclose(loss, ref)
+    loss.backward(); ref.backward()
+    assert torch.allclose(z.grad, z2.grad)
+
+
+def test_bce_with_logits_int_labels():
+    import torch
+    from openembedding_amd.ops.dispatch import bce_with_logits
+    z = torch.zeros(4, requires_grad=True)
+    y = torch.tensor([0, 1, 0, 1])            # non-float labels accepted
+    loss = bce_with_logits(z, y)
+    assert torch.isfinite(loss)
