@@ -116,3 +116,31 @@ def test_context_parses_flags_config(tmp_path):
         ctx.finalize()
     finally:
         oe.flags.config = old
+
+
+def test_inject_patches_nn_embedding():
+    # the reference's laboratory/inject demo: unmodified model code builds
+    # PS-backed embeddings after install(); uninstall restores torch
+    import torch
+    import torch.nn as nn
+    import openembedding_amd.inject as inject
+    import openembedding_amd.torch as embed
+
+    inject.install(sparse_as_dense_size=64)
+    try:
+        big = nn.Embedding(1000, 8)
+        small = nn.Embedding(10, 8)
+        hashed = nn.Embedding(-1, 8)
+        padded = nn.Embedding(1000, 8, padding_idx=0)  # no PS equivalent
+        assert isinstance(big, embed.Embedding)
+        assert isinstance(hashed, embed.Embedding)
+        assert type(small).__name__ == "Embedding" and not isinstance(
+            small, embed.Embedding)
+        assert isinstance(padded, inject._original)
+        out = big(torch.tensor([[1, 2, 999]]))
+        assert out.shape == (1, 3, 8)
+    finally:
+        inject.uninstall()
+    plain = nn.Embedding(1000, 8)
+    assert isinstance(plain, inject._original)
+    assert nn.Embedding is inject._original
