@@ -101,3 +101,43 @@ def test_optimizer_class_surface():
     loss.backward()
     o2.step()
     assert m2.embedding.variable.sharded.shard.optimizer.category == "ftrl"
+
+
+@pytest.mark.parametrize("batch", [1, 3, 10, 50])
+def test_tiny_and_odd_batches(batch):
+    # the reference's one-batch edge cases (build.sh test: batch 100/50/10);
+    # odd shapes must survive the whole dedup/all_to_all/gather/reduce stack
+    torch.manual_seed(1)
+    model = MODELS["deepfm"](field_vocabs=FIELD_VOCABS, dim=4)
+    opt = embed.distributed_optimizer(
+        torch.optim.Adagrad([p for p in model.parameters() if p.numel()],
+                            lr=0.05))
+    for _ in range(3):
+        dense, sparse, labels = synthetic_batch(batch)
+        sparse = sparse % torch.tensor(FIELD_VOCABS)
+        opt.zero_grad()
+        loss = torch.nn.functional.binary_cross_entropy_with_logits(
+            model(dense, sparse), labels)
+        loss.backward()
+        opt.step()
+    assert torch.isfinite(loss)
+
+
+def test_repeated_key_batch():
+    # an entire batch hitting ONE key per field: counts path + once-per-key
+    # optimizer application under maximal duplication
+    torch.manual_seed(2)
+    model = MODELS["deepfm"](field_vocabs=FIELD_VOCABS, dim=4)
+    opt = embed.distributed_optimizer(
+        torch.optim.Adagrad([p for p in model.parameters() if p.numel()],
+                            lr=0.05))
+    dense = torch.rand(64, 13)
+    sparse = torch.zeros(64, 26, dtype=torch.int64)    # all-duplicate keys
+    labels = torch.ones(64)
+    for _ in range(2):
+        opt.zero_grad()
+        loss = torch.nn.functional.binary_cross_entropy_with_logits(
+            model(dense, sparse), labels)
+        loss.backward()
+        opt.step()
+    assert torch.isfinite(loss)
