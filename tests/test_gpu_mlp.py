@@ -151,3 +151,41 @@ def test_fused_mlp_partial_fold():
     g = torch.randn(M, device=DEV)
     (out_folded * g).sum().backward()
     assert torch.allclose(partial.grad, g, atol=1e-6)
+
+
+def test_prebound_bias_kernel_matches_gemv_fallback():
+    # the flat-optimizer path (pre-bound bf16 .grad views) computes bias
+    # grads + head wgrad in k_mlp3_bias_bwd; it must match the GEMV
+    # fallback (non-prebound branch) to bf16 rounding
+    from openembedding_amd.models.ctr import _FusedMLP3Fn
+    torch.manual_seed(9)
+    M, K0, H = 1024, 247, 400
+    K0p = (K0 + 31) // 32 * 32
+    Hp = (H + 31) // 32 * 32
+    x0 = torch.zeros(M, K0p, device=DEV, dtype=torch.bfloat16)
+    x0[:, :K0] = (torch.randn(M, K0, device=DEV) * 0.1).to(torch.bfloat16)
+    shapes = [(H, K0), (H,), (H, H), (H,), (H, H), (H,), (1, H), (1,)]
+    base = [(torch.randn(*s, device=DEV)
+             * (0.05 if len(s) == 2 else 0.1)).to(torch.bfloat16)
+            for s in shapes]
+    g = torch.randn(M, device=DEV)
+
+    def run(prebind):
+        params = [t.clone().requires_grad_(True) for t in base]
+        if prebind:
+            for p in params:
+                p.grad = torch.zeros_like(p)
+        z = lambda *s: torch.zeros(*s, device=DEV, dtype=torch.bfloat16)  # noqa: E731
+        bufs = {"w1p": z(H, K0p), "w2p": z(H, Hp), "w3p": z(H, Hp),
+                "w3tp": z(H, Hp), "w2tp": z(H, Hp), "w1tp": z(K0p, Hp)}
+        out = _FusedMLP3Fn.apply(x0.clone().requires_grad_(True), None,
+                                 *params, bufs)
+        (out * g).sum().backward()
+        return [p.grad.float() for p in params]
+
+    fused = run(True)
+    gemv = run(False)
+    names = ["w1", "b1", "w2", "b2", "w3", "b3", "w4", "b4"]
+    for n, a, b in zip(names, fused, gemv):
+        assert torch.allclose(a, b, atol=5e-3, rtol=3e-2), (
+            n, (a - b).abs().max().item())
