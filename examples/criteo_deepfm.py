@@ -32,6 +32,9 @@ def main():
     p.add_argument("--lr", type=float, default=0.005)
     p.add_argument("--checkpoint", default="",
                    help="save server+dense model here at the end")
+    p.add_argument("--data", default="",
+                   help="train from a Criteo-format TSV instead of "
+                        "synthetic batches (openembedding_amd.data)")
     args = p.parse_args()
 
     ctx = embed.get_context()
@@ -43,21 +46,45 @@ def main():
     lossf = torch.nn.BCEWithLogitsLoss()
 
     gen = torch.Generator().manual_seed(1 + ctx.rank)
+
+    def batches():
+        if args.data:
+            from openembedding_amd.data import BackgroundLoader, CriteoTSV
+            # every rank reads the file; rank r trains rows r, r+W, ...
+            # (batch-level round-robin — the reference sharded csv the
+            # same way in its horovod examples)
+            for i, b in enumerate(BackgroundLoader(
+                    CriteoTSV(args.data, args.batch), depth=4)):
+                if i % ctx.world_size == ctx.rank:
+                    yield b
+        else:
+            while True:
+                yield synthetic_batch(args.batch, generator=gen)
+
     t0 = time.perf_counter()
+    src = batches()
     for step in range(args.steps):
-        dense, sparse, labels = synthetic_batch(args.batch, generator=gen)
+        try:
+            dense, sparse, labels = next(src)
+        except StopIteration:
+            break
         dense, sparse, labels = (dense.to(ctx.device), sparse.to(ctx.device),
                                  labels.to(ctx.device))
         opt.zero_grad()
         loss = lossf(model(dense, sparse), labels)
         loss.backward()
         opt.step()
+        last = step
         if ctx.rank == 0 and (step + 1) % 20 == 0:
             dt = time.perf_counter() - t0
             sps = args.batch * ctx.world_size * (step + 1) / dt
             print(f"step {step + 1}: loss={loss.item():.4f} "
                   f"{sps:,.0f} samples/s")
 
+    if ctx.rank == 0:
+        dt = time.perf_counter() - t0
+        print(f"done: {last + 1} steps, final loss={loss.item():.4f}, "
+              f"{(last + 1) * args.batch * ctx.world_size / dt:,.0f} samples/s")
     if args.checkpoint:
         wrapped = embed.Model(model)
         wrapped.save_weights(args.checkpoint)
