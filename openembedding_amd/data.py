@@ -62,6 +62,7 @@ class CriteoTSV:
     def __init__(self, path: str, batch_size: int,
                  field_vocabs: Optional[List[int]] = None,
                  hash_mode: bool = False,
+                 vocab_maps: Optional[List[dict]] = None,
                  drop_last: bool = False):
         self.path = path
         self.batch_size = int(batch_size)
@@ -69,6 +70,7 @@ class CriteoTSV:
         if len(self.field_vocabs) != N_SPARSE:
             raise ValueError(f"need {N_SPARSE} field vocabs")
         self.hash_mode = hash_mode
+        self.vocab_maps = vocab_maps  # from build_vocabs: token->id, 0 = OOV
         self.drop_last = drop_last
 
     def _emit(self, rows) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
@@ -90,17 +92,52 @@ class CriteoTSV:
                          if v else 0.0 for v in parts[1:1 + N_DENSE]]
                 cats = []
                 for fidx, tok in enumerate(parts[1 + N_DENSE:]):
-                    h = _hash_token(fidx, tok)
-                    if self.hash_mode:
-                        cats.append(h)
+                    if self.vocab_maps is not None:
+                        cats.append(self.vocab_maps[fidx].get(tok, 0))
+                    elif self.hash_mode:
+                        cats.append(_hash_token(fidx, tok))
                     else:
-                        cats.append(h % self.field_vocabs[fidx])
+                        cats.append(_hash_token(fidx, tok)
+                                    % self.field_vocabs[fidx])
                 rows.append((label, dense, cats))
                 if len(rows) == self.batch_size:
                     yield self._emit(rows)
                     rows = []
         if rows and not self.drop_last:
             yield self._emit(rows)
+
+
+def build_vocabs(path: str, min_count: int = 2
+                 ) -> Tuple[List[int], List[dict]]:
+    """One scan over a Criteo TSV: per-field token→id maps with a
+    frequency threshold (tokens seen < min_count collapse into the
+    field's OOV id 0 — the reference's criteo_preprocess.py recipe).
+
+    Returns (field_vocabs, field_maps): `field_vocabs[f]` is the table
+    size to give `CombinedEmbedding`, `field_maps[f][token] -> id` for
+    encoding; feed both back via ``CriteoTSV(..., vocab_maps=field_maps)``.
+    """
+    from collections import Counter
+    counters = [Counter() for _ in range(N_SPARSE)]
+    with open(path) as f:
+        for line in f:
+            parts = line.rstrip("\n").split("\t")
+            if len(parts) != 1 + N_DENSE + N_SPARSE:
+                continue
+            for fidx, tok in enumerate(parts[1 + N_DENSE:]):
+                counters[fidx][tok] += 1
+    maps: List[dict] = []
+    sizes: List[int] = []
+    for c in counters:
+        m = {}
+        nxt = 1                      # 0 reserved for OOV / rare
+        for tok, cnt in sorted(c.items()):
+            if cnt >= min_count:
+                m[tok] = nxt
+                nxt += 1
+        maps.append(m)
+        sizes.append(max(nxt, 2))
+    return sizes, maps
 
 
 class BackgroundLoader:

@@ -86,3 +86,48 @@ def test_end_to_end_train_from_file(tmp_path):
         loss.backward()
         opt.step()
     assert torch.isfinite(loss)
+
+
+def test_build_vocabs_frequency_threshold(tmp_path):
+    from openembedding_amd.data import build_vocabs
+    p = tmp_path / "v.tsv"
+    rows = []
+    for i in range(8):
+        dense = ["1"] * N_DENSE
+        # field 0: token "aa" appears 8x, "b<i>" once each
+        cats = ["aa" if i < 8 else "x"] + [f"t{i}c{c}" for c in range(1, N_SPARSE)]
+        rows.append("\t".join(["1"] + dense + cats))
+    p.write_text("\n".join(rows) + "\n")
+    sizes, maps = build_vocabs(str(p), min_count=2)
+    assert maps[0] == {"aa": 1} and sizes[0] == 2        # rare tokens -> OOV 0
+    assert maps[1] == {} and sizes[1] == 2               # all rare
+    batch = next(iter(CriteoTSV(str(p), 8, field_vocabs=sizes,
+                                vocab_maps=maps)))
+    sparse = batch[1]
+    assert set(sparse[:, 0].tolist()) == {1}             # "aa" -> id 1
+    assert set(sparse[:, 1].tolist()) == {0}             # rare -> OOV
+    assert int(sparse.max()) < max(sizes)
+
+
+def test_pulling_over_file_loader(tmp_path):
+    # the reference's full pipeline: dataset thread -> prefetched pull ->
+    # train (pulling() + BackgroundLoader + CriteoTSV)
+    import openembedding_amd.torch as embed
+    from openembedding_amd.models import MODELS
+    p = tmp_path / "train.tsv"
+    _write_sample(p, rows=24)
+    fv = [50] * N_SPARSE
+    model = MODELS["wdl"](field_vocabs=fv, dim=4)
+    opt = embed.distributed_optimizer(
+        torch.optim.Adagrad([q for q in model.parameters() if q.numel()],
+                            lr=0.05))
+    n = 0
+    for dense, sparse, labels in embed.pulling(
+            BackgroundLoader(CriteoTSV(str(p), 8, field_vocabs=fv)), model):
+        opt.zero_grad()
+        loss = torch.nn.functional.binary_cross_entropy_with_logits(
+            model(dense, sparse), labels)
+        loss.backward()
+        opt.step()
+        n += 1
+    assert n == 3 and torch.isfinite(loss)
