@@ -93,12 +93,9 @@ def _offsets(splits: Sequence[int]) -> List[int]:
 
 
 def all_reduce_sum(t: torch.Tensor) -> torch.Tensor:
+    """Shim kept for API symmetry with the gloo-aware helpers above
+    (dense-grad reduction itself lives in torch/__init__.py where it
+    overlaps the sparse commit)."""
     if dist_ready():
         dist.all_reduce(t, op=dist.ReduceOp.SUM)
-    return t
-
-
-def broadcast(t: torch.Tensor, src: int = 0) -> torch.Tensor:
-    if dist_ready():
-        dist.broadcast(t, src=src)
     return t
