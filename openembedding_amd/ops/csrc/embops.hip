@@ -285,10 +285,11 @@ __global__ void k_ht_lookup(u64* __restrict__ tk, int* __restrict__ tv,
             if (prev == EMPTY) {
                 int s = atomicAdd(nrows, 1);
                 if (s >= row_cap) {  // slab full (host sizing bug or a
-                    tk[h] = k;       // graph-captured insert): report miss
-                    slot = -1;       // instead of corrupting memory
-                    break;
-                }
+                    tv[h] = -1;      // graph-captured insert): this call AND
+                    slot = -1;       // every later lookup of this key must
+                    break;           // see a miss — a never-written tv[h]
+                }                    // here would later read as a garbage
+                                     // slot index (OOB row access)
                 tv[h] = s;
                 slot_keys[s] = (i64)k;
                 slot = s;
