@@ -1,0 +1,89 @@
+"""End-to-end user journey on CPU: train -> checkpoint -> resume -> more
+train -> export -> serve. One scenario chaining every major subsystem the
+way the reference's e2e example suite did (build.sh test: train, dump,
+load-at-epoch, serving round trip)."""
+
+import os
+
+import pytest
+import torch
+
+import openembedding_amd.torch as embed
+from openembedding_amd.models import DeepFM, synthetic_batch
+
+FV = [50, 3, 1000, 40, 7] + [100] * 21
+
+
+def _fresh_context():
+    import openembedding_amd.context as cm
+    import openembedding_amd.torch as api
+    if cm._context is not None:
+        cm._context.finalize()
+        cm._context = None
+    api._tracked.clear()
+
+
+def _train(model, opt, steps, seed):
+    gen = torch.Generator().manual_seed(seed)
+    lossf = torch.nn.BCEWithLogitsLoss()
+    for _ in range(steps):
+        dense, sparse, labels = synthetic_batch(128, generator=gen,
+                                                field_vocabs=FV)
+        opt.zero_grad()
+        loss = lossf(model(dense, sparse), labels)
+        loss.backward()
+        opt.step()
+    return loss
+
+
+def test_full_lifecycle(tmp_path):
+    _fresh_context()
+    torch.manual_seed(7)
+    model = DeepFM(field_vocabs=FV, dim=4)
+    opt = embed.distributed_optimizer(
+        torch.optim.Adagrad([p for p in model.parameters() if p.numel()],
+                            lr=0.05))
+    _train(model, opt, 5, seed=1)
+
+    # checkpoint both halves (torch weights + server model)
+    wrapped = embed.Model(model)
+    ckpt = str(tmp_path / "ckpt")
+    wrapped.save_weights(ckpt)
+    assert os.path.exists(ckpt) and os.path.isdir(ckpt + ".openembedding")
+
+    # keep training, then roll back to the checkpoint
+    probe = torch.arange(0, 100, 7) + model.embedding.field_offsets[2]
+    rows_at_ckpt = model.embedding.variable.sparse_read(probe).clone()
+    _train(model, opt, 3, seed=2)
+    assert not torch.allclose(
+        rows_at_ckpt, model.embedding.variable.sparse_read(probe))
+    wrapped.load_weights(ckpt)
+    torch.testing.assert_close(
+        rows_at_ckpt, model.embedding.variable.sparse_read(probe))
+
+    # resume: training continues from the restored state
+    _train(model, opt, 3, seed=2)
+
+    # PS-free export loads WITHOUT openembedding_amd machinery
+    export = str(tmp_path / "standalone.pt")
+    embed.save_as_original_model(model, export)
+    blob = torch.load(export, map_location="cpu", weights_only=True)
+    assert any("embedding" in k for k in blob["state_dict"])
+
+    # serve the dump and compare a served row with the live table
+    pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from openembedding_amd.serving import ModelController, make_app
+    ctx = embed.get_context()
+    uri = str(tmp_path / "dump")
+    embed.save_server_model(uri)
+    controller = ModelController()
+    controller.create_model(uri)
+    sign = f"{ctx.model_uuid}-{ctx.model_version}"
+    client = TestClient(make_app(controller))
+    live = model.embedding.variable.sparse_read(probe[:4])
+    r = client.post(f"/models/{sign}/variables/0/pull",
+                    json={"indices": probe[:4].tolist()})
+    assert r.status_code == 200
+    served = torch.tensor(r.json()["weights"])
+    torch.testing.assert_close(served, live)
