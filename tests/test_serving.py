@@ -2,6 +2,7 @@
 (reference client/ModelController.cpp, entry/controller.cc; read-only pull
 semantics EmbeddingPullOperator.cpp:179-181)."""
 
+import os
 import time
 
 import pytest
@@ -123,3 +124,42 @@ def test_metrics_endpoint(dumped_model):
     body = client.get("/metrics").text
     assert "openembedding_metric" in body
     assert "openembedding_models 1" in body
+
+
+@pytest.mark.timeout(120)
+def test_cli_server_boots(dumped_model, tmp_path):
+    # `python -m openembedding_amd.serving --model-uri U` must boot, serve
+    # the model list over real HTTP and die cleanly (reference controller
+    # daemon entry, entry/controller.cc main)
+    import socket
+    import subprocess
+    import sys
+    import time as _t
+    import urllib.request
+
+    uri, sign, _idx, _expect = dumped_model
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "openembedding_amd.serving",
+         "--host", "127.0.0.1", "--port", str(port),
+         "--device", "cpu", "--model-uri", uri],
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    try:
+        body = None
+        for _ in range(100):
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}/models", timeout=2) as r:
+                    body = r.read().decode()
+                break
+            except Exception:
+                assert proc.poll() is None, (
+                    "server died: " + proc.stdout.read().decode()[-2000:])
+                _t.sleep(0.2)
+        assert body is not None and sign in body
+    finally:
+        proc.terminate()
+        proc.wait(timeout=20)
