@@ -22,6 +22,9 @@ Public layout:
   openembedding_amd.ops         -- HIP kernels (gfx950) + CPU reference ops
   openembedding_amd.models      -- DeepFM / WDL / xDeepFM / LR model zoo
   openembedding_amd.checkpoint  -- dump/load in the reference's shard-file layout
+  openembedding_amd.serving     -- model registry + REST controller (ref entry/controller.cc)
+  openembedding_amd.data        -- Criteo TSV pipeline (ref csv datasets + criteo_preprocess.py)
+  openembedding_amd.inject      -- global nn.Embedding auto-patch (ref laboratory/inject)
 """
 
 __version__ = "0.1.0"
