@@ -32,6 +32,9 @@ def main():
     p.add_argument("--lr", type=float, default=0.005)
     p.add_argument("--checkpoint", default="",
                    help="save server+dense model here at the end")
+    p.add_argument("--eval-batches", type=int, default=0,
+                   help="after training, report AUC/logloss over this many "
+                        "held-out synthetic batches")
     p.add_argument("--data", default="",
                    help="train from a Criteo-format TSV instead of "
                         "synthetic batches (openembedding_amd.data)")
@@ -85,6 +88,20 @@ def main():
         dt = time.perf_counter() - t0
         print(f"done: {last + 1} steps, final loss={loss.item():.4f}, "
               f"{(last + 1) * args.batch * ctx.world_size / dt:,.0f} samples/s")
+    if args.eval_batches and ctx.rank == 0:
+        from openembedding_amd.models.metrics import StreamingAUC, \
+            StreamingLogLoss
+        auc, ll = StreamingAUC(), StreamingLogLoss()
+        egen = torch.Generator().manual_seed(991)
+        with torch.no_grad():
+            for _ in range(args.eval_batches):
+                dense, sparse, labels = synthetic_batch(args.batch,
+                                                        generator=egen)
+                out = model(dense.to(ctx.device), sparse.to(ctx.device))
+                auc.update(out, labels.to(ctx.device))
+                ll.update(out, labels.to(ctx.device))
+        print(f"eval: auc={auc.compute():.4f} logloss={ll.compute():.4f}")
+
     if args.checkpoint:
         wrapped = embed.Model(model)
         wrapped.save_weights(args.checkpoint)
