@@ -747,7 +747,7 @@ def should_persist_server_model() -> bool:
 def persist_server_model(path: str):
     """Lightweight checkpoint of the capacity tier (reference
     exb_persist_model): flush dirty cache rows to the host tier, dump, and
-    commit the watermark. For untired variables this equals
+    commit the watermark. For untiered variables this equals
     save_server_model."""
     ctx = get_context()
     for v in ctx.variables.values():
