@@ -1,0 +1,60 @@
+"""bench.py contract tests (CPU): the driver parses one JSON line from
+rank 0 with a fixed schema; variant flags must not break it."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+REQUIRED = {"metric", "value", "unit", "n_gpus", "steps", "warmup",
+            "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+            "dtype", "data", "config"}
+
+
+def _run_bench(*flags, timeout=420):
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "64", *flags],
+        cwd=REPO, capture_output=True, text=True, timeout=timeout)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")][-1]
+    return json.loads(line)
+
+
+def test_default_contract():
+    j = _run_bench()
+    assert REQUIRED <= set(j)
+    assert j["n_gpus"] == 1 and j["steps"] == 2 and j["warmup"] == 1
+    assert j["higher_is_better"] is True and j["scaling"] == "weak"
+    assert j["value"] > 0 and j["ms_per_step"] > 0
+    # value is the whole-job aggregate: batch * n_gpus * steps / elapsed
+    assert abs(j["value"] - 64 * 1 * 1000.0 / j["ms_per_step"]) / j["value"] < 0.01
+    assert j["config"]["model"] == "deepfm-dim9"
+    assert j["config"]["global_batch"] == 64
+    # vs_baseline only for the exact headline config (batch 4096)
+    assert j["vs_baseline"] is None
+
+
+def test_hash_mode_flag():
+    j = _run_bench("--hash")
+    assert j["config"]["table"] == "hash"
+
+
+def test_cache_tier_flag():
+    j = _run_bench("--cache-mb", "16")
+    assert j["config"]["table"] == "hash" and j["config"]["cache_mb"] == 16
+
+
+@pytest.mark.parametrize("model", ["wdl", "lr"])
+def test_model_variants(model):
+    j = _run_bench("--model", model)
+    assert j["config"]["model"].startswith(model)
+
+
+def test_xdeepfm_variant():
+    j = _run_bench("--model", "xdeepfm", "--dim", "4")
+    assert j["config"]["model"] == "xdeepfm-dim4"
