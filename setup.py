@@ -7,7 +7,7 @@ with the repo snapshot to GPU boxes; it is git-ignored)."""
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -34,7 +34,8 @@ ext = CUDAExtension(
 setup(
     name="openembedding_amd",
     version="0.1.0",
-    packages=["openembedding_amd"],
+    packages=find_packages(include=["openembedding_amd",
+                                    "openembedding_amd.*"]),
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension},
 )
