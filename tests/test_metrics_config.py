@@ -144,3 +144,18 @@ def test_inject_patches_nn_embedding():
     plain = nn.Embedding(1000, 8)
     assert isinstance(plain, inject._original)
     assert nn.Embedding is inject._original
+
+
+def test_top_level_parity_surface():
+    # the reference's module-level API shape (openembedding/__init__.py):
+    # flags, Master, Server, version — all present and minimally functional
+    import openembedding_amd as oe
+    assert oe.version == oe.__version__
+    assert hasattr(oe.flags, "config")
+    assert hasattr(oe.flags, "master_endpoint")
+    assert hasattr(oe.flags, "num_workers")
+    assert hasattr(oe.flags, "wait_num_servers")
+    m = oe.Master(bind_ip="127.0.0.1", port=7777)
+    assert m.running and m.endpoint == "127.0.0.1:7777"
+    s = oe.Server(master_endpoint=m.endpoint)
+    assert s.join() is None
