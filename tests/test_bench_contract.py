@@ -58,3 +58,17 @@ def test_model_variants(model):
 def test_xdeepfm_variant():
     j = _run_bench("--model", "xdeepfm", "--dim", "4")
     assert j["config"]["model"] == "xdeepfm-dim4"
+
+
+def test_benchmark_grid_script():
+    # the grid driver must run cells and write JSON lines (CPU, tiny cells)
+    out = os.path.join(REPO, "gpurun_out", "grid_test.jsonl")
+    r = subprocess.run(
+        [sys.executable, "scripts/benchmark_grid.py", "--models", "lr,wdl",
+         "--dims", "4", "--gpus", "1", "--steps", "2", "--warmup", "1",
+         "--batch", "64", "--out", out],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    rows = [json.loads(l) for l in open(out)]
+    assert len(rows) == 2
+    assert {row["config"]["model"] for row in rows} == {"lr-dim4", "wdl-dim4"}
