@@ -87,8 +87,10 @@ class ServedModel:
             sh = shards.get(hdr["variable_id"])
             if sh is None or not len(keys):
                 continue
-            kt = torch.from_numpy(np.ascontiguousarray(keys))
-            wt = torch.from_numpy(np.ascontiguousarray(w))
+            # frombuffer arrays are read-only; copy before wrapping (torch
+            # tensors over read-only memory are undefined behavior)
+            kt = torch.from_numpy(keys.copy())
+            wt = torch.from_numpy(w.copy())
             sh.import_rows(kt.to(sh.device), wt.to(sh.device))
         self.variables = {vid: ServedVariable(sh)
                           for vid, sh in shards.items()}
