@@ -62,6 +62,12 @@ def main():
     p.add_argument("--hash", action="store_true",
                    help="store embeddings in the open-addressed hash table "
                         "(lazy rows) instead of the array table")
+    p.add_argument("--prefetch", action="store_true",
+                   help="route batches through the pulling() prefetch "
+                        "pipeline (overlaps the next batch's embedding "
+                        "pull with this batch's compute; eager path only "
+                        "— intended for multi-rank runs; not yet the "
+                        "default: see docs/benchmark.md)")
     p.add_argument("--cache-mb", type=int, default=0,
                    help="capacity tier: device row-cache budget in MB, cold "
                         "rows spill to host DRAM (implies --hash)")
@@ -150,15 +156,29 @@ def main():
                   f"falling back to eager", flush=True)
             graph = None
 
-    def step(i):
-        dense, sparse, labels = pool[i % len(pool)]
-        if graph is not None:
-            static[0].copy_(dense)
-            static[1].copy_(sparse)
-            static[2].copy_(labels)
-            graph.replay()
-            return None
-        return run_step(dense, sparse, labels)
+    if args.prefetch and graph is None:
+        # dataset-side prefetch: batch t+1's embedding pull (unique +
+        # all_to_all + gather) runs on the side stream while batch t's
+        # dense compute runs — the reference's pipeline overlap
+        def batch_stream(total):
+            for i in range(total):
+                yield pool[i % len(pool)]
+
+        stream = embed.pulling(batch_stream(args.warmup + args.steps), model)
+
+        def step(i):
+            dense, sparse, labels = next(stream)
+            return run_step(dense, sparse, labels)
+    else:
+        def step(i):
+            dense, sparse, labels = pool[i % len(pool)]
+            if graph is not None:
+                static[0].copy_(dense)
+                static[1].copy_(sparse)
+                static[2].copy_(labels)
+                graph.replay()
+                return None
+            return run_step(dense, sparse, labels)
 
     for i in range(args.warmup):
         step(i)
@@ -218,6 +238,7 @@ def main():
                 "fp32 accumulators; embeddings, FM math, loss fp32"
                 if (args.amp == "native" and on_gpu) else "all fp32"),
             "graph": graph is not None,
+            "prefetch": bool(args.prefetch and graph is None),
             "table": "hash" if args.hash else "array",
             "cache_mb": args.cache_mb,
             "parallelism": (f"dense-dp{n_gpus} + embedding sharded "

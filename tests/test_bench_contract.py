@@ -72,3 +72,9 @@ def test_benchmark_grid_script():
     rows = [json.loads(l) for l in open(out)]
     assert len(rows) == 2
     assert {row["config"]["model"] for row in rows} == {"lr-dim4", "wdl-dim4"}
+
+
+def test_prefetch_flag():
+    j = _run_bench("--prefetch", "--graph", "off")
+    assert j["config"]["prefetch"] is True
+    assert j["value"] > 0
