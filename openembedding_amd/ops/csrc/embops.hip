@@ -271,6 +271,15 @@ __global__ void k_ht_lookup(u64* __restrict__ tk, int* __restrict__ tv,
     if (i >= n) return;
     if (u_dev && i >= *u_dev) return;
     u64 k = (u64)keys[i];
+    if (k == EMPTY) {
+        // key -1 is the table's empty marker and therefore RESERVED —
+        // exactly like the reference, which constructs every variable
+        // with empty_key = -1 (EmbeddingVariable.cpp:21,38,78). Define
+        // it as a miss instead of matching empty probe slots.
+        slots[i] = -1;
+        if (new_mask) new_mask[i] = 0;
+        return;
+    }
     u64 h = splitmix64(k) & (u64)mask;
     i64 slot = -1;
     unsigned char is_new = 0;
