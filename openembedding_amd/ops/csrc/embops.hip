@@ -141,7 +141,14 @@ __global__ void k_unique_insert(const i64* __restrict__ keys, long n,
     u64 k = 0;
     int lh = -1;           // LDS slot of this element's key
     bool lds_winner = false, lds_ok = false;
-    if (i < n) {
+    if (i < n && (u64)keys[i] == EMPTY) {
+        // RESERVED key -1 (the scratch table's empty marker, same value the
+        // reference reserves): give every occurrence the overflow-winner
+        // encoding — its own unique entry, resolved to slot -1 (zeros)
+        // downstream — instead of vacuously matching empty probe slots.
+        is_first[i] = 1;
+        slot_of[i] = -1;
+    } else if (i < n) {
         k = (u64)keys[i];
         u64 hh = splitmix64(k);
         lh = (int)(hh & (ULDS - 1));
@@ -343,6 +350,11 @@ __global__ void k_array_touch(unsigned char* __restrict__ valid,
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     if (u_dev && i >= *u_dev) { slots[i] = -1; return; }
+    if (keys[i] < 0) {  // invalid for a bounded vocabulary; C++ trunc
+        slots[i] = -1;  // division would silently map -1 to slot 0
+        new_mask[i] = 0;
+        return;
+    }
     i64 s = keys[i] / shard_num;
     if (s < 0 || s >= cap) s = -1;  // guarded; host validates separately
     slots[i] = s;
