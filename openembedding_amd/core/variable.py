@@ -157,6 +157,15 @@ class VariableShard:
             idx = self._index
             nxt = self._nrows
             for i, k in enumerate(keys.tolist()):
+                if k == -1:
+                    # reserved: the hash table's empty marker — the
+                    # reference reserves the same value (empty_key = -1,
+                    # EmbeddingVariable.cpp:21). The GPU backend resolves
+                    # it to a zero row; the CPU oracle fails loudly so the
+                    # mistake surfaces in development.
+                    raise ValueError(
+                        "key -1 is reserved in hash mode (table empty "
+                        "marker; same reservation as the reference)")
                 s = idx.get(k)
                 if s is None:
                     s = nxt

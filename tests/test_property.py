@@ -10,8 +10,11 @@ import numpy as np
 import torch
 from hypothesis import given, settings, strategies as st
 
+# key -1 is reserved in hash mode (table empty marker, same as the
+# reference's empty_key=-1) — excluded from fuzz ranges
 keys_strategy = st.lists(
-    st.integers(min_value=-2**62, max_value=2**62), min_size=1, max_size=300)
+    st.integers(min_value=-2**62, max_value=2**62).filter(lambda k: k != -1),
+    min_size=1, max_size=300)
 
 
 @settings(max_examples=60, deadline=None)
@@ -79,8 +82,10 @@ def test_checkpoint_reshard_partition(keys, world):
 
 
 @settings(max_examples=15, deadline=None)
-@given(keys=st.lists(st.integers(min_value=-2**62, max_value=2**62),
-                     min_size=1, max_size=80, unique=True),
+@given(keys=st.lists(
+           st.integers(min_value=-2**62, max_value=2**62
+                       ).filter(lambda k: k != -1),
+           min_size=1, max_size=80, unique=True),
        dim=st.integers(min_value=1, max_value=16),
        opt=st.sampled_from(["adagrad", "adam", "sgd"]))
 def test_checkpoint_roundtrip_fuzz(keys, dim, opt, tmp_path_factory):

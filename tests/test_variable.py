@@ -196,3 +196,20 @@ def test_single_key_many_duplicates():
     got = sh.pull_readonly(torch.tensor([0]))
     # default optimizer: w -= lr * summed_grad = -64
     assert torch.allclose(got, torch.full((1, 2), -64.0))
+
+
+def test_hash_mode_key_minus_one_reserved():
+    # the table's empty marker — reserved exactly like the reference's
+    # empty_key = -1; the CPU oracle fails loudly (the GPU kernel resolves
+    # it to a zero row instead, see ops/csrc/embops.hip k_ht_lookup)
+    import pytest as _pytest
+    from openembedding_amd.core import VariableMeta, VariableShard
+    meta = VariableMeta(variable_id=1, embedding_dim=4,
+                        vocabulary_size=1 << 63)
+    s = VariableShard(meta, 0, 1, device="cpu", seed=1)
+    s.set_initializer("uniform", minval=-1, maxval=1)
+    with _pytest.raises(ValueError, match="reserved"):
+        s.pull(torch.tensor([5, -1, 7]))
+    # read-only path: defined miss (zeros), no raise
+    out = s.pull_readonly(torch.tensor([-1]))
+    assert torch.equal(out, torch.zeros(1, 4))
