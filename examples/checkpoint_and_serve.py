@@ -14,6 +14,11 @@ import tempfile
 
 import torch
 
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(
+    _os.path.abspath(__file__))))  # run from a source checkout
+
 import openembedding_amd.torch as embed
 from openembedding_amd.models import DeepFM, synthetic_batch
 from openembedding_amd.serving import ModelController, make_app

@@ -14,6 +14,11 @@ replicated dense — the reference's "cache" policy).
 import torch
 import torch.nn as nn
 
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(
+    _os.path.abspath(__file__))))  # run from a source checkout
+
 import openembedding_amd.torch as embed
 
 
