@@ -93,11 +93,14 @@ def main():
         dt = time.perf_counter() - t0
         print(f"done: {last + 1} steps, final loss={loss.item():.4f}, "
               f"{(last + 1) * args.batch * ctx.world_size / dt:,.0f} samples/s")
-    if args.eval_batches and ctx.rank == 0:
+    if args.eval_batches:
+        # ALL ranks evaluate (model forward is a collective embedding pull
+        # at world>1 — a rank-0-only loop would deadlock the others), then
+        # sync() merges the per-rank accumulator state.
         from openembedding_amd.models.metrics import StreamingAUC, \
             StreamingLogLoss
         auc, ll = StreamingAUC(), StreamingLogLoss()
-        egen = torch.Generator().manual_seed(991)
+        egen = torch.Generator().manual_seed(991 + ctx.rank)
         with torch.no_grad():
             for _ in range(args.eval_batches):
                 dense, sparse, labels = synthetic_batch(args.batch,
@@ -105,7 +108,11 @@ def main():
                 out = model(dense.to(ctx.device), sparse.to(ctx.device))
                 auc.update(out, labels.to(ctx.device))
                 ll.update(out, labels.to(ctx.device))
-        print(f"eval: auc={auc.compute():.4f} logloss={ll.compute():.4f}")
+        auc.sync()
+        ll.sync()
+        if ctx.rank == 0:
+            print(f"eval: auc={auc.compute():.4f} "
+                  f"logloss={ll.compute():.4f}")
 
     if args.checkpoint:
         wrapped = embed.Model(model)

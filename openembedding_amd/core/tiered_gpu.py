@@ -194,7 +194,10 @@ class HipTieredVariableShard(HipVariableShard):
         out = super().pull(keys)
         slots = self._last_slots
         if slots is not None:
-            self._touch[slots.clamp(min=0)] = self.work_id
+            # stamp only resolved slots: a miss slot (-1) must not corrupt
+            # row 0's LRU order
+            valid = slots >= 0
+            self._touch[slots[valid]] = self.work_id
         return out
 
     def _lookup_or_insert(self, keys: torch.Tensor, u_dev=None):

@@ -13,6 +13,20 @@ import torch
 import torch.distributed as dist
 
 
+def _reduce_tensor(t: torch.Tensor) -> torch.Tensor:
+    """all_reduce(SUM) on a device the process-group backend accepts: the
+    nccl/RCCL backend rejects CPU tensors, so accumulator state kept on CPU
+    is staged through the current GPU for the reduction."""
+    if not (dist.is_available() and dist.is_initialized()):
+        return t
+    if dist.get_backend() == "nccl" and t.device.type != "cuda":
+        d = t.to("cuda")
+        dist.all_reduce(d, op=dist.ReduceOp.SUM)
+        return d.to(t.device)
+    dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    return t
+
+
 class StreamingAUC:
     """Histogram-bucketed ROC-AUC over logits (constant memory).
 
@@ -40,8 +54,8 @@ class StreamingAUC:
 
     def sync(self) -> "StreamingAUC":
         if dist.is_available() and dist.is_initialized():
-            for t in (self.pos, self.neg):
-                dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            self.pos = _reduce_tensor(self.pos)
+            self.neg = _reduce_tensor(self.neg)
         return self
 
     def compute(self) -> float:
@@ -80,7 +94,7 @@ class StreamingLogLoss:
     def sync(self) -> "StreamingLogLoss":
         if dist.is_available() and dist.is_initialized():
             t = torch.tensor([self.total, float(self.n)], dtype=torch.float64)
-            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            t = _reduce_tensor(t)
             self.total, self.n = float(t[0]), int(t[1])
         return self
 

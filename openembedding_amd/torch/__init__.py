@@ -93,15 +93,16 @@ class Variable:
         self.storage = storage
         self._prefetched: List = []  # FIFO of (match_id, out, handle, event)
 
-    def prefetch(self, indices: torch.Tensor, match_id=None) -> None:
+    def prefetch(self, indices: torch.Tensor, match_ref=None) -> None:
         """Issue the pull for a FUTURE batch now, on the prefetch stream
         (reference PrefetchPullWeights issued from the dataset thread,
-        exb_ops.cpp:109-205). ``match_id`` identifies the batch at consume
-        time (default: the identity of the indices tensor). Collective —
+        exb_ops.cpp:109-205). ``match_ref`` identifies the batch at consume
+        time (default: the indices tensor itself). Collective —
         every rank must prefetch the same variables in the same order."""
         ctx = get_context()
-        if match_id is None:
-            match_id = id(indices)
+        if match_ref is None:
+            match_ref = indices
+        match_id = id(match_ref)
         if ctx.device.type == "cuda":
             stream = ctx.prefetch_stream
             stream.wait_stream(torch.cuda.current_stream())
@@ -112,7 +113,10 @@ class Variable:
         else:
             out, handle = self.sharded.pull(indices)
             ev = None
-        self._prefetched.append((match_id, out, handle, ev))
+        # the queue holds a strong reference to the match object: without
+        # it CPython could reuse the id() of a dropped batch and a later
+        # unrelated tensor would consume the stale (out, handle) pair
+        self._prefetched.append((match_id, out, handle, ev, match_ref))
 
     def _take_prefetched(self, indices: torch.Tensor):
         """Pop the prefetched entry for these indices, or None. FIFO with
@@ -120,7 +124,8 @@ class Variable:
         Prefetch.h:34-44; object identity is exact here because pulling()
         passes the same tensors through)."""
         want = id(indices)
-        for pos, (match_id, out, handle, ev) in enumerate(self._prefetched):
+        for pos, (match_id, out, handle, ev, _ref) in enumerate(
+                self._prefetched):
             if match_id == want:
                 # drop skipped older entries along with the match
                 del self._prefetched[:pos + 1]
@@ -268,7 +273,7 @@ class CombinedEmbedding(Embedding):
         if self.sparse_as_dense:
             return
         keys = field_ids.to(torch.int64) + self.field_offsets
-        self.variable.prefetch(keys, match_id=id(field_ids))
+        self.variable.prefetch(keys, match_ref=field_ids)
 
 
 def _default_storage(ctx: Context) -> Storage:
@@ -377,16 +382,37 @@ class _FlatDenseAdagrad:
             }
             self.groups.append(grp)
         self.params = params
+        # p -> its view into the flat grad buffer (re-bound in zero_grad if
+        # model.zero_grad(set_to_none=True) detached it; checked in step)
+        self._grad_views = {p: p.grad for p in params}
         self._ext = None
         if dev.type == "cuda":
             from ..ops import require_hip
             self._ext = require_hip()
 
     def zero_grad(self):
+        for p, view in self._grad_views.items():
+            if p.grad is not view:
+                # model.zero_grad(set_to_none=True) (or a manual p.grad=None)
+                # detached the param from the flat buffer; re-bind — later
+                # backwards must accumulate into the flat buffer the
+                # allreduce and step read
+                p.grad = view
         for g in self.groups:
             g["flat_grad"].zero_()
 
+    def _check_bound(self):
+        for p, view in self._grad_views.items():
+            if p.grad is not view:
+                raise RuntimeError(
+                    "a dense param's .grad was detached from the flat "
+                    "buffer after backward (e.g. model.zero_grad("
+                    "set_to_none=True) mid-step); its gradient never "
+                    "reached the flat allreduce/step. Call "
+                    "optimizer.zero_grad() instead.")
+
     def step(self):
+        self._check_bound()
         for g in self.groups:
             if self._ext is not None:
                 self._ext.flat_adagrad(g["flat"], g["accum"], g["flat_grad"],
@@ -402,7 +428,11 @@ class _FlatDenseAdagrad:
                 g["flat"].addcdiv_(grad, std, value=-self.lr)
 
     def state_dict(self):
-        return {"groups": [{k: v for k, v in g.items() if k != "flat_grad"}
+        # detached clones: a caller that keeps training after save must not
+        # mutate the saved state through aliased tensors
+        return {"groups": [{k: (v.detach().clone()
+                                if torch.is_tensor(v) else v)
+                            for k, v in g.items() if k != "flat_grad"}
                            for g in self.groups], "lr": self.lr}
 
     def load_state_dict(self, sd):
@@ -480,11 +510,13 @@ class DistributedOptimizer:
 
     def _ensure_sparse_configured(self):
         cfg = self._sparse_config
+        translate_err = None
         if cfg is None:
             try:
                 cfg = _server_cfg_from_torch(self.optimizer)
-            except ValueError:
+            except ValueError as e:
                 cfg = None
+                translate_err = e
         for e in _tracked_live():
             if e.variable is None:
                 continue
@@ -496,7 +528,8 @@ class DistributedOptimizer:
                     raise RuntimeError(
                         "no sparse optimizer configured; pass sparse_config= "
                         "to distributed_optimizer or call set_optimizer on the "
-                        "Embedding variable")
+                        "Embedding variable"
+                        + (f" ({translate_err})" if translate_err else ""))
                 c = dict(cfg)
                 cat = c.pop("category")
                 e.variable.set_optimizer(cat, **c)
