@@ -54,6 +54,14 @@ void emb_mlp3_bias_bwd(const float*, const void*, const void*, const void*,
 void emb_bce_fwd(const float*, const float*, long, float*, hipStream_t_);
 void emb_bce_bwd(const float*, const float*, long, const float*, float*,
                  hipStream_t_);
+void emb_bucketize_pad(const i64*, long, const int*, long, long, i64*, int*,
+                       int*, int*, int*, hipStream_t_);
+void emb_gather_pad(const float*, const u64*, long, const int*, long, float*,
+                    hipStream_t_);
+void emb_scatter_out(const float*, const i64*, long, const int*, long, float*,
+                     hipStream_t_);
+void emb_split_payload(const float*, long, long, const int*, float*, u64*,
+                       hipStream_t_);
 void emb_flat_adagrad_f32(float*, float*, const float*, long, float, float,
                           hipStream_t_);
 void emb_flat_adagrad_bf16(float*, float*, const void*, void*, long, float,
@@ -220,6 +228,80 @@ std::tuple<torch::Tensor, torch::Tensor> reduce_by_inverse(
                           dim, ugrads.data_ptr<float>(),
                           (u64*)counts.data_ptr<i64>(), u, cur_stream());
     return {ugrads, counts};
+}
+
+// ---- padded all-to-all -------------------------------------------------
+
+void bucketize_pad(torch::Tensor uk_buf, OptTensor u_dev, int64_t world,
+                   int64_t cap, torch::Tensor send_keys,
+                   torch::Tensor send_src, torch::Tensor pos_of,
+                   torch::Tensor counts, torch::Tensor overflow) {
+    CHECK_GPU(uk_buf); CHECK_CONT(uk_buf); CHECK_CONT(send_keys);
+    CHECK_CONT(send_src); CHECK_CONT(pos_of);
+    TORCH_CHECK(send_keys.numel() == world * cap &&
+                send_src.numel() == world * cap &&
+                pos_of.numel() >= uk_buf.numel() &&
+                counts.numel() == world && overflow.numel() == 1,
+                "bucketize_pad buffer sizes");
+    TORCH_CHECK(send_src.dtype() == torch::kInt32 &&
+                pos_of.dtype() == torch::kInt32 &&
+                counts.dtype() == torch::kInt32 &&
+                overflow.dtype() == torch::kInt32, "int32 buffers expected");
+    const c10::cuda::CUDAGuard guard(uk_buf.device());
+    emb_bucketize_pad(uk_buf.data_ptr<i64>(), uk_buf.numel(), u_ptr(u_dev),
+                      world, cap, send_keys.data_ptr<i64>(),
+                      send_src.data_ptr<int>(), pos_of.data_ptr<int>(),
+                      counts.data_ptr<int>(), overflow.data_ptr<int>(),
+                      cur_stream());
+}
+
+torch::Tensor gather_pad(torch::Tensor ugrads, torch::Tensor counts,
+                         torch::Tensor send_src) {
+    CHECK_GPU(ugrads); CHECK_CONT(ugrads); CHECK_CONT(counts);
+    CHECK_CONT(send_src);
+    TORCH_CHECK(ugrads.dtype() == torch::kFloat32 &&
+                counts.dtype() == torch::kInt64 &&
+                send_src.dtype() == torch::kInt32, "gather_pad dtypes");
+    const c10::cuda::CUDAGuard guard(ugrads.device());
+    long total = send_src.numel();
+    long dim = ugrads.size(1);
+    auto send_p = torch::empty({total, dim + 1}, ugrads.options());
+    emb_gather_pad(ugrads.data_ptr<float>(), (const u64*)counts.data_ptr<i64>(),
+                   dim, send_src.data_ptr<int>(), total,
+                   send_p.data_ptr<float>(), cur_stream());
+    return send_p;
+}
+
+torch::Tensor scatter_out(torch::Tensor rows_recv, torch::Tensor inverse,
+                          torch::Tensor pos_of, int64_t n_elems) {
+    CHECK_GPU(rows_recv); CHECK_CONT(rows_recv); CHECK_CONT(inverse);
+    CHECK_CONT(pos_of);
+    TORCH_CHECK(rows_recv.dtype() == torch::kFloat32 &&
+                pos_of.dtype() == torch::kInt32, "scatter_out dtypes");
+    const c10::cuda::CUDAGuard guard(rows_recv.device());
+    long dim = rows_recv.size(1);
+    auto out = torch::empty({n_elems, dim}, rows_recv.options());
+    emb_scatter_out(rows_recv.data_ptr<float>(), inverse.data_ptr<i64>(),
+                    n_elems, pos_of.data_ptr<int>(), dim,
+                    out.data_ptr<float>(), cur_stream());
+    return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> split_payload(torch::Tensor g2c,
+                                                       OptTensor u_dev) {
+    CHECK_GPU(g2c); CHECK_CONT(g2c);
+    TORCH_CHECK(g2c.dtype() == torch::kFloat32, "split_payload dtype");
+    const c10::cuda::CUDAGuard guard(g2c.device());
+    long u = g2c.size(0);
+    long dim = g2c.size(1) - 1;
+    // tail rows beyond *u_dev stay uninitialized; every consumer
+    // (apply_optimizer) is u_dev-guarded and never reads them
+    auto grads = torch::empty({u, dim}, g2c.options());
+    auto counts = torch::empty({u}, g2c.options().dtype(torch::kInt64));
+    emb_split_payload(g2c.data_ptr<float>(), u, dim, u_ptr(u_dev),
+                      grads.data_ptr<float>(), (u64*)counts.data_ptr<i64>(),
+                      cur_stream());
+    return {grads, counts};
 }
 
 // ---- fused optimizers --------------------------------------------------
@@ -441,6 +523,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("reduce_by_inverse", &reduce_by_inverse,
           "grad reduce-by-key with counts (LDS-aggregated)");
     m.def("apply_optimizer", &apply_optimizer, "fused sparse optimizer step");
+    m.def("bucketize_pad", &bucketize_pad,
+          "owner-bucketize unique keys into a fixed padded [world, cap] "
+          "wire block (sync-free multi-rank route)");
+    m.def("gather_pad", &gather_pad,
+          "gather grads+counts payload into the padded send layout");
+    m.def("scatter_out", &scatter_out,
+          "padded pull: wire rows -> per-element output (fused dup scatter)");
+    m.def("split_payload", &split_payload,
+          "split owner-reduced payload into grads + int64 counts");
     m.def("ctr_head_fwd", &ctr_head_fwd,
           "fused CTR head fwd: deep_in assembly (+cast) + FM + first-order "
           "+ dense linear");

@@ -773,6 +773,89 @@ __global__ void k_bce_bwd(const float* __restrict__ z,
     g[i] = (s - y[i]) * inv_n * go[0];
 }
 
+// ------------------------------------------- padded all-to-all bucketize
+// The sync-free multi-rank wire format: instead of exact per-peer splits
+// (whose sizes need a device->host read every step), every rank ships a
+// FIXED [world, cap] key block to its peers, padded with the reserved key
+// -1. Padding flows through the whole owner pipeline as defined misses
+// (unique: own entry per occurrence; hash/array lookup: slot -1; gather:
+// zeros; optimizer: skipped), so no host knowledge is needed anywhere and
+// the full multi-rank step is hipGraph-capturable. `overflow` accumulates
+// keys dropped because a bucket exceeded cap; the host checks it outside
+// the hot loop and raises (capacity knob OEAMD_A2A_SLACK).
+// Reference analogue: EmbeddingPullOperator.cpp:67-79 shard bucketize (the
+// RPC fabric let the reference send exact sizes; xGMI all-to-all prefers
+// fixed shapes).
+
+__global__ void k_bucketize_pad(const i64* __restrict__ keys, long n,
+                                const int* __restrict__ u_dev, long world,
+                                long cap,
+                                i64* __restrict__ send_keys,
+                                int* __restrict__ send_src,
+                                int* __restrict__ pos_of,
+                                int* __restrict__ counts,
+                                int* __restrict__ overflow) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    pos_of[i] = -1;
+    if (u_dev && i >= *u_dev) return;   // beyond the live unique count
+    i64 k = keys[i];
+    if (k < 0) return;                  // reserved key: never shipped
+    long owner = (long)((u64)k % (u64)world);
+    int pos = atomicAdd(&counts[owner], 1);
+    if (pos >= cap) { atomicAdd(overflow, 1); return; }
+    long s = owner * cap + pos;
+    send_keys[s] = k;
+    send_src[s] = (int)i;
+    pos_of[i] = (int)s;
+}
+
+// push payload gather into the padded send layout: row = grads ‖ count
+// (count as f32: counts <= batch size, exact in fp32). Padded rows are
+// zero-filled; the owner drops them anyway (their keys are -1 -> slot -1).
+__global__ void k_gather_pad(const float* __restrict__ ugrads,
+                             const u64* __restrict__ counts, long dim,
+                             const int* __restrict__ send_src, long total,
+                             float* __restrict__ send_p) {
+    long dimp = dim + 1;
+    long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (e >= total * dimp) return;
+    long r = e / dimp, j = e % dimp;
+    int src = send_src[r];
+    float v = 0.0f;
+    if (src >= 0)
+        v = (j < dim) ? ugrads[(u64)src * dim + j] : (float)counts[src];
+    send_p[e] = v;
+}
+
+// final pull output: out[e] = returned row of e's unique (zeros for keys
+// that were never shipped: reserved/overflowed). Fuses the unique->element
+// duplicate scatter with the wire->unique permutation.
+__global__ void k_scatter_out(const float* __restrict__ rows_recv,
+                              const i64* __restrict__ inverse, long n_elems,
+                              const int* __restrict__ pos_of, long dim,
+                              float* __restrict__ out) {
+    long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (e >= n_elems * dim) return;
+    long r = e / dim, j = e % dim;
+    int pos = pos_of[inverse[r]];
+    out[e] = (pos >= 0) ? rows_recv[(u64)pos * dim + j] : 0.0f;
+}
+
+// split the owner-side reduced payload [u, dim+1] back into grads + counts
+__global__ void k_split_payload(const float* __restrict__ g2c, long u,
+                                long dim, const int* __restrict__ u_dev,
+                                float* __restrict__ grads,
+                                u64* __restrict__ counts) {
+    long dimp = dim + 1;
+    long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (e >= u * dimp) return;
+    long r = e / dimp, j = e % dimp;
+    if (u_dev && r >= *u_dev) return;
+    if (j < dim) grads[(u64)r * dim + j] = g2c[e];
+    else counts[r] = (u64)(g2c[e] + 0.5f);
+}
+
 // ============================================================== launchers
 
 extern "C" {
@@ -914,6 +997,43 @@ void emb_apply_optimizer(int opt, float* weights, float* state, long dim,
         }
         (void)G;
     }
+}
+
+void emb_bucketize_pad(const i64* keys, long n, const int* u_dev, long world,
+                       long cap, i64* send_keys, int* send_src, int* pos_of,
+                       int* counts, int* overflow, hipStream_t stream) {
+    long total = world * cap;
+    fill_u64((u64*)send_keys, total, EMPTY, stream);   // EMPTY == (i64)-1
+    fill_i32(send_src, total, -1, stream);
+    fill_i32(counts, world, 0, stream);
+    // overflow intentionally NOT cleared: it accumulates across steps and
+    // the host reads+clears it outside the hot loop
+    if (n) k_bucketize_pad<<<grid1d(n), BLOCK, 0, stream>>>(
+        keys, n, u_dev, world, cap, send_keys, send_src, pos_of, counts,
+        overflow);
+}
+
+void emb_gather_pad(const float* ugrads, const u64* counts, long dim,
+                    const int* send_src, long total, float* send_p,
+                    hipStream_t stream) {
+    if (total)
+        k_gather_pad<<<grid1d(total * (dim + 1)), BLOCK, 0, stream>>>(
+            ugrads, counts, dim, send_src, total, send_p);
+}
+
+void emb_scatter_out(const float* rows_recv, const i64* inverse, long n_elems,
+                     const int* pos_of, long dim, float* out,
+                     hipStream_t stream) {
+    if (n_elems)
+        k_scatter_out<<<grid1d(n_elems * dim), BLOCK, 0, stream>>>(
+            rows_recv, inverse, n_elems, pos_of, dim, out);
+}
+
+void emb_split_payload(const float* g2c, long u, long dim, const int* u_dev,
+                       float* grads, u64* counts, hipStream_t stream) {
+    if (u)
+        k_split_payload<<<grid1d(u * (dim + 1)), BLOCK, 0, stream>>>(
+            g2c, u, dim, u_dev, grads, counts);
 }
 
 void emb_flat_adagrad_f32(float* p, float* accum, const float* g, long n,

@@ -85,6 +85,21 @@ def all_to_all_v(inp: torch.Tensor, in_splits: Sequence[int],
     return out
 
 
+def all_to_all_equal(inp: torch.Tensor, world: int) -> torch.Tensor:
+    """Fixed-shape all-to-all: inp [world*cap, ...], chunk i of ``cap`` rows
+    goes to rank i; returns the same shape. No split sizes on the wire -> no
+    host sync, and the collective is hipGraph-capturable on nccl/RCCL."""
+    if not dist_ready():
+        return inp.clone()          # world-1 (forced-remote tests)
+    if backend() != "gloo":
+        out = torch.empty_like(inp)
+        dist.all_to_all_single(out, inp.contiguous())
+        return out
+    cap = inp.shape[0] // world
+    splits = [cap] * world
+    return all_to_all_v(inp, splits, splits)
+
+
 def _offsets(splits: Sequence[int]) -> List[int]:
     offs = [0]
     for s in splits:

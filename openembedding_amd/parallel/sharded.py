@@ -22,7 +22,8 @@ EmbeddingPullOperator.cpp:74-78, kept for checkpoint compatibility).
 from __future__ import annotations
 
 import dataclasses
-from typing import List, Optional
+import os
+from typing import Dict, List, Optional
 
 import torch
 
@@ -30,6 +31,40 @@ from ..core.variable import VariableShard
 from ..ops import dispatch as ops
 from ..utils.metrics import REGISTRY, stage_timer
 from . import comm
+
+
+class _PaddedPlan:
+    """Fixed-capacity wire plan of the sync-free padded all-to-all.
+
+    One per (variable, element count). Every rank ships a [world, cap] key
+    block padded with the reserved key -1 (defined miss everywhere:
+    ops/csrc/embops.hip k_bucketize_pad header comment), so the multi-rank
+    step needs ZERO device->host reads and is hipGraph-capturable. Buffers
+    are preallocated so a captured graph replays against stable addresses.
+
+    ``cap`` trades wire volume against skew headroom: per-peer unique
+    counts concentrate around u/world for modulo routing, so
+    cap = ceil(n * slack / world) (slack = OEAMD_A2A_SLACK, default 2)
+    bounds the overflow probability; overflowed keys are counted in
+    ``overflow`` and surface as a loud RuntimeError at the next host-side
+    check (ShardedVariable.check_padded_overflow), never as silent zeros.
+    """
+
+    def __init__(self, world: int, n: int, device: torch.device):
+        slack = float(os.environ.get("OEAMD_A2A_SLACK", "2.0"))
+        cap = int(-(-n * slack // world))          # ceil
+        cap = (cap + 255) // 256 * 256
+        self.cap = max(1, min(n, cap))
+        self.world = world
+        total = world * self.cap
+        self.send_keys = torch.full((total,), -1, dtype=torch.int64,
+                                    device=device)
+        self.send_src = torch.full((total,), -1, dtype=torch.int32,
+                                   device=device)
+        self.pos_of = torch.full((n,), -1, dtype=torch.int32, device=device)
+        self.counts = torch.zeros(world, dtype=torch.int32, device=device)
+        self.overflow = torch.zeros(1, dtype=torch.int32, device=device)
+        self.overflow_host = 0      # CPU-fallback bucketize accumulates here
 
 
 @dataclasses.dataclass
@@ -55,6 +90,10 @@ class PullHandle:
     # the matching push
     owner_u_dev: Optional[torch.Tensor] = None
     owner_slots: Optional[torch.Tensor] = None
+    # padded (fixed-capacity, zero-host-sync) multi-rank route:
+    padded: bool = False
+    plan: Optional["_PaddedPlan"] = None
+    owner_cpu: Optional[tuple] = None   # torch-fallback owner state
 
 
 class ShardedVariable:
@@ -69,6 +108,8 @@ class ShardedVariable:
         # EmbeddingPullOperator.cpp:208-247 — they diagnose all-to-all sizing)
         self.stat_pull_indices = 0
         self.stat_pull_unique = 0
+        self._plans: Dict[int, Optional[_PaddedPlan]] = {}
+        self._commits_since_check = 0
 
     # -------------------------------------------------------------- properties
 
@@ -115,9 +156,13 @@ class ShardedVariable:
         if (not remote and not readonly
                 and getattr(self.shard, "pull_bounded", None) is not None):
             return self._pull_local_bounded(indices, flat)
-        if (remote and not readonly
-                and getattr(self.shard, "pull_bounded", None) is not None):
-            return self._pull_remote_bounded(indices, flat)
+        if remote and not readonly:
+            if self._use_padded():
+                r = self._pull_remote_padded(indices, flat)
+                if r is not None:
+                    return r
+            if getattr(self.shard, "pull_bounded", None) is not None:
+                return self._pull_remote_bounded(indices, flat)
         unique, inverse = ops.unique_inverse(flat)
         self.stat_pull_unique += unique.numel()
         # reference pull_indices/pull_unique accumulators
@@ -191,6 +236,182 @@ class ShardedVariable:
         out = rows_u.index_select(0, inverse)
         return out.view(*indices.shape, self.shard.dim), h
 
+    # --------------------------------------------- padded (sync-free) route
+
+    # padded-mode override for tests: None = auto (env OEAMD_PADDED_A2A,
+    # else on iff the GPU bounded engine is present), True/False = forced
+    _padded = None
+
+    def _use_padded(self) -> bool:
+        if self._padded is not None:
+            return bool(self._padded)
+        env = os.environ.get("OEAMD_PADDED_A2A", "auto")
+        if env == "0":
+            return False
+        if env == "1":
+            return True
+        return getattr(self.shard, "pull_bounded", None) is not None
+
+    def _get_padded_plan(self, n: int,
+                         device: torch.device) -> Optional[_PaddedPlan]:
+        plan = self._plans.get(n, False)
+        if plan is not False:
+            return plan
+        import torch.distributed as dist
+        if comm.dist_ready() and dist.get_world_size() > 1:
+            # one-time agreement check: cap derives from n, so all ranks
+            # must see the same element count for this variable (the bench
+            # shape is rank-uniform; ragged shapes fall back to the exact
+            # route). One host sync, once per (variable, n).
+            dev = device if comm.backend() == "nccl" else "cpu"
+            t = torch.tensor([n, -n], dtype=torch.int64, device=dev)
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            if int(t[0]) != n or int(-t[1]) != n:
+                self._plans[n] = None
+                return None
+        plan = _PaddedPlan(self.world_size, n, device)
+        self._plans[n] = plan
+        return plan
+
+    def _pull_remote_padded(self, indices: torch.Tensor,
+                            flat: torch.Tensor):
+        """Multi-rank pull with ZERO host syncs: fixed [world, cap] key
+        blocks padded with the reserved key -1 ride two equal-split
+        all-to-alls; padding resolves to defined misses on the owner and is
+        dropped on return. hipGraph-capturable end to end."""
+        world = self.world_size
+        dim = self.shard.dim
+        dev = flat.device
+        n = flat.numel()
+        plan = self._get_padded_plan(n, dev)
+        if plan is None:
+            return None
+        gpu = getattr(self.shard, "pull_bounded", None) is not None
+        if gpu:
+            ext = self.shard.ext
+            uk_buf, inverse, u_dev = ext.unique_bounded(flat)
+            ext.bucketize_pad(uk_buf, u_dev, world, plan.cap,
+                              plan.send_keys, plan.send_src, plan.pos_of,
+                              plan.counts, plan.overflow)
+        else:
+            uk_buf, inverse = ops.unique_inverse(flat)
+            u_dev = None
+            self._bucketize_pad_torch(uk_buf, plan)
+        REGISTRY.add("pull_indices", n)
+        REGISTRY.add("pull_wire_keys", plan.send_keys.numel())
+        h = PullHandle(shape=indices.shape, unique=uk_buf, inverse=inverse,
+                       u_dev=u_dev, padded=True, plan=plan)
+        recv_keys = comm.all_to_all_equal(plan.send_keys, world)
+        if gpu:
+            uk2_buf, inv2, u2_dev = ext.unique_bounded(recv_keys)
+            rows_back, slots2 = self.shard.pull_bounded(uk2_buf, u2_dev,
+                                                        inv2)
+            h.owner_unique = uk2_buf
+            h.owner_inverse = inv2
+            h.owner_u_dev = u2_dev
+            h.owner_slots = slots2
+        else:
+            rows_back = self._owner_pull_padded_torch(recv_keys, h)
+        rows_recv = comm.all_to_all_equal(rows_back, world)
+        if gpu:
+            out = ext.scatter_out(rows_recv, inverse, plan.pos_of, n)
+        else:
+            pos = plan.pos_of.index_select(0, inverse).long()
+            out = rows_recv.index_select(0, pos.clamp(min=0))
+            out = out.masked_fill((pos < 0).unsqueeze(1), 0.0)
+        return out.view(*indices.shape, dim), h
+
+    def _bucketize_pad_torch(self, uk: torch.Tensor,
+                             plan: _PaddedPlan) -> None:
+        """Pure-torch bucketize into the padded wire layout (CPU engine /
+        gloo tests; same wire format as the HIP k_bucketize_pad)."""
+        world, cap = plan.world, plan.cap
+        plan.send_keys.fill_(-1)
+        plan.send_src.fill_(-1)
+        plan.pos_of.fill_(-1)
+        nz = (uk >= 0).nonzero(as_tuple=True)[0]
+        k = uk.index_select(0, nz)
+        own = k % world
+        order = torch.argsort(own, stable=True)
+        k_s, nz_s, own_s = k[order], nz[order], own[order]
+        cnt = torch.bincount(own_s, minlength=world)
+        start = torch.cumsum(cnt, 0) - cnt
+        pos = torch.arange(k_s.numel(), device=uk.device) - start[own_s]
+        ok = pos < cap
+        slot = (own_s * cap + pos)[ok]
+        plan.send_keys[slot] = k_s[ok]
+        plan.send_src[slot] = nz_s[ok].to(torch.int32)
+        plan.pos_of[nz_s[ok]] = slot.to(torch.int32)
+        plan.overflow_host += int((~ok).sum())
+
+    def _owner_pull_padded_torch(self, recv_keys: torch.Tensor,
+                                 h: PullHandle) -> torch.Tensor:
+        """Owner side of the padded route on the CPU engine: mask the -1
+        padding, exact-dedup the rest, gather (+lazy init), scatter back
+        into wire positions (padding returns zeros)."""
+        dim = self.shard.dim
+        valid = recv_keys >= 0
+        vk = recv_keys[valid]
+        uk2, inv2 = ops.unique_inverse(vk)
+        rows_back = torch.zeros(recv_keys.numel(), dim,
+                                dtype=self.shard.dtype,
+                                device=recv_keys.device)
+        if uk2.numel():
+            rows = self.shard.pull(uk2)
+            rows_back[valid] = rows.index_select(0, inv2)
+        h.owner_cpu = (valid, uk2, inv2)
+        return rows_back
+
+    def _push_padded(self, h: PullHandle, g: torch.Tensor) -> None:
+        dim = self.shard.dim
+        plan = h.plan
+        world = self.world_size
+        nbuf = h.unique.numel()
+        ugrads, counts = ops.reduce_by_inverse(h.inverse, g, nbuf)
+        if h.u_dev is not None:             # GPU engine
+            ext = self.shard.ext
+            send_p = ext.gather_pad(ugrads, counts, plan.send_src)
+            recv_p = comm.all_to_all_equal(send_p, world)
+            g2c, _ = ops.reduce_by_inverse(h.owner_inverse,
+                                           recv_p.contiguous(),
+                                           h.owner_unique.numel())
+            g2, c2 = ext.split_payload(g2c, h.owner_u_dev)
+            self.shard.push_slots(h.owner_unique, h.owner_u_dev,
+                                  h.owner_slots, g2, c2)
+            return
+        payload = torch.cat([ugrads,
+                             counts.to(ugrads.dtype).unsqueeze(1)], dim=1)
+        src = plan.send_src.long()
+        send_p = payload.index_select(0, src.clamp(min=0))
+        send_p = send_p.masked_fill((src < 0).unsqueeze(1), 0.0)
+        recv_p = comm.all_to_all_equal(send_p, world)
+        valid, uk2, inv2 = h.owner_cpu
+        if uk2.numel() == 0:
+            return
+        g2c = torch.zeros(uk2.numel(), dim + 1, dtype=recv_p.dtype,
+                          device=recv_p.device)
+        g2c.index_add_(0, inv2, recv_p[valid])
+        g2 = g2c[:, :dim].contiguous()
+        c2 = g2c[:, dim].round().to(torch.int64)
+        self.shard.push(uk2, g2, c2)
+
+    def check_padded_overflow(self) -> None:
+        """Raise if any padded wire block ever overflowed its capacity.
+        Called every 64 commits (outside graph capture) and explicitly by
+        bench.py after the timed loop; overflow means some keys silently
+        read zeros, so the run must fail loudly. Fix: raise OEAMD_A2A_SLACK
+        (or disable with OEAMD_PADDED_A2A=0)."""
+        for plan in self._plans.values():
+            if plan is None:
+                continue
+            ov = plan.overflow_host + int(plan.overflow.sum().item())
+            if ov:
+                raise RuntimeError(
+                    f"padded all-to-all overflow on variable "
+                    f"{self.variable_id}: {ov} keys were dropped (bucket "
+                    f"capacity {plan.cap}/peer). Raise OEAMD_A2A_SLACK "
+                    f"(default 2.0) or set OEAMD_PADDED_A2A=0.")
+
     def _pull_remote(self, h: PullHandle, readonly: bool) -> torch.Tensor:
         with stage_timer("pull", "remote"):
             return self._pull_remote_impl(h, readonly)
@@ -238,6 +459,9 @@ class ShardedVariable:
         """grads: [*shape, dim] gradient of the pulled weights."""
         dim = self.shard.dim
         g = grads.reshape(-1, dim)
+        if h.padded:
+            self._push_padded(h, g)
+            return
         u = h.unique.numel()
         ugrads, counts = ops.reduce_by_inverse(h.inverse, g, u)
         if h.bounded:
@@ -271,6 +495,14 @@ class ShardedVariable:
 
     def update_weights(self) -> None:
         self.shard.update_weights()
+        if self._plans:
+            self._commits_since_check += 1
+            if self._commits_since_check >= 64:
+                capturing = (torch.cuda.is_available()
+                             and torch.cuda.is_current_stream_capturing())
+                if not capturing:
+                    self._commits_since_check = 0
+                    self.check_padded_overflow()
 
     # ------------------------------------------------------- whole-table pulls
 
