@@ -1,31 +1,34 @@
-"""GPU capacity tier: HBM row-cache over a pinned host-DRAM backing store.
+"""GPU capacity tier v2: HBM row-cache over a pinned host-DRAM backing store.
 
-BASELINE.json config 5 ("host-DRAM cold-row offload tier, async hipMemcpy
-prefetch on side stream") and the reference's PMem tier re-based on the
-MI355X memory hierarchy (SURVEY §2.1 PmemEmbeddingTable / CacheItemPool):
+BASELINE.json config 5 ("host-DRAM cold-row offload tier, async prefetch on
+side stream") and the reference's PMem tier re-based on the MI355X memory
+hierarchy (SURVEY §2.1 PmemEmbeddingTable / CacheItemPool):
 
   HBM (288 GB)  = the cache tier: HipVariableShard's hash table + row slab,
                   bounded at ``cache_rows``; per-slot last-touch work_id
                   stamps (the reference's LRU work_id, PmemEmbeddingTable.h)
-  host DRAM     = the backing store: pinned tensors, rows move H2D/D2H with
-                  non_blocking copies (hipMemcpyAsync under the hood — on
-                  the prefetch side stream when driven through
-                  Variable.prefetch/pulling, which is what hides the
-                  fault-in latency off the critical path)
+  host DRAM     = the backing store: pinned slabs addressed by a SECOND
+                  device-side hash table (key -> host slot), so admission,
+                  fault-in and spill all run as HIP kernels reading/writing
+                  the pinned memory directly (zero-copy: only touched rows
+                  cross the host link) — no python dicts, no key lists.
 
-Key-set bookkeeping (which keys are spilled) is host-side — the fault-in
-decision needs host knowledge anyway because the batch's key list arrives
-from the host. The device never pays for it: the hot path (all keys cached)
-is exactly HipVariableShard's sync-free path plus one touch-stamp write.
+Hot path: exactly HipVariableShard's sync-free bounded path plus one
+fault-in kernel (no-op until something was spilled) and one touch-stamp
+scatter. Eviction runs at the COMMIT boundary (end of update_weights),
+after every pending gradient block was applied — so the table slots saved
+in live pull handles are never invalidated mid-step (the v1 design evicted
+at pull time and could corrupt prefetched handles). A slot-epoch guard
+turns any residual stale-handle use into a loud error.
 
-Eviction rebuilds the device hash table from the surviving keys (an
-open-addressed table has no cheap delete); evictions happen at batch
-granularity and move >= cache_rows/4 rows, so the rebuild amortizes.
+Driven through Variable.prefetch/pulling() the whole pull (fault-ins
+included) runs on the prefetch side stream, which is what hides the host
+link latency off the critical path.
 """
 
 from __future__ import annotations
 
-from typing import Dict
+from typing import Optional
 
 import torch
 
@@ -34,148 +37,214 @@ from .variable_gpu import HipVariableShard
 
 
 class HipTieredVariableShard(HipVariableShard):
-    def __init__(self, meta: VariableMeta, shard_id: int = 0, shard_num: int = 1,
-                 device: str = "cuda", seed: int = 0,
+    HOST_INITIAL_ROWS = 1 << 14
+    HOST_INITIAL_TABLE = 1 << 16
+
+    def __init__(self, meta: VariableMeta, shard_id: int = 0,
+                 shard_num: int = 1, device: str = "cuda", seed: int = 0,
                  cache_rows: int = 1 << 22):
         if not meta.use_hash_table:
             raise ValueError("the capacity tier requires hash mode")
         super().__init__(meta, shard_id, shard_num, device, seed)
         self.cache_rows = int(cache_rows)
         self.work_id = 0
-        self._touch = torch.zeros(self.weights.shape[0], dtype=torch.int64,
-                                  device=self.device)
-        self._host_weights = torch.zeros((0, self.dim), dtype=self.dtype)
-        self._host_state = torch.zeros((0, 0), dtype=self.dtype)
-        self._host_free: list = []
-        self._host_next = 0
-        self._host_of: Dict[int, int] = {}
+        dev = self.device
+        # +1: slot -1 maps to dummy index 0 so the stamp scatter never
+        # branches on the host (sync-free)
+        self._touch = torch.zeros(self.weights.shape[0] + 1,
+                                  dtype=torch.int64, device=dev)
+        # device-side host map: key -> host-slab slot
+        self._host_cap = self.HOST_INITIAL_TABLE
+        self.host_tk = torch.full((self._host_cap,), -1, dtype=torch.int64,
+                                  device=dev)
+        self.host_tv = torch.empty(self._host_cap, dtype=torch.int32,
+                                   device=dev)
+        self.host_nrows_dev = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.host_slot_keys = torch.empty(self.HOST_INITIAL_ROWS,
+                                          dtype=torch.int64, device=dev)
+        pin = torch.cuda.is_available()
+        self._host_weights = torch.zeros((self.HOST_INITIAL_ROWS, self.dim),
+                                         dtype=self.dtype, pin_memory=pin)
+        self._host_state = torch.zeros((self.HOST_INITIAL_ROWS, 0),
+                                       dtype=self.dtype, pin_memory=pin)
+        self._host_rows_upper = 0     # conservative >= live host rows
+        self._host_live = False       # False until the first spill
+        self._faulted = torch.zeros(1, dtype=torch.int32, device=dev)
+        self._evict_epoch = 0
+        self._last_batch_upper = 0
         self._checkpoint_work_id = None
         self._cache_full_since_ckpt = False
-        self._last_slots = None
 
-    # ------------------------------------------------------------ host slab
+    # ---------------------------------------------------------- host slabs
 
-    def _host_grow(self, need: int) -> None:
+    def _host_ensure(self, rows: int) -> None:
         cap = self._host_weights.shape[0]
-        if need <= cap:
+        if rows <= cap:
             return
-        new_cap = max(need, max(4096, cap * 2))
-        pin = self.device.type == "cuda" and torch.cuda.is_available()
-        hw = torch.zeros((new_cap, self.dim), dtype=self.dtype, pin_memory=pin)
+        new_cap = max(rows, cap * 2)
+        # fault-in/spill kernels hold raw pointers into the old slabs:
+        # drain the device before swapping them out
+        torch.cuda.synchronize()
+        pin = torch.cuda.is_available()
+        hw = torch.zeros((new_cap, self.dim), dtype=self.dtype,
+                         pin_memory=pin)
         hw[:cap] = self._host_weights
         self._host_weights = hw
         hs = torch.zeros((new_cap, self._host_state.shape[1]),
                          dtype=self.dtype, pin_memory=pin)
         hs[:cap] = self._host_state
         self._host_state = hs
+        sk = torch.empty(new_cap, dtype=torch.int64, device=self.device)
+        sk[:self.host_slot_keys.shape[0]] = self.host_slot_keys
+        self.host_slot_keys = sk
 
-    def _host_alloc(self, key: int) -> int:
-        hs = self._host_of.get(key)
-        if hs is None:
-            hs = self._host_free.pop() if self._host_free else self._host_next
-            if hs == self._host_next:
-                self._host_next += 1
-            self._host_grow(self._host_next)
-            self._host_of[key] = hs
-        return hs
+    def _host_maybe_rehash(self) -> None:
+        if self._host_rows_upper * 2 <= self._host_cap:
+            return
+        new_cap = self._host_cap
+        while self._host_rows_upper * 2 > new_cap:
+            new_cap *= 2
+        tk_new = torch.empty(new_cap, dtype=torch.int64, device=self.device)
+        tv_new = torch.empty(new_cap, dtype=torch.int32, device=self.device)
+        self.ext.ht_rehash(self.host_tk, self.host_tv, tk_new, tv_new)
+        self.host_tk, self.host_tv = tk_new, tv_new
+        self._host_cap = new_cap
 
     def set_optimizer(self, category: str, **cfg) -> None:
         super().set_optimizer(category, **cfg)
         if self._host_state.shape[1] != self.state_dim:
-            n = self._host_weights.shape[0]
-            pin = self.device.type == "cuda" and torch.cuda.is_available()
-            hs = torch.zeros((n, self.state_dim), dtype=self.dtype,
-                             pin_memory=pin)
-            if self._host_of and self.state_dim:
-                used = torch.tensor(sorted(self._host_of.values()),
-                                    dtype=torch.int64)
-                row = self._make_state_init_row().cpu()
-                hs[used] = row.expand(used.numel(), -1)
-            self._host_state = hs
+            if self._host_live:
+                raise RuntimeError("cannot change optimizer state layout "
+                                   "after rows were spilled to the host "
+                                   "tier")
+            pin = torch.cuda.is_available()
+            self._host_state = torch.zeros(
+                (self._host_weights.shape[0], self.state_dim),
+                dtype=self.dtype, pin_memory=pin)
 
     def _ensure_rows(self, need: int) -> None:
         super()._ensure_rows(need)
-        if self._touch.numel() < self.weights.shape[0]:
-            t = torch.zeros(self.weights.shape[0], dtype=torch.int64,
+        if self._touch.numel() < self.weights.shape[0] + 1:
+            t = torch.zeros(self.weights.shape[0] + 1, dtype=torch.int64,
                             device=self.device)
             t[:self._touch.numel()] = self._touch
             self._touch = t
 
-    # -------------------------------------------------- fault-in / eviction
+    def reserve_rows(self, rows: int) -> None:
+        # CombinedEmbedding pre-sizes for its whole key space; the tier's
+        # point is bounding HBM at cache_rows, so clamp the reservation
+        super().reserve_rows(min(int(rows), self.cache_rows))
 
-    def _tier_admit(self, keys: torch.Tensor) -> None:
-        """Host-side batch admission: fault spilled keys back into the HBM
-        cache and evict cold rows when over budget. Runs BEFORE the normal
-        HIP pull; one host sync (keys.tolist) — hidden by the prefetch
-        stream when driven through pulling()."""
-        kl = keys.tolist()
-        from_host = [k for k in dict.fromkeys(kl) if k in self._host_of]
-        self._sync_nrows()
-        need = self._nrows_exact + len(kl)  # upper bound of new cache rows
-        if need > self.cache_rows:
-            self._evict(set(kl))
-        if from_host:
-            hslots = torch.tensor([self._host_of[k] for k in from_host],
-                                  dtype=torch.int64)
-            kt = torch.tensor(from_host, dtype=torch.int64,
-                              device=self.device)
-            w = self._host_weights[hslots].to(self.device, non_blocking=True)
-            s = (self._host_state[hslots].to(self.device, non_blocking=True)
-                 if self.state_dim else None)
-            slots, _ = self._lookup_or_insert(kt)
-            self.weights[slots] = w
-            if s is not None:
-                self.state[slots] = s
-            for k in from_host:
-                self._host_free.append(self._host_of.pop(k))
+    # ------------------------------------------------------------ hot path
+
+    def _lookup_or_insert(self, keys: torch.Tensor, u_dev=None):
+        if self._in_graph_capture():
+            raise RuntimeError("the capacity tier is not hipGraph-"
+                               "capturable (evictions need host control); "
+                               "run eager with prefetch overlap")
+        slots, new_mask = super()._lookup_or_insert(keys, u_dev)
+        if self._host_live:
+            self.ext.fault_in(keys, u_dev, self.host_tk, self.host_tv,
+                              self._host_weights,
+                              self._host_state if self.state_dim else None,
+                              self.weights, self.state, slots, new_mask,
+                              self._faulted)
+        # LRU stamp: slot -1 lands on dummy index 0 (sync-free)
+        self._touch.scatter_(0, slots + 1, self.work_id)
+        slots._oe_epoch = self._evict_epoch
+        self._last_batch_upper = max(self._last_batch_upper, keys.numel())
+        return slots, new_mask
+
+    def push_slots(self, keys_buf, u_dev, slots, grads, counts) -> None:
+        if getattr(slots, "_oe_epoch", self._evict_epoch) != self._evict_epoch:
+            raise RuntimeError(
+                "stale pull handle: the cache was compacted (eviction) "
+                "after this pull; pull handles must be pushed before the "
+                "commit that follows them")
+        super().push_slots(keys_buf, u_dev, slots, grads, counts)
+
+    def pull_readonly(self, keys: torch.Tensor) -> torch.Tensor:
+        slots = self._lookup_readonly(keys)
+        empty = torch.empty(0, dtype=torch.uint8, device=self.device)
+        out = self.ext.gather_init(self.weights, self.state, slots, empty,
+                                   keys, None, 0, 0.0, 0.0, 0.0, self.seed,
+                                   torch.empty(0, dtype=self.dtype,
+                                               device=self.device), True,
+                                   None)
+        if self._host_live:
+            self.ext.gather_host(keys, slots, self.host_tk, self.host_tv,
+                                 self._host_weights, out)
+        return out
+
+    def update_weights(self) -> None:
+        super().update_weights()
+        self.work_id += 1
+        # evict at the commit boundary: all pending gradient blocks were
+        # just applied, so no live handle's slots are invalidated
+        if self._nrows_upper + self._last_batch_upper > self.cache_rows:
+            self._evict()
+
+    # ------------------------------------------------------------- eviction
 
     def _sync_nrows(self) -> None:
         self._nrows_exact = int(self.nrows_dev.item())
         self._nrows_upper = self._nrows_exact
 
-    def _evict(self, pinned_keys) -> None:
-        """Move the coldest rows (not in ``pinned_keys``) to host and rebuild
-        the device table from the survivors."""
-        self._cache_full_since_ckpt = True
+    def _spill(self, evict_slots: torch.Tensor) -> None:
+        """Copy cache rows (by slot) into the host tier (insert keys into
+        the device-side host map, then one spill kernel)."""
+        n_ev = evict_slots.numel()
+        if n_ev == 0:
+            return
+        keys = self.host_keys_of_slots(evict_slots)
+        self._host_ensure(self._host_rows_upper + n_ev)
+        self._host_maybe_rehash()
+        hslots, _ = self.ext.ht_lookup(self.host_tk, self.host_tv, keys,
+                                       self.host_nrows_dev,
+                                       self.host_slot_keys, True, None)
+        self._host_rows_upper += n_ev
+        self.ext.spill_rows(evict_slots, hslots, self.weights, self.state,
+                            self._host_weights,
+                            self._host_state if self.state_dim else None)
+        self._host_live = True
+        # tighten the host-rows bound (one sync, eviction is off the hot
+        # path by construction)
+        self._host_rows_upper = int(self.host_nrows_dev.item())
+
+    def host_keys_of_slots(self, slots: torch.Tensor) -> torch.Tensor:
+        return self.slot_keys.index_select(0, slots)
+
+    def _evict(self) -> None:
+        self._sync_nrows()
         n = self._nrows_exact
-        if n == 0:
+        margin = self._last_batch_upper
+        if n + margin <= self.cache_rows:
             return
-        n_evict_target = max(n - (3 * self.cache_rows) // 4,
-                             self.cache_rows // 4)
-        order = torch.argsort(self._touch[:n], stable=True)
-        keys_sorted = self.slot_keys[:n].index_select(0, order).cpu()
-        evict, evict_pos = [], []
-        for pos, k in enumerate(keys_sorted.tolist()):
-            if len(evict) >= n_evict_target:
-                break
-            if k in pinned_keys:
-                continue
-            evict.append(k)
-            evict_pos.append(pos)
-        if not evict:
+        keep_target = max(1, (3 * self.cache_rows) // 4 - margin)
+        n_evict = n - keep_target
+        if n_evict <= 0:
             return
-        evict_slots = order[torch.tensor(evict_pos, dtype=torch.int64,
-                                         device=self.device)]
-        w = self.weights[evict_slots].cpu()
-        s = self.state[evict_slots].cpu() if self.state_dim else None
-        for i, k in enumerate(evict):
-            hs = self._host_alloc(k)
-            self._host_weights[hs] = w[i]
-            if s is not None:
-                self._host_state[hs] = s[i]
+        self._cache_full_since_ckpt = True
+        touches = self._touch[1:n + 1]
+        # exact count (not a threshold mask): with work_id ties — every row
+        # touched every step — a mask would evict the whole cache
+        evict_slots = torch.topk(touches, n_evict, largest=False,
+                                 sorted=False).indices
+        evict_mask = torch.zeros(n, dtype=torch.bool, device=self.device)
+        evict_mask[evict_slots] = True
+        self._spill(evict_slots)
         # rebuild the cache with the survivors, compacted
-        mask = torch.ones(n, dtype=torch.bool, device=self.device)
-        mask[evict_slots] = False
-        keep = mask.nonzero(as_tuple=True)[0]
-        kw = self.weights[keep].clone()
-        ks = self.state[keep].clone() if self.state_dim else None
-        kk = self.slot_keys[:n][keep].clone()
-        kt = self._touch[:n][keep].clone()
-        m = keep.numel()
+        keep_slots = (~evict_mask).nonzero(as_tuple=True)[0]
+        kk = self.slot_keys.index_select(0, keep_slots)
+        kw = self.weights.index_select(0, keep_slots)
+        ks = (self.state.index_select(0, keep_slots)
+              if self.state_dim else None)
+        kt = touches.index_select(0, keep_slots)
         self.tk.fill_(-1)
         self.nrows_dev.zero_()
         self._nrows_upper = 0
         self._nrows_exact = 0
+        m = keep_slots.numel()
         if m:
             slots, _ = self.ext.ht_lookup(self.tk, self.tv, kk,
                                           self.nrows_dev, self.slot_keys,
@@ -183,49 +252,19 @@ class HipTieredVariableShard(HipVariableShard):
             self.weights[slots] = kw
             if ks is not None:
                 self.state[slots] = ks
-            self._touch[slots] = kt
+            self._touch.zero_()
+            self._touch.scatter_(0, slots + 1, kt)
             self._nrows_exact = m
             self._nrows_upper = m
+        else:
+            self._touch.zero_()
+        self._evict_epoch += 1
 
-    # -------------------------------------------------------------- training
+    # ----------------------------------------------------------- statistics
 
-    def pull(self, keys: torch.Tensor) -> torch.Tensor:
-        self._tier_admit(keys)
-        out = super().pull(keys)
-        slots = self._last_slots
-        if slots is not None:
-            # stamp only resolved slots: a miss slot (-1) must not corrupt
-            # row 0's LRU order
-            valid = slots >= 0
-            self._touch[slots[valid]] = self.work_id
-        return out
-
-    def _lookup_or_insert(self, keys: torch.Tensor, u_dev=None):
-        slots, new_mask = super()._lookup_or_insert(keys, u_dev)
-        self._last_slots = slots
-        return slots, new_mask
-
-    # bounded-path opt-out: the tier needs the host key list per batch, so
-    # the sharded engine must route through the exact path (it checks
-    # `getattr(shard, "pull_bounded", None)`)
-    pull_bounded = None
-    push_slots = None
-
-    def pull_readonly(self, keys: torch.Tensor) -> torch.Tensor:
-        out = super().pull_readonly(keys)
-        kl = keys.tolist()
-        rows = [(i, self._host_of[k]) for i, k in enumerate(kl)
-                if k in self._host_of]
-        if rows:
-            idx = torch.tensor([i for i, _ in rows], dtype=torch.int64,
-                               device=self.device)
-            hs = torch.tensor([h for _, h in rows], dtype=torch.int64)
-            out[idx] = self._host_weights[hs].to(self.device)
-        return out
-
-    def update_weights(self) -> None:
-        super().update_weights()
-        self.work_id += 1
+    def fault_count(self) -> int:
+        """Rows faulted in from the host tier since start (one sync)."""
+        return int(self._faulted.item())
 
     # ----------------------------------------------------------- persistence
 
@@ -233,17 +272,13 @@ class HipTieredVariableShard(HipVariableShard):
         return self._cache_full_since_ckpt and self._checkpoint_work_id is None
 
     def persist(self) -> int:
+        """Lightweight checkpoint: flush every cached row to the host tier
+        (reference persist; rows stay cached and keep training)."""
         self._sync_nrows()
         n = self._nrows_exact
         if n:
-            keys = self.slot_keys[:n].cpu().tolist()
-            w = self.weights[:n].cpu()
-            s = self.state[:n].cpu() if self.state_dim else None
-            for sl, k in enumerate(keys):
-                hs = self._host_alloc(k)
-                self._host_weights[hs] = w[sl]
-                if s is not None:
-                    self._host_state[hs] = s[sl]
+            self._spill(torch.arange(n, dtype=torch.int64,
+                                     device=self.device))
         self._checkpoint_work_id = self.work_id
         self._cache_full_since_ckpt = False
         return self.work_id
@@ -251,33 +286,45 @@ class HipTieredVariableShard(HipVariableShard):
     def checkpoint_committed(self) -> None:
         self._checkpoint_work_id = None
 
+    def _host_only(self):
+        """(keys, host_slots) of host-tier rows not currently cached."""
+        hn = int(self.host_nrows_dev.item())
+        if hn == 0:
+            e = torch.empty(0, dtype=torch.int64, device=self.device)
+            return e, e
+        hkeys = self.host_slot_keys[:hn]
+        cache_slots, _ = self.ext.ht_lookup(self.tk, self.tv, hkeys,
+                                            self.nrows_dev, self.slot_keys,
+                                            False, None)
+        mask = cache_slots < 0
+        return hkeys[mask], torch.arange(hn, device=self.device)[mask]
+
     @property
     def num_rows(self) -> int:
-        cached = set(self.slot_keys[:int(self.nrows_dev.item())]
-                     .cpu().tolist())
-        return len(cached) + sum(1 for k in self._host_of if k not in cached)
+        n = int(self.nrows_dev.item())
+        hkeys, _ = self._host_only()
+        return n + hkeys.numel()
 
     def export_rows(self, include_state: bool = True):
         keys_c, w_c, s_c = super().export_rows(include_state)
-        cached = set(keys_c.cpu().tolist())
-        items = sorted(((k, h) for k, h in self._host_of.items()
-                        if k not in cached), key=lambda kv: kv[1])
-        if not items:
+        hkeys, hslots = self._host_only()
+        if hkeys.numel() == 0:
             return keys_c, w_c, s_c
-        hkeys = torch.tensor([k for k, _ in items], dtype=torch.int64,
-                             device=self.device)
-        hslots = torch.tensor([h for _, h in items], dtype=torch.int64)
+        hs_cpu = hslots.cpu()
         keys = torch.cat([keys_c, hkeys])
-        w = torch.cat([w_c, self._host_weights[hslots].to(self.device)])
+        w = torch.cat([w_c, self._host_weights[hs_cpu].to(self.device)])
         s = None
         if include_state and self.state_dim:
-            s = torch.cat([s_c, self._host_state[hslots].to(self.device)])
+            s = torch.cat([s_c, self._host_state[hs_cpu].to(self.device)])
         return keys, w, s
 
     def clear(self) -> None:
         super().clear()
-        self._host_of = {}
-        self._host_free = []
-        self._host_next = 0
+        self.host_tk.fill_(-1)
+        self.host_nrows_dev.zero_()
+        self._host_rows_upper = 0
+        self._host_live = False
+        self._touch.zero_()
         self._cache_full_since_ckpt = False
         self._checkpoint_work_id = None
+        self._evict_epoch += 1

@@ -56,6 +56,13 @@ void emb_bce_bwd(const float*, const float*, long, const float*, float*,
                  hipStream_t_);
 void emb_bucketize_pad(const i64*, long, const int*, long, long, i64*, int*,
                        int*, int*, int*, hipStream_t_);
+void emb_fault_in(const i64*, long, const int*, const u64*, const int*, long,
+                  const float*, const float*, float*, float*, long, long,
+                  const i64*, unsigned char*, int*, hipStream_t_);
+void emb_spill_rows(const i64*, const i64*, long, const float*, const float*,
+                    float*, float*, long, long, hipStream_t_);
+void emb_gather_host(const i64*, long, const i64*, const u64*, const int*,
+                     long, const float*, float*, long, hipStream_t_);
 void emb_gather_pad(const float*, const u64*, long, const int*, long, float*,
                     hipStream_t_);
 void emb_scatter_out(const float*, const i64*, long, const int*, long, float*,
@@ -228,6 +235,70 @@ std::tuple<torch::Tensor, torch::Tensor> reduce_by_inverse(
                           dim, ugrads.data_ptr<float>(),
                           (u64*)counts.data_ptr<i64>(), u, cur_stream());
     return {ugrads, counts};
+}
+
+// ---- capacity tier v2 --------------------------------------------------
+// Pinned host tensors are device-accessible on ROCm (hipHostMalloc-backed
+// via the torch pinned allocator): the fault-in/spill kernels read/write
+// them directly — only touched rows cross the host link.
+
+#define CHECK_PINNED(t) \
+    TORCH_CHECK((t).is_pinned(), #t " must be a pinned host tensor")
+
+void fault_in(torch::Tensor keys, OptTensor u_dev, torch::Tensor htk,
+              torch::Tensor htv, torch::Tensor host_w, OptTensor host_s,
+              torch::Tensor weights, torch::Tensor state, torch::Tensor slots,
+              torch::Tensor new_mask, torch::Tensor faulted) {
+    CHECK_GPU(keys); CHECK_CONT(keys); CHECK_CONT(host_w);
+    CHECK_PINNED(host_w);
+    const c10::cuda::CUDAGuard guard(keys.device());
+    long sd = state.numel() ? state.size(1) : 0;
+    const float* hs = nullptr;
+    if (sd) {
+        TORCH_CHECK(host_s.has_value(), "state tier needs host_s");
+        CHECK_PINNED(*host_s);
+        hs = host_s->data_ptr<float>();
+    }
+    emb_fault_in(keys.data_ptr<i64>(), keys.numel(), u_ptr(u_dev),
+                 (const u64*)htk.data_ptr<i64>(), htv.data_ptr<int>(),
+                 htk.numel(), host_w.data_ptr<float>(), hs,
+                 weights.data_ptr<float>(),
+                 sd ? state.data_ptr<float>() : nullptr, weights.size(1), sd,
+                 slots.data_ptr<i64>(), new_mask.data_ptr<uint8_t>(),
+                 faulted.data_ptr<int>(), cur_stream());
+}
+
+void spill_rows(torch::Tensor cache_slots, torch::Tensor host_slots,
+                torch::Tensor weights, torch::Tensor state,
+                torch::Tensor host_w, OptTensor host_s) {
+    CHECK_GPU(cache_slots); CHECK_CONT(cache_slots); CHECK_CONT(host_slots);
+    CHECK_PINNED(host_w);
+    const c10::cuda::CUDAGuard guard(cache_slots.device());
+    long sd = state.numel() ? state.size(1) : 0;
+    float* hs = nullptr;
+    if (sd) {
+        TORCH_CHECK(host_s.has_value(), "state tier needs host_s");
+        CHECK_PINNED(*host_s);
+        hs = host_s->data_ptr<float>();
+    }
+    emb_spill_rows(cache_slots.data_ptr<i64>(), host_slots.data_ptr<i64>(),
+                   cache_slots.numel(), weights.data_ptr<float>(),
+                   sd ? state.data_ptr<float>() : nullptr,
+                   host_w.data_ptr<float>(), hs, weights.size(1), sd,
+                   cur_stream());
+}
+
+void gather_host(torch::Tensor keys, torch::Tensor cache_slots,
+                 torch::Tensor htk, torch::Tensor htv, torch::Tensor host_w,
+                 torch::Tensor out) {
+    CHECK_GPU(keys); CHECK_CONT(keys); CHECK_CONT(out);
+    CHECK_PINNED(host_w);
+    const c10::cuda::CUDAGuard guard(keys.device());
+    emb_gather_host(keys.data_ptr<i64>(), keys.numel(),
+                    cache_slots.data_ptr<i64>(),
+                    (const u64*)htk.data_ptr<i64>(), htv.data_ptr<int>(),
+                    htk.numel(), host_w.data_ptr<float>(),
+                    out.data_ptr<float>(), out.size(1), cur_stream());
 }
 
 // ---- padded all-to-all -------------------------------------------------
@@ -523,6 +594,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("reduce_by_inverse", &reduce_by_inverse,
           "grad reduce-by-key with counts (LDS-aggregated)");
     m.def("apply_optimizer", &apply_optimizer, "fused sparse optimizer step");
+    m.def("fault_in", &fault_in,
+          "tier v2: copy host-resident rows into fresh cache slots, "
+          "clearing their lazy-init mask");
+    m.def("spill_rows", &spill_rows,
+          "tier v2: copy cache rows into the pinned host slab");
+    m.def("gather_host", &gather_host,
+          "tier v2: read-only host-row gather for cache misses");
     m.def("bucketize_pad", &bucketize_pad,
           "owner-bucketize unique keys into a fixed padded [world, cap] "
           "wire block (sync-free multi-rank route)");

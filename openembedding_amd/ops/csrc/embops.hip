@@ -856,6 +856,113 @@ __global__ void k_split_payload(const float* __restrict__ g2c, long u,
     else counts[r] = (u64)(g2c[e] + 0.5f);
 }
 
+// ------------------------------------------------------- capacity tier v2
+// HBM row-cache over a pinned host-DRAM backing store (the reference's
+// DRAM-cache-over-PMem design, PmemEmbeddingTable.h:237-270, re-based on
+// the MI355X memory hierarchy). A SECOND device-side hash table maps
+// spilled keys -> host-slab slots, so the fault-in decision runs on
+// device with no host key lists; pinned host memory is read/written
+// directly by these kernels (zero-copy gather: only touched rows cross
+// the PCIe/host link).
+
+__device__ __forceinline__ long host_probe(const u64* __restrict__ htk,
+                                           const int* __restrict__ htv,
+                                           long mask, u64 k) {
+    u64 h = splitmix64(k) & (u64)mask;
+    for (long p = 0; p <= mask; ++p) {
+        u64 cur = htk[h];
+        if (cur == k) return (long)htv[h];
+        if (cur == EMPTY) return -1;
+        h = (h + 1) & (u64)mask;
+    }
+    return -1;
+}
+
+// fault-in: rows that missed the cache (new_mask=1) but live in the host
+// tier are copied host->cache and their new_mask cleared, so the
+// downstream gather treats them as existing rows (no lazy re-init).
+// G lanes per row.
+template <int G>
+__global__ void k_fault_in(const i64* __restrict__ keys, long n,
+                           const int* __restrict__ u_dev,
+                           const u64* __restrict__ htk,
+                           const int* __restrict__ htv, long hmask,
+                           const float* __restrict__ host_w,
+                           const float* __restrict__ host_s,
+                           float* __restrict__ weights,
+                           float* __restrict__ state, long dim, long sd,
+                           const i64* __restrict__ slots,
+                           unsigned char* __restrict__ new_mask,
+                           int* __restrict__ faulted) {
+    long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) / G;
+    int lane = threadIdx.x % G;
+    if (g >= n) return;
+    if (u_dev && g >= *u_dev) return;
+    if (!new_mask[g]) return;
+    i64 slot = slots[g];
+    if (slot < 0) return;
+    u64 k = (u64)keys[g];
+    if (k == EMPTY) return;
+    long hs = host_probe(htk, htv, hmask, k);
+    if (hs < 0) return;                       // truly new: lazy init
+    float* wrow = weights + (u64)slot * dim;
+    const float* hw = host_w + (u64)hs * dim;
+    for (long j = lane; j < dim; j += G) wrow[j] = hw[j];
+    if (sd > 0) {
+        float* srow = state + (u64)slot * sd;
+        const float* hsrow = host_s + (u64)hs * sd;
+        for (long j = lane; j < sd; j += G) srow[j] = hsrow[j];
+    }
+    if (lane == 0) {
+        new_mask[g] = 0;
+        atomicAdd(faulted, 1);
+    }
+}
+
+// spill: copy cache rows (by slot) into host-slab rows (by host slot)
+template <int G>
+__global__ void k_spill_rows(const i64* __restrict__ cache_slots,
+                             const i64* __restrict__ host_slots, long n,
+                             const float* __restrict__ weights,
+                             const float* __restrict__ state,
+                             float* __restrict__ host_w,
+                             float* __restrict__ host_s, long dim, long sd) {
+    long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) / G;
+    int lane = threadIdx.x % G;
+    if (g >= n) return;
+    i64 cs = cache_slots[g], hs = host_slots[g];
+    if (cs < 0 || hs < 0) return;
+    const float* wrow = weights + (u64)cs * dim;
+    float* hw = host_w + (u64)hs * dim;
+    for (long j = lane; j < dim; j += G) hw[j] = wrow[j];
+    if (sd > 0) {
+        const float* srow = state + (u64)cs * sd;
+        float* hsrow = host_s + (u64)hs * sd;
+        for (long j = lane; j < sd; j += G) hsrow[j] = srow[j];
+    }
+}
+
+// read-only host gather (serving/export): rows that missed the cache
+// (slot -1) but exist in the host tier are read straight into `out`
+template <int G>
+__global__ void k_gather_host(const i64* __restrict__ keys, long n,
+                              const i64* __restrict__ cache_slots,
+                              const u64* __restrict__ htk,
+                              const int* __restrict__ htv, long hmask,
+                              const float* __restrict__ host_w,
+                              float* __restrict__ out, long dim) {
+    long g = ((long)blockIdx.x * blockDim.x + threadIdx.x) / G;
+    int lane = threadIdx.x % G;
+    if (g >= n) return;
+    if (cache_slots[g] >= 0) return;          // cache hit already gathered
+    u64 k = (u64)keys[g];
+    if (k == EMPTY) return;
+    long hs = host_probe(htk, htv, hmask, k);
+    if (hs < 0) return;                       // unknown key: stays zeros
+    const float* hw = host_w + (u64)hs * dim;
+    for (long j = lane; j < dim; j += G) out[(u64)g * dim + j] = hw[j];
+}
+
 // ============================================================== launchers
 
 extern "C" {
@@ -997,6 +1104,50 @@ void emb_apply_optimizer(int opt, float* weights, float* state, long dim,
         }
         (void)G;
     }
+}
+
+void emb_fault_in(const i64* keys, long n, const int* u_dev, const u64* htk,
+                  const int* htv, long hcap, const float* host_w,
+                  const float* host_s, float* weights, float* state,
+                  long dim, long sd, const i64* slots,
+                  unsigned char* new_mask, int* faulted,
+                  hipStream_t stream) {
+    if (!n) return;
+    if (dim <= 32)
+        k_fault_in<16><<<grid1d(n * 16), BLOCK, 0, stream>>>(
+            keys, n, u_dev, htk, htv, hcap - 1, host_w, host_s, weights,
+            state, dim, sd, slots, new_mask, faulted);
+    else
+        k_fault_in<64><<<grid1d(n * 64), BLOCK, 0, stream>>>(
+            keys, n, u_dev, htk, htv, hcap - 1, host_w, host_s, weights,
+            state, dim, sd, slots, new_mask, faulted);
+}
+
+void emb_spill_rows(const i64* cache_slots, const i64* host_slots, long n,
+                    const float* weights, const float* state, float* host_w,
+                    float* host_s, long dim, long sd, hipStream_t stream) {
+    if (!n) return;
+    if (dim <= 32)
+        k_spill_rows<16><<<grid1d(n * 16), BLOCK, 0, stream>>>(
+            cache_slots, host_slots, n, weights, state, host_w, host_s, dim,
+            sd);
+    else
+        k_spill_rows<64><<<grid1d(n * 64), BLOCK, 0, stream>>>(
+            cache_slots, host_slots, n, weights, state, host_w, host_s, dim,
+            sd);
+}
+
+void emb_gather_host(const i64* keys, long n, const i64* cache_slots,
+                     const u64* htk, const int* htv, long hcap,
+                     const float* host_w, float* out, long dim,
+                     hipStream_t stream) {
+    if (!n) return;
+    if (dim <= 32)
+        k_gather_host<16><<<grid1d(n * 16), BLOCK, 0, stream>>>(
+            keys, n, cache_slots, htk, htv, hcap - 1, host_w, out, dim);
+    else
+        k_gather_host<64><<<grid1d(n * 64), BLOCK, 0, stream>>>(
+            keys, n, cache_slots, htk, htv, hcap - 1, host_w, out, dim);
 }
 
 void emb_bucketize_pad(const i64* keys, long n, const int* u_dev, long world,

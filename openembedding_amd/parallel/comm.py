@@ -30,7 +30,7 @@ def all_to_all_lengths(send_counts: torch.Tensor) -> torch.Tensor:
         return send_counts.clone()  # world-1 (forced-remote tests)
     world = dist.get_world_size()
     if backend() == "gloo":
-        cpu = send_counts.to("cpu")
+        cpu = send_counts.detach().cpu()
         gathered = [torch.zeros_like(cpu) for _ in range(world)]
         dist.all_gather(gathered, cpu)
         rank = dist.get_rank()
@@ -57,6 +57,12 @@ def all_to_all_v(inp: torch.Tensor, in_splits: Sequence[int],
                                output_split_sizes=list(out_splits),
                                input_split_sizes=list(in_splits))
         return out
+    if inp.is_cuda:
+        # gloo transports CPU buffers only: stage D2H/H2D so the GPU engine
+        # can run multi-process on one device (scripts/rccl_2rank_1gpu.py —
+        # RCCL itself refuses two ranks on one GPU)
+        res = all_to_all_v(inp.cpu(), in_splits, out_splits)
+        return res.to(inp.device)
     # gloo fallback: batched isend/irecv (no alltoall in ProcessGroupGloo)
     in_offs = _offsets(in_splits)
     out_offs = _offsets(out_splits)

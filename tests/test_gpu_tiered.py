@@ -1,6 +1,9 @@
-"""GPU capacity tier (HBM cache over pinned host DRAM): equivalence with
-the untired HIP shard under eviction pressure, fault-back, persistence
-(BASELINE.json config 5; reference PMem tier semantics)."""
+"""GPU capacity tier v2 (HBM cache over pinned host DRAM): equivalence with
+the untired HIP shard under eviction pressure, fault-back, LRU order,
+persistence (BASELINE.json config 5; reference PMem tier semantics,
+PmemEmbeddingTable.h). All bookkeeping is device-side in v2 — these tests
+also pin the contract that the hot path stays on the bounded sync-free
+route."""
 
 import pytest
 import torch
@@ -46,7 +49,8 @@ def test_equivalence_under_eviction():
         ukt, wt = _step(t, keys)
         ukr, wr = _step(r, keys)
         assert torch.equal(wt, wr), f"diverged at step {step}"
-    assert len(t._host_of) > 0, "no eviction happened — raise pressure"
+    assert t._host_live, "no eviction happened — raise pressure"
+    assert t.fault_count() > 0, "no fault-in happened — raise pressure"
     allk = torch.arange(256, dtype=torch.int64, device=DEV)
     assert torch.equal(t.pull_readonly(allk), r.pull_readonly(allk))
     assert t.num_rows == r.num_rows
@@ -56,9 +60,36 @@ def test_fault_back_roundtrip():
     t, _ = _mk(cache_rows=8)
     k1 = torch.arange(0, 8, dtype=torch.int64, device=DEV)
     w1 = t.pull(k1).clone()
-    t.pull(torch.arange(8, 16, dtype=torch.int64, device=DEV))
-    assert len(t._host_of) > 0
+    # v2 evicts at the commit boundary: push+commit the next batches to
+    # force k1 out of the cache
+    for lo in (8, 16):
+        _step(t, torch.arange(lo, lo + 8, dtype=torch.int64, device=DEV))
+    assert t._host_live
     assert torch.equal(t.pull(k1), w1)
+    assert t.fault_count() > 0
+
+
+def test_lru_order_evicts_coldest():
+    """Rows with recent touch stamps must survive eviction (device LRU)."""
+    t, _ = _mk(cache_rows=16, dim=4)
+    hot = torch.arange(0, 8, dtype=torch.int64, device=DEV)
+    cold = torch.arange(100, 116, dtype=torch.int64, device=DEV)
+    t.pull(torch.cat([hot, cold]))          # 24 cached rows
+    # white-box stamps: hot recently touched, cold stale
+    slots_hot, _ = t.ext.ht_lookup(t.tk, t.tv, hot, t.nrows_dev,
+                                   t.slot_keys, False, None)
+    t._touch[1:25] = 1
+    t._touch[slots_hot + 1] = 5
+    t._last_batch_upper = 0                 # isolate the LRU decision
+    t._evict()                              # keep_target = 12 -> evict 12
+    assert t._host_live
+    assert int((t._lookup_readonly(hot) >= 0).sum()) == 8, \
+        "recently-touched rows must survive"
+    assert int((t._lookup_readonly(cold) >= 0).sum()) == 4, \
+        "the 12 coldest rows must be evicted"
+    # and the evicted rows are still readable through the host tier
+    r = t.pull_readonly(cold)
+    assert torch.isfinite(r).all() and float(r.abs().sum()) > 0
 
 
 def test_persist_and_export():
@@ -77,15 +108,74 @@ def test_persist_and_export():
     assert torch.equal(st[ot], sr[orr])
 
 
-def test_sharded_engine_routes_exact_path():
-    """The engine must not take the bounded path for a tiered shard."""
+def test_sharded_engine_keeps_bounded_path():
+    """v2 keeps the sync-free bounded route for tiered shards (v1 opted
+    out into the exact path — the measured 2x tier overhead)."""
     from openembedding_amd.parallel.sharded import ShardedVariable
 
-    t, _ = _mk(cache_rows=16)
+    t, r = _mk(cache_rows=4096)
     v = ShardedVariable(t)
     keys = torch.randint(0, 64, (40,), dtype=torch.int64).to(DEV)
     out, h = v.pull(keys)
-    assert not h.bounded
+    assert h.bounded, "tiered shard must ride the bounded sync-free path"
     v.push(h, torch.ones_like(out))
     v.update_weights()
     torch.cuda.synchronize()
+    # equivalence with the untired shard on the same flow
+    vr = ShardedVariable(r)
+    out_r, h_r = vr.pull(keys)
+    assert torch.equal(out.cpu(), out_r.cpu())
+    vr.push(h_r, torch.ones_like(out_r))
+    vr.update_weights()
+    after_t = t.pull_readonly(torch.unique(keys))
+    after_r = r.pull_readonly(torch.unique(keys))
+    assert torch.equal(after_t, after_r)
+
+
+def test_stale_handle_raises():
+    """A pull handle held across an eviction must fail loudly, not apply
+    gradients to recycled slots (advisor finding, round 1)."""
+    from openembedding_amd.parallel.sharded import ShardedVariable
+
+    t, _ = _mk(cache_rows=8)
+    v = ShardedVariable(t)
+    keys = torch.arange(0, 8, dtype=torch.int64, device=DEV)
+    out, h = v.pull(keys)          # handle with saved slots
+    # trigger eviction via other keys
+    for lo in (8, 16):
+        _step(t, torch.arange(lo, lo + 8, dtype=torch.int64, device=DEV))
+    assert t._host_live
+    with pytest.raises(RuntimeError, match="stale pull handle"):
+        v.push(h, torch.ones_like(out))
+        v.update_weights()
+
+
+def test_deepfm_cache_mb_runs():
+    """End-to-end: DeepFM hash-mode with a tight --cache-mb budget steps
+    without error and trains (loss finite)."""
+    import openembedding_amd as oe
+    from openembedding_amd import context as ctx_mod
+    import openembedding_amd.torch as embed
+    from openembedding_amd.models import DeepFM, synthetic_batch
+
+    old = oe.flags.config
+    oe.flags.config = "server:\n  cache_size: 8\n"
+    try:
+        ctx = ctx_mod.Context(device=DEV)
+        ctx_mod._context = ctx
+        model = DeepFM(dim=8, hash_mode=True).to(DEV)
+        opt = embed.distributed_optimizer(
+            torch.optim.Adagrad(model.parameters(), lr=0.01))
+        lossf = torch.nn.BCEWithLogitsLoss()
+        g = torch.Generator().manual_seed(3)
+        for _ in range(6):
+            dense, sparse, labels = synthetic_batch(256, generator=g)
+            opt.zero_grad()
+            loss = lossf(model(dense.to(DEV), sparse.to(DEV)),
+                         labels.to(DEV))
+            loss.backward()
+            opt.step()
+        assert torch.isfinite(loss)
+    finally:
+        oe.flags.config = old
+        ctx_mod._context = None
