@@ -238,11 +238,19 @@ class HipVariableShard(VariableShard):
             raise RuntimeError("update_weights called before set_optimizer")
         opt_id = _OPT_IDS[self.optimizer.category]
         cfg = _opt_cfg_vector(self.optimizer)
+        blocks = self._pending_bounded
+        self._pending_bounded = []
+        if len(blocks) > 1:
+            # several pulls committed together: reference semantics
+            # (MpscGradientReducer.h:30-53) are ONE optimizer step per
+            # unique key over the summed grads+counts — merge the blocks
+            # (sync-free: tail-mask, concat, re-dedup) instead of applying
+            # them sequentially
+            blocks = [self._merge_bounded(blocks)]
         # bounded blocks: slots known from pull, zero-sync apply
-        for keys_buf, u_dev, slots, grads, counts in self._pending_bounded:
+        for keys_buf, u_dev, slots, grads, counts in blocks:
             self.ext.apply_optimizer(opt_id, self.weights, self.state, slots,
                                      grads.contiguous(), counts, cfg, u_dev)
-        self._pending_bounded = []
         if self._pending:
             if len(self._pending) == 1:
                 keys, grads, counts = self._pending[0]
@@ -270,6 +278,29 @@ class HipVariableShard(VariableShard):
                     self.weights.shape[0] * 3) // 4:
                 self._nrows_exact = int(self.nrows_dev.item())
                 self._nrows_upper = self._nrows_exact
+
+    def _merge_bounded(self, blocks):
+        """Merge several bounded gradient blocks into one: mask each
+        block's garbage tail with the reserved key -1 / zero payload, then
+        concatenate and re-dedup. -1 entries dedup into their own uids with
+        slot -1 and are skipped by the apply kernel. Zero host syncs."""
+        ks, ps = [], []
+        for keys_buf, u_dev, slots, grads, counts in blocks:
+            self.ext.mask_tail(keys_buf, grads, counts, u_dev)
+            ks.append(keys_buf)
+            ps.append(torch.cat([grads,
+                                 counts.to(grads.dtype).unsqueeze(1)],
+                                dim=1))
+        keys = torch.cat(ks)
+        payload = torch.cat(ps)
+        uk, inv, u_dev = self.ext.unique_bounded(keys)
+        g2c, _ = self.ext.reduce_by_inverse(inv, payload.contiguous(),
+                                            keys.numel())
+        g2, c2 = self.ext.split_payload(g2c, u_dev)
+        slots, new_mask = self._lookup_or_insert(uk, u_dev)
+        # init rows that were pushed without a prior pull
+        self._gather(uk, slots, new_mask, False, u_dev=u_dev)
+        return uk, u_dev, slots, g2, c2
 
     @staticmethod
     def _in_graph_capture() -> bool:

@@ -56,6 +56,7 @@ void emb_bce_bwd(const float*, const float*, long, const float*, float*,
                  hipStream_t_);
 void emb_bucketize_pad(const i64*, long, const int*, long, long, i64*, int*,
                        int*, int*, int*, hipStream_t_);
+void emb_mask_tail(i64*, float*, u64*, long, long, const int*, hipStream_t_);
 void emb_fault_in(const i64*, long, const int*, const u64*, const int*, long,
                   const float*, const float*, float*, float*, long, long,
                   const i64*, unsigned char*, int*, hipStream_t_);
@@ -235,6 +236,17 @@ std::tuple<torch::Tensor, torch::Tensor> reduce_by_inverse(
                           dim, ugrads.data_ptr<float>(),
                           (u64*)counts.data_ptr<i64>(), u, cur_stream());
     return {ugrads, counts};
+}
+
+void mask_tail(torch::Tensor keys, torch::Tensor grads, torch::Tensor counts,
+               OptTensor u_dev) {
+    CHECK_GPU(keys); CHECK_CONT(keys); CHECK_CONT(grads); CHECK_CONT(counts);
+    TORCH_CHECK(grads.size(0) == keys.numel()
+                && counts.numel() == keys.numel(), "mask_tail sizes");
+    const c10::cuda::CUDAGuard guard(keys.device());
+    emb_mask_tail(keys.data_ptr<i64>(), grads.data_ptr<float>(),
+                  (u64*)counts.data_ptr<i64>(), keys.numel(), grads.size(1),
+                  u_ptr(u_dev), cur_stream());
 }
 
 // ---- capacity tier v2 --------------------------------------------------
@@ -594,6 +606,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("reduce_by_inverse", &reduce_by_inverse,
           "grad reduce-by-key with counts (LDS-aggregated)");
     m.def("apply_optimizer", &apply_optimizer, "fused sparse optimizer step");
+    m.def("mask_tail", &mask_tail,
+          "zero the garbage tail of a bounded block (enables multi-block "
+          "merge by concatenation)");
     m.def("fault_in", &fault_in,
           "tier v2: copy host-resident rows into fresh cache slots, "
           "clearing their lazy-init mask");

@@ -276,7 +276,14 @@ __global__ void k_ht_lookup(u64* __restrict__ tk, int* __restrict__ tv,
                             long row_cap) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    if (u_dev && i >= *u_dev) return;
+    if (u_dev && i >= *u_dev) {
+        // beyond the live unique count: write a defined miss — consumers
+        // are u_dev-guarded, but a garbage slot index must never reach a
+        // host-side scatter (the tier's LRU stamp faulted on exactly that)
+        slots[i] = -1;
+        if (new_mask) new_mask[i] = 0;
+        return;
+    }
     u64 k = (u64)keys[i];
     if (k == EMPTY) {
         // key -1 is the table's empty marker and therefore RESERVED —
@@ -349,7 +356,7 @@ __global__ void k_array_touch(unsigned char* __restrict__ valid,
                               const int* __restrict__ u_dev) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    if (u_dev && i >= *u_dev) { slots[i] = -1; return; }
+    if (u_dev && i >= *u_dev) { slots[i] = -1; new_mask[i] = 0; return; }
     if (keys[i] < 0) {  // invalid for a bounded vocabulary; C++ trunc
         slots[i] = -1;  // division would silently map -1 to slot 0
         new_mask[i] = 0;
@@ -856,6 +863,23 @@ __global__ void k_split_payload(const float* __restrict__ g2c, long u,
     else counts[r] = (u64)(g2c[e] + 0.5f);
 }
 
+// zero the garbage tail of a bounded (buffer + device-count) gradient
+// block so blocks can be merged by concatenation: a commit with several
+// pulls of one variable must apply the optimizer ONCE per unique key over
+// the summed gradients (reference MpscGradientReducer.h:30-53 semantics),
+// not once per block.
+__global__ void k_mask_tail(i64* __restrict__ keys, float* __restrict__ grads,
+                            u64* __restrict__ counts, long n, long dim,
+                            const int* __restrict__ u_dev) {
+    long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (e >= n * dim) return;
+    long i = e / dim, j = e % dim;
+    long live = u_dev ? (long)*u_dev : n;
+    if (i < live) return;
+    grads[e] = 0.f;
+    if (j == 0) { keys[i] = -1; counts[i] = 0; }
+}
+
 // ------------------------------------------------------- capacity tier v2
 // HBM row-cache over a pinned host-DRAM backing store (the reference's
 // DRAM-cache-over-PMem design, PmemEmbeddingTable.h:237-270, re-based on
@@ -1104,6 +1128,14 @@ void emb_apply_optimizer(int opt, float* weights, float* state, long dim,
         }
         (void)G;
     }
+}
+
+void emb_mask_tail(i64* keys, float* grads, u64* counts, long n, long dim,
+                   const int* u_dev, hipStream_t stream) {
+    if (n)
+        k_mask_tail<<<grid1d(n * dim), BLOCK, 0, stream>>>(keys, grads,
+                                                           counts, n, dim,
+                                                           u_dev);
 }
 
 void emb_fault_in(const i64* keys, long n, const int* u_dev, const u64* htk,

@@ -149,6 +149,32 @@ def test_scatter_out_kernel():
     torch.testing.assert_close(out[5], rows[9])
 
 
+def test_multi_pull_commit_merges_blocks():
+    """Two pulls of overlapping keys committed together must apply the
+    optimizer ONCE per unique key over summed grads (reference
+    MpscGradientReducer semantics) — on both the CPU oracle and the GPU
+    bounded path (which queues one block per pull and must merge them)."""
+    batches = [(torch.arange(0, 64, dtype=torch.int64),
+                torch.full((64, DIM), 0.5)),
+               (torch.arange(32, 96, dtype=torch.int64),  # overlap 32..64
+                torch.full((64, DIM), 0.25))]
+    results = []
+    for device in ("cpu", DEV):
+        st, var = _sharded(device, padded=False, force_remote=False)
+        hs = []
+        for keys, grads in batches:
+            _, h = var.pull(keys.to(device))
+            hs.append((h, grads))
+        for h, grads in hs:
+            var.push(h, grads.to(device))
+        st.update_weights()
+        probe = torch.arange(0, 96, dtype=torch.int64, device=device)
+        after, _ = var.pull(probe, readonly=True)
+        results.append(after.cpu())
+    torch.testing.assert_close(results[0], results[1], rtol=1e-5,
+                               atol=1e-6)
+
+
 def test_padded_step_graph_capturable():
     """The padded route must capture into a hipGraph (array mode; replays
     train fresh data through static buffers)."""
