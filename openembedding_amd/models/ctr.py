@@ -11,6 +11,7 @@ and return logits [B].
 
 from __future__ import annotations
 
+import math
 from typing import List, Optional, Sequence
 
 import torch
@@ -420,6 +421,10 @@ def convert_mlp_bf16(model: _CTRBase) -> _CTRBase:
               and seq[0].out_features <= 416)
         model.fused_mlp = ok
     model.head_bf16 = True
+    if hasattr(model, "cin"):
+        # CIN layer GEMMs in bf16 (fp32 accumulation inside the library
+        # GEMM); weights and outputs stay fp32
+        model.cin.compute_dtype = torch.bfloat16
     return model
 
 
