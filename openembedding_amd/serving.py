@@ -73,16 +73,24 @@ class ServedModel:
     def load(self, device: str = "cpu") -> None:
         """Stream the dump into read-only tables. All shards of the dump are
         merged into one local table (shard_num=1 view — serving is per-key
-        lookup, the training shard layout is irrelevant)."""
+        lookup, the training shard layout is irrelevant).
+
+        On a cuda device the tables are HipVariableShards, so serving pulls
+        run the same k_ht_lookup / k_gather_init HIP kernels as training
+        (the round-1 CPU-table fallback bypassed every kernel)."""
         meta = read_meta(self.uri)
         self.meta = meta
+        shard_cls = VariableShard
+        if device.startswith("cuda"):
+            from .core.variable_gpu import HipVariableShard
+            shard_cls = HipVariableShard
         shards: Dict[int, VariableShard] = {}
         for mvar in meta["variables"]:
             vm = VariableMeta(variable_id=mvar["variable_id"],
                               embedding_dim=mvar["embedding_dim"],
                               vocabulary_size=mvar["vocabulary_size"])
-            shards[vm.variable_id] = VariableShard(vm, shard_id=0, shard_num=1,
-                                                   device=device)
+            shards[vm.variable_id] = shard_cls(vm, shard_id=0, shard_num=1,
+                                               device=device)
         for hdr, keys, w, _s in iter_blocks(self.uri, meta):
             sh = shards.get(hdr["variable_id"])
             if sh is None or not len(keys):
@@ -217,6 +225,56 @@ class ModelController:
 
     def shutdown_node(self, node_id: int = 0) -> None:
         self.shutdown_requested = True
+
+
+class ServingClient:
+    """Minimal HTTP pull client with replica failover.
+
+    The documented HA deployment runs N serving processes behind this
+    client (or any load balancer); it is the reference's client-side
+    replica machinery — pick_one_replica + retry-until-success
+    (EmbeddingPullOperator.cpp:50-58, c_api_test.h:117-121) — over HTTP:
+    requests rotate round-robin across endpoints and fail over to the next
+    replica on connection error or non-200."""
+
+    def __init__(self, endpoints: List[str], timeout: float = 5.0,
+                 attempts_per_endpoint: int = 2):
+        if not endpoints:
+            raise ValueError("need at least one endpoint")
+        self.endpoints = [e.rstrip("/") for e in endpoints]
+        self.timeout = timeout
+        self.attempts = attempts_per_endpoint * len(self.endpoints)
+        self._i = 0
+
+    def _request(self, method: str, path: str, json_body=None):
+        import requests
+
+        last: Optional[Exception] = None
+        for _ in range(self.attempts):
+            ep = self.endpoints[self._i % len(self.endpoints)]
+            self._i += 1
+            try:
+                r = requests.request(method, ep + path, json=json_body,
+                                     timeout=self.timeout)
+                if r.status_code == 200:
+                    return r.json()
+                last = RuntimeError(f"{ep}{path}: HTTP {r.status_code} "
+                                    f"{r.text[:200]}")
+            except Exception as e:  # noqa: BLE001 - any replica failure
+                last = e
+        raise RuntimeError(f"all {len(self.endpoints)} replicas failed: "
+                           f"{last!r}")
+
+    def pull(self, sign: str, variable_id: int, indices) -> List:
+        if torch.is_tensor(indices):
+            indices = indices.tolist()
+        r = self._request("POST",
+                          f"/models/{sign}/variables/{variable_id}/pull",
+                          {"indices": indices})
+        return r["weights"]
+
+    def show_models(self) -> List[dict]:
+        return self._request("GET", "/models")
 
 
 def make_app(controller: Optional[ModelController] = None,
