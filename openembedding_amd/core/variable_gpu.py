@@ -319,7 +319,11 @@ class HipVariableShard(VariableShard):
 
     def export_rows(self, include_state: bool = True):
         if self.meta.use_hash_table:
-            n = self.num_rows
+            # cached-row count, NOT self.num_rows: the tier subclass
+            # overrides num_rows to include host-resident rows, which are
+            # exported by ITS export_rows — using it here would read stale
+            # slab entries beyond the live count
+            n = int(self.nrows_dev.item())
             keys = self.slot_keys[:n].clone()
             w = self.weights[:n].clone()
             s = self.state[:n].clone() if (include_state and self.state_dim) else None

@@ -307,44 +307,43 @@ class _CINLayerFn(torch.autograd.Function):
     (bf16 on the native-MFMA path, fp32 for matched-precision runs)."""
 
     @staticmethod
+    def _vmat(x0p, xkp, B, F, H, d):
+        # V [d, B, F*H] = per-slice outer products, built from the
+        # d-leading contiguous layouts (one broadcast mul, coalesced)
+        return (x0p.unsqueeze(3) * xkp.unsqueeze(2)).reshape(d, B, F * H)
+
+    @staticmethod
     def forward(ctx, x0, xk, W, cdt):
         B, F, d = x0.shape
         H = xk.shape[1]
-        O = W.shape[0]
-        x0c = x0.to(cdt)
-        xkc = xk.to(cdt)
-        Wt = W.to(cdt).t().contiguous()          # [F*H, O]
-        out = torch.empty(B, O, d, device=x0.device, dtype=torch.float32)
-        for dd in range(d):
-            v = (x0c[:, :, dd].unsqueeze(2)
-                 * xkc[:, :, dd].unsqueeze(1)).reshape(B, F * H)
-            out[:, :, dd] = (v @ Wt).float()
-        ctx.save_for_backward(x0c, xkc, W)
+        # d-leading contiguous copies: every downstream op is a dense
+        # d-batched GEMM with unit strides (the per-slice strided variant
+        # measured slower than the einsum path it replaced)
+        x0p = x0.permute(2, 0, 1).to(cdt).contiguous()   # [d, B, F]
+        xkp = xk.permute(2, 0, 1).to(cdt).contiguous()   # [d, B, H]
+        Wt = W.to(cdt).t().contiguous()                  # [F*H, O]
+        v = _CINLayerFn._vmat(x0p, xkp, B, F, H, d)      # [d, B, F*H]
+        out = torch.matmul(v, Wt)                        # [d, B, O]
+        ctx.save_for_backward(x0p, xkp, W)
         ctx.cdt = cdt
-        return out
+        return out.permute(1, 2, 0).float()              # [B, O, d]
 
     @staticmethod
     def backward(ctx, dout):
-        x0c, xkc, W = ctx.saved_tensors
+        x0p, xkp, W = ctx.saved_tensors
         cdt = ctx.cdt
-        B, F, d = x0c.shape
-        H = xkc.shape[1]
-        Wc = W.to(cdt)                            # [O, F*H]
-        dW = torch.zeros_like(W, dtype=torch.float32)
-        dx0 = torch.empty_like(x0c, dtype=torch.float32)
-        dxk = torch.empty_like(xkc, dtype=torch.float32)
-        for dd in range(d):
-            g = dout[:, :, dd].to(cdt)            # [B, O]
-            v = (x0c[:, :, dd].unsqueeze(2)
-                 * xkc[:, :, dd].unsqueeze(1)).reshape(B, F * H)
-            dW += (g.t() @ v).float()
-            p = (g @ Wc).view(B, F, H)            # [B, F, H]
-            dx0[:, :, dd] = torch.bmm(
-                p, xkc[:, :, dd].unsqueeze(2)).squeeze(2).float()
-            dxk[:, :, dd] = torch.bmm(
-                p.transpose(1, 2), x0c[:, :, dd].unsqueeze(2)
-            ).squeeze(2).float()
-        return dx0, dxk, dW.to(W.dtype), None
+        d, B, F = x0p.shape
+        H = xkp.shape[2]
+        Wc = W.to(cdt)                                    # [O, F*H]
+        g = dout.permute(2, 0, 1).to(cdt).contiguous()    # [d, B, O]
+        v = _CINLayerFn._vmat(x0p, xkp, B, F, H, d)       # rebuilt, not saved
+        # dW = sum_d g_d^T @ v_d  (d-batched GEMM, fp32 accumulate)
+        dW = torch.matmul(g.transpose(1, 2), v).float().sum(0)
+        p = torch.matmul(g, Wc).view(d, B, F, H)          # [d, B, F, H]
+        dx0 = (p * xkp.unsqueeze(2)).sum(3)               # [d, B, F]
+        dxk = (p * x0p.unsqueeze(3)).sum(2)               # [d, B, H]
+        return (dx0.permute(1, 2, 0).float(),
+                dxk.permute(1, 2, 0).float(), dW.to(W.dtype), None)
 
 
 class CIN(nn.Module):
