@@ -37,6 +37,10 @@ def test_serve_from_gpu(tmp_path):
     c.create_model(uri)
     var = c.manager.find_model_variable(sign, 0)
     assert var.shard.device.type == "cuda"
+    # the GPU read path must be the HIP-kernel table, not the CPU dict
+    # fallback (round-1 verdict: serving pulls bypassed every kernel)
+    from openembedding_amd.core.variable_gpu import HipVariableShard
+    assert isinstance(var.shard, HipVariableShard)
     got = var.pull_weights(probe)
     assert torch.allclose(got, live, atol=1e-6)
 
