@@ -663,11 +663,30 @@ def load_server_model(path: str):
     load_model(get_context(), path)
 
 
-def save_as_original_model(model: nn.Module, path: str):
-    """Export a fully materialized, PS-free copy of the model: every PS
-    Embedding is replaced by a plain nn.Embedding with pulled rows
-    (reference exb.py:506-547 clone-and-materialize). Saved with torch.save;
-    loadable without openembedding_amd."""
+def save_as_original_model(model: nn.Module, path: str,
+                           format: str = "saved_model"):
+    """Export a fully materialized, PS-free copy of the model (reference
+    exb.py:506-547 clone-and-materialize).
+
+    format="saved_model" (default, the north-star contract): a TensorFlow
+    SavedModel directory (saved_model.pb + variables bundle) that TF /
+    TF-Serving loads with no OpenEmbedding runtime — supported for the
+    model zoo (LR/WDL/DeepFM/xDeepFM); see export_tf.py.
+
+    format="torch": a torch.save state dict with every PS Embedding
+    materialized into a dense weight; loadable without openembedding_amd.
+    """
+    if format == "saved_model":
+        from ..export_tf import export_saved_model
+        try:
+            return export_saved_model(model, path)
+        except (NotImplementedError, AttributeError) as e:
+            raise NotImplementedError(
+                f"SavedModel export covers the model zoo "
+                f"(LR/WDL/DeepFM/xDeepFM); for custom modules use "
+                f"save_as_original_model(..., format='torch') ({e})")
+    if format != "torch":
+        raise ValueError(f"unknown export format {format!r}")
     state = {}
     for name, mod in model.named_modules():
         if isinstance(mod, Embedding) and not mod.sparse_as_dense:

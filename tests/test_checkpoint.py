@@ -92,7 +92,8 @@ def test_save_as_original_model(tmp_path):
     out = e(torch.tensor([1, 2]))
     out.sum().backward()
     embed.get_context().update_all_weights()
-    embed.save_as_original_model(model, str(tmp_path / "orig.pt"))
+    embed.save_as_original_model(model, str(tmp_path / "orig.pt"),
+                                 format="torch")
     blob = torch.load(tmp_path / "orig.pt", weights_only=True)
     assert blob["format"] == "openembedding_amd.original"
     w = blob["state_dict"]["emb.weight"]

@@ -64,9 +64,15 @@ def test_full_lifecycle(tmp_path):
     # resume: training continues from the restored state
     _train(model, opt, 3, seed=2)
 
-    # PS-free export loads WITHOUT openembedding_amd machinery
+    # PS-free exports load WITHOUT openembedding_amd machinery:
+    # default = TF SavedModel directory (north-star format) ...
+    sm = str(tmp_path / "standalone_sm")
+    embed.save_as_original_model(model, sm)
+    assert os.path.exists(os.path.join(sm, "saved_model.pb"))
+    assert os.path.exists(os.path.join(sm, "variables", "variables.index"))
+    # ... and the torch materialization explicitly
     export = str(tmp_path / "standalone.pt")
-    embed.save_as_original_model(model, export)
+    embed.save_as_original_model(model, export, format="torch")
     blob = torch.load(export, map_location="cpu", weights_only=True)
     assert any("embedding" in k for k in blob["state_dict"])
 
