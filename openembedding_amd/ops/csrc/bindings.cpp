@@ -70,6 +70,10 @@ void emb_scatter_out(const float*, const i64*, long, const int*, long, float*,
                      hipStream_t_);
 void emb_split_payload(const float*, long, long, const int*, float*, u64*,
                        hipStream_t_);
+void emb_flat_step_scalars(float*, float, float, hipStream_t_);
+void emb_flat_opt(int, void*, float*, float*, float*, const void*,
+                  const float*, long, int, float, float, float, float,
+                  hipStream_t_);
 void emb_flat_adagrad_f32(float*, float*, const float*, long, float, float,
                           hipStream_t_);
 void emb_flat_adagrad_bf16(float*, float*, const void*, void*, long, float,
@@ -480,6 +484,38 @@ void flat_adagrad(torch::Tensor param, torch::Tensor accum,
     }
 }
 
+void flat_step_scalars(torch::Tensor sc, double b1, double b2) {
+    CHECK_GPU(sc); CHECK_CONT(sc);
+    TORCH_CHECK(sc.numel() >= 3 && sc.dtype() == torch::kFloat32,
+                "step scalars must be float32 [3]");
+    const c10::cuda::CUDAGuard guard(sc.device());
+    emb_flat_step_scalars(sc.data_ptr<float>(), (float)b1, (float)b2,
+                          cur_stream());
+}
+
+void flat_opt(int64_t opt, torch::Tensor param, OptTensor master,
+              OptTensor s1, OptTensor s2, torch::Tensor grad, OptTensor sc,
+              double lr, double c0, double c1, double c2) {
+    CHECK_GPU(param); CHECK_CONT(param); CHECK_CONT(grad);
+    const c10::cuda::CUDAGuard guard(param.device());
+    long n = param.numel();
+    bool bf16 = param.dtype() == torch::kBFloat16;
+    float* mp = nullptr;
+    if (bf16) {
+        TORCH_CHECK(master.has_value(), "bf16 flat_opt needs fp32 master");
+        mp = master->data_ptr<float>();
+    } else {
+        TORCH_CHECK(param.dtype() == torch::kFloat32, "flat_opt dtype");
+    }
+    emb_flat_opt((int)opt, param.data_ptr(), mp,
+                 s1.has_value() ? s1->data_ptr<float>() : nullptr,
+                 s2.has_value() ? s2->data_ptr<float>() : nullptr,
+                 grad.data_ptr(),
+                 sc.has_value() ? sc->data_ptr<float>() : nullptr,
+                 n, bf16 ? 1 : 0, (float)lr, (float)c0, (float)c1,
+                 (float)c2, cur_stream());
+}
+
 // ---- fused MLP bias grads ---------------------------------------------
 
 void mlp3_bias_bwd(torch::Tensor dout, torch::Tensor dz1, torch::Tensor dz2,
@@ -639,4 +675,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bce_bwd", &bce_bwd, "fused BCE-with-logits backward");
     m.def("flat_adagrad", &flat_adagrad,
           "fused flat-buffer Adagrad (f32, or bf16 weights + f32 master)");
+    m.def("flat_opt", &flat_opt,
+          "fused flat-buffer dense optimizer (0=sgd, 1=adagrad, 2=adam)");
+    m.def("flat_step_scalars", &flat_step_scalars,
+          "device-side step counter + Adam bias-correction factors "
+          "(capture-safe)");
 }

@@ -750,6 +750,78 @@ __global__ void k_flat_adagrad_bf16(float* __restrict__ master,
     p[i] = (oebf16)m;
 }
 
+// Generic flat dense optimizers (sgd / adagrad / adam), f32 params or
+// bf16 params + fp32 master. Semantics match torch.optim defaults
+// (dampening 0, no weight decay / amsgrad). Adam's bias correction reads
+// DEVICE scalars prepared by k_flat_step_scalars so a hipGraph-captured
+// step keeps correcting (a host-side 1-beta^t would freeze at capture).
+
+enum { FLAT_SGD = 0, FLAT_ADAGRAD = 1, FLAT_ADAM = 2 };
+
+__global__ void k_flat_step_scalars(float* __restrict__ sc, float b1,
+                                    float b2) {
+    if (threadIdx.x || blockIdx.x) return;
+    float t = sc[0] + 1.0f;
+    sc[0] = t;
+    sc[1] = 1.0f - powf(b1, t);
+    sc[2] = 1.0f - powf(b2, t);
+}
+
+template <int OPT>
+__device__ __forceinline__ float flat_opt_step(
+        float w, float gi, float* __restrict__ s1, float* __restrict__ s2,
+        long i, const float* __restrict__ sc, float lr, float c0, float c1,
+        float c2) {
+    if (OPT == FLAT_ADAGRAD) {
+        float a = s1[i] + gi * gi;
+        s1[i] = a;
+        return w - lr * gi / (sqrtf(a) + c0);
+    }
+    if (OPT == FLAT_SGD) {
+        if (c0 == 0.0f) return w - lr * gi;     // plain
+        float b = c0 * s1[i] + gi;              // momentum, dampening 0
+        s1[i] = b;
+        float d = (c1 != 0.0f) ? gi + c0 * b : b;   // c1 = nesterov
+        return w - lr * d;
+    }
+    // FLAT_ADAM: c0=beta1, c1=beta2, c2=eps; sc[1]=1-b1^t, sc[2]=1-b2^t
+    float m = c0 * s1[i] + (1.0f - c0) * gi;
+    float v = c1 * s2[i] + (1.0f - c1) * gi * gi;
+    s1[i] = m;
+    s2[i] = v;
+    float mh = m / sc[1];
+    float vh = v / sc[2];
+    return w - lr * mh / (sqrtf(vh) + c2);
+}
+
+template <int OPT>
+__global__ void k_flat_opt_f32(float* __restrict__ p,
+                               float* __restrict__ s1,
+                               float* __restrict__ s2,
+                               const float* __restrict__ g,
+                               const float* __restrict__ sc, long n,
+                               float lr, float c0, float c1, float c2) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    p[i] = flat_opt_step<OPT>(p[i], g[i], s1, s2, i, sc, lr, c0, c1, c2);
+}
+
+template <int OPT>
+__global__ void k_flat_opt_bf16(float* __restrict__ master,
+                                float* __restrict__ s1,
+                                float* __restrict__ s2,
+                                const oebf16* __restrict__ g,
+                                oebf16* __restrict__ p,
+                                const float* __restrict__ sc, long n,
+                                float lr, float c0, float c1, float c2) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    float m = flat_opt_step<OPT>(master[i], (float)g[i], s1, s2, i, sc, lr,
+                                 c0, c1, c2);
+    master[i] = m;
+    p[i] = (oebf16)m;
+}
+
 // ---- fused BCE-with-logits (mean) --------------------------------------
 // torch's BCEWithLogitsLoss costs ~5 launches per step in the captured
 // train graph (log_sigmoid, mean reduce, grad fill, sigmoid-sub-scale
@@ -1230,6 +1302,34 @@ void emb_flat_adagrad_bf16(float* master, float* accum, const void* g,
                            hipStream_t stream) {
     if (n) k_flat_adagrad_bf16<<<grid1d(n), BLOCK, 0, stream>>>(
         master, accum, (const oebf16*)g, (oebf16*)p, n, lr, eps);
+}
+
+void emb_flat_step_scalars(float* sc, float b1, float b2,
+                           hipStream_t stream) {
+    k_flat_step_scalars<<<1, 64, 0, stream>>>(sc, b1, b2);
+}
+
+#define LAUNCH_FLAT(OPT)                                                    \
+    do {                                                                    \
+        if (bf16)                                                           \
+            k_flat_opt_bf16<OPT><<<grid1d(n), BLOCK, 0, stream>>>(          \
+                master, s1, s2, (const oebf16*)g, (oebf16*)p, sc, n, lr,    \
+                c0, c1, c2);                                                \
+        else                                                                \
+            k_flat_opt_f32<OPT><<<grid1d(n), BLOCK, 0, stream>>>(          \
+                (float*)p, s1, s2, (const float*)g, sc, n, lr, c0, c1, c2); \
+    } while (0)
+
+void emb_flat_opt(int opt, void* p, float* master, float* s1, float* s2,
+                  const void* g, const float* sc, long n, int bf16,
+                  float lr, float c0, float c1, float c2,
+                  hipStream_t stream) {
+    if (!n) return;
+    switch (opt) {
+        case FLAT_SGD: LAUNCH_FLAT(FLAT_SGD); break;
+        case FLAT_ADAGRAD: LAUNCH_FLAT(FLAT_ADAGRAD); break;
+        case FLAT_ADAM: LAUNCH_FLAT(FLAT_ADAM); break;
+    }
 }
 
 void emb_bce_fwd(const float* z, const float* y, long n, float* loss,

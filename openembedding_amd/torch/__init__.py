@@ -328,23 +328,34 @@ def _server_cfg_from_torch(opt: torch.optim.Optimizer) -> Dict:
     raise AssertionError(cat)
 
 
-class _FlatDenseAdagrad:
+class _FlatDenseOptimizer:
     """Dense params re-based onto ONE flat buffer: zero_grad is one fill,
     the distributed allreduce is one collective on one tensor, and the
-    Adagrad step is 3 flat kernels (torch's multi-tensor apply was ~60us +
-    ~60us of per-param grad fills per step on the DeepFM profile).
-    Numerically identical to torch.optim.Adagrad with default lr_decay=0,
-    weight_decay=0."""
+    optimizer step is a handful of flat kernels (torch's multi-tensor
+    apply was ~60us + ~60us of per-param grad fills per step on the DeepFM
+    profile). Supports plain Adagrad, SGD (momentum/nesterov) and Adam —
+    numerically identical to the torch.optim defaults (no weight decay /
+    lr_decay / amsgrad); other optimizers keep the non-flat path."""
+
+    _FLAT_IDS = {"SGD": 0, "Adagrad": 1, "Adam": 2}
 
     def __init__(self, optimizer: torch.optim.Optimizer):
         g = optimizer.param_groups[0]
-        if (type(optimizer).__name__ != "Adagrad"
+        name = type(optimizer).__name__
+        if (name not in self._FLAT_IDS
                 or len(optimizer.param_groups) != 1
-                or g.get("lr_decay", 0) or g.get("weight_decay", 0)):
-            raise ValueError("flat dense path supports plain Adagrad "
-                             "(single group, no lr_decay/weight_decay)")
+                or g.get("lr_decay", 0) or g.get("weight_decay", 0)
+                or g.get("dampening", 0) or g.get("amsgrad", False)):
+            raise ValueError(
+                "flat dense path supports plain Adagrad/SGD/Adam (single "
+                "group, no weight_decay/lr_decay/dampening/amsgrad)")
+        self.kind = name
+        self.opt_id = self._FLAT_IDS[name]
         self.lr = g["lr"]
-        self.eps = g.get("eps", 1e-10)
+        self.eps = g.get("eps", 1e-10 if name == "Adagrad" else 1e-8)
+        self.momentum = float(g.get("momentum", 0.0))
+        self.nesterov = bool(g.get("nesterov", False))
+        self.betas = g.get("betas", (0.9, 0.999))
         init_acc = g.get("initial_accumulator_value", 0.0)
         params = [p for p in g["params"]
                   if p.requires_grad and p.numel() > 0]
@@ -375,12 +386,30 @@ class _FlatDenseAdagrad:
                 "dtype": dt,
                 "flat": flat,
                 "flat_grad": flat_grad,
-                "accum": torch.full((total,), float(init_acc),
-                                    device=dev, dtype=torch.float32),
                 "master": (flat.to(torch.float32)
                            if dt == torch.bfloat16 else None),
+                "s1": None,
+                "s2": None,
             }
+            if self.kind == "Adagrad":
+                grp["s1"] = torch.full((total,), float(init_acc),
+                                       device=dev, dtype=torch.float32)
+            elif self.kind == "SGD" and self.momentum:
+                grp["s1"] = torch.zeros(total, device=dev,
+                                        dtype=torch.float32)
+            elif self.kind == "Adam":
+                grp["s1"] = torch.zeros(total, device=dev,
+                                        dtype=torch.float32)
+                grp["s2"] = torch.zeros(total, device=dev,
+                                        dtype=torch.float32)
+            grp["accum"] = grp["s1"]   # legacy alias (round-1 state dicts)
             self.groups.append(grp)
+        # device step counter + Adam bias-correction factors: computed ON
+        # DEVICE so a hipGraph-captured step keeps advancing t
+        self.step_scalars = (torch.zeros(3, device=dev,
+                                         dtype=torch.float32)
+                             if self.kind == "Adam" else None)
+        self._t = 0      # host mirror for the torch fallback
         self.params = params
         # p -> its view into the flat grad buffer (re-bound in zero_grad if
         # model.zero_grad(set_to_none=True) detached it; checked in step)
@@ -413,34 +442,70 @@ class _FlatDenseAdagrad:
 
     def step(self):
         self._check_bound()
-        for g in self.groups:
-            if self._ext is not None:
-                self._ext.flat_adagrad(g["flat"], g["accum"], g["flat_grad"],
-                                       g["master"], self.lr, self.eps)
-                continue
-            grad = g["flat_grad"].to(torch.float32)
-            g["accum"].addcmul_(grad, grad)
-            std = g["accum"].sqrt().add_(self.eps)
-            if g["master"] is not None:
-                g["master"].addcdiv_(grad, std, value=-self.lr)
-                g["flat"].copy_(g["master"])
+        self._t += 1
+        if self._ext is not None:
+            if self.step_scalars is not None:
+                self._ext.flat_step_scalars(self.step_scalars,
+                                            self.betas[0], self.betas[1])
+            if self.kind == "SGD":
+                cfg = (self.momentum, 1.0 if self.nesterov else 0.0, 0.0)
+            elif self.kind == "Adagrad":
+                cfg = (self.eps, 0.0, 0.0)
             else:
-                g["flat"].addcdiv_(grad, std, value=-self.lr)
+                cfg = (self.betas[0], self.betas[1], self.eps)
+            for g in self.groups:
+                self._ext.flat_opt(self.opt_id, g["flat"], g["master"],
+                                   g["s1"], g["s2"], g["flat_grad"],
+                                   self.step_scalars, self.lr, *cfg)
+            return
+        for g in self.groups:
+            grad = g["flat_grad"].to(torch.float32)
+            w = g["master"] if g["master"] is not None else g["flat"]
+            if self.kind == "Adagrad":
+                g["s1"].addcmul_(grad, grad)
+                std = g["s1"].sqrt().add_(self.eps)
+                w.addcdiv_(grad, std, value=-self.lr)
+            elif self.kind == "SGD":
+                if self.momentum:
+                    g["s1"].mul_(self.momentum).add_(grad)
+                    d = (grad + self.momentum * g["s1"]
+                         if self.nesterov else g["s1"])
+                else:
+                    d = grad
+                w.add_(d, alpha=-self.lr)
+            else:  # Adam
+                b1, b2 = self.betas
+                g["s1"].mul_(b1).add_(grad, alpha=1 - b1)
+                g["s2"].mul_(b2).addcmul_(grad, grad, value=1 - b2)
+                mh = g["s1"] / (1 - b1 ** self._t)
+                vh = g["s2"] / (1 - b2 ** self._t)
+                w.addcdiv_(mh, vh.sqrt().add_(self.eps), value=-self.lr)
+            if g["master"] is not None:
+                g["flat"].copy_(g["master"])
 
     def state_dict(self):
         # detached clones: a caller that keeps training after save must not
         # mutate the saved state through aliased tensors
         return {"groups": [{k: (v.detach().clone()
                                 if torch.is_tensor(v) else v)
-                            for k, v in g.items() if k != "flat_grad"}
-                           for g in self.groups], "lr": self.lr}
+                            for k, v in g.items()
+                            if k not in ("flat_grad", "accum")}
+                           for g in self.groups],
+                "lr": self.lr, "kind": self.kind, "t": self._t,
+                "step_scalars": (self.step_scalars.detach().clone()
+                                 if self.step_scalars is not None
+                                 else None)}
 
     def load_state_dict(self, sd):
         for g, s in zip(self.groups, sd["groups"]):
             g["flat"].copy_(s["flat"])
-            g["accum"].copy_(s["accum"])
-            if g["master"] is not None and s.get("master") is not None:
-                g["master"].copy_(s["master"])
+            for k in ("s1", "s2", "master"):
+                src = s.get(k, s.get("accum") if k == "s1" else None)
+                if g.get(k) is not None and src is not None:
+                    g[k].copy_(src)
+        self._t = int(sd.get("t", 0))
+        if self.step_scalars is not None and sd.get("step_scalars") is not None:
+            self.step_scalars.copy_(sd["step_scalars"])
 
 
 class DistributedOptimizer:
@@ -462,9 +527,9 @@ class DistributedOptimizer:
         self._sparse_config = sparse_config
         self._configured_vars = set()
         self.ctx = get_context()
-        self._flat: Optional[_FlatDenseAdagrad] = None
+        self._flat: Optional[_FlatDenseOptimizer] = None
         if flatten_dense:
-            self._flat = _FlatDenseAdagrad(optimizer)
+            self._flat = _FlatDenseOptimizer(optimizer)
 
     # behave like the wrapped optimizer
     def __getattr__(self, name):

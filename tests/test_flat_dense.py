@@ -8,12 +8,15 @@ import openembedding_amd.torch as embed
 from openembedding_amd.models import DeepFM, synthetic_batch
 
 
-def _train(flatten, n=5):
+def _train(flatten, n=5, opt_ctor=None):
     torch.manual_seed(3)
     model = DeepFM(dim=4)
+    if opt_ctor is None:
+        opt_ctor = lambda ps: torch.optim.Adagrad(ps, lr=0.01)  # noqa: E731
     opt = embed.distributed_optimizer(
-        torch.optim.Adagrad(model.parameters(), lr=0.01),
-        flatten_dense=flatten)
+        opt_ctor(model.parameters()),
+        flatten_dense=flatten,
+        sparse_config=dict(category="adagrad", learning_rate=0.01))
     lossf = torch.nn.BCEWithLogitsLoss()
     gen = torch.Generator().manual_seed(9)
     losses = []
@@ -55,6 +58,29 @@ def test_flat_state_roundtrip():
     blob = opt.state_dict()
     assert "flat_dense" in blob
     opt.load_state_dict(blob)  # no-throw, idempotent
+
+
+@pytest.mark.parametrize("ctor", [
+    lambda ps: torch.optim.SGD(ps, lr=0.05),
+    lambda ps: torch.optim.SGD(ps, lr=0.05, momentum=0.9),
+    lambda ps: torch.optim.SGD(ps, lr=0.05, momentum=0.9, nesterov=True),
+    lambda ps: torch.optim.Adam(ps, lr=0.003),
+], ids=["sgd", "sgd-mom", "sgd-nesterov", "adam"])
+def test_flat_matches_plain_sgd_adam(ctor):
+    """round-2: the flat path must cover SGD/Adam too, identically to the
+    per-param torch step (the round-1 Adagrad-only gate made every other
+    optimizer fall into the slow cat/allreduce path)."""
+    _reset()
+    plain_losses, plain_sd, _ = _train(False, opt_ctor=ctor)
+    _reset()
+    flat_losses, flat_sd, _ = _train(True, opt_ctor=ctor)
+    for a, b in zip(plain_losses, flat_losses):
+        assert a == pytest.approx(b, rel=1e-5), (plain_losses, flat_losses)
+    for k in plain_sd:
+        if k.endswith("grad_hook"):
+            continue
+        assert torch.allclose(plain_sd[k], flat_sd[k], atol=1e-6), k
+    _reset()
 
 
 def test_flat_rejects_unsupported():

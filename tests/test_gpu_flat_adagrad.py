@@ -68,3 +68,48 @@ def test_native_bf16_deepfm_trains():
     assert all(torch.isfinite(torch.tensor(losses)))
     # training moves the loss (random labels -> toward ~0.56 entropy region)
     assert losses[-1] < losses[0]
+
+
+@pytest.mark.parametrize("kind", ["sgd", "sgd-mom", "sgd-nesterov", "adam"])
+def test_flat_opt_kernels_match_torch(kind):
+    """round-2 generic flat kernels (flat_opt): SGD/momentum/nesterov and
+    Adam vs per-tensor torch.optim on cuda, f32 and bf16+master forms."""
+    from openembedding_amd.ops import require_hip
+    ext = require_hip()
+    torch.manual_seed(2)
+    n, lr = 4096, 0.02
+    opt_id = {"sgd": 0, "sgd-mom": 0, "sgd-nesterov": 0, "adam": 2}[kind]
+    mom = 0.9 if "mom" in kind or "nesterov" in kind else 0.0
+    nest = 1.0 if "nesterov" in kind else 0.0
+    b1, b2, eps = 0.9, 0.999, 1e-8
+
+    for bf16 in (False, True):
+        master = torch.randn(n, device=DEV)
+        p = master.to(torch.bfloat16) if bf16 else master.clone()
+        ref = torch.nn.Parameter(master.clone())
+        if opt_id == 0:
+            topt = torch.optim.SGD([ref], lr=lr, momentum=mom,
+                                   nesterov=bool(nest))
+            cfg = (mom, nest, 0.0)
+            s1 = torch.zeros(n, device=DEV) if mom else None
+            s2, sc = None, None
+        else:
+            topt = torch.optim.Adam([ref], lr=lr, betas=(b1, b2), eps=eps)
+            cfg = (b1, b2, eps)
+            s1 = torch.zeros(n, device=DEV)
+            s2 = torch.zeros(n, device=DEV)
+            sc = torch.zeros(3, device=DEV)
+        g = torch.Generator(device=DEV).manual_seed(5)
+        for _ in range(4):
+            grad32 = torch.randn(n, device=DEV, generator=g)
+            grad = grad32.to(torch.bfloat16) if bf16 else grad32
+            if sc is not None:
+                ext.flat_step_scalars(sc, b1, b2)
+            ext.flat_opt(opt_id, p, master if bf16 else None, s1, s2,
+                         grad, sc, lr, *cfg)
+            ref.grad = grad.to(torch.float32)  # same bf16-read grad
+            topt.step()
+        w = master if bf16 else p
+        assert torch.allclose(w, ref.detach(), atol=1e-5, rtol=1e-5), kind
+        if bf16:
+            assert torch.equal(p, w.to(torch.bfloat16))
