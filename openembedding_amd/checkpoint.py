@@ -70,6 +70,11 @@ def dump_model(ctx, uri: str, include_optimizer: bool = True) -> None:
             meta["variables"].append({
                 "variable_id": var.shard.meta.variable_id,
                 "storage_ordinal": st.storage_id,
+                # reference ModelVariableMeta field (Meta.h:77-99): with it
+                # present, this JSON is a superset of the reference's
+                # ModelOfflineMeta schema (model_sign + variables[datatype,
+                # embedding_dim, vocabulary_size, storage_name] + version)
+                "storage_name": str(st.storage_id),
                 "datatype": var.shard.meta.datatype_str(),
                 "embedding_dim": var.shard.dim,
                 "vocabulary_size": var.shard.meta.vocabulary_size,

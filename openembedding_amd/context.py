@@ -136,7 +136,14 @@ class Context:
         meta = VariableMeta(variable_id=vid, embedding_dim=embedding_dim,
                             dtype=dtype, vocabulary_size=vocabulary_size)
         cache_mb = self.config.server.cache_size_mb
-        if self.device.type == "cuda":
+        if self.device.type == "cuda" and dtype == torch.float64:
+            # f64 parity (reference registers f32 AND f64 variables,
+            # EmbeddingVariable.cpp:277-279): the HIP kernels are f32-only,
+            # so f64 tables run the torch-op engine on the same cuda device
+            # — correct, slower, and documented as such
+            shard_cls = VariableShard
+            kw = {}
+        elif self.device.type == "cuda":
             if cache_mb > 0 and meta.use_hash_table:
                 from .core.tiered_gpu import HipTieredVariableShard
                 row_bytes = 4 * (embedding_dim + 64)

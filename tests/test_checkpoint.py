@@ -242,3 +242,22 @@ def test_hash_mode_roundtrip_with_state(tmp_path):
     after = v.shard.pull_readonly(keys)
     assert not torch.equal(after, before_w)
     ctx.finalize()
+
+
+def test_model_meta_reference_schema(tmp_path):
+    """model_meta is a superset of the reference's ModelOfflineMeta JSON
+    (Meta.h:104-145): model_sign + version "0.2" + variables with
+    datatype/embedding_dim/vocabulary_size/storage_name."""
+    import json
+    e = embed.Embedding(50, 4)
+    e.variable.set_optimizer("sgd", learning_rate=0.1)
+    embed.save_server_model(str(tmp_path / "m"))
+    meta = json.load(open(tmp_path / "m" / "model_meta"))
+    assert meta["version"] == "0.2"
+    assert isinstance(meta["model_sign"], str) and meta["model_sign"]
+    assert meta["variables"]
+    for v in meta["variables"]:
+        for k in ("datatype", "embedding_dim", "vocabulary_size",
+                  "storage_name"):
+            assert k in v, k
+        assert isinstance(v["storage_name"], str)

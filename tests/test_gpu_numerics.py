@@ -244,3 +244,28 @@ def test_fused_bce_matches_torch():
         (loss * 3.0).backward()
         (ref * 3.0).backward()
         assert torch.allclose(z.grad, z2.grad, atol=1e-6, rtol=1e-5)
+
+
+def test_float64_variable_on_gpu():
+    """f64 parity (reference registers f32 and f64,
+    EmbeddingVariable.cpp:277-279): f64 variables train on cuda through
+    the torch-op engine and match the CPU f64 oracle exactly."""
+    from openembedding_amd import context as ctx_mod
+
+    results = []
+    for device in ("cpu", DEV):
+        ctx_mod._context = None
+        ctx = ctx_mod.Context(device=device)
+        st = ctx.create_storage()
+        var = st.create_variable(1000, 6, dtype=torch.float64)
+        var.set_initializer("uniform", minval=-1.0, maxval=1.0)
+        var.set_optimizer("adagrad", learning_rate=0.1)
+        keys = torch.arange(0, 64, dtype=torch.int64, device=device)
+        out, h = var.pull(keys)
+        assert out.dtype == torch.float64
+        var.push(h, torch.ones_like(out))
+        st.update_weights()
+        after, _ = var.pull(keys, readonly=True)
+        results.append(after.cpu())
+    ctx_mod._context = None
+    torch.testing.assert_close(results[0], results[1])
