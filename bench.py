@@ -118,6 +118,15 @@ def measure(args, embed, amp_mode, steps, warmup, pool, device, world, rank):
             print(f"[bench] hipGraph capture unavailable ({e!r}); "
                   f"falling back to eager", flush=True)
             graph = None
+        if world > 1 and dist.is_initialized():
+            # unanimity: a rank whose capture failed must not meet captured
+            # ranks' collectives with eager ones — all fall back together
+            ok = torch.tensor([1.0 if graph is not None else 0.0],
+                              device=device if dist.get_backend() == "nccl"
+                              else "cpu")
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if float(ok.item()) == 0.0:
+                graph = None
 
     # prefetch: explicit --prefetch, or the default at world>1 when the
     # step is not graph-captured (the reference's pipeline overlap)
