@@ -313,27 +313,85 @@ class _CINLayerFn(torch.autograd.Function):
         return (x0p.unsqueeze(3) * xkp.unsqueeze(2)).reshape(d, B, F * H)
 
     @staticmethod
+    def _hip_ok(x0, W, cdt):
+        if not (x0.is_cuda and cdt == torch.bfloat16 and hip_available()):
+            return False
+        O, K = W.shape
+        F = x0.shape[1]
+        H = K // F
+        n = x0.shape[0] * x0.shape[2]    # columns; dw operands pad to 32
+        return (F <= 32 and H <= 128 and 16 <= O <= 128 and O % 16 == 0
+                and n % 32 == 0)
+
+    @staticmethod
+    def _bufs(W, F, H):
+        # padded bf16 weight mirrors, cached on the parameter and
+        # refreshed (copy_) every forward — same pattern as the fused MLP
+        O, K = W.shape
+        Kp = (K + 31) // 32 * 32
+        Op = (O + 31) // 32 * 32
+        bufs = getattr(W, "_cin_bufs", None)
+        if bufs is None or bufs["wp"].shape != (O, Kp):
+            bufs = {
+                "wp": torch.zeros(O, Kp, dtype=torch.bfloat16,
+                                  device=W.device),
+                "wt": torch.zeros(Kp, Op, dtype=torch.bfloat16,
+                                  device=W.device),
+            }
+            W._cin_bufs = bufs
+        bufs["wp"][:, :K].copy_(W)
+        bufs["wt"][:K, :O].copy_(W.t())
+        return bufs
+
+    @staticmethod
     def forward(ctx, x0, xk, W, cdt):
         B, F, d = x0.shape
         H = xk.shape[1]
-        # d-leading contiguous copies: every downstream op is a dense
-        # d-batched GEMM with unit strides (the per-slice strided variant
-        # measured slower than the einsum path it replaced)
+        hip = _CINLayerFn._hip_ok(x0, W, cdt)
+        if hip:
+            # implicit-GEMM HIP kernels (ops/csrc/cin.hip): V is built in
+            # LDS per block and never touches HBM (the torch path below
+            # materializes ~245 MB per build; rocprof showed those builds
+            # + re-reads as ~70% of the xDeepFM step)
+            from ..ops import require_hip
+            ext = require_hip()
+            x0p = x0.permute(2, 0, 1).reshape(d * B, F).contiguous()
+            xkp = xk.permute(2, 0, 1).reshape(d * B, H).contiguous()
+            bufs = _CINLayerFn._bufs(W, F, H)
+            out = ext.cin_fwd(x0p, xkp, bufs["wp"])       # [d*B, O]
+            ctx.save_for_backward(x0p, xkp, W)
+            ctx.meta = (B, F, H, d, True)
+            return out.view(d, B, -1).permute(1, 2, 0).float()
         x0p = x0.permute(2, 0, 1).to(cdt).contiguous()   # [d, B, F]
         xkp = xk.permute(2, 0, 1).to(cdt).contiguous()   # [d, B, H]
         Wt = W.to(cdt).t().contiguous()                  # [F*H, O]
         v = _CINLayerFn._vmat(x0p, xkp, B, F, H, d)      # [d, B, F*H]
         out = torch.matmul(v, Wt)                        # [d, B, O]
         ctx.save_for_backward(x0p, xkp, W)
+        ctx.meta = (B, F, H, d, False)
         ctx.cdt = cdt
         return out.permute(1, 2, 0).float()              # [B, O, d]
 
     @staticmethod
     def backward(ctx, dout):
         x0p, xkp, W = ctx.saved_tensors
+        B, F, H, d, hip = ctx.meta
+        if hip:
+            from ..ops import require_hip
+            ext = require_hip()
+            N = d * B
+            bufs = W._cin_bufs
+            doutp = dout.permute(2, 0, 1).reshape(N, -1).contiguous()
+            dx0p, dxkp = ext.cin_dx(doutp, bufs["wt"], x0p, xkp)
+            # transposed bf16 operands for the weight-grad GEMM
+            dzt = doutp.t().contiguous().to(torch.bfloat16)
+            x0t = x0p.t().contiguous().to(torch.bfloat16)
+            xkt = xkp.t().contiguous().to(torch.bfloat16)
+            dW = ext.cin_dw(dzt, x0t, xkt, W.shape[0], 8)
+            return (dx0p.view(d, B, F).permute(1, 2, 0),
+                    dxkp.view(d, B, H).permute(1, 2, 0),
+                    dW.to(W.dtype), None)
         cdt = ctx.cdt
-        d, B, F = x0p.shape
-        H = xkp.shape[2]
         Wc = W.to(cdt)                                    # [O, F*H]
         g = dout.permute(2, 0, 1).to(cdt).contiguous()    # [d, B, O]
         v = _CINLayerFn._vmat(x0p, xkp, B, F, H, d)       # rebuilt, not saved

@@ -78,6 +78,12 @@ void emb_flat_adagrad_f32(float*, float*, const float*, long, float, float,
                           hipStream_t_);
 void emb_flat_adagrad_bf16(float*, float*, const void*, void*, long, float,
                            float, hipStream_t_);
+void emb_cin_fwd(const float*, const float*, const void*, float*, long,
+                 long, long, long, long, hipStream_t_);
+void emb_cin_dw(const void*, const void*, const void*, float*, long, long,
+                long, long, long, hipStream_t_);
+void emb_cin_dx(const float*, const void*, const float*, const float*,
+                float*, float*, long, long, long, long, long, hipStream_t_);
 void emb_mlp3_fwd(const void*, long, long, const void*, const void*,
                   const void*, const void*, const void*, const void*,
                   const void*, const void*, const float*, long, long, void*,
@@ -543,6 +549,63 @@ void mlp3_bias_bwd(torch::Tensor dout, torch::Tensor dz1, torch::Tensor dz2,
                       dw4.data_ptr(), db4.data_ptr(), cur_stream());
 }
 
+// ---- CIN implicit-GEMM -------------------------------------------------
+
+torch::Tensor cin_fwd(torch::Tensor x0p, torch::Tensor xkp,
+                      torch::Tensor wp) {
+    CHECK_GPU(x0p); CHECK_CONT(x0p); CHECK_CONT(xkp); CHECK_CONT(wp);
+    TORCH_CHECK(x0p.dtype() == torch::kFloat32
+                && xkp.dtype() == torch::kFloat32
+                && wp.dtype() == torch::kBFloat16, "cin_fwd dtypes");
+    const c10::cuda::CUDAGuard guard(x0p.device());
+    long N = x0p.size(0), F = x0p.size(1), H = xkp.size(1);
+    long O = wp.size(0), Kp = wp.size(1);
+    TORCH_CHECK(F <= 32 && H <= 128 && O <= 128 && O % 16 == 0
+                && Kp % 32 == 0 && xkp.size(0) == N, "cin_fwd shapes");
+    auto out = torch::empty({N, O}, x0p.options());
+    emb_cin_fwd(x0p.data_ptr<float>(), xkp.data_ptr<float>(), wp.data_ptr(),
+                out.data_ptr<float>(), N, F, H, O, Kp, cur_stream());
+    return out;
+}
+
+torch::Tensor cin_dw(torch::Tensor dzt, torch::Tensor x0t,
+                     torch::Tensor xkt, int64_t O, int64_t n_split) {
+    CHECK_GPU(dzt); CHECK_CONT(dzt); CHECK_CONT(x0t); CHECK_CONT(xkt);
+    TORCH_CHECK(dzt.dtype() == torch::kBFloat16
+                && x0t.dtype() == torch::kBFloat16
+                && xkt.dtype() == torch::kBFloat16, "cin_dw dtypes");
+    const c10::cuda::CUDAGuard guard(dzt.device());
+    long Np = dzt.size(1), F = x0t.size(0), H = xkt.size(0);
+    TORCH_CHECK(Np % 32 == 0 && x0t.size(1) == Np && xkt.size(1) == Np
+                && H <= 128 && O <= 128, "cin_dw shapes");
+    auto dw = torch::zeros({O, F * H},
+                           dzt.options().dtype(torch::kFloat32));
+    emb_cin_dw(dzt.data_ptr(), x0t.data_ptr(), xkt.data_ptr(),
+               dw.data_ptr<float>(), Np, F, H, O, n_split, cur_stream());
+    return dw;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> cin_dx(
+        torch::Tensor doutp, torch::Tensor wt, torch::Tensor x0p,
+        torch::Tensor xkp) {
+    CHECK_GPU(doutp); CHECK_CONT(doutp); CHECK_CONT(wt); CHECK_CONT(x0p);
+    CHECK_CONT(xkp);
+    TORCH_CHECK(doutp.dtype() == torch::kFloat32
+                && wt.dtype() == torch::kBFloat16, "cin_dx dtypes");
+    const c10::cuda::CUDAGuard guard(doutp.device());
+    long N = doutp.size(0), O = doutp.size(1);
+    long F = x0p.size(1), H = xkp.size(1), Op = wt.size(1);
+    TORCH_CHECK(Op % 32 == 0 && O <= Op && F <= 32 && H <= 128
+                && wt.size(0) >= F * H, "cin_dx shapes");
+    auto dx0 = torch::empty({N, F}, x0p.options());
+    auto dxk = torch::empty({N, H}, xkp.options());
+    emb_cin_dx(doutp.data_ptr<float>(), wt.data_ptr(),
+               x0p.data_ptr<float>(), xkp.data_ptr<float>(),
+               dx0.data_ptr<float>(), dxk.data_ptr<float>(), N, F, H, O,
+               Op, cur_stream());
+    return {dx0, dxk};
+}
+
 // ---- fused BCE-with-logits --------------------------------------------
 
 torch::Tensor bce_fwd(torch::Tensor logits, torch::Tensor labels) {
@@ -671,6 +734,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused 3-hidden-layer MLP forward (bf16 MFMA, bias+ReLU fused)");
     m.def("mlp3_bias_bwd", &mlp3_bias_bwd,
           "MLP bias grads + head wgrad in one pass over the dz mirrors");
+    m.def("cin_fwd", &cin_fwd,
+          "CIN layer forward: implicit outer-product MFMA GEMM");
+    m.def("cin_dw", &cin_dw,
+          "CIN weight grad: per-field split-K MFMA with on-the-fly operand");
+    m.def("cin_dx", &cin_dx,
+          "CIN input grads: P=W^T dZ GEMM fused with the xk/x0 contractions");
     m.def("bce_fwd", &bce_fwd, "fused BCE-with-logits forward (mean)");
     m.def("bce_bwd", &bce_bwd, "fused BCE-with-logits backward");
     m.def("flat_adagrad", &flat_adagrad,

@@ -1,0 +1,291 @@
+// CIN implicit-GEMM kernels (gfx950, bf16 MFMA 16x16x32).
+//
+// One CIN layer is  out[n, o] = sum_{f,h} W[o, f*H+h] * x0[n, f] * xk[n, h]
+// over columns n = dd*B + b (d-leading layout, matching models/ctr.py's
+// _CINLayerFn). The operand V[n, k] = x0[n, k/H] * xk[n, k%H] is an outer
+// product per column — torch paths must materialize it (245 MB bf16 per
+// build at the benchmark shape; the round-1 einsum made it 490 MB fp32).
+// Here V is built on the fly in LDS per 128-column block, so the only
+// HBM traffic is x0/xk/W/out — measured profile said the V builds +
+// re-reads were ~1.2 ms of the 1.67 ms xDeepFM step.
+//
+// Fragment maps are the ones verified by mlp.hip/tests:
+//   A[m][k]: m = lane&15, k = (lane>>4)*8 + e   (A = W[o][k] row-major)
+//   B[k][n]: n = lane&15, k = (lane>>4)*8 + e   (B read from V[n][k] rows)
+//   C/D:     col = lane&15, row = (lane>>4)*4 + r
+//
+// Geometry: block = 128 columns x all O outputs; 8 waves; K streamed in
+// chunks of CIN_KC built into LDS (bf16, padded row stride breaks the
+// 16-lane fragment-read bank cycle). Requires O % 16 == 0, O <= 128,
+// F <= 32, H <= 128 (the zoo shapes; ctr.py falls back to the chunked
+// torch path otherwise).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+typedef __hip_bfloat16 cbf16;
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define CIN_COLS 128          // columns (b,dd pairs) per block
+#define CIN_KC 160            // K chunk built per LDS round (5 MFMA steps)
+#define CIN_VLD (CIN_KC + 8)  // LDS row stride: 84 dwords, 16-lane clean
+#define CIN_WAVES 8
+
+static __device__ __forceinline__ bf16x8 cin_ld_frag(const cbf16* p) {
+    return *reinterpret_cast<const bf16x8*>(__builtin_assume_aligned(p, 16));
+}
+
+extern "C" __global__ __launch_bounds__(64 * CIN_WAVES, 1)
+void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
+               const float* __restrict__ xkp,   // [N, H]
+               const cbf16* __restrict__ w,     // [O, Kp] padded bf16
+               float* __restrict__ out,         // [N, O]
+               long N, long F, long H, long O, long Kp) {
+    __shared__ cbf16 vtile[CIN_COLS * CIN_VLD];
+    __shared__ float x0c[CIN_COLS * 32];        // F <= 32
+    __shared__ float xkc[CIN_COLS * 128];       // H <= 128
+    const int wave = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const long n0 = (long)blockIdx.x * CIN_COLS;
+    const long K = F * H;
+
+    // stage this block's x0/xk columns (coalesced: F/H contiguous per col)
+    for (long i = threadIdx.x; i < CIN_COLS * F; i += blockDim.x) {
+        long c = i / F, f = i % F;
+        x0c[c * 32 + f] = (n0 + c < N) ? x0p[(n0 + c) * F + f] : 0.f;
+    }
+    for (long i = threadIdx.x; i < CIN_COLS * H; i += blockDim.x) {
+        long c = i / H, h = i % H;
+        xkc[c * 128 + h] = (n0 + c < N) ? xkp[(n0 + c) * H + h] : 0.f;
+    }
+
+    // accumulators: wave owns o rows [wave*16, wave*16+16) (O <= 128 ->
+    // one pass with 8 waves) x all 8 column subtiles of the block
+    const long o0 = (long)wave * 16;
+    f32x4 acc[8];
+    #pragma unroll
+    for (int t = 0; t < 8; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (long k0 = 0; k0 < Kp; k0 += CIN_KC) {
+        __syncthreads();
+        // build V chunk: vtile[c][kk] = x0c[c][k/H] * xkc[c][k%H]
+        for (long i = threadIdx.x; i < CIN_COLS * CIN_KC; i += blockDim.x) {
+            long c = i / CIN_KC, kk = i % CIN_KC;
+            long k = k0 + kk;
+            float v = 0.f;
+            if (k < K) v = x0c[c * 32 + k / H] * xkc[c * 128 + k % H];
+            vtile[c * CIN_VLD + kk] = (cbf16)v;
+        }
+        __syncthreads();
+        if (o0 >= O) continue;      // narrow O: spare waves still build V
+        const long kc_lim = (Kp - k0 < CIN_KC) ? (Kp - k0) : CIN_KC;
+        const long koff = (lane >> 4) * 8;
+        const cbf16* pa = w + (o0 + (lane & 15)) * Kp + k0 + koff;
+        const cbf16* pb = vtile + (lane & 15) * CIN_VLD + koff;
+        for (long ks = 0; ks < kc_lim; ks += 32) {
+            bf16x8 a = cin_ld_frag(pa + ks);
+            #pragma unroll
+            for (int t = 0; t < 8; ++t) {
+                bf16x8 b = cin_ld_frag(pb + t * 16 * CIN_VLD + ks);
+                acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a, b, acc[t], 0, 0, 0);
+            }
+        }
+    }
+
+    // epilogue: each lane holds 4 consecutive o values per subtile
+    if (o0 < O) {
+        #pragma unroll
+        for (int t = 0; t < 8; ++t) {
+            long n = n0 + t * 16 + (lane & 15);
+            if (n >= N) continue;
+            long ob = o0 + (lane >> 4) * 4;
+            float* dst = out + n * O + ob;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) dst[r] = acc[t][r];
+        }
+    }
+}
+
+extern "C" void emb_cin_fwd(const float* x0p, const float* xkp,
+                            const void* w, float* out, long N, long F,
+                            long H, long O, long Kp, hipStream_t stream) {
+    if (!N) return;
+    long grid = (N + CIN_COLS - 1) / CIN_COLS;
+    k_cin_fwd<<<(int)grid, 64 * CIN_WAVES, 0, stream>>>(
+        x0p, xkp, (const cbf16*)w, out, N, F, H, O, Kp);
+}
+
+// ------------------------------------------------------------ weight grad
+// dW[o, f*H+h] = sum_n dZ[n,o] x0[n,f] xk[n,h]. Reframed per f as a GEMM
+// over n with the B operand built on the fly: B'[n][h] = xkt[h][n] *
+// x0t[f][n]. Grid = (f, n-split); per block: all [O x H] tiles accumulate
+// over the block's n-range in registers, then one fp32 atomicAdd pass into
+// dW (split-K). Operands dZt [O, Np], xkt [H, Np], x0t [F, Np] are bf16
+// transposed copies (library transposes, ~20 MB total); Np % 32 == 0 with
+// zero tails.
+
+extern "C" __global__ __launch_bounds__(64 * CIN_WAVES, 1)
+void k_cin_dw(const cbf16* __restrict__ dzt,   // [O, Np]
+              const cbf16* __restrict__ x0t,   // [F, Np]
+              const cbf16* __restrict__ xkt,   // [H, Np]
+              float* __restrict__ dw,          // [O, F*H] fp32 accum
+              long Np, long F, long H, long O, long n_split) {
+    __shared__ cbf16 btile[128 * 40];   // [h][32 n], stride 40: bank-clean
+    const int wave = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const long f = blockIdx.x;
+    const long split = blockIdx.y;
+    const long nsz = ((Np / 32 + n_split - 1) / n_split) * 32;
+    const long nbeg = split * nsz;
+    const long nend = (nbeg + nsz < Np) ? (nbeg + nsz) : Np;
+    if (nbeg >= Np) return;
+    const long o0 = (long)wave * 16;
+
+    f32x4 acc[8];
+    #pragma unroll
+    for (int t = 0; t < 8; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    const long koff = (lane >> 4) * 8;
+    for (long n0s = nbeg; n0s < nend; n0s += 32) {
+        __syncthreads();
+        // B' chunk: [H][32] = xkt rows * x0 row f (broadcast over h)
+        for (long i = threadIdx.x; i < H * 32; i += blockDim.x) {
+            long h = i / 32, nn = i % 32;
+            float v = (float)xkt[h * Np + n0s + nn]
+                      * (float)x0t[f * Np + n0s + nn];
+            btile[h * 40 + nn] = (cbf16)v;
+        }
+        __syncthreads();
+        if (o0 >= O) continue;
+        bf16x8 a = cin_ld_frag(dzt + (o0 + (lane & 15)) * Np + n0s + koff);
+        #pragma unroll
+        for (int t = 0; t < 8; ++t) {
+            if (t * 16 >= H) break;
+            bf16x8 b = cin_ld_frag(btile + ((long)t * 16 + (lane & 15)) * 40
+                                   + koff);
+            acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
+                                                             0, 0, 0);
+        }
+    }
+    if (o0 >= O) return;
+    #pragma unroll
+    for (int t = 0; t < 8; ++t) {
+        if (t * 16 >= H) break;
+        long h = t * 16 + (lane & 15);
+        long ob = o0 + (lane >> 4) * 4;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r)
+            atomicAdd(&dw[(ob + r) * (F * H) + f * H + h], acc[t][r]);
+    }
+}
+
+extern "C" void emb_cin_dw(const void* dzt, const void* x0t, const void* xkt,
+                           float* dw, long Np, long F, long H, long O,
+                           long n_split, hipStream_t stream) {
+    if (!Np) return;
+    dim3 grid((unsigned)F, (unsigned)n_split);
+    k_cin_dw<<<grid, 64 * CIN_WAVES, 0, stream>>>(
+        (const cbf16*)dzt, (const cbf16*)x0t, (const cbf16*)xkt, dw, Np, F,
+        H, O, n_split);
+}
+
+// --------------------------------------------------------- input gradients
+// Per column n:  P[k] = sum_o W[o,k] dZ[n,o]   (GEMM over O, K-tiled)
+//   dx0[n,f] = sum_h P[f*H+h] xk[n,h]
+//   dxk[n,h] = sum_f P[f*H+h] x0[n,f]
+// P never touches HBM: each P subtile is consumed from registers into LDS
+// fp32 accumulators (atomic — lanes of one wave can collide on (c,f)).
+// Wt [K, Op] is the bf16 transposed weight copy; block = 64 columns,
+// waves round-robin the K tiles (each Wt row read once per block).
+
+#define CDX_COLS 64
+
+extern "C" __global__ __launch_bounds__(64 * CIN_WAVES, 1)
+void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
+              const cbf16* __restrict__ wt,     // [K, Op]
+              const float* __restrict__ x0p,    // [N, F]
+              const float* __restrict__ xkp,    // [N, H]
+              float* __restrict__ dx0p,         // [N, F] out
+              float* __restrict__ dxkp,         // [N, H] out
+              long N, long F, long H, long O, long Op) {
+    __shared__ cbf16 dzc[CDX_COLS * 136];       // [c][o], stride 136
+    __shared__ float x0c[CDX_COLS * 32];
+    __shared__ float xkc[CDX_COLS * 128];
+    __shared__ float dx0a[CDX_COLS * 32];
+    __shared__ float dxka[CDX_COLS * 128];
+    const int wave = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const long n0 = (long)blockIdx.x * CDX_COLS;
+    const long K = F * H;
+
+    for (long i = threadIdx.x; i < CDX_COLS * 32; i += blockDim.x)
+        dx0a[i] = 0.f;
+    for (long i = threadIdx.x; i < CDX_COLS * 128; i += blockDim.x)
+        dxka[i] = 0.f;
+    for (long i = threadIdx.x; i < CDX_COLS * F; i += blockDim.x) {
+        long c = i / F, ff = i % F;
+        x0c[c * 32 + ff] = (n0 + c < N) ? x0p[(n0 + c) * F + ff] : 0.f;
+    }
+    for (long i = threadIdx.x; i < CDX_COLS * H; i += blockDim.x) {
+        long c = i / H, h = i % H;
+        xkc[c * 128 + h] = (n0 + c < N) ? xkp[(n0 + c) * H + h] : 0.f;
+    }
+    for (long i = threadIdx.x; i < CDX_COLS * 128; i += blockDim.x) {
+        long c = i / 128, o = i % 128;
+        float v = (o < O && n0 + c < N) ? doutp[(n0 + c) * O + o] : 0.f;
+        dzc[c * 136 + o] = (cbf16)v;
+    }
+    __syncthreads();
+
+    const long koff = (lane >> 4) * 8;
+    const long ktiles = (K + 15) / 16;
+    for (long kt = wave; kt < ktiles; kt += CIN_WAVES) {
+        const long k0 = kt * 16;
+        #pragma unroll
+        for (int cs = 0; cs < CDX_COLS / 16; ++cs) {
+            f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+            const cbf16* pa = wt + (k0 + (lane & 15)) * Op + koff;
+            const cbf16* pb = dzc + ((long)cs * 16 + (lane & 15)) * 136
+                              + koff;
+            for (long oc = 0; oc < Op; oc += 32) {
+                bf16x8 a = cin_ld_frag(pa + oc);
+                bf16x8 b = cin_ld_frag(pb + oc);
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc,
+                                                              0, 0, 0);
+            }
+            // consume: lane holds P[k0 + (lane>>4)*4 + r][cs*16 + lane&15]
+            long c = (long)cs * 16 + (lane & 15);
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                long k = k0 + (lane >> 4) * 4 + r;
+                if (k >= K) break;
+                float p = acc[r];
+                long ff = k / H, h = k % H;
+                atomicAdd(&dx0a[c * 32 + ff], p * xkc[c * 128 + h]);
+                atomicAdd(&dxka[c * 128 + h], p * x0c[c * 32 + ff]);
+            }
+        }
+    }
+    __syncthreads();
+    for (long i = threadIdx.x; i < CDX_COLS * F; i += blockDim.x) {
+        long c = i / F, ff = i % F;
+        if (n0 + c < N) dx0p[(n0 + c) * F + ff] = dx0a[c * 32 + ff];
+    }
+    for (long i = threadIdx.x; i < CDX_COLS * H; i += blockDim.x) {
+        long c = i / H, h = i % H;
+        if (n0 + c < N) dxkp[(n0 + c) * H + h] = dxka[c * 128 + h];
+    }
+}
+
+extern "C" void emb_cin_dx(const float* doutp, const void* wt,
+                           const float* x0p, const float* xkp, float* dx0p,
+                           float* dxkp, long N, long F, long H, long O,
+                           long Op, hipStream_t stream) {
+    if (!N) return;
+    long grid = (N + CDX_COLS - 1) / CDX_COLS;
+    k_cin_dx<<<(int)grid, 64 * CIN_WAVES, 0, stream>>>(
+        doutp, (const cbf16*)wt, x0p, xkp, dx0p, dxkp, N, F, H, O, Op);
+}

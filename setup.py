@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "openembedding_amd/ops/csrc/embops.hip",
         "openembedding_amd/ops/csrc/ctrhead.hip",
         "openembedding_amd/ops/csrc/mlp.hip",
+        "openembedding_amd/ops/csrc/cin.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3"],
