@@ -44,8 +44,8 @@ def reduce_by_inverse(inverse: torch.Tensor, grads: torch.Tensor, u: int
 
     inverse int64 [n], grads [n, dim] -> (ugrads [u, dim], counts int64 [u]).
     """
-    if _use_hip(grads):
-        ext = require_hip()
+    if _use_hip(grads) and grads.dtype == torch.float32:
+        ext = require_hip()   # f64 variables use the torch path below
         if ext is not None:
             return ext.reduce_by_inverse(inverse, grads, u)
     ugrads = torch.zeros((u, grads.shape[1]), dtype=grads.dtype,

@@ -80,7 +80,9 @@ class _PullPushFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_out: torch.Tensor):
-        ctx.var.push(ctx.handle, grad_out.contiguous().to(torch.float32))
+        # cast to the VARIABLE's dtype (f32 normally; f64 tables keep f64)
+        ctx.var.push(ctx.handle,
+                     grad_out.contiguous().to(ctx.var.shard.dtype))
         return torch.zeros(0, device=grad_out.device), None, None, None
 
 
