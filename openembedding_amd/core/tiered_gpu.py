@@ -72,6 +72,7 @@ class HipTieredVariableShard(HipVariableShard):
         self._faulted = torch.zeros(1, dtype=torch.int32, device=dev)
         self._evict_epoch = 0
         self._last_batch_upper = 0
+        self._evict_streak = 0      # consecutive commits that evicted
         self._checkpoint_work_id = None
         self._cache_full_since_ckpt = False
 
@@ -183,6 +184,8 @@ class HipTieredVariableShard(HipVariableShard):
         # just applied, so no live handle's slots are invalidated
         if self._nrows_upper + self._last_batch_upper > self.cache_rows:
             self._evict()
+        else:
+            self._evict_streak = 0
 
     # ------------------------------------------------------------- eviction
 
@@ -220,7 +223,13 @@ class HipTieredVariableShard(HipVariableShard):
         margin = self._last_batch_upper
         if n + margin <= self.cache_rows:
             return
-        keep_target = max(1, (3 * self.cache_rows) // 4 - margin)
+        # steady thrash (working set >> cache: an eviction every commit):
+        # evict deeper so the fixed rebuild cost amortizes over more
+        # admitted rows (reference PmemEmbeddingTable evicts per-row and
+        # has no rebuild; an epoch rebuild must batch harder instead)
+        self._evict_streak += 1
+        keep_frac = 4 if self._evict_streak >= 4 else (4 * 3)
+        keep_target = max(1, (self.cache_rows * keep_frac) // 16 - margin)
         n_evict = n - keep_target
         if n_evict <= 0:
             return

@@ -84,14 +84,19 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
         const long koff = (lane >> 4) * 8;
         const cbf16* pa = w + (o0 + (lane & 15)) * Kp + k0 + koff;
         const cbf16* pb = vtile + (lane & 15) * CIN_VLD + koff;
+        // W-stream double buffer: the next k-step's global load is in
+        // flight while this step's 8 MFMAs consume the LDS fragments
+        bf16x8 a0 = cin_ld_frag(pa);
         for (long ks = 0; ks < kc_lim; ks += 32) {
-            bf16x8 a = cin_ld_frag(pa + ks);
+            bf16x8 a1;
+            if (ks + 32 < kc_lim) a1 = cin_ld_frag(pa + ks + 32);
             #pragma unroll
             for (int t = 0; t < 8; ++t) {
                 bf16x8 b = cin_ld_frag(pb + t * 16 * CIN_VLD + ks);
                 acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a, b, acc[t], 0, 0, 0);
+                    a0, b, acc[t], 0, 0, 0);
             }
+            a0 = a1;
         }
     }
 
@@ -149,13 +154,16 @@ void k_cin_dw(const cbf16* __restrict__ dzt,   // [O, Np]
     for (int t = 0; t < 8; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
     const long koff = (lane >> 4) * 8;
+    const long Hp16 = (H + 15) / 16 * 16;   // partial h-tile rows zeroed
     for (long n0s = nbeg; n0s < nend; n0s += 32) {
         __syncthreads();
         // B' chunk: [H][32] = xkt rows * x0 row f (broadcast over h)
-        for (long i = threadIdx.x; i < H * 32; i += blockDim.x) {
+        for (long i = threadIdx.x; i < Hp16 * 32; i += blockDim.x) {
             long h = i / 32, nn = i % 32;
-            float v = (float)xkt[h * Np + n0s + nn]
-                      * (float)x0t[f * Np + n0s + nn];
+            float v = 0.f;
+            if (h < H)
+                v = (float)xkt[h * Np + n0s + nn]
+                    * (float)x0t[f * Np + n0s + nn];
             btile[h * 40 + nn] = (cbf16)v;
         }
         __syncthreads();
@@ -173,8 +181,8 @@ void k_cin_dw(const cbf16* __restrict__ dzt,   // [O, Np]
     if (o0 >= O) return;
     #pragma unroll
     for (int t = 0; t < 8; ++t) {
-        if (t * 16 >= H) break;
         long h = t * 16 + (lane & 15);
+        if (h >= H) continue;   // partial tile: never cross into f+1's h=0
         long ob = o0 + (lane >> 4) * 4;
         #pragma unroll
         for (int r = 0; r < 4; ++r)
@@ -250,11 +258,24 @@ void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
             const cbf16* pa = wt + (k0 + (lane & 15)) * Op + koff;
             const cbf16* pb = dzc + ((long)cs * 16 + (lane & 15)) * 136
                               + koff;
-            for (long oc = 0; oc < Op; oc += 32) {
-                bf16x8 a = cin_ld_frag(pa + oc);
-                bf16x8 b = cin_ld_frag(pb + oc);
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc,
-                                                              0, 0, 0);
+            if (Op == 128) {      // all Wt loads in flight before the MFMAs
+                bf16x8 a[4], b[4];
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    a[j] = cin_ld_frag(pa + 32 * j);
+                    b[j] = cin_ld_frag(pb + 32 * j);
+                }
+                #pragma unroll
+                for (int j = 0; j < 4; ++j)
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a[j], b[j], acc, 0, 0, 0);
+            } else {
+                for (long oc = 0; oc < Op; oc += 32) {
+                    bf16x8 a = cin_ld_frag(pa + oc);
+                    bf16x8 b = cin_ld_frag(pb + oc);
+                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a, b, acc, 0, 0, 0);
+                }
             }
             // consume: lane holds P[k0 + (lane>>4)*4 + r][cs*16 + lane&15]
             long c = (long)cs * 16 + (lane & 15);
