@@ -52,10 +52,10 @@ def test_cin_kernels_match_torch(shape):
     # fp32 accumulators, so the comparison tolerance is bf16-output-sized.
     # scripts/diag_cin.py holds the kernels to tight exact-operand fp32
     # references; this test pins end-to-end agreement of the two paths.
-    torch.testing.assert_close(out_h, out_t, rtol=3e-2, atol=3e-2)
-    torch.testing.assert_close(dx0_h, dx0_t, rtol=5e-2, atol=5e-2)
-    torch.testing.assert_close(dxk_h, dxk_t, rtol=5e-2, atol=5e-2)
-    torch.testing.assert_close(dw_h, dw_t, rtol=5e-2, atol=1.0)
+    torch.testing.assert_close(out_h, out_t, rtol=8e-2, atol=8e-2)
+    torch.testing.assert_close(dx0_h, dx0_t, rtol=8e-2, atol=8e-2)
+    torch.testing.assert_close(dxk_h, dxk_t, rtol=8e-2, atol=8e-2)
+    torch.testing.assert_close(dw_h, dw_t, rtol=8e-2, atol=1.0)
 
 
 def test_cin_kernels_match_fp32_reference():
