@@ -46,18 +46,28 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
     __shared__ cbf16 vtile[CIN_COLS * CIN_VLD];
     __shared__ float x0c[CIN_COLS * 32];        // F <= 32
     __shared__ float xkc[CIN_COLS * 128];       // H <= 128
+    __shared__ short ftab[CIN_KC], htab[CIN_KC];  // k -> (f, h) per chunk
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
     const long n0 = (long)blockIdx.x * CIN_COLS;
-    const long K = F * H;
+    const int K = (int)(F * H);
+    const int Fi = (int)F, Hi = (int)H;
+
+    // NOTE all per-element index math below is INT32: 64-bit div/mod is a
+    // multi-thousand-cycle emulation on CDNA and was the original
+    // bottleneck of this kernel (~270 us/call before, profile r2g)
 
     // stage this block's x0/xk columns (coalesced: F/H contiguous per col)
-    for (long i = threadIdx.x; i < CIN_COLS * F; i += blockDim.x) {
-        long c = i / F, f = i % F;
+    for (int base = 0; base < CIN_COLS * Fi; base += (int)blockDim.x) {
+        int i = base + (int)threadIdx.x;
+        if (i >= CIN_COLS * Fi) break;
+        int c = i / Fi, f = i - c * Fi;
         x0c[c * 32 + f] = (n0 + c < N) ? x0p[(n0 + c) * F + f] : 0.f;
     }
-    for (long i = threadIdx.x; i < CIN_COLS * H; i += blockDim.x) {
-        long c = i / H, h = i % H;
+    for (int base = 0; base < CIN_COLS * Hi; base += (int)blockDim.x) {
+        int i = base + (int)threadIdx.x;
+        if (i >= CIN_COLS * Hi) break;
+        int c = i / Hi, h = i - c * Hi;
         xkc[c * 128 + h] = (n0 + c < N) ? xkp[(n0 + c) * H + h] : 0.f;
     }
 
@@ -70,12 +80,26 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
 
     for (long k0 = 0; k0 < Kp; k0 += CIN_KC) {
         __syncthreads();
-        // build V chunk: vtile[c][kk] = x0c[c][k/H] * xkc[c][k%H]
-        for (long i = threadIdx.x; i < CIN_COLS * CIN_KC; i += blockDim.x) {
-            long c = i / CIN_KC, kk = i % CIN_KC;
-            long k = k0 + kk;
+        // k -> (f, h) lookup for this chunk: one int32 div per entry,
+        // then the build loop is pure LDS reads + one multiply
+        if (threadIdx.x < CIN_KC) {
+            int k = (int)k0 + (int)threadIdx.x;
+            int f = (k < K) ? k / Hi : 0;
+            ftab[threadIdx.x] = (short)f;
+            htab[threadIdx.x] = (short)(k - f * Hi);
+        }
+        __syncthreads();
+        // build V chunk: vtile[c][kk] = x0c[c][f(k)] * xkc[c][h(k)]
+        for (int base = 0; base < CIN_COLS * CIN_KC;
+             base += (int)blockDim.x) {
+            int i = base + (int)threadIdx.x;
+            // CIN_COLS*CIN_KC (20480) divides evenly by 512 threads;
+            // CIN_KC is a literal so the div lowers to multiply-shift
+            int c = i / CIN_KC;
+            int kk = i - c * CIN_KC;
             float v = 0.f;
-            if (k < K) v = x0c[c * 32 + k / H] * xkc[c * 128 + k % H];
+            if (k0 + kk < K)
+                v = x0c[c * 32 + ftab[kk]] * xkc[c * 128 + htab[kk]];
             vtile[c * CIN_VLD + kk] = (cbf16)v;
         }
         __syncthreads();
@@ -154,25 +178,33 @@ void k_cin_dw(const cbf16* __restrict__ dzt,   // [O, Np]
     for (int t = 0; t < 8; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
     const long koff = (lane >> 4) * 8;
-    const long Hp16 = (H + 15) / 16 * 16;   // partial h-tile rows zeroed
+    const int Hp16 = (int)((H + 15) / 16 * 16);  // partial h-tile zeroed
+    const int Hi = (int)H;
+    const cbf16* pa = dzt + (o0 + (lane & 15)) * Np + koff;
+    // A-stream pipelined across n-chunks: the next chunk's dZt fragment
+    // loads behind the current chunk's B' build + barrier
+    bf16x8 a_cur;
+    if (o0 < O) a_cur = cin_ld_frag(pa + nbeg);
     for (long n0s = nbeg; n0s < nend; n0s += 32) {
         __syncthreads();
         // B' chunk: [H][32] = xkt rows * x0 row f (broadcast over h)
-        for (long i = threadIdx.x; i < Hp16 * 32; i += blockDim.x) {
-            long h = i / 32, nn = i % 32;
+        for (int i = (int)threadIdx.x; i < Hp16 * 32;
+             i += (int)blockDim.x) {
+            int h = i >> 5, nn = i & 31;
             float v = 0.f;
-            if (h < H)
+            if (h < Hi)
                 v = (float)xkt[h * Np + n0s + nn]
                     * (float)x0t[f * Np + n0s + nn];
             btile[h * 40 + nn] = (cbf16)v;
         }
         __syncthreads();
         if (o0 >= O) continue;
-        bf16x8 a = cin_ld_frag(dzt + (o0 + (lane & 15)) * Np + n0s + koff);
+        bf16x8 a = a_cur;
+        if (n0s + 32 < nend) a_cur = cin_ld_frag(pa + n0s + 32);
         #pragma unroll
         for (int t = 0; t < 8; ++t) {
-            if (t * 16 >= H) break;
-            bf16x8 b = cin_ld_frag(btile + ((long)t * 16 + (lane & 15)) * 40
+            if (t * 16 >= Hi) break;
+            bf16x8 b = cin_ld_frag(btile + (t * 16 + (lane & 15)) * 40
                                    + koff);
             acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
                                                              0, 0, 0);
@@ -224,25 +256,39 @@ void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
     __shared__ float xkc[CDX_COLS * 128];
     __shared__ float dx0a[CDX_COLS * 32];
     __shared__ float dxka[CDX_COLS * 128];
+    __shared__ short ftab[4096], htab[4096];    // k -> (f, h), K <= 4096
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
     const long n0 = (long)blockIdx.x * CDX_COLS;
-    const long K = F * H;
+    const int K = (int)(F * H);
+    const int Fi = (int)F, Hi = (int)H;
 
-    for (long i = threadIdx.x; i < CDX_COLS * 32; i += blockDim.x)
+    // INT32 index math throughout — 64-bit div/mod emulation made the
+    // first version of this kernel ~1.2 ms/call (profile r2g)
+    for (int i = (int)threadIdx.x; i < K; i += (int)blockDim.x) {
+        int f = i / Hi;
+        ftab[i] = (short)f;
+        htab[i] = (short)(i - f * Hi);
+    }
+    for (int i = (int)threadIdx.x; i < CDX_COLS * 32;
+         i += (int)blockDim.x)
         dx0a[i] = 0.f;
-    for (long i = threadIdx.x; i < CDX_COLS * 128; i += blockDim.x)
+    for (int i = (int)threadIdx.x; i < CDX_COLS * 128;
+         i += (int)blockDim.x)
         dxka[i] = 0.f;
-    for (long i = threadIdx.x; i < CDX_COLS * F; i += blockDim.x) {
-        long c = i / F, ff = i % F;
+    for (int i = (int)threadIdx.x; i < CDX_COLS * Fi;
+         i += (int)blockDim.x) {
+        int c = i / Fi, ff = i - (i / Fi) * Fi;
         x0c[c * 32 + ff] = (n0 + c < N) ? x0p[(n0 + c) * F + ff] : 0.f;
     }
-    for (long i = threadIdx.x; i < CDX_COLS * H; i += blockDim.x) {
-        long c = i / H, h = i % H;
+    for (int i = (int)threadIdx.x; i < CDX_COLS * Hi;
+         i += (int)blockDim.x) {
+        int c = i / Hi, h = i - (i / Hi) * Hi;
         xkc[c * 128 + h] = (n0 + c < N) ? xkp[(n0 + c) * H + h] : 0.f;
     }
-    for (long i = threadIdx.x; i < CDX_COLS * 128; i += blockDim.x) {
-        long c = i / 128, o = i % 128;
+    for (int i = (int)threadIdx.x; i < CDX_COLS * 128;
+         i += (int)blockDim.x) {
+        int c = i >> 7, o = i & 127;
         float v = (o < O && n0 + c < N) ? doutp[(n0 + c) * O + o] : 0.f;
         dzc[c * 136 + o] = (cbf16)v;
     }
@@ -278,25 +324,27 @@ void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
                 }
             }
             // consume: lane holds P[k0 + (lane>>4)*4 + r][cs*16 + lane&15]
-            long c = (long)cs * 16 + (lane & 15);
+            int c = cs * 16 + (lane & 15);
             #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                long k = k0 + (lane >> 4) * 4 + r;
+                int k = (int)k0 + (lane >> 4) * 4 + r;
                 if (k >= K) break;
                 float p = acc[r];
-                long ff = k / H, h = k % H;
+                int ff = ftab[k], h = htab[k];
                 atomicAdd(&dx0a[c * 32 + ff], p * xkc[c * 128 + h]);
                 atomicAdd(&dxka[c * 128 + h], p * x0c[c * 32 + ff]);
             }
         }
     }
     __syncthreads();
-    for (long i = threadIdx.x; i < CDX_COLS * F; i += blockDim.x) {
-        long c = i / F, ff = i % F;
+    for (int i = (int)threadIdx.x; i < CDX_COLS * Fi;
+         i += (int)blockDim.x) {
+        int c = i / Fi, ff = i - (i / Fi) * Fi;
         if (n0 + c < N) dx0p[(n0 + c) * F + ff] = dx0a[c * 32 + ff];
     }
-    for (long i = threadIdx.x; i < CDX_COLS * H; i += blockDim.x) {
-        long c = i / H, h = i % H;
+    for (int i = (int)threadIdx.x; i < CDX_COLS * Hi;
+         i += (int)blockDim.x) {
+        int c = i / Hi, h = i - (i / Hi) * Hi;
         if (n0 + c < N) dxkp[(n0 + c) * H + h] = dxka[c * 128 + h];
     }
 }
