@@ -236,55 +236,55 @@ extern "C" void emb_cin_dw(const void* dzt, const void* x0t, const void* xkt,
 // Per column n:  P[k] = sum_o W[o,k] dZ[n,o]   (GEMM over O, K-tiled)
 //   dx0[n,f] = sum_h P[f*H+h] xk[n,h]
 //   dxk[n,h] = sum_f P[f*H+h] x0[n,f]
-// P never touches HBM: each P subtile is consumed from registers into LDS
-// fp32 accumulators (atomic — lanes of one wave can collide on (c,f)).
-// Wt [K, Op] is the bf16 transposed weight copy; block = 64 columns,
-// waves round-robin the K tiles (each Wt row read once per block).
+// P never touches HBM: K is walked one FIELD at a time (k in [f*H,(f+1)*H)
+// is exactly h), each wave computes whole k-tiles of the chunk's P into an
+// LDS staging tile, and the consume phase is plain bank-padded LDS
+// arithmetic — NO atomics (the first version accumulated through LDS
+// float atomics and measured 1.2 ms/call: contention turns them into
+// serialized CAS traffic). Wt [Kp, Op] is the bf16 transposed weight copy;
+// block = 64 columns.
 
 #define CDX_COLS 64
+#define CDX_X0LD 36     // x0/dx0 LDS row stride (bank-cycling pad)
+#define CDX_XKLD 132    // xk/dxk LDS row stride
+#define CDX_PLD 72      // pchunk row stride
 
 extern "C" __global__ __launch_bounds__(64 * CIN_WAVES, 1)
 void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
-              const cbf16* __restrict__ wt,     // [K, Op]
+              const cbf16* __restrict__ wt,     // [Kp, Op]
               const float* __restrict__ x0p,    // [N, F]
               const float* __restrict__ xkp,    // [N, H]
               float* __restrict__ dx0p,         // [N, F] out
               float* __restrict__ dxkp,         // [N, H] out
               long N, long F, long H, long O, long Op) {
-    __shared__ cbf16 dzc[CDX_COLS * 136];       // [c][o], stride 136
-    __shared__ float x0c[CDX_COLS * 32];
-    __shared__ float xkc[CDX_COLS * 128];
-    __shared__ float dx0a[CDX_COLS * 32];
-    __shared__ float dxka[CDX_COLS * 128];
-    __shared__ short ftab[4096], htab[4096];    // k -> (f, h), K <= 4096
+    __shared__ cbf16 dzc[CDX_COLS * 136];        // [c][o], stride 136
+    __shared__ float x0c[CDX_COLS * CDX_X0LD];   // [c][f]
+    __shared__ float xkc[CDX_COLS * CDX_XKLD];   // [c][h]
+    __shared__ float dx0a[CDX_COLS * CDX_X0LD];  // [c][f] accum
+    __shared__ float dxka[CDX_COLS * CDX_XKLD];  // [c][h] accum
+    __shared__ float pchunk[128 * CDX_PLD];      // [h][c] P staging
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
     const long n0 = (long)blockIdx.x * CDX_COLS;
-    const int K = (int)(F * H);
     const int Fi = (int)F, Hi = (int)H;
 
-    // INT32 index math throughout — 64-bit div/mod emulation made the
-    // first version of this kernel ~1.2 ms/call (profile r2g)
-    for (int i = (int)threadIdx.x; i < K; i += (int)blockDim.x) {
-        int f = i / Hi;
-        ftab[i] = (short)f;
-        htab[i] = (short)(i - f * Hi);
-    }
-    for (int i = (int)threadIdx.x; i < CDX_COLS * 32;
+    for (int i = (int)threadIdx.x; i < CDX_COLS * CDX_X0LD;
          i += (int)blockDim.x)
         dx0a[i] = 0.f;
-    for (int i = (int)threadIdx.x; i < CDX_COLS * 128;
+    for (int i = (int)threadIdx.x; i < CDX_COLS * CDX_XKLD;
          i += (int)blockDim.x)
         dxka[i] = 0.f;
     for (int i = (int)threadIdx.x; i < CDX_COLS * Fi;
          i += (int)blockDim.x) {
         int c = i / Fi, ff = i - (i / Fi) * Fi;
-        x0c[c * 32 + ff] = (n0 + c < N) ? x0p[(n0 + c) * F + ff] : 0.f;
+        x0c[c * CDX_X0LD + ff] = (n0 + c < N) ? x0p[(n0 + c) * F + ff]
+                                              : 0.f;
     }
     for (int i = (int)threadIdx.x; i < CDX_COLS * Hi;
          i += (int)blockDim.x) {
         int c = i / Hi, h = i - (i / Hi) * Hi;
-        xkc[c * 128 + h] = (n0 + c < N) ? xkp[(n0 + c) * H + h] : 0.f;
+        xkc[c * CDX_XKLD + h] = (n0 + c < N) ? xkp[(n0 + c) * H + h]
+                                             : 0.f;
     }
     for (int i = (int)threadIdx.x; i < CDX_COLS * 128;
          i += (int)blockDim.x) {
@@ -292,62 +292,81 @@ void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
         float v = (o < O && n0 + c < N) ? doutp[(n0 + c) * O + o] : 0.f;
         dzc[c * 136 + o] = (cbf16)v;
     }
-    __syncthreads();
 
     const long koff = (lane >> 4) * 8;
-    const long ktiles = (K + 15) / 16;
-    for (long kt = wave; kt < ktiles; kt += CIN_WAVES) {
-        const long k0 = kt * 16;
-        #pragma unroll
-        for (int cs = 0; cs < CDX_COLS / 16; ++cs) {
-            f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    const int htiles = (Hi + 15) / 16;
+    for (int f = 0; f < Fi; ++f) {
+        __syncthreads();   // previous consume done before P overwrite
+        // this field's P [H x cols]: wave w owns k-tiles w, w+8, ...
+        for (int kt = wave; kt < htiles; kt += CIN_WAVES) {
+            const long k0 = (long)f * Hi + kt * 16;
             const cbf16* pa = wt + (k0 + (lane & 15)) * Op + koff;
-            const cbf16* pb = dzc + ((long)cs * 16 + (lane & 15)) * 136
-                              + koff;
-            if (Op == 128) {      // all Wt loads in flight before the MFMAs
-                bf16x8 a[4], b[4];
-                #pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    a[j] = cin_ld_frag(pa + 32 * j);
-                    b[j] = cin_ld_frag(pb + 32 * j);
-                }
+            bf16x8 a[4];
+            if (Op == 128) {   // Wt fragments hoisted across col subtiles
                 #pragma unroll
                 for (int j = 0; j < 4; ++j)
-                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a[j], b[j], acc, 0, 0, 0);
-            } else {
-                for (long oc = 0; oc < Op; oc += 32) {
-                    bf16x8 a = cin_ld_frag(pa + oc);
-                    bf16x8 b = cin_ld_frag(pb + oc);
-                    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a, b, acc, 0, 0, 0);
+                    a[j] = cin_ld_frag(pa + 32 * j);
+            }
+            #pragma unroll
+            for (int cs = 0; cs < CDX_COLS / 16; ++cs) {
+                f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+                const cbf16* pb = dzc + (cs * 16 + (lane & 15)) * 136
+                                  + koff;
+                if (Op == 128) {
+                    #pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        bf16x8 b = cin_ld_frag(pb + 32 * j);
+                        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            a[j], b, acc, 0, 0, 0);
+                    }
+                } else {
+                    for (long oc = 0; oc < Op; oc += 32) {
+                        bf16x8 av = cin_ld_frag(pa + oc);
+                        bf16x8 b = cin_ld_frag(pb + oc);
+                        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            av, b, acc, 0, 0, 0);
+                    }
+                }
+                // P subtile store: lane rows h = kt*16 + (lane>>4)*4 + r
+                int c = cs * 16 + (lane & 15);
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    int h = kt * 16 + (lane >> 4) * 4 + r;
+                    if (h < Hi) pchunk[h * CDX_PLD + c] = acc[r];
                 }
             }
-            // consume: lane holds P[k0 + (lane>>4)*4 + r][cs*16 + lane&15]
-            int c = cs * 16 + (lane & 15);
-            #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                int k = (int)k0 + (lane >> 4) * 4 + r;
-                if (k >= K) break;
-                float p = acc[r];
-                int ff = ftab[k], h = htab[k];
-                atomicAdd(&dx0a[c * 32 + ff], p * xkc[c * 128 + h]);
-                atomicAdd(&dxka[c * 128 + h], p * x0c[c * 32 + ff]);
-            }
+        }
+        __syncthreads();
+        // consume — plain bank-padded LDS, no atomics:
+        //   dxk[c][h] += P[h][c] * x0[c][f]      (elementwise)
+        //   dx0[c][f]  = sum_h P[h][c] * xk[c][h] (per-column dot)
+        for (int i = (int)threadIdx.x; i < CDX_COLS * Hi;
+             i += (int)blockDim.x) {
+            int c = i / Hi, h = i - (i / Hi) * Hi;
+            dxka[c * CDX_XKLD + h] += pchunk[h * CDX_PLD + c]
+                                      * x0c[c * CDX_X0LD + f];
+        }
+        if (threadIdx.x < CDX_COLS) {
+            int c = (int)threadIdx.x;
+            float s = 0.f;
+            for (int h = 0; h < Hi; ++h)
+                s += pchunk[h * CDX_PLD + c] * xkc[c * CDX_XKLD + h];
+            dx0a[c * CDX_X0LD + f] = s;
         }
     }
     __syncthreads();
     for (int i = (int)threadIdx.x; i < CDX_COLS * Fi;
          i += (int)blockDim.x) {
         int c = i / Fi, ff = i - (i / Fi) * Fi;
-        if (n0 + c < N) dx0p[(n0 + c) * F + ff] = dx0a[c * 32 + ff];
+        if (n0 + c < N) dx0p[(n0 + c) * F + ff] = dx0a[c * CDX_X0LD + ff];
     }
     for (int i = (int)threadIdx.x; i < CDX_COLS * Hi;
          i += (int)blockDim.x) {
         int c = i / Hi, h = i - (i / Hi) * Hi;
-        if (n0 + c < N) dxkp[(n0 + c) * H + h] = dxka[c * 128 + h];
+        if (n0 + c < N) dxkp[(n0 + c) * H + h] = dxka[c * CDX_XKLD + h];
     }
 }
+
 
 extern "C" void emb_cin_dx(const float* doutp, const void* wt,
                            const float* x0p, const float* xkp, float* dx0p,
