@@ -29,23 +29,23 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define CIN_COLS 128          // columns (b,dd pairs) per block
-#define CIN_KC 160            // K chunk built per LDS round (5 MFMA steps)
-#define CIN_VLD (CIN_KC + 8)  // LDS row stride: 84 dwords, 16-lane clean
+#define CIN_KC 128            // K chunk built per LDS round (4 MFMA steps)
+#define CIN_VLD (CIN_KC + 8)  // LDS row stride: 68 dwords, 16-lane clean
 #define CIN_WAVES 8
 
 static __device__ __forceinline__ bf16x8 cin_ld_frag(const cbf16* p) {
     return *reinterpret_cast<const bf16x8*>(__builtin_assume_aligned(p, 16));
 }
 
-extern "C" __global__ __launch_bounds__(64 * CIN_WAVES, 1)
+extern "C" __global__ __launch_bounds__(64 * CIN_WAVES, 2)
 void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
                const float* __restrict__ xkp,   // [N, H]
                const cbf16* __restrict__ w,     // [O, Kp] padded bf16
                float* __restrict__ out,         // [N, O]
                long N, long F, long H, long O, long Kp) {
     __shared__ cbf16 vtile[CIN_COLS * CIN_VLD];
-    __shared__ float x0c[CIN_COLS * 32];        // F <= 32
-    __shared__ float xkc[CIN_COLS * 128];       // H <= 128
+    __shared__ cbf16 x0c[CIN_COLS * 32];          // F <= 32, bf16 staged
+    __shared__ cbf16 xkc[CIN_COLS * 128];         // H <= 128
     __shared__ short ftab[CIN_KC], htab[CIN_KC];  // k -> (f, h) per chunk
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
@@ -53,26 +53,29 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
     const int K = (int)(F * H);
     const int Fi = (int)F, Hi = (int)H;
 
-    // NOTE all per-element index math below is INT32: 64-bit div/mod is a
-    // multi-thousand-cycle emulation on CDNA and was the original
-    // bottleneck of this kernel (~270 us/call before, profile r2g)
+    // All per-element index math is INT32 (64-bit div/mod emulation cost
+    // ~10x here), operands staged as bf16 (pre-rounded once — the torch
+    // path rounds identically), and the V build writes PACKED bf16x8: the
+    // first version was ISSUE-bound on the scalar build (PMC: 35% active
+    // issue, MFMA a small fraction) at 1 block/CU; this shape runs 2
+    // blocks/CU with ~3x fewer issue slots per element.
 
-    // stage this block's x0/xk columns (coalesced: F/H contiguous per col)
     for (int base = 0; base < CIN_COLS * Fi; base += (int)blockDim.x) {
         int i = base + (int)threadIdx.x;
         if (i >= CIN_COLS * Fi) break;
         int c = i / Fi, f = i - c * Fi;
-        x0c[c * 32 + f] = (n0 + c < N) ? x0p[(n0 + c) * F + f] : 0.f;
+        x0c[c * 32 + f] = (cbf16)((n0 + c < N) ? x0p[(n0 + c) * F + f]
+                                               : 0.f);
     }
     for (int base = 0; base < CIN_COLS * Hi; base += (int)blockDim.x) {
         int i = base + (int)threadIdx.x;
         if (i >= CIN_COLS * Hi) break;
         int c = i / Hi, h = i - c * Hi;
-        xkc[c * 128 + h] = (n0 + c < N) ? xkp[(n0 + c) * H + h] : 0.f;
+        xkc[c * 128 + h] = (cbf16)((n0 + c < N) ? xkp[(n0 + c) * H + h]
+                                                : 0.f);
     }
 
-    // accumulators: wave owns o rows [wave*16, wave*16+16) (O <= 128 ->
-    // one pass with 8 waves) x all 8 column subtiles of the block
+    // wave owns o rows [wave*16, wave*16+16) x all 8 column subtiles
     const long o0 = (long)wave * 16;
     f32x4 acc[8];
     #pragma unroll
@@ -80,8 +83,6 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
 
     for (long k0 = 0; k0 < Kp; k0 += CIN_KC) {
         __syncthreads();
-        // k -> (f, h) lookup for this chunk: one int32 div per entry,
-        // then the build loop is pure LDS reads + one multiply
         if (threadIdx.x < CIN_KC) {
             int k = (int)k0 + (int)threadIdx.x;
             int f = (k < K) ? k / Hi : 0;
@@ -89,18 +90,23 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
             htab[threadIdx.x] = (short)(k - f * Hi);
         }
         __syncthreads();
-        // build V chunk: vtile[c][kk] = x0c[c][f(k)] * xkc[c][h(k)]
-        for (int base = 0; base < CIN_COLS * CIN_KC;
-             base += (int)blockDim.x) {
-            int i = base + (int)threadIdx.x;
-            // CIN_COLS*CIN_KC (20480) divides evenly by 512 threads;
-            // CIN_KC is a literal so the div lowers to multiply-shift
-            int c = i / CIN_KC;
-            int kk = i - c * CIN_KC;
-            float v = 0.f;
-            if (k0 + kk < K)
-                v = x0c[c * 32 + ftab[kk]] * xkc[c * 128 + htab[kk]];
-            vtile[c * CIN_VLD + kk] = (cbf16)v;
+        // packed build: 8 V elements -> one 16 B LDS store
+        const int k0i = (int)k0;
+        for (int g = (int)threadIdx.x; g < CIN_COLS * (CIN_KC / 8);
+             g += (int)blockDim.x) {
+            int c = g >> 4;               // CIN_KC/8 == 16 groups per row
+            int kko = (g & 15) << 3;
+            bf16x8 vv;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                int kk = kko + j;
+                float v = 0.f;
+                if (k0i + kk < K)
+                    v = (float)x0c[c * 32 + ftab[kk]]
+                        * (float)xkc[c * 128 + htab[kk]];
+                vv[j] = (__bf16)v;
+            }
+            *reinterpret_cast<bf16x8*>(vtile + c * CIN_VLD + kko) = vv;
         }
         __syncthreads();
         if (o0 >= O) continue;      // narrow O: spare waves still build V
@@ -108,8 +114,6 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
         const long koff = (lane >> 4) * 8;
         const cbf16* pa = w + (o0 + (lane & 15)) * Kp + k0 + koff;
         const cbf16* pb = vtile + (lane & 15) * CIN_VLD + koff;
-        // W-stream double buffer: the next k-step's global load is in
-        // flight while this step's 8 MFMAs consume the LDS fragments
         bf16x8 a0 = cin_ld_frag(pa);
         for (long ks = 0; ks < kc_lim; ks += 32) {
             bf16x8 a1;
@@ -187,15 +191,23 @@ void k_cin_dw(const cbf16* __restrict__ dzt,   // [O, Np]
     if (o0 < O) a_cur = cin_ld_frag(pa + nbeg);
     for (long n0s = nbeg; n0s < nend; n0s += 32) {
         __syncthreads();
-        // B' chunk: [H][32] = xkt rows * x0 row f (broadcast over h)
-        for (int i = (int)threadIdx.x; i < Hp16 * 32;
-             i += (int)blockDim.x) {
-            int h = i >> 5, nn = i & 31;
-            float v = 0.f;
-            if (h < Hi)
-                v = (float)xkt[h * Np + n0s + nn]
-                    * (float)x0t[f * Np + n0s + nn];
-            btile[h * 40 + nn] = (cbf16)v;
+        // B' chunk: [H][32] = xkt rows * x0 row f (broadcast over h),
+        // packed: two bf16x8 loads -> one bf16x8 store per 8 elements
+        for (int g = (int)threadIdx.x; g < Hp16 * 4;
+             g += (int)blockDim.x) {
+            int h = g >> 2, no = (g & 3) << 3;
+            bf16x8 vv;
+            if (h < Hi) {
+                bf16x8 xk8 = cin_ld_frag(xkt + h * Np + n0s + no);
+                bf16x8 x08 = cin_ld_frag(x0t + f * Np + n0s + no);
+                #pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    vv[j] = (__bf16)((float)xk8[j] * (float)x08[j]);
+            } else {
+                #pragma unroll
+                for (int j = 0; j < 8; ++j) vv[j] = (__bf16)0.f;
+            }
+            *reinterpret_cast<bf16x8*>(btile + h * 40 + no) = vv;
         }
         __syncthreads();
         if (o0 >= O) continue;
@@ -263,6 +275,7 @@ void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
     __shared__ float dx0a[CDX_COLS * CDX_X0LD];  // [c][f] accum
     __shared__ float dxka[CDX_COLS * CDX_XKLD];  // [c][h] accum
     __shared__ float pchunk[128 * CDX_PLD];      // [h][c] P staging
+    __shared__ float dotred[8 * 72];             // dx0 dot partials
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
     const long n0 = (long)blockIdx.x * CDX_COLS;
@@ -337,20 +350,27 @@ void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
             }
         }
         __syncthreads();
-        // consume — plain bank-padded LDS, no atomics:
-        //   dxk[c][h] += P[h][c] * x0[c][f]      (elementwise)
-        //   dx0[c][f]  = sum_h P[h][c] * xk[c][h] (per-column dot)
-        for (int i = (int)threadIdx.x; i < CDX_COLS * Hi;
-             i += (int)blockDim.x) {
-            int c = i / Hi, h = i - (i / Hi) * Hi;
-            dxka[c * CDX_XKLD + h] += pchunk[h * CDX_PLD + c]
-                                      * x0c[c * CDX_X0LD + f];
+        // consume — fused, balanced, no atomics: thread (hr, c) walks
+        // h = hr, hr+8, ..., updating dxk elementwise and accumulating a
+        // dx0 partial; partials reduce across the 8 h-groups through LDS
+        {
+            int c = (int)threadIdx.x & 63;
+            int hr = (int)threadIdx.x >> 6;
+            float x0f = x0c[c * CDX_X0LD + f];
+            float s = 0.f;
+            for (int h = hr; h < Hi; h += 8) {
+                float p = pchunk[h * CDX_PLD + c];
+                dxka[c * CDX_XKLD + h] += p * x0f;
+                s += p * xkc[c * CDX_XKLD + h];
+            }
+            dotred[hr * 72 + c] = s;
         }
+        __syncthreads();
         if (threadIdx.x < CDX_COLS) {
             int c = (int)threadIdx.x;
             float s = 0.f;
-            for (int h = 0; h < Hi; ++h)
-                s += pchunk[h * CDX_PLD + c] * xkc[c * CDX_XKLD + h];
+            #pragma unroll
+            for (int hr = 0; hr < 8; ++hr) s += dotred[hr * 72 + c];
             dx0a[c * CDX_X0LD + f] = s;
         }
     }

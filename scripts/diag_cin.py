@@ -33,8 +33,10 @@ def pad_w(W, Kp, Op=None):
 
 
 def vref(x0p, xkp, F, H):
-    # kernel rounding: fp32 product -> bf16 (inputs NOT pre-rounded)
-    v = x0p.unsqueeze(2) * xkp.unsqueeze(1)          # [N, F, H] fp32
+    # kernel rounding: inputs staged bf16, fp32 product, bf16 store
+    # (identical to the chunked-torch path's roundings)
+    v = (x0p.to(bf).float().unsqueeze(2)
+         * xkp.to(bf).float().unsqueeze(1))          # [N, F, H]
     return v.reshape(x0p.shape[0], F * H).to(bf).float()
 
 
