@@ -41,8 +41,9 @@ def main():
                    help="after training, report AUC/logloss over this many "
                         "held-out synthetic batches")
     p.add_argument("--data", default="",
-                   help="train from a Criteo-format TSV instead of "
-                        "synthetic batches (openembedding_amd.data)")
+                   help="train from a Criteo-format file instead of "
+                        "synthetic batches (openembedding_amd.data); "
+                        "*.tfrecord selects the TFRecord reader")
     args = p.parse_args()
 
     ctx = embed.get_context()
@@ -57,12 +58,16 @@ def main():
 
     def batches():
         if args.data:
-            from openembedding_amd.data import BackgroundLoader, CriteoTSV
+            from openembedding_amd.data import (BackgroundLoader,
+                                                CriteoTFRecord, CriteoTSV)
+            reader_cls = (CriteoTFRecord
+                          if args.data.endswith(".tfrecord")
+                          else CriteoTSV)
             # every rank reads the file; rank r trains rows r, r+W, ...
             # (batch-level round-robin — the reference sharded csv the
             # same way in its horovod examples)
             for i, b in enumerate(BackgroundLoader(
-                    CriteoTSV(args.data, args.batch), depth=4)):
+                    reader_cls(args.data, args.batch), depth=4)):
                 if i % ctx.world_size == ctx.rank:
                     yield b
         else:
