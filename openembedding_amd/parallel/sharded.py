@@ -254,24 +254,33 @@ class ShardedVariable:
 
     def _get_padded_plan(self, n: int,
                          device: torch.device) -> Optional[_PaddedPlan]:
-        plan = self._plans.get(n, False)
-        if plan is not False:
-            return plan
-        import torch.distributed as dist
-        if comm.dist_ready() and dist.get_world_size() > 1:
-            # one-time agreement check: cap derives from n, so all ranks
-            # must see the same element count for this variable (the bench
-            # shape is rank-uniform; ragged shapes fall back to the exact
-            # route). One host sync, once per (variable, n).
-            dev = device if comm.backend() == "nccl" else "cpu"
-            t = torch.tensor([n, -n], dtype=torch.int64, device=dev)
-            dist.all_reduce(t, op=dist.ReduceOp.MIN)
-            if int(t[0]) != n or int(-t[1]) != n:
-                self._plans[n] = None
-                return None
-        plan = _PaddedPlan(self.world_size, n, device)
-        self._plans[n] = plan
-        return plan
+        # TWO plans per shape, rotated per pull: a pull's wire buffers
+        # (send_src/pos_of) must survive until its push consumes them, and
+        # one prefetched pull may be outstanding while the current batch
+        # pushes — two live handles max under the issue-after-commit
+        # prefetch discipline (deeper hand-rolled prefetch would need a
+        # deeper rotation; pulling() never issues more than one ahead)
+        pair = self._plans.get(n, False)
+        if pair is None:
+            return None
+        if pair is False:
+            import torch.distributed as dist
+            if comm.dist_ready() and dist.get_world_size() > 1:
+                # one-time agreement check: cap derives from n, so all
+                # ranks must see the same element count for this variable
+                # (the bench shape is rank-uniform; ragged shapes fall
+                # back to the exact route). One host sync per (var, n).
+                dev = device if comm.backend() == "nccl" else "cpu"
+                t = torch.tensor([n, -n], dtype=torch.int64, device=dev)
+                dist.all_reduce(t, op=dist.ReduceOp.MIN)
+                if int(t[0]) != n or int(-t[1]) != n:
+                    self._plans[n] = None
+                    return None
+            pair = [_PaddedPlan(self.world_size, n, device),
+                    _PaddedPlan(self.world_size, n, device), 0]
+            self._plans[n] = pair
+        pair[2] ^= 1
+        return pair[pair[2]]
 
     def _pull_remote_padded(self, indices: torch.Tensor,
                             flat: torch.Tensor):
@@ -401,16 +410,18 @@ class ShardedVariable:
         bench.py after the timed loop; overflow means some keys silently
         read zeros, so the run must fail loudly. Fix: raise OEAMD_A2A_SLACK
         (or disable with OEAMD_PADDED_A2A=0)."""
-        for plan in self._plans.values():
-            if plan is None:
+        for pair in self._plans.values():
+            if pair is None:
                 continue
-            ov = plan.overflow_host + int(plan.overflow.sum().item())
-            if ov:
-                raise RuntimeError(
-                    f"padded all-to-all overflow on variable "
-                    f"{self.variable_id}: {ov} keys were dropped (bucket "
-                    f"capacity {plan.cap}/peer). Raise OEAMD_A2A_SLACK "
-                    f"(default 2.0) or set OEAMD_PADDED_A2A=0.")
+            for plan in pair[:2]:
+                ov = plan.overflow_host + int(plan.overflow.sum().item())
+                if ov:
+                    raise RuntimeError(
+                        f"padded all-to-all overflow on variable "
+                        f"{self.variable_id}: {ov} keys were dropped "
+                        f"(bucket capacity {plan.cap}/peer). Raise "
+                        f"OEAMD_A2A_SLACK (default 2.0) or set "
+                        f"OEAMD_PADDED_A2A=0.")
 
     def _pull_remote(self, h: PullHandle, readonly: bool) -> torch.Tensor:
         with stage_timer("pull", "remote"):
