@@ -230,3 +230,67 @@ def test_reserved_key_zero_row():
     out, h = var.pull(keys)
     assert float(out[1].abs().sum()) == 0.0
     assert float(out[0].abs().sum()) > 0.0
+
+
+@pytest.mark.timeout(240)
+def test_padded_fuzz_vs_exact():
+    """Property fuzz of the padded route (CPU fallback, forced-remote):
+    random key mixes — duplicates, hot keys, the reserved -1, shapes not
+    divisible by anything — must match the exact route bit-for-bit."""
+    from openembedding_amd.context import Context
+
+    g = torch.Generator().manual_seed(42)
+    for trial in range(10):
+        n = int(torch.randint(1, 300, (1,), generator=g))
+        hot = int(torch.randint(1, 10, (1,), generator=g))
+        keys = torch.randint(0, 50, (n,), generator=g, dtype=torch.int64)
+        keys[torch.randint(0, n, (min(hot, n),), generator=g)] = 7
+        grads = torch.randn(n, DIM, generator=g)
+        results = []
+        for padded in (False, True):
+            ctx = Context(device="cpu")
+            st, var = _make_var(ctx)
+            var._force_remote = True
+            var._padded = padded
+            out, h = var.pull(keys)
+            var.push(h, grads)
+            st.update_weights()
+            after, _ = var.pull(keys, readonly=True)
+            var.check_padded_overflow()
+            results.append((out, after))
+        torch.testing.assert_close(results[0][0], results[1][0])
+        torch.testing.assert_close(results[0][1], results[1][1])
+
+
+@pytest.mark.timeout(240)
+def test_padded_two_outstanding_pulls():
+    """Two pulls before their pushes (the prefetch pattern): the plan
+    rotation must keep both handles' wire maps alive."""
+    from openembedding_amd.context import Context
+
+    ctx = Context(device="cpu")
+    st, var = _make_var(ctx)
+    var._force_remote = True
+    var._padded = True
+    k1 = torch.tensor([3, 5, 7, 5], dtype=torch.int64)
+    k2 = torch.tensor([5, 9, 11, 3], dtype=torch.int64)
+    out1, h1 = var.pull(k1)
+    out2, h2 = var.pull(k2)        # same shape -> would reuse h1's plan
+    assert h1.plan is not h2.plan, "outstanding pulls must not share a plan"
+    var.push(h1, torch.ones_like(out1))
+    var.push(h2, torch.full_like(out2, 2.0))
+    st.update_weights()
+
+    # reference semantics: one optimizer step per unique key over summed
+    # grads — compare with the exact path doing the same two-pull commit
+    ctx2 = Context(device="cpu")
+    st2, var2 = _make_var(ctx2)
+    o1, g1 = var2.pull(k1)
+    o2, g2 = var2.pull(k2)
+    var2.push(g1, torch.ones_like(o1))
+    var2.push(g2, torch.full_like(o2, 2.0))
+    st2.update_weights()
+    probe = torch.tensor([3, 5, 7, 9, 11], dtype=torch.int64)
+    a, _ = var.pull(probe, readonly=True)
+    b, _ = var2.pull(probe, readonly=True)
+    torch.testing.assert_close(a, b)
