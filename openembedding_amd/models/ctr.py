@@ -116,9 +116,26 @@ class _FusedMLP3Fn(torch.autograd.Function):
         params = ctx._params
         if all(p.grad is not None for p in params):
             w1g, b1g, w2g, b2g, w3g, b3g, w4g, b4g = (p.grad for p in params)
-            w1g.addmm_(dz1.t(), x0[:, :K0])
-            w2g.addmm_(dz2.t(), a1)
-            w3g.addmm_(dz3.t(), a2)
+            M, H = dz1.shape
+            K0p = x0.shape[1]
+            if (w1g.dtype == torch.bfloat16 and w1g.is_contiguous()
+                    and w2g.is_contiguous() and w3g.is_contiguous()
+                    and M % 32 == 0):
+                # all three wgrads in ONE MFMA launch (+= via fp32 scratch,
+                # self-cleaning finisher) — the hipBLASLt trio ran ~59 us
+                # per step on 64x16 macro tiles (profiles/, round 2)
+                scratch = bufs.get("wscratch")
+                want = H * K0p + 2 * H * H
+                if scratch is None or scratch.numel() != want:
+                    scratch = torch.zeros(want, dtype=torch.float32,
+                                          device=dz1.device)
+                    bufs["wscratch"] = scratch
+                ext.mlp3_wgrad(dz1, dz2, dz3, x0, a1, a2, scratch,
+                               w1g, w2g, w3g)
+            else:
+                w1g.addmm_(dz1.t(), x0[:, :K0])
+                w2g.addmm_(dz2.t(), a1)
+                w3g.addmm_(dz3.t(), a2)
             dpart = dout if ctx._has_partial else None
             if b1g.dtype == torch.bfloat16 and b1g.is_contiguous() \
                     and b2g.is_contiguous() and b3g.is_contiguous() \

@@ -51,6 +51,9 @@ void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
 void emb_mlp3_bias_bwd(const float*, const void*, const void*, const void*,
                        const void*, long, long, float*, void*, void*, void*,
                        void*, void*, hipStream_t_);
+void emb_mlp3_wgrad(const void*, const void*, const void*, const void*,
+                    const void*, const void*, long, long, long, long,
+                    float*, void*, void*, void*, hipStream_t_);
 void emb_bce_fwd(const float*, const float*, long, float*, hipStream_t_);
 void emb_bce_bwd(const float*, const float*, long, const float*, float*,
                  hipStream_t_);
@@ -606,6 +609,30 @@ std::tuple<torch::Tensor, torch::Tensor> cin_dx(
     return {dx0, dxk};
 }
 
+void mlp3_wgrad(torch::Tensor dz1, torch::Tensor dz2, torch::Tensor dz3,
+                torch::Tensor x0, torch::Tensor a1, torch::Tensor a2,
+                torch::Tensor scratch, torch::Tensor dw1, torch::Tensor dw2,
+                torch::Tensor dw3) {
+    CHECK_GPU(dz1); CHECK_CONT(dz1); CHECK_CONT(dz2); CHECK_CONT(dz3);
+    CHECK_CONT(x0); CHECK_CONT(a1); CHECK_CONT(a2); CHECK_CONT(scratch);
+    TORCH_CHECK(dz1.dtype() == torch::kBFloat16
+                && dw1.dtype() == torch::kBFloat16
+                && scratch.dtype() == torch::kFloat32, "mlp3_wgrad dtypes");
+    long M = dz1.size(0), H = dz1.size(1), K0p = x0.size(1);
+    long K0 = dw1.size(1);
+    TORCH_CHECK(M % 32 == 0, "mlp3_wgrad: M must be x32");
+    TORCH_CHECK(scratch.numel() >= H * K0p + 2 * H * H
+                && dw1.size(0) == H && K0 <= K0p && dw2.numel() == H * H
+                && dw3.numel() == H * H
+                && dw1.is_contiguous() && dw2.is_contiguous()
+                && dw3.is_contiguous(), "mlp3_wgrad shapes");
+    const c10::cuda::CUDAGuard guard(dz1.device());
+    emb_mlp3_wgrad(dz1.data_ptr(), dz2.data_ptr(), dz3.data_ptr(),
+                   x0.data_ptr(), a1.data_ptr(), a2.data_ptr(), M, H, K0p,
+                   K0, scratch.data_ptr<float>(), dw1.data_ptr(),
+                   dw2.data_ptr(), dw3.data_ptr(), cur_stream());
+}
+
 // ---- fused BCE-with-logits --------------------------------------------
 
 torch::Tensor bce_fwd(torch::Tensor logits, torch::Tensor labels) {
@@ -732,6 +759,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused dgrad chain backward of the 3-layer MLP");
     m.def("mlp3_fwd", &mlp3_fwd,
           "fused 3-hidden-layer MLP forward (bf16 MFMA, bias+ReLU fused)");
+    m.def("mlp3_wgrad", &mlp3_wgrad,
+          "all three MLP weight grads in one MFMA launch (+= into bf16 "
+          "grads via fp32 scratch)");
     m.def("mlp3_bias_bwd", &mlp3_bias_bwd,
           "MLP bias grads + head wgrad in one pass over the dz mirrors");
     m.def("cin_fwd", &cin_fwd,

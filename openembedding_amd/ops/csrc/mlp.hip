@@ -392,3 +392,151 @@ extern "C" void emb_mlp3_bwd(const float* dout, long M, long K0p,
         (const mbf16*)w1t, H, Hp, (mbf16*)dz1, (mbf16*)dz2, (mbf16*)dz3,
         (mbf16*)dx0);
 }
+
+// ---------------------------------------------------------- fused wgrads
+// dW_L[i,j] += sum_m dz_L[m,i] * a_{L-1}[m,j] for the 3 hidden layers in
+// ONE launch. hipBLASLt ran these [H x M]@[M x H'] shapes at ~64 TF
+// (64x16 macro tiles, 182 workgroups): here a block computes a 64x64
+// fp32 tile with both operands LDS-transposed on load, M split 4 ways
+// into an fp32 scratch that a finisher folds into the bf16 grads
+// (+=, self-cleaning — the same capture-safe pattern as the bias pass).
+
+#define WG_KC 64            // m-chunk staged per LDS round
+#define WG_LD 72            // LDS row stride (16 B-aligned frag rows)
+
+extern "C" __global__ __launch_bounds__(512, 2)
+void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
+                  const mbf16* __restrict__ dz2,   // [M, H]
+                  const mbf16* __restrict__ dz3,   // [M, H]
+                  const mbf16* __restrict__ x0,    // [M, K0p]
+                  const mbf16* __restrict__ a1,    // [M, H]
+                  const mbf16* __restrict__ a2,    // [M, H]
+                  long M, long H, long K0p,
+                  long t1, long tk, long th,       // L1 tiles, K0p/H tiles
+                  float* __restrict__ scratch,     // [H*K0p + 2*H*H]
+                  long m_split) {
+    // block -> (layer, i-tile, j-tile)
+    long bt = blockIdx.x;
+    const mbf16 *dz, *a;
+    float* out;
+    long ti, tj, jdim;
+    if (bt < t1) {                    // L1: dz1^T @ x0 -> [H, K0p]
+        dz = dz1; a = x0; out = scratch; jdim = K0p;
+        ti = bt / tk; tj = bt - ti * tk;
+    } else if (bt < t1 + th * th) {   // L2: dz2^T @ a1 -> [H, H]
+        bt -= t1;
+        dz = dz2; a = a1; out = scratch + H * K0p; jdim = H;
+        ti = bt / th; tj = bt - ti * th;
+    } else {                          // L3: dz3^T @ a2 -> [H, H]
+        bt -= t1 + th * th;
+        dz = dz3; a = a2; out = scratch + H * K0p + H * H; jdim = H;
+        ti = bt / th; tj = bt - ti * th;
+    }
+    const long i0 = ti * 64, j0 = tj * 64;
+    const long msz = ((M / 32 + m_split - 1) / m_split) * 32;
+    const long mbeg = (long)blockIdx.y * msz;
+    const long mend = (mbeg + msz < M) ? (mbeg + msz) : M;
+    if (mbeg >= M) return;
+
+    __shared__ mbf16 ldsA[64 * WG_LD];   // dz^T tile  [i][m]
+    __shared__ mbf16 ldsB[64 * WG_LD];   // a^T tile   [j][m]
+    const int wave = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+
+    // wave w owns output subtiles {w, w+8} of the 4x4 16x16 grid
+    f32x4 acc[2];
+    acc[0] = f32x4{0.f, 0.f, 0.f, 0.f};
+    acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    for (long m0 = mbeg; m0 < mend; m0 += WG_KC) {
+        __syncthreads();
+        // stage transposed: read rows m (64 consecutive columns
+        // coalesced), write LDS [col][m] — 512 threads x 8 elements
+        const long mlim = (mend - m0 < WG_KC) ? (mend - m0) : WG_KC;
+        for (int e = (int)threadIdx.x; e < 64 * WG_KC;
+             e += (int)blockDim.x) {
+            int mm = e >> 6, cc = e & 63;
+            mbf16 va = (mbf16)0.0f, vb = (mbf16)0.0f;
+            if (mm < mlim) {
+                long m = m0 + mm;
+                if (i0 + cc < H) va = dz[m * H + i0 + cc];
+                if (j0 + cc < jdim)
+                    vb = a[m * jdim + j0 + cc];
+            }
+            ldsA[cc * WG_LD + mm] = va;
+            ldsB[cc * WG_LD + mm] = vb;
+        }
+        __syncthreads();
+        const long koff = (lane >> 4) * 8;
+        #pragma unroll
+        for (int t = 0; t < 2; ++t) {
+            int st = wave + t * 8;            // subtile id 0..15
+            int si = (st >> 2) * 16, sj = (st & 3) * 16;
+            const mbf16* pa = ldsA + (si + (lane & 15)) * WG_LD + koff;
+            const mbf16* pb = ldsB + (sj + (lane & 15)) * WG_LD + koff;
+            for (int ks = 0; ks < WG_KC; ks += 32) {
+                bf16x8 av = ld_frag(pa + ks);
+                bf16x8 bv = ld_frag(pb + ks);
+                acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    av, bv, acc[t], 0, 0, 0);
+            }
+        }
+    }
+    #pragma unroll
+    for (int t = 0; t < 2; ++t) {
+        int st = wave + t * 8;
+        int si = (st >> 2) * 16, sj = (st & 3) * 16;
+        long i = i0 + si + (lane >> 4) * 4;   // C row = A's m-dim = i
+        long j = j0 + sj + (lane & 15);
+        if (j >= jdim) continue;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            if (i + r >= H) break;
+            atomicAdd(&out[(i + r) * jdim + j], acc[t][r]);
+        }
+    }
+}
+
+extern "C" __global__ void k_mlp3_wgrad_finish(
+        float* __restrict__ scratch, long H, long K0p, long K0,
+        mbf16* __restrict__ dw1, mbf16* __restrict__ dw2,
+        mbf16* __restrict__ dw3) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    long n1 = H * K0p, n23 = H * H;
+    if (i >= n1 + 2 * n23) return;
+    float v = scratch[i];
+    scratch[i] = 0.f;                    // self-cleaning for the next step
+    mbf16* dst;
+    if (i < n1) {
+        long row = i / K0p, col = i - row * K0p;
+        if (col >= K0) return;           // x0 pad columns: zero, unfolded
+        dst = dw1 + row * K0 + col;      // the PARAM is [H, K0] unpadded
+    } else if (i < n1 + n23) {
+        dst = dw2 + (i - n1);
+    } else {
+        dst = dw3 + (i - n1 - n23);
+    }
+    *dst = (mbf16)((float)*dst + v);     // += matches torch beta=1 accum
+}
+
+extern "C" void emb_mlp3_wgrad(const void* dz1, const void* dz2,
+                               const void* dz3, const void* x0,
+                               const void* a1, const void* a2,
+                               long M, long H, long K0p, long K0,
+                               float* scratch,
+                               void* dw1, void* dw2, void* dw3,
+                               hipStream_t stream) {
+    if (!M) return;
+    const long m_split = 4;
+    long th = (H + 63) / 64, tk = (K0p + 63) / 64;
+    long t1 = th * tk;
+    long tiles = t1 + 2 * th * th;
+    dim3 grid((unsigned)tiles, (unsigned)m_split);
+    k_mlp3_wgrad<<<grid, 512, 0, stream>>>(
+        (const mbf16*)dz1, (const mbf16*)dz2, (const mbf16*)dz3,
+        (const mbf16*)x0, (const mbf16*)a1, (const mbf16*)a2, M, H, K0p,
+        t1, tk, th, scratch, m_split);
+    long total = H * K0p + 2 * H * H;
+    k_mlp3_wgrad_finish<<<(int)((total + 255) / 256), 256, 0, stream>>>(
+        scratch, H, K0p, K0, (mbf16*)dw1, (mbf16*)dw2, (mbf16*)dw3);
+}

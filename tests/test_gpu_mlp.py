@@ -189,3 +189,40 @@ def test_prebound_bias_kernel_matches_gemv_fallback():
     for n, a, b in zip(names, fused, gemv):
         assert torch.allclose(a, b, atol=5e-3, rtol=3e-2), (
             n, (a - b).abs().max().item())
+
+
+def test_mlp3_wgrad_matches_addmm():
+    """Fused 3-wgrad MFMA kernel vs the hipBLASLt addmm_ trio it replaces
+    (same bf16 operands; fp32 accumulation both sides)."""
+    from openembedding_amd.ops import require_hip
+    ext = require_hip()
+    torch.manual_seed(4)
+    M, H, K0, K0p = 1024, 400, 247, 256
+    bf = torch.bfloat16
+    dz1 = torch.randn(M, H, device=DEV).to(bf)
+    dz2 = torch.randn(M, H, device=DEV).to(bf)
+    dz3 = torch.randn(M, H, device=DEV).to(bf)
+    x0 = torch.zeros(M, K0p, device=DEV, dtype=bf)
+    x0[:, :K0] = torch.randn(M, K0, device=DEV).to(bf)
+    a1 = torch.randn(M, H, device=DEV).to(bf)
+    a2 = torch.randn(M, H, device=DEV).to(bf)
+    base = [torch.randn(H, K0, device=DEV).to(bf).contiguous(),
+            torch.randn(H, H, device=DEV).to(bf).contiguous(),
+            torch.randn(H, H, device=DEV).to(bf).contiguous()]
+    got = [b.clone() for b in base]
+    ref = [b.clone() for b in base]
+    scratch = torch.zeros(H * K0p + 2 * H * H, device=DEV)
+    ext.mlp3_wgrad(dz1, dz2, dz3, x0, a1, a2, scratch, *got)
+    ref[0].addmm_(dz1.t(), x0[:, :K0])
+    ref[1].addmm_(dz2.t(), a1)
+    ref[2].addmm_(dz3.t(), a2)
+    for g, r in zip(got, ref):
+        torch.testing.assert_close(g.float(), r.float(), rtol=2e-2,
+                                   atol=2e-2)
+    # scratch self-cleaned for the next (captured) step
+    assert float(scratch.abs().sum()) == 0.0
+    # second call accumulates again (beta=1 semantics)
+    ext.mlp3_wgrad(dz1, dz2, dz3, x0, a1, a2, scratch, *got)
+    ref[1].addmm_(dz2.t(), a1)
+    torch.testing.assert_close(got[1].float(), ref[1].float(), rtol=3e-2,
+                               atol=3e-2)
