@@ -401,8 +401,8 @@ extern "C" void emb_mlp3_bwd(const float* dout, long M, long K0p,
 // into an fp32 scratch that a finisher folds into the bf16 grads
 // (+=, self-cleaning — the same capture-safe pattern as the bias pass).
 
-#define WG_KC 64            // m-chunk staged per LDS round
-#define WG_LD 72            // LDS row stride (16 B-aligned frag rows)
+#define WG_KC 128           // m-chunk staged per LDS round
+#define WG_LD (WG_KC + 8)   // LDS row stride (16 B-aligned frag rows)
 
 extern "C" __global__ __launch_bounds__(512, 2)
 void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
@@ -438,8 +438,8 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
     const long mend = (mbeg + msz < M) ? (mbeg + msz) : M;
     if (mbeg >= M) return;
 
-    __shared__ mbf16 ldsA[64 * WG_LD];   // dz^T tile  [i][m]
-    __shared__ mbf16 ldsB[64 * WG_LD];   // a^T tile   [j][m]
+    __shared__ mbf16 ldsA[64 * (WG_KC + 8)];   // dz^T tile  [i][m]
+    __shared__ mbf16 ldsB[64 * (WG_KC + 8)];   // a^T tile   [j][m]
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
 
@@ -527,7 +527,7 @@ extern "C" void emb_mlp3_wgrad(const void* dz1, const void* dz2,
                                void* dw1, void* dw2, void* dw3,
                                hipStream_t stream) {
     if (!M) return;
-    const long m_split = 4;
+    const long m_split = 8;
     long th = (H + 63) / 64, tk = (K0p + 63) / 64;
     long t1 = th * tk;
     long tiles = t1 + 2 * th * th;
