@@ -120,7 +120,11 @@ class _FusedMLP3Fn(torch.autograd.Function):
             K0p = x0.shape[1]
             if (w1g.dtype == torch.bfloat16 and w1g.is_contiguous()
                     and w2g.is_contiguous() and w3g.is_contiguous()
-                    and M % 32 == 0):
+                    and M % 32 == 0 and K0p <= 512):
+                # K0p <= 512: at the dim9 shapes the fused kernel beats the
+                # hipBLASLt trio (12.29 -> 12.62M); at dim64 (K0p 1696)
+                # hipBLASLt's wide-N tiles win (4.90 vs 4.75M) — measured,
+                # profiles r2m-r2o
                 # all three wgrads in ONE MFMA launch (+= via fp32 scratch,
                 # self-cleaning finisher) — the hipBLASLt trio ran ~59 us
                 # per step on 64x16 macro tiles (profiles/, round 2)
