@@ -62,6 +62,16 @@ def main():
     assert torch.allclose(live, served, atol=1e-6), "served rows differ!"
     print("serving matches training: OK")
 
+    # TF-Serving-loadable SavedModel export (the reference's
+    # save_as_original_model contract): a standalone directory TF loads
+    # with no OpenEmbedding runtime
+    sm = tempfile.mkdtemp(prefix="oe_savedmodel_")
+    embed.save_as_original_model(model, sm)
+    import os
+    assert os.path.exists(os.path.join(sm, "saved_model.pb"))
+    print(f"SavedModel exported to {sm} "
+          f"(saved_model.pb + variables bundle)")
+
 
 if __name__ == "__main__":
     main()
