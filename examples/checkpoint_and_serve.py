@@ -27,12 +27,16 @@ from openembedding_amd.serving import ModelController, make_app
 def main():
     ctx = embed.get_context()
     torch.manual_seed(0)
-    model = DeepFM(dim=9).to(ctx.device)
+    # small per-field vocabularies keep the SavedModel materialization
+    # below quick (full Criteo cardinalities work the same, just bigger)
+    fv = [1000] * 26
+    model = DeepFM(field_vocabs=fv, dim=9).to(ctx.device)
     opt = embed.distributed_optimizer(
         torch.optim.Adagrad(model.parameters(), lr=0.01))
     lossf = torch.nn.BCEWithLogitsLoss()
     for _ in range(5):
-        dense, sparse, labels = synthetic_batch(256, device=str(ctx.device))
+        dense, sparse, labels = synthetic_batch(
+            256, field_vocabs=fv, device=str(ctx.device))
         opt.zero_grad()
         loss = lossf(model(dense, sparse), labels)
         loss.backward()
