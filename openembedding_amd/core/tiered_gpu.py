@@ -201,11 +201,13 @@ class HipTieredVariableShard(HipVariableShard):
             return
         keys = self.host_keys_of_slots(evict_slots)
         self._host_ensure(self._host_rows_upper + n_ev)
+        # count the incoming rows BEFORE the load-factor check so the
+        # insert below never runs into a >50%-full probe table
+        self._host_rows_upper += n_ev
         self._host_maybe_rehash()
         hslots, _ = self.ext.ht_lookup(self.host_tk, self.host_tv, keys,
                                        self.host_nrows_dev,
                                        self.host_slot_keys, True, None)
-        self._host_rows_upper += n_ev
         self.ext.spill_rows(evict_slots, hslots, self.weights, self.state,
                             self._host_weights,
                             self._host_state if self.state_dim else None)

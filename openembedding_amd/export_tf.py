@@ -64,7 +64,7 @@ class GraphBuilder:
         self.add(name, "VariableV2",
                  dtype=tp.attr_type(dt),
                  shape=tp.attr_shape(arr.shape),
-                 container=tp.f_bytes(2, b"") and _attr_s(b""),
+                 container=_attr_s(b""),
                  shared_name=_attr_s(b""))
         self.add(name + "/read", "Identity", [name], T=tp.attr_type(dt))
         self.variables[name] = (dt, list(arr.shape),
