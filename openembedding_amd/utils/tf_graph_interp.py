@@ -1,9 +1,18 @@
-"""Numpy interpreter for exported GraphDefs (test-side, TF-free).
+"""Numpy interpreter for exported SavedModel GraphDefs (TF-free).
 
 Executes the op set openembedding_amd.export_tf emits, decoding the wire
-format with the same generic decoder the writer round-trips through. An
-unknown op or malformed attr fails loudly — that is the point: it is an
-independent consumer of the export."""
+format with the generic decoder in utils/tfproto.py. An unknown op or
+malformed attr fails loudly — it is an INDEPENDENT consumer of the export
+(the round-trip tests run predictions through it against the live torch
+engine), and doubles as a way to smoke-test an export on machines without
+TensorFlow:
+
+    from openembedding_amd.utils.tf_graph_interp import GraphInterp
+    g = GraphInterp("/path/to/saved_model")
+    ins, outs = g.signature_io()
+    logits, probs = g.run({"dense": d, "sparse": s},
+                          [outs["logits"], outs["probabilities"]])
+"""
 
 from __future__ import annotations
 

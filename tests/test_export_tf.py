@@ -41,7 +41,7 @@ def _train_some(model, steps=2, batch=32):
 
 def _check_export(model, tmp_path, rtol=2e-4, atol=2e-5):
     from openembedding_amd.export_tf import export_saved_model
-    from _tf_graph_interp import GraphInterp
+    from openembedding_amd.utils.tf_graph_interp import GraphInterp
     from openembedding_amd.models import synthetic_batch
 
     path = str(tmp_path / "saved_model")
@@ -107,7 +107,7 @@ def test_bundle_crc_detects_corruption(tmp_path):
     blob = bytearray(open(data_file, "rb").read())
     blob[3] ^= 0xFF
     open(data_file, "wb").write(bytes(blob))
-    from _tf_graph_interp import parse_saved_model
+    from openembedding_amd.utils.tf_graph_interp import parse_saved_model
     with pytest.raises(ValueError, match="crc"):
         parse_saved_model(path)
 
