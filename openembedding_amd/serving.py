@@ -104,6 +104,69 @@ class ServedModel:
                           for vid, sh in shards.items()}
         self.status = ModelStatus.NORMAL
 
+    def export_block(self, variable_id: int, start: int, count: int):
+        """Row block [start, start+count) of a variable in export order —
+        the replica-to-replica restore wire (reference
+        EmbeddingRestoreOperator.cpp:19-106 pulled shard content from live
+        replicas in server_block_num_items batches; here the batches ride
+        HTTP). The export snapshot is cached on first use (tables are
+        read-only while serving)."""
+        if not hasattr(self, "_export_cache"):
+            self._export_cache = {}
+        blk = self._export_cache.get(variable_id)
+        if blk is None:
+            sh = self.variables[variable_id].shard
+            keys, w, _s = sh.export_rows(include_state=False)
+            blk = (keys.cpu(), w.cpu())
+            self._export_cache[variable_id] = blk
+        keys, w = blk
+        end = min(start + count, keys.numel())
+        return {"total": int(keys.numel()),
+                "keys": keys[start:end].tolist(),
+                "weights": w[start:end].tolist()}
+
+    def load_from_replica(self, base_url: str, sign: str,
+                          device: str = "cpu",
+                          block: int = 65536) -> None:
+        """Coordinated restore: rebuild this model's tables by paging rows
+        out of a LIVE replica (no dump needed — the reference's
+        restore-from-replica path for a replacement server)."""
+        import requests
+
+        meta = requests.get(f"{base_url}/models/{sign}", timeout=10).json()
+        self.meta = {"model_sign": meta["model_sign"],
+                     "variables": meta["variables"], "version": "0.2"}
+        shard_cls = VariableShard
+        if device.startswith("cuda"):
+            from .core.variable_gpu import HipVariableShard
+            shard_cls = HipVariableShard
+        shards: Dict[int, VariableShard] = {}
+        for mvar in meta["variables"]:
+            vm = VariableMeta(variable_id=mvar["variable_id"],
+                              embedding_dim=mvar["embedding_dim"],
+                              vocabulary_size=mvar["vocabulary_size"])
+            sh = shard_cls(vm, shard_id=0, shard_num=1, device=device)
+            start = 0
+            while True:
+                r = requests.get(
+                    f"{base_url}/models/{sign}/variables/"
+                    f"{vm.variable_id}/rows",
+                    params={"start": start, "count": block},
+                    timeout=60).json()
+                if r["keys"]:
+                    kt = torch.tensor(r["keys"], dtype=torch.int64,
+                                      device=sh.device)
+                    wt = torch.tensor(r["weights"], dtype=torch.float32,
+                                      device=sh.device)
+                    sh.import_rows(kt, wt)
+                start += block
+                if start >= r["total"]:
+                    break
+            shards[vm.variable_id] = sh
+        self.variables = {vid: ServedVariable(sh)
+                          for vid, sh in shards.items()}
+        self.status = ModelStatus.NORMAL
+
     def describe(self) -> dict:
         return {
             "model_sign": self.sign,
@@ -189,6 +252,34 @@ class ModelController:
             if model.status == ModelStatus.ERROR:
                 self.manager._remove(sign)
                 raise RuntimeError(f"load failed: {model.error}")
+        else:
+            threading.Thread(target=_load, daemon=True).start()
+        return model
+
+    def restore_model_from_replica(self, base_url: str, sign: str,
+                                   wait: bool = True) -> ServedModel:
+        """Coordinated restore from a live replica (reference
+        EmbeddingRestoreOperator replica path / server --restore): page
+        the rows out of ``base_url`` instead of a dump URI."""
+        existing = self.manager.get(sign)
+        if existing is not None and existing.status != ModelStatus.ERROR:
+            raise ValueError(f"model {sign!r} already exists")
+        model = ServedModel(sign, f"replica://{base_url}")
+        self.manager._register(model)
+
+        def _load():
+            try:
+                model.load_from_replica(base_url, sign,
+                                        device=self.manager.device)
+            except Exception as e:  # noqa: BLE001
+                model.status = ModelStatus.ERROR
+                model.error = repr(e)
+
+        if wait:
+            _load()
+            if model.status == ModelStatus.ERROR:
+                self.manager._remove(sign)
+                raise RuntimeError(f"replica restore failed: {model.error}")
         else:
             threading.Thread(target=_load, daemon=True).start()
         return model
@@ -298,9 +389,12 @@ def make_app(controller: Optional[ModelController] = None,
     app.state.controller = controller
 
     class CreateModelReq(BaseModel):
-        model_uri: str
+        model_uri: Optional[str] = None
         sign: Optional[str] = None
         wait: bool = True
+        # coordinated restore: rebuild from a live replica instead of a
+        # dump (requires sign; reference server --restore semantics)
+        from_replica: Optional[str] = None
 
     class PullReq(BaseModel):
         indices: list
@@ -308,13 +402,35 @@ def make_app(controller: Optional[ModelController] = None,
     @app.post("/models")
     def create_model(req: CreateModelReq):
         try:
-            m = controller.create_model(req.model_uri, sign=req.sign,
-                                        wait=req.wait)
+            if req.from_replica:
+                if not req.sign:
+                    raise HTTPException(status_code=422,
+                                        detail="from_replica needs sign")
+                m = controller.restore_model_from_replica(
+                    req.from_replica, req.sign, wait=req.wait)
+            else:
+                if not req.model_uri:
+                    raise HTTPException(status_code=422,
+                                        detail="model_uri required")
+                m = controller.create_model(req.model_uri, sign=req.sign,
+                                            wait=req.wait)
         except ValueError as e:
             raise HTTPException(status_code=409, detail=str(e))
         except (FileNotFoundError, RuntimeError) as e:
             raise HTTPException(status_code=400, detail=str(e))
         return m.describe()
+
+    @app.get("/models/{sign}/variables/{variable_id}/rows")
+    def export_rows(sign: str, variable_id: int, start: int = 0,
+                    count: int = 65536):
+        """Replica-restore wire: row blocks in export order."""
+        m = controller.manager.get(sign)
+        if m is None or m.status != ModelStatus.NORMAL:
+            raise HTTPException(status_code=404, detail=f"model {sign!r}")
+        if variable_id not in m.variables:
+            raise HTTPException(status_code=404,
+                                detail=f"variable {variable_id}")
+        return m.export_block(variable_id, start, count)
 
     @app.get("/models")
     def list_models():

@@ -39,6 +39,30 @@ def _start_server(port, dump):
         stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
 
 
+def _start_empty_server(port):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    return subprocess.Popen(
+        [sys.executable, "-m", "openembedding_amd.serving",
+         "--host", "127.0.0.1", "--port", str(port), "--device", "cpu"],
+        cwd=REPO, env=env,
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+
+
+def _wait_http(port, timeout=60):
+    import requests
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        try:
+            r = requests.get(f"http://127.0.0.1:{port}/models", timeout=2)
+            if r.status_code == 200:
+                return True
+        except Exception:  # noqa: BLE001
+            pass
+        time.sleep(0.3)
+    return False
+
+
 def _wait_ready(port, timeout=60):
     import requests
     t0 = time.time()
@@ -108,6 +132,23 @@ def test_failover_kill_restart(tmp_path):
         torch.testing.assert_close(w, want, rtol=1e-5, atol=1e-6)
         for _ in range(4):
             check_pull()
+
+        # coordinated restore: a THIRD replica reconstructs the model from
+        # a LIVE replica over the wire — no dump involved (reference
+        # EmbeddingRestoreOperator.cpp:19-106 replica path)
+        import requests
+        p3 = _free_port()
+        proc3 = _start_empty_server(p3)
+        procs.append(proc3)
+        assert _wait_http(p3)
+        r = requests.post(
+            f"http://127.0.0.1:{p3}/models",
+            json={"from_replica": f"http://127.0.0.1:{ports[1]}",
+                  "sign": sign}, timeout=120)
+        assert r.status_code == 200, r.text
+        restored = ServingClient([f"http://127.0.0.1:{p3}"])
+        w3 = torch.tensor(restored.pull(sign, vid, probe))
+        torch.testing.assert_close(w3, want, rtol=1e-5, atol=1e-6)
     finally:
         for p in procs:
             if p.poll() is None:
