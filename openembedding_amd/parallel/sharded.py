@@ -517,11 +517,13 @@ class ShardedVariable:
 
     # ------------------------------------------------------- whole-table pulls
 
-    def pull_dense(self, start: int, count: int) -> torch.Tensor:
+    def pull_dense(self, start: int, count: int,
+                   readonly: bool = False) -> torch.Tensor:
         """Materialize rows [start, start+count) of a bounded-vocab variable on
         every rank (export path; reference save_as_original_model bulk pull,
-        exb.py:506-547). Collective."""
+        exb.py:506-547 — which pulled through the TRAINING path, so missing
+        rows instantiate their initializer values). Collective."""
         keys = torch.arange(start, start + count, dtype=torch.int64,
                             device=self.shard.device)
-        out, _ = self.pull(keys, readonly=True)
+        out, _ = self.pull(keys, readonly=readonly)
         return out

@@ -707,14 +707,20 @@ def distributed_model(model: nn.Module, sparse_as_dense_size: int = 64
     return model
 
 
-def sparse_read_as_dense(variable: Variable, vocab: int) -> torch.Tensor:
+def sparse_read_as_dense(variable: Variable, vocab: int,
+                         readonly: bool = False) -> torch.Tensor:
     """Materialize rows [0, vocab) of a variable on every rank
-    (reference sparse_read_as_dense / save_as_original_model bulk pull)."""
+    (reference sparse_read_as_dense / save_as_original_model bulk pull).
+
+    Default readonly=False matches the reference: its export pulled
+    through the TRAINING path, so never-touched rows materialize with
+    their initializer values (exb.py:529-538), not zeros."""
     chunks = []
     step = max(1, (1 << 20) // max(1, variable.embedding_dim))
     for start in range(0, vocab, step):
         n = min(step, vocab - start)
-        chunks.append(variable.sharded.pull_dense(start, n))
+        chunks.append(variable.sharded.pull_dense(start, n,
+                                                  readonly=readonly))
     return torch.cat(chunks)
 
 
