@@ -166,7 +166,7 @@ void k_cin_dw(const cbf16* __restrict__ dzt,   // [O, Np]
               const cbf16* __restrict__ xkt,   // [H, Np]
               float* __restrict__ dw,          // [O, F*H] fp32 accum
               long Np, long F, long H, long O, long n_split) {
-    __shared__ cbf16 btile[128 * 40];   // [h][32 n], stride 40: bank-clean
+    __shared__ cbf16 btile[128 * 72];   // [h][64 n], stride 72 (16B rows)
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
     const long f = blockIdx.x;
@@ -185,19 +185,20 @@ void k_cin_dw(const cbf16* __restrict__ dzt,   // [O, Np]
     const int Hp16 = (int)((H + 15) / 16 * 16);  // partial h-tile zeroed
     const int Hi = (int)H;
     const cbf16* pa = dzt + (o0 + (lane & 15)) * Np + koff;
-    // A-stream pipelined across n-chunks: the next chunk's dZt fragment
-    // loads behind the current chunk's B' build + barrier
+    // A-stream pipelined across 64-wide n-chunks (2 MFMA k-steps each):
+    // the next step's dZt fragment loads behind the build + barrier
     bf16x8 a_cur;
     if (o0 < O) a_cur = cin_ld_frag(pa + nbeg);
-    for (long n0s = nbeg; n0s < nend; n0s += 32) {
+    for (long n0s = nbeg; n0s < nend; n0s += 64) {
         __syncthreads();
-        // B' chunk: [H][32] = xkt rows * x0 row f (broadcast over h),
+        // B' chunk: [H][64] = xkt rows * x0 row f (broadcast over h),
         // packed: two bf16x8 loads -> one bf16x8 store per 8 elements
-        for (int g = (int)threadIdx.x; g < Hp16 * 4;
+        const long nrem = nend - n0s;
+        for (int g = (int)threadIdx.x; g < Hp16 * 8;
              g += (int)blockDim.x) {
-            int h = g >> 2, no = (g & 3) << 3;
+            int h = g >> 3, no = (g & 7) << 3;
             bf16x8 vv;
-            if (h < Hi) {
+            if (h < Hi && no < nrem) {
                 bf16x8 xk8 = cin_ld_frag(xkt + h * Np + n0s + no);
                 bf16x8 x08 = cin_ld_frag(x0t + f * Np + n0s + no);
                 #pragma unroll
@@ -207,19 +208,23 @@ void k_cin_dw(const cbf16* __restrict__ dzt,   // [O, Np]
                 #pragma unroll
                 for (int j = 0; j < 8; ++j) vv[j] = (__bf16)0.f;
             }
-            *reinterpret_cast<bf16x8*>(btile + h * 40 + no) = vv;
+            *reinterpret_cast<bf16x8*>(btile + h * 72 + no) = vv;
         }
         __syncthreads();
         if (o0 >= O) continue;
-        bf16x8 a = a_cur;
-        if (n0s + 32 < nend) a_cur = cin_ld_frag(pa + n0s + 32);
-        #pragma unroll
-        for (int t = 0; t < 8; ++t) {
-            if (t * 16 >= Hi) break;
-            bf16x8 b = cin_ld_frag(btile + (t * 16 + (lane & 15)) * 40
-                                   + koff);
-            acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t],
-                                                             0, 0, 0);
+        const long ks_lim = (nrem < 64) ? nrem : 64;
+        for (long ks = 0; ks < ks_lim; ks += 32) {
+            bf16x8 a = a_cur;
+            if (n0s + ks + 32 < nend)
+                a_cur = cin_ld_frag(pa + n0s + ks + 32);
+            #pragma unroll
+            for (int t = 0; t < 8; ++t) {
+                if (t * 16 >= Hi) break;
+                bf16x8 b = cin_ld_frag(btile + (t * 16 + (lane & 15)) * 72
+                                       + koff + ks);
+                acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a, b, acc[t], 0, 0, 0);
+            }
         }
     }
     if (o0 >= O) return;
@@ -308,6 +313,15 @@ void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
 
     const long koff = (lane >> 4) * 8;
     const int htiles = (Hi + 15) / 16;
+    // Wt fragment prefetch: the NEXT field's loads are issued before the
+    // consume barrier, so their L2 latency hides behind the consume phase
+    bf16x8 a_pre[4];
+    const bool owns_tile = wave < htiles;   // one k-tile per wave per field
+    if (Op == 128 && owns_tile) {
+        const cbf16* pa0 = wt + ((long)wave * 16 + (lane & 15)) * Op + koff;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) a_pre[j] = cin_ld_frag(pa0 + 32 * j);
+    }
     for (int f = 0; f < Fi; ++f) {
         __syncthreads();   // previous consume done before P overwrite
         // this field's P [H x cols]: wave w owns k-tiles w, w+8, ...
@@ -315,10 +329,15 @@ void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
             const long k0 = (long)f * Hi + kt * 16;
             const cbf16* pa = wt + (k0 + (lane & 15)) * Op + koff;
             bf16x8 a[4];
-            if (Op == 128) {   // Wt fragments hoisted across col subtiles
-                #pragma unroll
-                for (int j = 0; j < 4; ++j)
-                    a[j] = cin_ld_frag(pa + 32 * j);
+            if (Op == 128) {
+                if (kt == wave) {
+                    #pragma unroll
+                    for (int j = 0; j < 4; ++j) a[j] = a_pre[j];
+                } else {
+                    #pragma unroll
+                    for (int j = 0; j < 4; ++j)
+                        a[j] = cin_ld_frag(pa + 32 * j);
+                }
             }
             #pragma unroll
             for (int cs = 0; cs < CDX_COLS / 16; ++cs) {
@@ -348,6 +367,12 @@ void k_cin_dx(const float* __restrict__ doutp,  // [N, O] fp32
                     if (h < Hi) pchunk[h * CDX_PLD + c] = acc[r];
                 }
             }
+        }
+        if (Op == 128 && owns_tile && f + 1 < Fi) {
+            const cbf16* pan = wt + ((long)(f + 1) * Hi + wave * 16
+                                     + (lane & 15)) * Op + koff;
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) a_pre[j] = cin_ld_frag(pan + 32 * j);
         }
         __syncthreads();
         // consume — fused, balanced, no atomics: thread (hr, c) walks
