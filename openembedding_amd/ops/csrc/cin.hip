@@ -81,6 +81,13 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
     #pragma unroll
     for (int t = 0; t < 8; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
+    // first chunk's W fragment issued before any barrier; later chunks'
+    // first fragments are issued during the previous chunk's MFMA phase,
+    // so the W stream's L2 latency hides behind build + barriers
+    const long koff0 = (lane >> 4) * 8;
+    bf16x8 a_pre;
+    if (o0 < O)
+        a_pre = cin_ld_frag(w + (o0 + (lane & 15)) * Kp + koff0);
     for (long k0 = 0; k0 < Kp; k0 += CIN_KC) {
         __syncthreads();
         if (threadIdx.x < CIN_KC) {
@@ -111,13 +118,14 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
         __syncthreads();
         if (o0 >= O) continue;      // narrow O: spare waves still build V
         const long kc_lim = (Kp - k0 < CIN_KC) ? (Kp - k0) : CIN_KC;
-        const long koff = (lane >> 4) * 8;
-        const cbf16* pa = w + (o0 + (lane & 15)) * Kp + k0 + koff;
-        const cbf16* pb = vtile + (lane & 15) * CIN_VLD + koff;
-        bf16x8 a0 = cin_ld_frag(pa);
+        const cbf16* pa = w + (o0 + (lane & 15)) * Kp + k0 + koff0;
+        const cbf16* pb = vtile + (lane & 15) * CIN_VLD + koff0;
+        bf16x8 a0 = a_pre;
         for (long ks = 0; ks < kc_lim; ks += 32) {
             bf16x8 a1;
             if (ks + 32 < kc_lim) a1 = cin_ld_frag(pa + ks + 32);
+            else if (k0 + CIN_KC < Kp)      // next chunk's first fragment
+                a1 = cin_ld_frag(pa + CIN_KC);
             #pragma unroll
             for (int t = 0; t < 8; ++t) {
                 bf16x8 b = cin_ld_frag(pb + t * 16 * CIN_VLD + ks);
@@ -126,6 +134,7 @@ void k_cin_fwd(const float* __restrict__ x0p,   // [N, F]  (d-leading cols)
             }
             a0 = a1;
         }
+        a_pre = a0;                         // carries the prefetched frag
     }
 
     // epilogue: each lane holds 4 consecutive o values per subtile
