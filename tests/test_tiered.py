@@ -140,3 +140,76 @@ def test_clear_resets_both_tiers():
     assert t.num_rows == 0 and not t._host_of
     assert torch.equal(t.pull_readonly(torch.tensor([1])),
                        torch.zeros(1, 4))
+
+
+def test_tiered_world2_matches_untiered(tmp_path):
+    """CPU capacity tier at world 2 (gloo): a tiered run over the sharded
+    route must match an untiered one bit-for-bit (the tier is a storage
+    policy, not a semantics change)."""
+    import os
+    import socket
+
+    import torch.distributed as dist
+    import torch.multiprocessing as mp
+
+    def _free_port():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        return p
+
+    port = _free_port()
+    mp.spawn(_tier_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
+    for v in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
+              "MASTER_PORT"):
+        os.environ.pop(v, None)
+    a = torch.load(tmp_path / "after_tiered_0.pt", weights_only=True)
+    b = torch.load(tmp_path / "after_plain_0.pt", weights_only=True)
+    torch.testing.assert_close(a, b)
+
+
+def _tier_worker(rank, world, port, tmp):
+    import os
+
+    import torch.distributed as dist
+
+    import openembedding_amd as oe
+    from openembedding_amd.context import Context
+
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    g = torch.Generator().manual_seed(31)
+    batches = [(torch.randint(0, 40000, (600,), generator=g,
+                              dtype=torch.int64),
+                torch.randn(600, 4, generator=g)) for _ in range(6)]
+    # ~3.4k unique keys vs the 1024-row cache floor -> real evictions
+    probe = torch.unique(torch.cat([k for k, _ in batches]))
+    results = {}
+    for label, cfg in (("tiered", "server:\n  cache_size: 1\n"),
+                       ("plain", "")):
+        old = oe.flags.config
+        oe.flags.config = cfg
+        try:
+            ctx = Context(device="cpu")
+            st = ctx.create_storage()
+            var = st.create_variable(1 << 63, 4)   # hash mode (tier req.)
+            var.set_initializer("uniform", minval=-1, maxval=1)
+            var.set_optimizer("adagrad", learning_rate=0.1)
+            for keys, grads in batches:
+                out, h = var.pull(keys)
+                var.push(h, grads)
+                st.update_weights()
+            after, _ = var.pull(probe, readonly=True)
+            results[label] = after
+        finally:
+            oe.flags.config = old
+    if rank == 0:
+        torch.save(results["tiered"], os.path.join(tmp, "after_tiered_0.pt"))
+        torch.save(results["plain"], os.path.join(tmp, "after_plain_0.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
