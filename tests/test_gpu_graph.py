@@ -90,11 +90,14 @@ def test_graph_replay_matches_eager():
                            for p in model2.dnn.parameters()])
     # float atomics (head backward dw/db, reduce-by-key) make two separate
     # runs differ at rounding level, and 12 Adagrad steps amplify a 1-ulp
-    # grad flip into ~1e-4-scale weight deltas (observed flaking at 1e-4);
-    # a replay bug (e.g. a non-replayed scratch reset) shows up orders of
-    # magnitude above 1e-3
-    assert torch.allclose(got_rows, ref_rows, atol=1e-3, rtol=1e-3)
-    assert torch.allclose(got_dense, ref_dense, atol=1e-3, rtol=1e-3)
+    # grad flip into ~1e-4..1e-3-scale weight deltas (observed flaking at
+    # 1e-4 and, rarely, at 1e-3 depending on which tests ran before); a
+    # replay bug (e.g. a non-replayed scratch reset) shows up orders of
+    # magnitude above this
+    assert torch.allclose(got_rows, ref_rows, atol=5e-3, rtol=5e-3), \
+        float((got_rows - ref_rows).abs().max())
+    assert torch.allclose(got_dense, ref_dense, atol=5e-3, rtol=5e-3), \
+        float((got_dense - ref_dense).abs().max())
 
 
 def test_graph_replay_hash_mode_with_reservation():
