@@ -37,7 +37,7 @@ void emb_gather_init(float*, float*, long, long, const i64*,
                      float*, int, float, float, float, u64, const float*,
                      const int*, hipStream_t_);
 void emb_reduce_by_inverse(const i64*, const float*, long, long, float*, u64*,
-                           long, float*, u64*, hipStream_t_);
+                           long, hipStream_t_);
 void emb_apply_optimizer(int, float*, float*, long, long, const i64*, long,
                          const float*, const u64*, const float*, const int*,
                          hipStream_t_);
@@ -245,22 +245,9 @@ std::tuple<torch::Tensor, torch::Tensor> reduce_by_inverse(
     long dim = grads.size(1);
     auto ugrads = torch::empty({u, dim}, grads.options());
     auto counts = torch::empty({u}, grads.options().dtype(torch::kInt64));
-    float* part = nullptr;
-    u64* cpart = nullptr;
-    torch::Tensor part_t, cpart_t;
-    if (dim <= 128) {
-        // flush-partition scratch (4x): hot-row atomic contention fix —
-        // transient, pooled by the caching allocator / capture pool
-        part_t = torch::empty({4 * u, dim}, grads.options());
-        cpart_t = torch::empty({4 * u},
-                               grads.options().dtype(torch::kInt64));
-        part = part_t.data_ptr<float>();
-        cpart = (u64*)cpart_t.data_ptr<i64>();
-    }
     emb_reduce_by_inverse(inverse.data_ptr<i64>(), grads.data_ptr<float>(), n,
                           dim, ugrads.data_ptr<float>(),
-                          (u64*)counts.data_ptr<i64>(), u, part, cpart,
-                          cur_stream());
+                          (u64*)counts.data_ptr<i64>(), u, cur_stream());
     return {ugrads, counts};
 }
 

@@ -463,22 +463,12 @@ __global__ void k_gather_init(float* __restrict__ weights,
 // version did `dim` serial LDS atomics per element with strided reads —
 // 112us avg in the DeepFM profile; this shape removes both problems).
 // Lane 0 of the group probes the LDS hash and broadcasts the slot.
-// Flush partitioning (round 2): even with the LDS pre-aggregation, hot
-// uids are atomically flushed by every block — rows of 3-vocab fields are
-// contended by all ~400 blocks. The flush goes to one of RB_PARTS
-// partition buffers (block % RB_PARTS) and a cheap merge pass sums them:
-// hot-row contention drops by the partition count for ~2 extra passes
-// over small buffers.
-#define RB_PARTS 4
-
 template <int H, int G>
 __global__ void k_reduce_lds(const i64* __restrict__ inverse,
                              const float* __restrict__ grads,
-                             long n, long dim, long u,
-                             float* __restrict__ part,   // [RB_PARTS, u, dim]
-                             u64* __restrict__ cpart) {  // [RB_PARTS, u]
-    float* __restrict__ ugrads = part + (blockIdx.x % RB_PARTS) * u * dim;
-    u64* __restrict__ counts = cpart + (blockIdx.x % RB_PARTS) * u;
+                             long n, long dim,
+                             float* __restrict__ ugrads,
+                             u64* __restrict__ counts) {
     extern __shared__ char smem[];
     int* luid = (int*)smem;
     int* lcnt = (int*)(smem + H * 4);
@@ -547,25 +537,6 @@ __global__ void k_reduce_lds(const i64* __restrict__ inverse,
         float* ug = ugrads + (u64)uid * dim;
         const float* acc = lacc + (u64)h * dim;
         for (long j = lane; j < dim; j += G) atomicAdd(&ug[j], acc[j]);
-    }
-}
-
-__global__ void k_reduce_merge(const float* __restrict__ part,
-                               const u64* __restrict__ cpart, long u,
-                               long dim, float* __restrict__ ugrads,
-                               u64* __restrict__ counts) {
-    long e = (long)blockIdx.x * blockDim.x + threadIdx.x;
-    if (e < u * dim) {
-        float s = 0.f;
-        #pragma unroll
-        for (int p = 0; p < RB_PARTS; ++p) s += part[p * u * dim + e];
-        ugrads[e] = s;
-    }
-    if (e < u) {
-        u64 c = 0;
-        #pragma unroll
-        for (int p = 0; p < RB_PARTS; ++p) c += cpart[p * u + e];
-        counts[e] = c;
     }
 }
 
@@ -1162,43 +1133,31 @@ void emb_gather_init(float* weights, float* state, long dim, long sd,
 
 void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
                            long dim, float* ugrads, u64* counts, long u,
-                           float* part, u64* cpart, hipStream_t stream) {
-    if (u == 0) return;
-    if (n == 0 || !part) {
-        // no partition scratch (caller opted out / large-dim path):
-        // legacy direct-accumulation layout
-        fill_f32(ugrads, u * dim, 0.0f, stream);
-        fill_u64(counts, u, 0ull, stream);
-        if (n == 0) return;
-    }
+                           hipStream_t stream) {
+    fill_f32(ugrads, u * dim, 0.0f, stream);
+    fill_u64(counts, u, 0ull, stream);
+    if (n == 0) return;
     int grid = cdiv(n, BLOCK);
-    if (dim <= 128 && part) {
-        fill_f32(part, (long)RB_PARTS * u * dim, 0.0f, stream);
-        fill_u64(cpart, (long)RB_PARTS * u, 0ull, stream);
-        if (dim <= 16) {
-            const int H = 512;
-            size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 40 KiB
-            k_reduce_lds<H, 16><<<grid, BLOCK, smem, stream>>>(
-                inverse, grads, n, dim, u, part, cpart);
-        } else if (dim <= 64) {
-            const int H = 128;
-            size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 34 KiB
-            k_reduce_lds<H, 64><<<grid, BLOCK, smem, stream>>>(
-                inverse, grads, n, dim, u, part, cpart);
-        } else {
-            const int H = 64;
-            size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 33 KiB
-            k_reduce_lds<H, 64><<<grid, BLOCK, smem, stream>>>(
-                inverse, grads, n, dim, u, part, cpart);
-        }
-        long total = u * dim > u ? u * dim : u;
-        k_reduce_merge<<<grid1d(total), BLOCK, 0, stream>>>(
-            part, cpart, u, dim, ugrads, counts);
-        return;
+    if (dim <= 16) {
+        const int H = 512;
+        size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 40 KiB
+        k_reduce_lds<H, 16><<<grid, BLOCK, smem, stream>>>(
+            inverse, grads, n, dim, ugrads, counts);
+    } else if (dim <= 64) {
+        const int H = 128;
+        size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 34 KiB
+        k_reduce_lds<H, 64><<<grid, BLOCK, smem, stream>>>(
+            inverse, grads, n, dim, ugrads, counts);
+    } else if (dim <= 128) {
+        const int H = 64;
+        size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 33 KiB
+        k_reduce_lds<H, 64><<<grid, BLOCK, smem, stream>>>(
+            inverse, grads, n, dim, ugrads, counts);
+    } else {
+        k_reduce_grads<<<grid1d(n * dim), BLOCK, 0, stream>>>(inverse, grads,
+                                                              n, dim, ugrads);
+        k_reduce_counts<<<grid1d(n), BLOCK, 0, stream>>>(inverse, n, counts);
     }
-    k_reduce_grads<<<grid1d(n * dim), BLOCK, 0, stream>>>(inverse, grads,
-                                                          n, dim, ugrads);
-    k_reduce_counts<<<grid1d(n), BLOCK, 0, stream>>>(inverse, n, counts);
 }
 
 #define LAUNCH_OPT(G, OPT)                                                  \
