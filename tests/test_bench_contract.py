@@ -78,3 +78,23 @@ def test_prefetch_flag():
     j = _run_bench("--prefetch", "--graph", "off")
     assert j["config"]["prefetch"] is True
     assert j["value"] > 0
+
+
+@pytest.mark.timeout(420)
+def test_torchrun_world2_contract():
+    """The driver's SCALE launch shape: torchrun --nproc-per-node N
+    bench.py --gpus N. Runs on CPU/gloo here; rank 0 must print the one
+    JSON line with whole-job value and multi-rank config."""
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr=127.0.0.1",
+         "--master-port=29655", "bench.py", "--gpus", "2", "--steps", "2",
+         "--warmup", "1", "--batch", "32", "--data-pool", "2"],
+        cwd=REPO, capture_output=True, text=True, timeout=390)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines()
+            if ln.strip().startswith("{") and '"metric"' in ln][-1]
+    j = json.loads(line)
+    assert j["n_gpus"] == 2
+    assert j["config"]["global_batch"] == 64          # whole-job
+    assert "all_to_all" in j["config"]["parallelism"]
