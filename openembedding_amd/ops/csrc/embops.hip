@@ -1138,19 +1138,23 @@ void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
     fill_u64(counts, u, 0ull, stream);
     if (n == 0) return;
     int grid = cdiv(n, BLOCK);
+    // H must cover the block's worst-case distinct uids (BLOCK = 256):
+    // an undersized hash overflows most elements to DIRECT global atomics
+    // — measured 360 us/call at dim 65 with H=64 (75% overflow) vs the
+    // LDS-aggregated path's ~tens of us (profiles/bench_deepfm_dim64)
     if (dim <= 16) {
         const int H = 512;
         size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 40 KiB
         k_reduce_lds<H, 16><<<grid, BLOCK, smem, stream>>>(
             inverse, grads, n, dim, ugrads, counts);
-    } else if (dim <= 64) {
-        const int H = 128;
-        size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 34 KiB
+    } else if (dim <= 96) {
+        const int H = 256;
+        size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 98 KiB
         k_reduce_lds<H, 64><<<grid, BLOCK, smem, stream>>>(
             inverse, grads, n, dim, ugrads, counts);
     } else if (dim <= 128) {
-        const int H = 64;
-        size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 33 KiB
+        const int H = 128;
+        size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 66 KiB
         k_reduce_lds<H, 64><<<grid, BLOCK, smem, stream>>>(
             inverse, grads, n, dim, ugrads, counts);
     } else {
