@@ -1148,9 +1148,12 @@ void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
         k_reduce_lds<H, 16><<<grid, BLOCK, smem, stream>>>(
             inverse, grads, n, dim, ugrads, counts);
     } else if (dim <= 96) {
+        // G=32: halves the per-group serial element chain vs G=64 (each
+        // group walks its G elements in order); ceil(96/32)=3 fragments
+        // fit the 4-slot staging
         const int H = 256;
         size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 98 KiB
-        k_reduce_lds<H, 64><<<grid, BLOCK, smem, stream>>>(
+        k_reduce_lds<H, 32><<<grid, BLOCK, smem, stream>>>(
             inverse, grads, n, dim, ugrads, counts);
     } else if (dim <= 128) {
         const int H = 128;
