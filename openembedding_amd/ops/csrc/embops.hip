@@ -463,7 +463,7 @@ __global__ void k_gather_init(float* __restrict__ weights,
 // version did `dim` serial LDS atomics per element with strided reads —
 // 112us avg in the DeepFM profile; this shape removes both problems).
 // Lane 0 of the group probes the LDS hash and broadcasts the slot.
-template <int H, int G>
+template <int H, int G, int GF = 4>
 __global__ void k_reduce_lds(const i64* __restrict__ inverse,
                              const float* __restrict__ grads,
                              long n, long dim,
@@ -493,10 +493,10 @@ __global__ void k_reduce_lds(const i64* __restrict__ inverse,
         long e = base + t;
         uids[t] = (e < n) ? (int)inverse[e] : -1;
     }
-    float gfrag[2][4];  // up to 4 grad values per lane (dim <= 4*G)
+    float gfrag[2][GF];  // up to GF grad values per lane (dim <= GF*G)
     const int nj = (int)((dim + G - 1) / G);
     #pragma unroll
-    for (int j = 0; j < 4; ++j)
+    for (int j = 0; j < GF; ++j)
         gfrag[0][j] = (j < nj && base < n && lane + j * G < dim)
                           ? grads[(u64)base * dim + lane + j * G] : 0.0f;
     for (int t = 0; t < G; ++t) {
@@ -505,7 +505,7 @@ __global__ void k_reduce_lds(const i64* __restrict__ inverse,
         long en = base + t + 1;
         if (en < n) {
             #pragma unroll
-            for (int j = 0; j < 4; ++j)
+            for (int j = 0; j < GF; ++j)
                 gfrag[(t + 1) & 1][j] = (j < nj && lane + j * G < dim)
                     ? grads[(u64)en * dim + lane + j * G] : 0.0f;
         }
@@ -525,7 +525,7 @@ __global__ void k_reduce_lds(const i64* __restrict__ inverse,
         float* dstb = (h >= 0) ? (lacc + (u64)h * dim)
                                : (ugrads + (u64)uid * dim);
         #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < GF; ++j)
             if (j < nj && lane + j * G < dim)
                 atomicAdd(&dstb[lane + j * G], gfrag[t & 1][j]);
     }
@@ -1148,12 +1148,12 @@ void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
         k_reduce_lds<H, 16><<<grid, BLOCK, smem, stream>>>(
             inverse, grads, n, dim, ugrads, counts);
     } else if (dim <= 96) {
-        // G=32: halves the per-group serial element chain vs G=64 (each
-        // group walks its G elements in order); ceil(96/32)=3 fragments
-        // fit the 4-slot staging
+        // G=16: quarter-length serial element chain vs G=64 (each group
+        // walks its G elements in order); GF=8 staging slots cover
+        // ceil(96/16)=6 fragments
         const int H = 256;
         size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 98 KiB
-        k_reduce_lds<H, 32><<<grid, BLOCK, smem, stream>>>(
+        k_reduce_lds<H, 16, 8><<<grid, BLOCK, smem, stream>>>(
             inverse, grads, n, dim, ugrads, counts);
     } else if (dim <= 128) {
         const int H = 128;
