@@ -1143,9 +1143,11 @@ void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
     // — measured 360 us/call at dim 65 with H=64 (75% overflow) vs the
     // LDS-aggregated path's ~tens of us (profiles/bench_deepfm_dim64)
     if (dim <= 16) {
+        // G=8: halves the serial per-group element chain vs G=16
+        // (nj = ceil(16/8) = 2 fragments)
         const int H = 512;
         size_t smem = H * 8 + (size_t)H * dim * 4;   // <= 40 KiB
-        k_reduce_lds<H, 16><<<grid, BLOCK, smem, stream>>>(
+        k_reduce_lds<H, 8><<<grid, BLOCK, smem, stream>>>(
             inverse, grads, n, dim, ugrads, counts);
     } else if (dim <= 96) {
         // G=16: quarter-length serial element chain vs G=64 (each group
