@@ -27,7 +27,7 @@ Public layout:
   openembedding_amd.inject      -- global nn.Embedding auto-patch (ref laboratory/inject)
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 version = __version__
 

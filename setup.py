@@ -34,7 +34,7 @@ ext = CUDAExtension(
 
 setup(
     name="openembedding_amd",
-    version="0.1.0",
+    version="0.2.0",
     packages=find_packages(include=["openembedding_amd",
                                     "openembedding_amd.*"]),
     ext_modules=[ext],
