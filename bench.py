@@ -128,10 +128,13 @@ def measure(args, embed, amp_mode, steps, warmup, pool, device, world, rank):
             if float(ok.item()) == 0.0:
                 graph = None
 
-    # prefetch: explicit --prefetch, or the default at world>1 when the
-    # step is not graph-captured (the reference's pipeline overlap)
-    want_prefetch = args.prefetch or (world > 1 and graph is None
-                                      and args.model != "lr")
+    # prefetch: opt-in only. Measured (gpurun r2ag): the prefetch stream
+    # costs ~80 us/step of stream-switch overhead at world 1 — more than
+    # the ~20 us padded-wire time it could hide at world>1 (the
+    # reference's prefetch paid off against millisecond RPC latencies;
+    # xGMI is not that). Graph capture is the multi-rank fast path; plain
+    # eager is the fallback.
+    want_prefetch = args.prefetch
     if want_prefetch and graph is None:
         def batch_stream(total):
             for i in range(total):
