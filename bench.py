@@ -18,8 +18,8 @@ Baseline: the reference's published DeepFM dim9 throughput on 8x Tesla T4
 Multi-rank: the embedding all-to-all runs the padded sync-free route
 (parallel/sharded.py _pull_remote_padded) so the whole step — collectives
 included — is hipGraph-capturable; if capture is unavailable the bench
-falls back to eager with the pulling() prefetch pipeline overlapping the
-next batch's pull with this batch's compute.
+falls back to plain eager (the pulling() prefetch pipeline is opt-in via
+--prefetch: measured net-negative against the ~20 us xGMI wire).
 """
 
 import argparse
