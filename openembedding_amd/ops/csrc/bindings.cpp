@@ -74,7 +74,7 @@ void emb_scatter_out(const float*, const i64*, long, const int*, long, float*,
 void emb_split_payload(const float*, long, long, const int*, float*, u64*,
                        hipStream_t_);
 void emb_flat_step_scalars(float*, float, float, hipStream_t_);
-void emb_flat_opt(int, void*, float*, float*, float*, const void*,
+void emb_flat_opt(int, void*, float*, float*, float*, void*,
                   const float*, long, int, float, float, float, float,
                   hipStream_t_);
 void emb_flat_adagrad_f32(float*, float*, const float*, long, float, float,

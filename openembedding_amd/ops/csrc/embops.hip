@@ -794,23 +794,28 @@ __device__ __forceinline__ float flat_opt_step(
     return w - lr * mh / (sqrtf(vh) + c2);
 }
 
+// Both variants zero the grad element after consuming it: the next
+// backward's AccumulateGrad lands on a clean buffer, so the caller's
+// zero_grad() is a steady-state no-op (saves one FillFunctor launch per
+// dtype group per captured step).
 template <int OPT>
 __global__ void k_flat_opt_f32(float* __restrict__ p,
                                float* __restrict__ s1,
                                float* __restrict__ s2,
-                               const float* __restrict__ g,
+                               float* __restrict__ g,
                                const float* __restrict__ sc, long n,
                                float lr, float c0, float c1, float c2) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     p[i] = flat_opt_step<OPT>(p[i], g[i], s1, s2, i, sc, lr, c0, c1, c2);
+    g[i] = 0.0f;
 }
 
 template <int OPT>
 __global__ void k_flat_opt_bf16(float* __restrict__ master,
                                 float* __restrict__ s1,
                                 float* __restrict__ s2,
-                                const oebf16* __restrict__ g,
+                                oebf16* __restrict__ g,
                                 oebf16* __restrict__ p,
                                 const float* __restrict__ sc, long n,
                                 float lr, float c0, float c1, float c2) {
@@ -820,6 +825,7 @@ __global__ void k_flat_opt_bf16(float* __restrict__ master,
                                  c0, c1, c2);
     master[i] = m;
     p[i] = (oebf16)m;
+    g[i] = (oebf16)0.0f;
 }
 
 // ---- fused BCE-with-logits (mean) --------------------------------------
@@ -1322,15 +1328,15 @@ void emb_flat_step_scalars(float* sc, float b1, float b2,
     do {                                                                    \
         if (bf16)                                                           \
             k_flat_opt_bf16<OPT><<<grid1d(n), BLOCK, 0, stream>>>(          \
-                master, s1, s2, (const oebf16*)g, (oebf16*)p, sc, n, lr,    \
+                master, s1, s2, (oebf16*)g, (oebf16*)p, sc, n, lr,          \
                 c0, c1, c2);                                                \
         else                                                                \
             k_flat_opt_f32<OPT><<<grid1d(n), BLOCK, 0, stream>>>(          \
-                (float*)p, s1, s2, (const float*)g, sc, n, lr, c0, c1, c2); \
+                (float*)p, s1, s2, (float*)g, sc, n, lr, c0, c1, c2);       \
     } while (0)
 
 void emb_flat_opt(int opt, void* p, float* master, float* s1, float* s2,
-                  const void* g, const float* sc, long n, int bf16,
+                  void* g, const float* sc, long n, int bf16,
                   float lr, float c0, float c1, float c2,
                   hipStream_t stream) {
     if (!n) return;

@@ -331,7 +331,8 @@ def _server_cfg_from_torch(opt: torch.optim.Optimizer) -> Dict:
 
 
 class _FlatDenseOptimizer:
-    """Dense params re-based onto ONE flat buffer: zero_grad is one fill,
+    """Dense params re-based onto ONE flat buffer: zero_grad is a no-op in
+    steady state (the step kernel re-zeroes the grad buffer it consumes),
     the distributed allreduce is one collective on one tensor, and the
     optimizer step is a handful of flat kernels (torch's multi-tensor
     apply was ~60us + ~60us of per-param grad fills per step on the DeepFM
@@ -412,6 +413,7 @@ class _FlatDenseOptimizer:
                                          dtype=torch.float32)
                              if self.kind == "Adam" else None)
         self._t = 0      # host mirror for the torch fallback
+        self._grads_zeroed = True   # flat_grad starts zeroed
         self.params = params
         # p -> its view into the flat grad buffer (re-bound in zero_grad if
         # model.zero_grad(set_to_none=True) detached it; checked in step)
@@ -429,6 +431,15 @@ class _FlatDenseOptimizer:
                 # backwards must accumulate into the flat buffer the
                 # allreduce and step read
                 p.grad = view
+        if self._grads_zeroed:
+            # step() left the flat buffer zeroed (k_flat_opt writes g[i]=0
+            # after consuming it), so the first zero_grad of the next
+            # iteration is a no-op — one fewer fill per dtype group inside
+            # the captured step. The flag is consumed here: a second
+            # zero_grad in the same cycle (e.g. discarding an extra
+            # backward's grads) does the real fill.
+            self._grads_zeroed = False
+            return
         for g in self.groups:
             g["flat_grad"].zero_()
 
@@ -459,6 +470,7 @@ class _FlatDenseOptimizer:
                 self._ext.flat_opt(self.opt_id, g["flat"], g["master"],
                                    g["s1"], g["s2"], g["flat_grad"],
                                    self.step_scalars, self.lr, *cfg)
+            self._grads_zeroed = True   # kernel wrote g[i]=0 after consume
             return
         for g in self.groups:
             grad = g["flat_grad"].to(torch.float32)
@@ -484,6 +496,8 @@ class _FlatDenseOptimizer:
                 w.addcdiv_(mh, vh.sqrt().add_(self.eps), value=-self.lr)
             if g["master"] is not None:
                 g["flat"].copy_(g["master"])
+            g["flat_grad"].zero_()
+        self._grads_zeroed = True
 
     def state_dict(self):
         # detached clones: a caller that keeps training after save must not

@@ -113,3 +113,23 @@ def test_bce_with_logits_int_labels():
     y = torch.tensor([0, 1, 0, 1])            # non-float labels accepted
     loss = bce_with_logits(z, y)
     assert torch.isfinite(loss)
+
+
+def test_zero_grad_noop_after_step_still_correct():
+    """step() re-zeroes the flat grad buffer, so the first zero_grad of the
+    next iteration is skipped; training must match the non-flat path, and
+    a zero_grad used to DISCARD an extra backward (second zero_grad in the
+    same cycle) must really clear the buffer."""
+    _reset()
+    losses, sd, opt = _train(True)
+    f = opt._flat
+    # after the final step(): buffer zeroed, flag armed
+    assert f._grads_zeroed
+    assert all(float(g["flat_grad"].abs().max()) == 0 for g in f.groups)
+    f.zero_grad()                       # consumes the flag (no fill needed)
+    assert not f._grads_zeroed
+    # stray backward -> second zero_grad must do the real fill
+    g0 = f.groups[0]
+    g0["flat_grad"].fill_(1.0)
+    f.zero_grad()
+    assert float(g0["flat_grad"].abs().max()) == 0
