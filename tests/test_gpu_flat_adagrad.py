@@ -103,11 +103,14 @@ def test_flat_opt_kernels_match_torch(kind):
         for _ in range(4):
             grad32 = torch.randn(n, device=DEV, generator=g)
             grad = grad32.to(torch.bfloat16) if bf16 else grad32
+            # snapshot BEFORE flat_opt: the kernel zeroes the grad buffer
+            # after consuming it (zero_grad no-op contract)
+            ref.grad = grad.to(torch.float32).clone()  # same bf16-read grad
             if sc is not None:
                 ext.flat_step_scalars(sc, b1, b2)
             ext.flat_opt(opt_id, p, master if bf16 else None, s1, s2,
                          grad, sc, lr, *cfg)
-            ref.grad = grad.to(torch.float32)  # same bf16-read grad
+            assert float(grad.abs().max()) == 0.0  # kernel re-zeroed grads
             topt.step()
         w = master if bf16 else p
         assert torch.allclose(w, ref.detach(), atol=1e-5, rtol=1e-5), kind
