@@ -65,9 +65,10 @@ class _FusedMLP3Fn(torch.autograd.Function):
         from ..ops import require_hip
         ext = require_hip()
         # refresh padded weight copies (zero tails allocated once in bufs)
-        bufs["w1p"][:, :w1.shape[1]].copy_(w1)
-        bufs["w2p"][:, :w2.shape[1]].copy_(w2)
-        bufs["w3p"][:, :w3.shape[1]].copy_(w3)
+        # in ONE launch — three aten copy_ were ~3.8 us of launch/ramp
+        # each for <1 MB moved (profiles/step_attrib_r2.txt)
+        ext.mlp3_pack(w1, bufs["w1p"], False, w2, bufs["w2p"], False,
+                      w3, bufs["w3p"], False)
         w4f = w4.reshape(-1).contiguous()
         # partial (the head's first-order+FM+dense logits) folds into the
         # final-dot epilogue, replacing the `partial + dnn` add kernel
@@ -88,10 +89,10 @@ class _FusedMLP3Fn(torch.autograd.Function):
         x0, w1, w2, w3, w4f, a1, a2, a3 = ctx.saved_tensors
         bufs = ctx.bufs
         K0 = w1.shape[1]
-        # padded transposed weights for the fused dgrad chain
-        bufs["w3tp"][:, :w3.shape[0]].copy_(w3.t())
-        bufs["w2tp"][:, :w2.shape[0]].copy_(w2.t())
-        bufs["w1tp"][:, :w1.shape[0]].copy_(bufs["w1p"].t())
+        # padded transposed weights for the fused dgrad chain (one launch;
+        # w1tp takes w1p as source so its K0p..K0 tail rows stay zero)
+        ext.mlp3_pack(w3, bufs["w3tp"], True, w2, bufs["w2tp"], True,
+                      bufs["w1p"], bufs["w1tp"], True)
         dx0, dz1, dz2, dz3 = ext.mlp3_bwd(
             dout.contiguous(), a1, a2, a3, w4f,
             bufs["w3tp"], bufs["w2tp"], bufs["w1tp"])

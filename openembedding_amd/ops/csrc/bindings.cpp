@@ -54,6 +54,10 @@ void emb_mlp3_bias_bwd(const float*, const void*, const void*, const void*,
 void emb_mlp3_wgrad(const void*, const void*, const void*, const void*,
                     const void*, const void*, long, long, long, long,
                     float*, void*, void*, void*, hipStream_t_);
+void emb_mlp3_pack(const void*, void*, long, long, long, long, int,
+                   const void*, void*, long, long, long, long, int,
+                   const void*, void*, long, long, long, long, int,
+                   hipStream_t_);
 void emb_bce_fwd(const float*, const float*, long, float*, hipStream_t_);
 void emb_bce_bwd(const float*, const float*, long, const float*, float*,
                  hipStream_t_);
@@ -633,6 +637,36 @@ void mlp3_wgrad(torch::Tensor dz1, torch::Tensor dz2, torch::Tensor dz3,
                    dw2.data_ptr(), dw3.data_ptr(), cur_stream());
 }
 
+void mlp3_pack(torch::Tensor s1, torch::Tensor d1, bool t1,
+               torch::Tensor s2, torch::Tensor d2, bool t2,
+               torch::Tensor s3, torch::Tensor d3, bool t3) {
+    // one launch refreshing three padded weight copies (plain or
+    // transposed); src is the [rows, cols] weight, dst the padded buffer
+    torch::Tensor ss[3] = {s1, s2, s3}, dd[3] = {d1, d2, d3};
+    long a[3][6];
+    int tt[3] = {t1, t2, t3};
+    for (int i = 0; i < 3; i++) {
+        CHECK_GPU(ss[i]); CHECK_CONT(ss[i]);
+        TORCH_CHECK(ss[i].dtype() == torch::kBFloat16
+                    && dd[i].dtype() == torch::kBFloat16
+                    && ss[i].dim() == 2 && dd[i].dim() == 2
+                    && dd[i].stride(1) == 1, "mlp3_pack dtypes/shapes");
+        long r = ss[i].size(0), c = ss[i].size(1);
+        long dr = tt[i] ? c : r, dc = tt[i] ? r : c;
+        TORCH_CHECK(dd[i].size(0) >= dr && dd[i].size(1) >= dc,
+                    "mlp3_pack: dst too small");
+        a[i][0] = r; a[i][1] = c;
+        a[i][2] = ss[i].stride(0); a[i][3] = dd[i].stride(0);
+    }
+    const c10::cuda::CUDAGuard guard(s1.device());
+    emb_mlp3_pack(s1.data_ptr(), d1.data_ptr(), a[0][0], a[0][1], a[0][2],
+                  a[0][3], tt[0],
+                  s2.data_ptr(), d2.data_ptr(), a[1][0], a[1][1], a[1][2],
+                  a[1][3], tt[1],
+                  s3.data_ptr(), d3.data_ptr(), a[2][0], a[2][1], a[2][2],
+                  a[2][3], tt[2], cur_stream());
+}
+
 // ---- fused BCE-with-logits --------------------------------------------
 
 torch::Tensor bce_fwd(torch::Tensor logits, torch::Tensor labels) {
@@ -757,6 +791,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ctr_head_bwd", &ctr_head_bwd, "fused CTR head backward");
     m.def("mlp3_bwd", &mlp3_bwd,
           "fused dgrad chain backward of the 3-layer MLP");
+    m.def("mlp3_pack", &mlp3_pack,
+          "refresh three padded (or transposed) weight copies in one "
+          "launch");
     m.def("mlp3_fwd", &mlp3_fwd,
           "fused 3-hidden-layer MLP forward (bf16 MFMA, bias+ReLU fused)");
     m.def("mlp3_wgrad", &mlp3_wgrad,

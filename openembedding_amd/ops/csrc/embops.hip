@@ -65,6 +65,35 @@ static inline void fill_f32(float* p, long n, float v, hipStream_t s) {
     if (n) k_fill<float><<<(int)((n + 255) / 256), 256, 0, s>>>(p, n, v);
 }
 
+// One launch resetting all of emb_unique's scratch (table + counter +
+// is_first) — three separate fills cost ~2 extra kernel ramps per pull in
+// the captured train step.
+__global__ void k_unique_reset(u64* __restrict__ tk, long cap,
+                               unsigned char* __restrict__ is_first, long n,
+                               int* __restrict__ counter) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < cap) tk[i] = ~0ull;   // EMPTY
+    if (i < n) is_first[i] = 0;
+    if (i == 0) *counter = 0;
+}
+
+// Same idea for the reduce-by-key scratch (payload accumulator + counts).
+__global__ void k_reduce_reset(float* __restrict__ ugrads, long ne,
+                               u64* __restrict__ counts, long u) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < ne) ugrads[i] = 0.0f;
+    if (i < u) counts[i] = 0ull;
+}
+
+// And for the padded all-to-all send blocks (keys + src + per-peer counts).
+__global__ void k_pad_reset(u64* __restrict__ send_keys,
+                            int* __restrict__ send_src, long total,
+                            int* __restrict__ counts, long world) {
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < total) { send_keys[i] = ~0ull; send_src[i] = -1; }
+    if (i < world) counts[i] = 0;
+}
+
 // ------------------------------------------------------------------ rng
 
 DEV u64 splitmix64(u64 z) {
@@ -1072,12 +1101,16 @@ extern "C" {
 void emb_unique(const i64* keys, long n, u64* tk, int* tv, long cap,
                 int* slot_of, unsigned char* is_first, i64* unique_keys,
                 i64* inverse, int* counter, hipStream_t stream) {
-    // fill kernels, not hipMemsetAsync: memsets issued here were NOT
+    // fill kernel, not hipMemsetAsync: memsets issued here were NOT
     // replayed inside hipGraph captures (scratch kept stale keys across
-    // replays until the probe loops hung); kernels always capture
-    fill_u64(tk, cap, EMPTY, stream);
-    fill_i32(counter, 1, 0, stream);
-    fill_u8(is_first, n, 0, stream);
+    // replays until the probe loops hung); kernels always capture. One
+    // fused launch — the three separate fills were each ~4.5 us of
+    // launch/ramp in the step (profiles/step_attrib_r2.txt)
+    {
+        long mx = cap > n ? cap : n;
+        k_unique_reset<<<(int)((mx + 255) / 256), 256, 0, stream>>>(
+            tk, cap, is_first, n, counter);
+    }
     long mask = cap - 1;
     k_unique_insert<<<grid1d(n), BLOCK, 0, stream>>>(keys, n, tk, mask,
                                                      slot_of, is_first);
@@ -1140,8 +1173,12 @@ void emb_gather_init(float* weights, float* state, long dim, long sd,
 void emb_reduce_by_inverse(const i64* inverse, const float* grads, long n,
                            long dim, float* ugrads, u64* counts, long u,
                            hipStream_t stream) {
-    fill_f32(ugrads, u * dim, 0.0f, stream);
-    fill_u64(counts, u, 0ull, stream);
+    {
+        long ne = u * dim;
+        long mx = ne > u ? ne : u;
+        if (mx) k_reduce_reset<<<(int)((mx + 255) / 256), 256, 0, stream>>>(
+            ugrads, ne, counts, u);
+    }
     if (n == 0) return;
     int grid = cdiv(n, BLOCK);
     // H must cover the block's worst-case distinct uids (BLOCK = 256):
@@ -1273,9 +1310,9 @@ void emb_bucketize_pad(const i64* keys, long n, const int* u_dev, long world,
                        long cap, i64* send_keys, int* send_src, int* pos_of,
                        int* counts, int* overflow, hipStream_t stream) {
     long total = world * cap;
-    fill_u64((u64*)send_keys, total, EMPTY, stream);   // EMPTY == (i64)-1
-    fill_i32(send_src, total, -1, stream);
-    fill_i32(counts, world, 0, stream);
+    // one fused reset launch: keys EMPTY ((i64)-1), src -1, counts 0
+    if (total) k_pad_reset<<<grid1d(total), BLOCK, 0, stream>>>(
+        (u64*)send_keys, send_src, total, counts, world);
     // overflow intentionally NOT cleared: it accumulates across steps and
     // the host reads+clears it outside the hot loop
     if (n) k_bucketize_pad<<<grid1d(n), BLOCK, 0, stream>>>(

@@ -540,3 +540,55 @@ extern "C" void emb_mlp3_wgrad(const void* dz1, const void* dz2,
     k_mlp3_wgrad_finish<<<(int)((total + 255) / 256), 256, 0, stream>>>(
         scratch, H, K0p, K0, (mbf16*)dw1, (mbf16*)dw2, (mbf16*)dw3);
 }
+
+// ---- fused weight repack ----------------------------------------------
+// The train step refreshes six padded weight copies per iteration (three
+// plain pads before mlp3_fwd, three transposed pads before mlp3_bwd); as
+// separate aten::copy_ launches they cost ~3.8 us EACH while moving <1 MB
+// (profiles/step_attrib_r2.txt) — pure launch/ramp. One kernel packs all
+// three matrices of a phase. gridDim.y selects the matrix; the valid
+// region only is written (padded tails were zeroed once at alloc and
+// never change).
+
+struct PackOne {
+    const mbf16* src;   // [rows, src_ld] row-major, valid [rows, cols]
+    mbf16* dst;         // plain: [rows, dst_ld]; trans: [cols, dst_ld]
+    int rows, cols, src_ld, dst_ld, trans;
+};
+
+extern "C" __global__ void k_mlp3_pack(PackOne p0, PackOne p1, PackOne p2) {
+    PackOne p = blockIdx.y == 0 ? p0 : (blockIdx.y == 1 ? p1 : p2);
+    long n = (long)p.rows * p.cols;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    if (p.trans) {
+        // iterate destination-linear: consecutive i -> consecutive r ->
+        // coalesced writes; strided reads ride L2 (matrices are ~100s KB)
+        int c = (int)(i / p.rows), r = (int)(i % p.rows);
+        p.dst[(long)c * p.dst_ld + r] = p.src[(long)r * p.src_ld + c];
+    } else {
+        int r = (int)(i / p.cols), c = (int)(i % p.cols);
+        p.dst[(long)r * p.dst_ld + c] = p.src[(long)r * p.src_ld + c];
+    }
+}
+
+extern "C" void emb_mlp3_pack(const void* w1, void* d1, long r1, long c1,
+                              long sld1, long dld1, int t1,
+                              const void* w2, void* d2, long r2, long c2,
+                              long sld2, long dld2, int t2,
+                              const void* w3, void* d3, long r3, long c3,
+                              long sld3, long dld3, int t3,
+                              hipStream_t stream) {
+    PackOne p0{(const mbf16*)w1, (mbf16*)d1, (int)r1, (int)c1, (int)sld1,
+               (int)dld1, t1};
+    PackOne p1{(const mbf16*)w2, (mbf16*)d2, (int)r2, (int)c2, (int)sld2,
+               (int)dld2, t2};
+    PackOne p2{(const mbf16*)w3, (mbf16*)d3, (int)r3, (int)c3, (int)sld3,
+               (int)dld3, t3};
+    long mx = r1 * c1;
+    if (r2 * c2 > mx) mx = r2 * c2;
+    if (r3 * c3 > mx) mx = r3 * c3;
+    if (!mx) return;
+    dim3 grid((unsigned)((mx + 255) / 256), 3);
+    k_mlp3_pack<<<grid, 256, 0, stream>>>(p0, p1, p2);
+}
