@@ -56,6 +56,22 @@ def main():
     print(prof.key_averages().table(sort_by="self_cuda_time_total",
                                     row_limit=48, max_name_column_width=64))
 
+    # second pass with python stacks: where do the remaining fill_/copy_
+    # launches come from?
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+                 with_stack=True) as prof2:
+        for i in range(8):
+            step(pool[i % 4])
+        torch.cuda.synchronize()
+    print("\n==== fill_/copy_/add_ by stack ====")
+    tbl = prof2.key_averages(group_by_stack_n=6)
+    for e in sorted(tbl, key=lambda e: -e.self_device_time_total):
+        if e.key.split(".")[-1] not in ("fill_", "copy_", "add_", "zero_"):
+            continue
+        print(f"\n{e.key}  n={e.count}  self_cuda={e.self_device_time_total:.0f}us")
+        for line in (e.stack or [])[:6]:
+            print("   ", line)
+
 
 if __name__ == "__main__":
     main()
