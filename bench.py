@@ -75,6 +75,9 @@ def measure(args, embed, amp_mode, steps, warmup, pool, device, world, rank):
     from openembedding_amd.ops.dispatch import bce_with_logits as lossf
 
     amp = amp_mode == "bf16" and on_gpu
+    # cached backward seed: loss.backward() re-fills a ones scalar every
+    # step (~4 us launch in the captured graph)
+    seed_one = torch.ones((), device=device) if on_gpu else None
 
     def run_step(dense, sparse, labels):
         opt.zero_grad(set_to_none=False)
@@ -84,7 +87,7 @@ def measure(args, embed, amp_mode, steps, warmup, pool, device, world, rank):
                             cache_enabled=False):
             out = model(dense, sparse)
         loss = lossf(out.float(), labels)
-        loss.backward()
+        loss.backward(gradient=seed_one)
         opt.step()
         return loss
 

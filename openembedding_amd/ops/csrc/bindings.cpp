@@ -463,8 +463,10 @@ ctr_head_bwd(torch::Tensor e_all, torch::Tensor dense, torch::Tensor w,
     bool out_bf16 = d_deep_in.dtype() == torch::kBFloat16;
     auto de_all = torch::empty_like(e_all);
     auto d_dense = torch::empty_like(dense);
-    auto dw = torch::zeros_like(w);
-    auto db = torch::zeros({1}, w.options());
+    // dw/db are zeroed by k_ctr_head_bwd_e before bwd_d accumulates
+    auto dw = torch::empty_like(w);
+    auto db = torch::empty({1}, w.options());
+    if (B == 0) { dw.zero_(); db.zero_(); }
     emb_ctr_head_bwd(e_all.data_ptr<float>(), dense.data_ptr<float>(),
                      w.data_ptr<float>(), d_deep_in.data_ptr(),
                      d_partial.data_ptr<float>(),

@@ -108,9 +108,16 @@ __global__ void k_ctr_head_bwd_e(const float* __restrict__ e_all,
                                  long B, long F, long dim,
                                  long nd, long out_stride,
                                  float* __restrict__ de_all,
-                                 int use_fm) {
+                                 int use_fm,
+                                 float* __restrict__ dw,
+                                 float* __restrict__ db) {
     const long D1 = dim + 1;
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    // zero the tiny dense-linear grad accumulators consumed by the
+    // bwd_d kernel that follows on this stream (two torch::zeros
+    // allocations were ~4 us of launch each for ~56 bytes)
+    if (i < nd) dw[i] = 0.0f;
+    if (i == nd) *db = 0.0f;
     if (i >= B * F * D1) return;
     const long c = i % D1;
     const long f = (i / D1) % F;
@@ -191,14 +198,14 @@ void emb_ctr_head_bwd(const float* e_all, const float* dense, const float* w,
     if (out_bf16) {
         k_ctr_head_bwd_e<bf16><<<(int)grid_e, block, 0, stream>>>(
             e_all, (const bf16*)d_deep_in, d_partial, s_in, B, F, dim, nd,
-            out_stride, de_all, use_fm);
+            out_stride, de_all, use_fm, dw, db);
         k_ctr_head_bwd_d<bf16><<<(int)grid_d, block, 0, stream>>>(
             dense, w, (const bf16*)d_deep_in, d_partial, B, F, dim, nd,
             out_stride, d_dense, dw, db);
     } else {
         k_ctr_head_bwd_e<float><<<(int)grid_e, block, 0, stream>>>(
             e_all, (const float*)d_deep_in, d_partial, s_in, B, F, dim, nd,
-            out_stride, de_all, use_fm);
+            out_stride, de_all, use_fm, dw, db);
         k_ctr_head_bwd_d<float><<<(int)grid_d, block, 0, stream>>>(
             dense, w, (const float*)d_deep_in, d_partial, B, F, dim, nd,
             out_stride, d_dense, dw, db);
