@@ -348,10 +348,7 @@ extern "C" void emb_mlp3_bias_bwd(const float* dout, const void* dz1,
                                   void* dw4, void* db4,
                                   hipStream_t stream) {
     if (M == 0) return;
-    // 8 rows/block -> 512 blocks at M=4096: 32 rows/block only filled 128
-    // of the 256 CUs (measured 24.1 -> see profiles/step_attrib_r2.txt);
-    // the extra per-column atomics ride the same scratch lines
-    const long rows_per_blk = 8;
+    const long rows_per_blk = 32;
     int ga = (int)((M + rows_per_blk - 1) / rows_per_blk);
     k_mlp3_bias_bwd<<<ga, 256, 0, stream>>>(
         dout, (const mbf16*)dz1, (const mbf16*)dz2, (const mbf16*)dz3,
