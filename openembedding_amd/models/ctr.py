@@ -129,13 +129,10 @@ class _FusedMLP3Fn(torch.autograd.Function):
                 # all three wgrads in ONE MFMA launch (+= via fp32 scratch,
                 # self-cleaning finisher) — the hipBLASLt trio ran ~59 us
                 # per step on 64x16 macro tiles (profiles/, round 2)
-                # 8 m-split slabs of plain stores (no atomics); every
-                # active slab is fully overwritten each step, so empty is
-                # fine — the finisher sums only active slabs
                 scratch = bufs.get("wscratch")
-                want = 8 * (H * K0p + 2 * H * H)
+                want = H * K0p + 2 * H * H
                 if scratch is None or scratch.numel() != want:
-                    scratch = torch.empty(want, dtype=torch.float32,
+                    scratch = torch.zeros(want, dtype=torch.float32,
                                           device=dz1.device)
                     bufs["wscratch"] = scratch
                 ext.mlp3_wgrad(dz1, dz2, dz3, x0, a1, a2, scratch,

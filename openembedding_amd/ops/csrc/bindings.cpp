@@ -627,7 +627,7 @@ void mlp3_wgrad(torch::Tensor dz1, torch::Tensor dz2, torch::Tensor dz3,
     long M = dz1.size(0), H = dz1.size(1), K0p = x0.size(1);
     long K0 = dw1.size(1);
     TORCH_CHECK(M % 32 == 0, "mlp3_wgrad: M must be x32");
-    TORCH_CHECK(scratch.numel() >= 8 * (H * K0p + 2 * H * H)
+    TORCH_CHECK(scratch.numel() >= H * K0p + 2 * H * H
                 && dw1.size(0) == H && K0 <= K0p && dw2.numel() == H * H
                 && dw3.numel() == H * H
                 && dw1.is_contiguous() && dw2.is_contiguous()

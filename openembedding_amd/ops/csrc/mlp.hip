@@ -397,10 +397,9 @@ extern "C" void emb_mlp3_bwd(const float* dout, long M, long K0p,
 // dW_L[i,j] += sum_m dz_L[m,i] * a_{L-1}[m,j] for the 3 hidden layers in
 // ONE launch. hipBLASLt ran these [H x M]@[M x H'] shapes at ~64 TF
 // (64x16 macro tiles, 182 workgroups): here a block computes a 64x64
-// fp32 tile with both operands LDS-transposed on load, M split 8 ways
-// into per-split fp32 scratch SLABS (plain stores — a tile has exactly
-// one writer per slab, so no atomics) that a finisher sums and folds
-// into the bf16 grads (+=, capture-safe like the bias pass).
+// fp32 tile with both operands LDS-transposed on load, M split 4 ways
+// into an fp32 scratch that a finisher folds into the bf16 grads
+// (+=, self-cleaning — the same capture-safe pattern as the bias pass).
 
 #define WG_KC 128           // m-chunk staged per LDS round
 #define WG_LD (WG_KC + 8)   // LDS row stride (16 B-aligned frag rows)
@@ -414,27 +413,23 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
                   const mbf16* __restrict__ a2,    // [M, H]
                   long M, long H, long K0p,
                   long t1, long tk, long th,       // L1 tiles, K0p/H tiles
-                  float* __restrict__ scratch,     // [m_split][H*K0p+2*H*H]
+                  float* __restrict__ scratch,     // [H*K0p + 2*H*H]
                   long m_split) {
-    // block -> (layer, i-tile, j-tile); blockIdx.y -> its own scratch SLAB
-    // (plain stores, no atomics: each tile has exactly one writer per
-    // slab; the finisher sums the active slabs)
+    // block -> (layer, i-tile, j-tile)
     long bt = blockIdx.x;
-    const long ssz = H * K0p + 2 * H * H;
-    float* slab = scratch + (long)blockIdx.y * ssz;
     const mbf16 *dz, *a;
     float* out;
     long ti, tj, jdim;
     if (bt < t1) {                    // L1: dz1^T @ x0 -> [H, K0p]
-        dz = dz1; a = x0; out = slab; jdim = K0p;
+        dz = dz1; a = x0; out = scratch; jdim = K0p;
         ti = bt / tk; tj = bt - ti * tk;
     } else if (bt < t1 + th * th) {   // L2: dz2^T @ a1 -> [H, H]
         bt -= t1;
-        dz = dz2; a = a1; out = slab + H * K0p; jdim = H;
+        dz = dz2; a = a1; out = scratch + H * K0p; jdim = H;
         ti = bt / th; tj = bt - ti * th;
     } else {                          // L3: dz3^T @ a2 -> [H, H]
         bt -= t1 + th * th;
-        dz = dz3; a = a2; out = slab + H * K0p + H * H; jdim = H;
+        dz = dz3; a = a2; out = scratch + H * K0p + H * H; jdim = H;
         ti = bt / th; tj = bt - ti * th;
     }
     const long i0 = ti * 64, j0 = tj * 64;
@@ -497,7 +492,7 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
             if (i + r >= H) break;
-            out[(i + r) * jdim + j] = acc[t][r];
+            atomicAdd(&out[(i + r) * jdim + j], acc[t][r]);
         }
     }
 }
@@ -505,15 +500,12 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
 extern "C" __global__ void k_mlp3_wgrad_finish(
         float* __restrict__ scratch, long H, long K0p, long K0,
         mbf16* __restrict__ dw1, mbf16* __restrict__ dw2,
-        mbf16* __restrict__ dw3, long nslab) {
+        mbf16* __restrict__ dw3) {
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     long n1 = H * K0p, n23 = H * H;
-    long ssz = n1 + 2 * n23;
-    if (i >= ssz) return;
-    // sum the active m-split slabs (each fully overwritten this step —
-    // no zeroing or atomics anywhere in the pass)
-    float v = 0.f;
-    for (long s = 0; s < nslab; ++s) v += scratch[s * ssz + i];
+    if (i >= n1 + 2 * n23) return;
+    float v = scratch[i];
+    scratch[i] = 0.f;                    // self-cleaning for the next step
     mbf16* dst;
     if (i < n1) {
         long row = i / K0p, col = i - row * K0p;
@@ -539,19 +531,14 @@ extern "C" void emb_mlp3_wgrad(const void* dz1, const void* dz2,
     long th = (H + 63) / 64, tk = (K0p + 63) / 64;
     long t1 = th * tk;
     long tiles = t1 + 2 * th * th;
-    // only slabs whose m-range is non-empty are launched / summed
-    long msz = ((M / 32 + m_split - 1) / m_split) * 32;
-    if (msz < 32) msz = 32;
-    long nslab = (M + msz - 1) / msz;
-    if (nslab > m_split) nslab = m_split;
-    dim3 grid((unsigned)tiles, (unsigned)nslab);
+    dim3 grid((unsigned)tiles, (unsigned)m_split);
     k_mlp3_wgrad<<<grid, 512, 0, stream>>>(
         (const mbf16*)dz1, (const mbf16*)dz2, (const mbf16*)dz3,
         (const mbf16*)x0, (const mbf16*)a1, (const mbf16*)a2, M, H, K0p,
         t1, tk, th, scratch, m_split);
     long total = H * K0p + 2 * H * H;
     k_mlp3_wgrad_finish<<<(int)((total + 255) / 256), 256, 0, stream>>>(
-        scratch, H, K0p, K0, (mbf16*)dw1, (mbf16*)dw2, (mbf16*)dw3, nslab);
+        scratch, H, K0p, K0, (mbf16*)dw1, (mbf16*)dw2, (mbf16*)dw3);
 }
 
 // ---- fused weight repack ----------------------------------------------

@@ -211,8 +211,7 @@ def test_mlp3_wgrad_matches_addmm():
             torch.randn(H, H, device=DEV).to(bf).contiguous()]
     got = [b.clone() for b in base]
     ref = [b.clone() for b in base]
-    # 8 m-split slabs (plain stores, no pre-zero needed -> empty)
-    scratch = torch.empty(8 * (H * K0p + 2 * H * H), device=DEV)
+    scratch = torch.zeros(H * K0p + 2 * H * H, device=DEV)
     ext.mlp3_wgrad(dz1, dz2, dz3, x0, a1, a2, scratch, *got)
     ref[0].addmm_(dz1.t(), x0[:, :K0])
     ref[1].addmm_(dz2.t(), a1)
@@ -220,6 +219,8 @@ def test_mlp3_wgrad_matches_addmm():
     for g, r in zip(got, ref):
         torch.testing.assert_close(g.float(), r.float(), rtol=2e-2,
                                    atol=2e-2)
+    # scratch self-cleaned for the next (captured) step
+    assert float(scratch.abs().sum()) == 0.0
     # second call accumulates again (beta=1 semantics)
     ext.mlp3_wgrad(dz1, dz2, dz3, x0, a1, a2, scratch, *got)
     ref[1].addmm_(dz2.t(), a1)
