@@ -226,3 +226,30 @@ def test_mlp3_wgrad_matches_addmm():
     ref[1].addmm_(dz2.t(), a1)
     torch.testing.assert_close(got[1].float(), ref[1].float(), rtol=3e-2,
                                atol=3e-2)
+
+
+def test_mlp3_pack_matches_copy():
+    """Fused weight repack kernel vs the aten copies it replaced: plain
+    pads and transposed pads, including the padded-source (w1p -> w1tp)
+    case and untouched zero tails."""
+    from openembedding_amd.ops import require_hip
+    ext = require_hip()
+    torch.manual_seed(7)
+    bf = torch.bfloat16
+    H, K0, K0p, Hp = 400, 247, 256, 416
+    w1 = torch.randn(H, K0, device=DEV).to(bf)
+    w2 = torch.randn(H, H, device=DEV).to(bf)
+    w3 = torch.randn(H, H, device=DEV).to(bf)
+    z = lambda *s: torch.zeros(*s, device=DEV, dtype=bf)  # noqa: E731
+    w1p, w2p, w3p = z(H, K0p), z(H, Hp), z(H, Hp)
+    ext.mlp3_pack(w1, w1p, False, w2, w2p, False, w3, w3p, False)
+    assert torch.equal(w1p[:, :K0], w1) and float(w1p[:, K0:].abs().sum()) == 0
+    assert torch.equal(w2p[:, :H], w2) and float(w2p[:, H:].abs().sum()) == 0
+    assert torch.equal(w3p[:, :H], w3)
+
+    w3tp, w2tp, w1tp = z(H, Hp), z(H, Hp), z(K0p, Hp)
+    ext.mlp3_pack(w3, w3tp, True, w2, w2tp, True, w1p, w1tp, True)
+    assert torch.equal(w3tp[:, :H], w3.t())
+    assert torch.equal(w2tp[:, :H], w2.t())
+    assert torch.equal(w1tp[:, :H], w1p.t())   # incl. zero K0..K0p rows
+    assert float(w1tp[:, H:].abs().sum()) == 0
