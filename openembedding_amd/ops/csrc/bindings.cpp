@@ -25,7 +25,7 @@ typedef struct ihipStream_t* hipStream_t_;
 
 extern "C" {
 void emb_unique(const i64*, long, u64*, int*, long, int*, unsigned char*,
-                i64*, i64*, int*, hipStream_t_);
+                i64*, i64*, int*, const i64*, int, hipStream_t_);
 void emb_ht_lookup(u64*, int*, long, const i64*, long, int*, i64*, i64*,
                    unsigned char*, int, const int*, long, hipStream_t_);
 void emb_ht_rehash(const u64*, const int*, long, u64*, int*, long,
@@ -128,9 +128,20 @@ const int* u_ptr(const OptTensor& u_dev) {
 // ---- unique ----------------------------------------------------------
 
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> unique_bounded(
-    torch::Tensor keys) {
+    torch::Tensor keys, OptTensor field_offsets) {
     CHECK_GPU(keys); CHECK_CONT(keys);
     TORCH_CHECK(keys.dtype() == torch::kInt64);
+    const i64* foff = nullptr;
+    int F = 0;
+    if (field_offsets.has_value() && field_offsets->numel()) {
+        CHECK_GPU(*field_offsets); CHECK_CONT(*field_offsets);
+        TORCH_CHECK(field_offsets->dtype() == torch::kInt64,
+                    "field_offsets must be int64");
+        TORCH_CHECK(keys.numel() % field_offsets->numel() == 0,
+                    "keys length must be a multiple of field_offsets");
+        foff = field_offsets->data_ptr<i64>();
+        F = (int)field_offsets->numel();
+    }
     const c10::cuda::CUDAGuard guard(keys.device());
     long n = keys.numel();
     auto opts_i64 = keys.options();
@@ -147,12 +158,13 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> unique_bounded(
     emb_unique(keys.data_ptr<i64>(), n, (u64*)tk.data_ptr<i64>(),
                tv.data_ptr<int>(), cap, slot_of.data_ptr<int>(),
                is_first.data_ptr<uint8_t>(), uk.data_ptr<i64>(),
-               inverse.data_ptr<i64>(), counter.data_ptr<int>(), cur_stream());
+               inverse.data_ptr<i64>(), counter.data_ptr<int>(), foff, F,
+               cur_stream());
     return {uk, inverse, counter};
 }
 
 std::tuple<torch::Tensor, torch::Tensor> unique_inverse(torch::Tensor keys) {
-    auto r = unique_bounded(keys);
+    auto r = unique_bounded(keys, c10::nullopt);
     long u = std::get<2>(r).item<int>();  // host sync
     return {std::get<0>(r).narrow(0, 0, u), std::get<1>(r)};
 }
@@ -758,7 +770,10 @@ mlp3_bwd(torch::Tensor dout, torch::Tensor a1, torch::Tensor a2,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("unique_inverse", &unique_inverse, "hash-based unique+inverse");
     m.def("unique_bounded", &unique_bounded,
-          "unique+inverse without host sync (n-sized buffer + device count)");
+          "unique+inverse without host sync (n-sized buffer + device "
+          "count); optional fused per-field key offsets",
+          pybind11::arg("keys"),
+          pybind11::arg("field_offsets") = pybind11::none());
     m.def("ht_lookup", &ht_lookup, "hash table lookup/insert");
     m.def("ht_rehash", &ht_rehash, "hash table rehash into larger table");
     m.def("array_touch", &array_touch,

@@ -157,10 +157,14 @@ DEV float init_value(int cat, float p0, float p1, float p2,
 #define UA_ITERS 8  // elements per thread in the uid-assign kernel
 #define UI_ILP 4    // parallel lookups per thread in the inverse kernel
 
+// foff/F: optional per-field key offsets fused into the read (keys is then
+// the RAW [B, F] field-id layout; flat key = keys[i] + foff[i % F]) — saves
+// the broadcast-add launch + 8B/elem intermediate every pull.
 __global__ void k_unique_insert(const i64* __restrict__ keys, long n,
                                 u64* __restrict__ tk, long mask,
                                 int* __restrict__ slot_of,
-                                unsigned char* __restrict__ is_first) {
+                                unsigned char* __restrict__ is_first,
+                                const i64* __restrict__ foff, int F) {
     __shared__ u64 lkeys[ULDS];
     __shared__ int lslot[ULDS];
     for (int t = threadIdx.x; t < ULDS; t += blockDim.x) lkeys[t] = EMPTY;
@@ -168,9 +172,14 @@ __global__ void k_unique_insert(const i64* __restrict__ keys, long n,
 
     long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
     u64 k = 0;
+    i64 kin = 0;
+    if (i < n) {
+        kin = keys[i];
+        if (foff) kin += foff[(int)i % F];   // n < 2^31: 32-bit mod
+    }
     int lh = -1;           // LDS slot of this element's key
     bool lds_winner = false, lds_ok = false;
-    if (i < n && (u64)keys[i] == EMPTY) {
+    if (i < n && (u64)kin == EMPTY) {
         // RESERVED key -1 (the scratch table's empty marker, same value the
         // reference reserves): give every occurrence the overflow-winner
         // encoding — its own unique entry, resolved to slot -1 (zeros)
@@ -178,7 +187,7 @@ __global__ void k_unique_insert(const i64* __restrict__ keys, long n,
         is_first[i] = 1;
         slot_of[i] = -1;
     } else if (i < n) {
-        k = (u64)keys[i];
+        k = (u64)kin;
         u64 hh = splitmix64(k);
         lh = (int)(hh & (ULDS - 1));
         for (int probes = 0; probes < 64; ++probes) {
@@ -224,7 +233,8 @@ __global__ void k_unique_assign(const i64* __restrict__ keys, long n,
                                 int* __restrict__ tv,
                                 i64* __restrict__ unique_keys,
                                 int* __restrict__ counter,
-                                i64* __restrict__ inverse) {
+                                i64* __restrict__ inverse,
+                                const i64* __restrict__ foff, int F) {
     // UA_ITERS elements per thread with ONE returning atomic per wave:
     // ballots for all iterations are taken first (independent loads in
     // flight), their popcounts summed, a single atomicAdd reserves the uid
@@ -253,7 +263,9 @@ __global__ void k_unique_assign(const i64* __restrict__ keys, long n,
         if (b & (1ull << lane)) {
             long i = i0 + t * stride;
             int uid = base + __popcll(b & ((1ull << lane) - 1ull));
-            unique_keys[uid] = keys[i];
+            i64 kin = keys[i];
+            if (foff) kin += foff[(int)i % F];
+            unique_keys[uid] = kin;
             int s = slot_of[i];
             if (s >= 0) tv[s] = uid;
             else inverse[i] = uid;  // overflow winner: direct assignment
@@ -1100,7 +1112,8 @@ extern "C" {
 
 void emb_unique(const i64* keys, long n, u64* tk, int* tv, long cap,
                 int* slot_of, unsigned char* is_first, i64* unique_keys,
-                i64* inverse, int* counter, hipStream_t stream) {
+                i64* inverse, int* counter, const i64* foff, int F,
+                hipStream_t stream) {
     // fill kernel, not hipMemsetAsync: memsets issued here were NOT
     // replayed inside hipGraph captures (scratch kept stale keys across
     // replays until the probe loops hung); kernels always capture. One
@@ -1113,12 +1126,13 @@ void emb_unique(const i64* keys, long n, u64* tk, int* tv, long cap,
     }
     long mask = cap - 1;
     k_unique_insert<<<grid1d(n), BLOCK, 0, stream>>>(keys, n, tk, mask,
-                                                     slot_of, is_first);
+                                                     slot_of, is_first,
+                                                     foff, F);
     int ga = grid1d((n + UA_ITERS - 1) / UA_ITERS);
     int gi = grid1d((n + UI_ILP - 1) / UI_ILP);
     k_unique_assign<<<ga, BLOCK, 0, stream>>>(keys, n, slot_of, is_first,
                                               tv, unique_keys, counter,
-                                              inverse);
+                                              inverse, foff, F);
     k_unique_inverse<<<gi, BLOCK, 0, stream>>>(n, slot_of, tv, inverse);
 }
 

@@ -144,25 +144,37 @@ class ShardedVariable:
     # identity) so the GPU bucketize/fan-out logic is validatable on one GPU
     _force_remote = False
 
-    def pull(self, indices: torch.Tensor, readonly: bool = False):
+    def pull(self, indices: torch.Tensor, readonly: bool = False,
+             field_offsets: torch.Tensor = None):
         """indices: int64 tensor of any shape -> (weights [*shape, dim], handle).
+
+        ``field_offsets`` (optional, [n_fields] int64, indices then the RAW
+        [batch, n_fields] field ids): fused into the GPU unique kernels —
+        flat key = id + offset[field] — saving the broadcast-add launch and
+        its intermediate every pull. Non-bounded paths fold it eagerly.
 
         Collective when world_size > 1: every rank must call it the same
         number of times per variable per step."""
+        bounded_ok = (not readonly
+                      and getattr(self.shard, "pull_bounded", None)
+                      is not None)
+        if field_offsets is not None and not bounded_ok:
+            indices = indices + field_offsets
+            field_offsets = None
         flat = indices.reshape(-1).to(torch.int64)
         n = flat.numel()
         self.stat_pull_indices += n
         remote = self.world_size > 1 or self._force_remote
-        if (not remote and not readonly
-                and getattr(self.shard, "pull_bounded", None) is not None):
-            return self._pull_local_bounded(indices, flat)
+        if not remote and bounded_ok:
+            return self._pull_local_bounded(indices, flat, field_offsets)
         if remote and not readonly:
             if self._use_padded():
-                r = self._pull_remote_padded(indices, flat)
+                r = self._pull_remote_padded(indices, flat, field_offsets)
                 if r is not None:
                     return r
-            if getattr(self.shard, "pull_bounded", None) is not None:
-                return self._pull_remote_bounded(indices, flat)
+            if bounded_ok:
+                return self._pull_remote_bounded(indices, flat,
+                                                 field_offsets)
         unique, inverse = ops.unique_inverse(flat)
         self.stat_pull_unique += unique.numel()
         # reference pull_indices/pull_unique accumulators
@@ -179,18 +191,20 @@ class ShardedVariable:
         out = out.view(*h.shape, self.shard.dim)
         return out, h
 
-    def _pull_local_bounded(self, indices: torch.Tensor, flat: torch.Tensor):
+    def _pull_local_bounded(self, indices: torch.Tensor, flat: torch.Tensor,
+                            field_offsets: torch.Tensor = None):
         """Single-GPU sync-free pull: bounded unique buffer + fused
         gather/init/scatter; zero host round-trips (hipGraph-capturable)."""
         ext = self.shard.ext
-        uk_buf, inverse, u_dev = ext.unique_bounded(flat)
+        uk_buf, inverse, u_dev = ext.unique_bounded(flat, field_offsets)
         out, slots = self.shard.pull_bounded(uk_buf, u_dev, inverse)
         h = PullHandle(shape=indices.shape, unique=uk_buf, inverse=inverse,
                        bounded=True, u_dev=u_dev, slots=slots)
         return out.view(*indices.shape, self.shard.dim), h
 
     def _pull_remote_bounded(self, indices: torch.Tensor,
-                             flat: torch.Tensor):
+                             flat: torch.Tensor,
+                             field_offsets: torch.Tensor = None):
         """Multi-rank pull on the GPU engine with ONE host sync per step
         (the combined split-size read): dedup, owner bucketize and the
         sentinel-last ordering all run device-side off the bounded unique
@@ -198,7 +212,8 @@ class ShardedVariable:
         world = self.world_size
         ext = self.shard.ext
         dev = flat.device
-        uk_buf, inverse, u_dev = ext.unique_bounded(flat)   # no sync
+        uk_buf, inverse, u_dev = ext.unique_bounded(flat,
+                                                    field_offsets)  # no sync
         n = uk_buf.numel()
         valid = torch.arange(n, device=dev) < u_dev.to(torch.int64)
         owner = torch.where(valid, uk_buf % world,
@@ -283,7 +298,8 @@ class ShardedVariable:
         return pair[pair[2]]
 
     def _pull_remote_padded(self, indices: torch.Tensor,
-                            flat: torch.Tensor):
+                            flat: torch.Tensor,
+                            field_offsets: torch.Tensor = None):
         """Multi-rank pull with ZERO host syncs: fixed [world, cap] key
         blocks padded with the reserved key -1 ride two equal-split
         all-to-alls; padding resolves to defined misses on the owner and is
@@ -298,7 +314,7 @@ class ShardedVariable:
         gpu = getattr(self.shard, "pull_bounded", None) is not None
         if gpu:
             ext = self.shard.ext
-            uk_buf, inverse, u_dev = ext.unique_bounded(flat)
+            uk_buf, inverse, u_dev = ext.unique_bounded(flat, field_offsets)
             ext.bucketize_pad(uk_buf, u_dev, world, plan.cap,
                               plan.send_keys, plan.send_src, plan.pos_of,
                               plan.counts, plan.overflow)

@@ -69,9 +69,9 @@ class _PullPushFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, hook: torch.Tensor, indices: torch.Tensor,
-                var: ShardedVariable, pre=None):
+                var: ShardedVariable, pre=None, field_offsets=None):
         if pre is None:
-            out, handle = var.pull(indices)
+            out, handle = var.pull(indices, field_offsets=field_offsets)
         else:  # prefetched (out, handle) from Variable.prefetch
             out, handle = pre
         ctx.var = var
@@ -83,7 +83,8 @@ class _PullPushFn(torch.autograd.Function):
         # cast to the VARIABLE's dtype (f32 normally; f64 tables keep f64)
         ctx.var.push(ctx.handle,
                      grad_out.contiguous().to(ctx.var.shard.dtype))
-        return torch.zeros(0, device=grad_out.device), None, None, None
+        return (torch.zeros(0, device=grad_out.device), None, None, None,
+                None)
 
 
 class Variable:
@@ -214,13 +215,21 @@ class Embedding(nn.Module):
         return self._forward_keys(indices, indices)
 
     def _forward_keys(self, keys: torch.Tensor,
-                      match_ref: torch.Tensor) -> torch.Tensor:
+                      match_ref: torch.Tensor,
+                      field_offsets: torch.Tensor = None) -> torch.Tensor:
+        # field_offsets: keys are RAW per-field ids; the GPU pull fuses the
+        # offset add into its unique kernels (CombinedEmbedding hot path)
         if self.sparse_as_dense:
+            if field_offsets is not None:
+                keys = keys + field_offsets
             return self.dense(keys)
         if torch.is_grad_enabled() and self.grad_hook.requires_grad:
             pre = self.variable._take_prefetched(match_ref)
             return _PullPushFn.apply(self.grad_hook, keys,
-                                     self.variable.sharded, pre)
+                                     self.variable.sharded, pre,
+                                     field_offsets)
+        if field_offsets is not None:
+            keys = keys + field_offsets
         return self.variable.sparse_read(keys)
 
     def prefetch(self, indices: torch.Tensor) -> None:
@@ -266,9 +275,15 @@ class CombinedEmbedding(Embedding):
                          device=get_context().device))
 
     def forward(self, field_ids: torch.Tensor) -> torch.Tensor:
-        keys = field_ids.to(torch.int64) + self.field_offsets
-        # prefetch matching is by the ORIGINAL field_ids tensor (keys is a
-        # fresh tensor every call)
+        ids = field_ids.to(torch.int64)
+        if ids.is_cuda and not self.sparse_as_dense:
+            # raw ids through; the GPU pull fuses `+ field_offsets` into
+            # its unique kernels (saves the broadcast-add launch + the
+            # [B, F] int64 intermediate every step). Prefetch matching is
+            # by the ORIGINAL field_ids tensor either way.
+            return self._forward_keys(ids, field_ids,
+                                      field_offsets=self.field_offsets)
+        keys = ids + self.field_offsets
         return self._forward_keys(keys, field_ids)
 
     def prefetch(self, field_ids: torch.Tensor) -> None:
