@@ -93,3 +93,40 @@ def test_full_lifecycle(tmp_path):
     assert r.status_code == 200
     served = torch.tensor(r.json()["weights"])
     torch.testing.assert_close(served, live)
+
+
+def test_untranslatable_optimizer_raises_actionably():
+    """NAdam (VERDICT weak-8): no server-side category exists for it
+    (the reference's 9 sparse optimizers have no nadam either) — the first
+    step must raise an error that NAMES the optimizer and says how to fix
+    it, and an explicit sparse_config must make the same setup work."""
+    import openembedding_amd.context as cm
+    import openembedding_amd.torch as api
+    import openembedding_amd.torch as embed
+    import pytest
+    import torch
+    from openembedding_amd.models import DeepFM, synthetic_batch
+
+    def reset():
+        if cm._context is not None:
+            cm._context.finalize()
+            cm._context = None
+        api._tracked.clear()
+
+    def one_step(opt_kwargs):
+        torch.manual_seed(0)
+        model = DeepFM(dim=4)
+        opt = embed.distributed_optimizer(
+            torch.optim.NAdam(model.parameters(), lr=0.01), **opt_kwargs)
+        dense, sparse, labels = synthetic_batch(32)
+        loss = torch.nn.BCEWithLogitsLoss()(model(dense, sparse), labels)
+        opt.zero_grad(set_to_none=False)
+        loss.backward()
+        opt.step()
+
+    reset()
+    with pytest.raises(RuntimeError, match="NAdam"):
+        one_step({})
+    reset()
+    one_step({"sparse_config": dict(category="adagrad", learning_rate=0.01)})
+    reset()
