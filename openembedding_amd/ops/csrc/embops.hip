@@ -888,6 +888,29 @@ __global__ void k_bce_fwd(const float* __restrict__ z,
     if ((threadIdx.x & 63) == 0 && v != 0.f) atomicAdd(loss, v);
 }
 
+// Single-block variant for bench-sized batches: grid-stride sum + LDS
+// cross-wave reduce + ONE plain store — no pre-zero fill, no atomics
+// (two launches -> one; the fill alone was ~4 us of launch/ramp).
+__global__ void k_bce_fwd_1blk(const float* __restrict__ z,
+                               const float* __restrict__ y, long n,
+                               float inv_n, float* __restrict__ loss) {
+    __shared__ float wsum[16];
+    float v = 0.f;
+    for (long i = threadIdx.x; i < n; i += blockDim.x) {
+        float zi = z[i];
+        v += (fmaxf(zi, 0.f) - zi * y[i]
+              + log1pf(__expf(-fabsf(zi)))) * inv_n;
+    }
+    for (int off = 32; off; off >>= 1) v += __shfl_down(v, off);
+    if ((threadIdx.x & 63) == 0) wsum[threadIdx.x >> 6] = v;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t = 0.f;
+        for (int w = 0; w < (int)(blockDim.x >> 6); ++w) t += wsum[w];
+        *loss = t;
+    }
+}
+
 __global__ void k_bce_bwd(const float* __restrict__ z,
                           const float* __restrict__ y, long n, float inv_n,
                           const float* __restrict__ go,
@@ -1400,9 +1423,15 @@ void emb_flat_opt(int opt, void* p, float* master, float* s1, float* s2,
 
 void emb_bce_fwd(const float* z, const float* y, long n, float* loss,
                  hipStream_t stream) {
+    if (!n) { fill_f32(loss, 1, 0.f, stream); return; }
+    if (n <= 65536) {   // one block reduces it faster than fill + atomics
+        k_bce_fwd_1blk<<<1, 1024, 0, stream>>>(z, y, n, 1.f / (float)n,
+                                               loss);
+        return;
+    }
     fill_f32(loss, 1, 0.f, stream);
-    int ga = grid1d(n);
-    if (n) k_bce_fwd<<<ga, BLOCK, 0, stream>>>(z, y, n, 1.f / (float)n, loss);
+    k_bce_fwd<<<grid1d(n), BLOCK, 0, stream>>>(z, y, n, 1.f / (float)n,
+                                               loss);
 }
 
 void emb_bce_bwd(const float* z, const float* y, long n, const float* go,
