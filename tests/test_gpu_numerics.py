@@ -235,7 +235,9 @@ def test_fused_bce_matches_torch():
     # fwd + bwd of the 2-kernel BCE vs torch.nn.BCEWithLogitsLoss (fp32)
     from openembedding_amd.ops.dispatch import bce_with_logits
     g = torch.Generator(device="cpu").manual_seed(7)
-    for n in (1, 63, 4096, 10000):
+    # <=65536 runs the single-block store variant, above it the
+    # fill+atomic grid variant — cover both
+    for n in (1, 63, 4096, 10000, 70000):
         z = (torch.randn(n, generator=g) * 4).to(DEV).requires_grad_(True)
         y = (torch.rand(n, generator=g) < 0.3).float().to(DEV)
         z2 = z.detach().clone().requires_grad_(True)
