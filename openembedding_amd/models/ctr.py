@@ -119,9 +119,25 @@ class _FusedMLP3Fn(torch.autograd.Function):
             w1g, b1g, w2g, b2g, w3g, b3g, w4g, b4g = (p.grad for p in params)
             M, H = dz1.shape
             K0p = x0.shape[1]
-            if (w1g.dtype == torch.bfloat16 and w1g.is_contiguous()
-                    and w2g.is_contiguous() and w3g.is_contiguous()
-                    and M % 32 == 0 and K0p <= 512):
+            wgrad_fused = (w1g.dtype == torch.bfloat16
+                           and w1g.is_contiguous() and w2g.is_contiguous()
+                           and w3g.is_contiguous()
+                           and M % 32 == 0 and K0p <= 512)
+            bias_fused = (b1g.dtype == torch.bfloat16
+                          and b1g.is_contiguous() and b2g.is_contiguous()
+                          and b3g.is_contiguous() and w4g.is_contiguous())
+            # when BOTH fused paths run, the wgrad kernel carries the three
+            # dz column sums (bias grads) from its LDS-staged tiles and the
+            # bias pass skips its 3x dz re-read (~10 MB/step)
+            bias_in_wgrad = wgrad_fused and bias_fused
+            bscratch = None
+            if bias_fused:
+                bscratch = bufs.get("bscratch")
+                if bscratch is None or bscratch.numel() != 4 * H + 1:
+                    bscratch = torch.zeros(4 * H + 1, dtype=torch.float32,
+                                           device=a1.device)
+                    bufs["bscratch"] = bscratch
+            if wgrad_fused:
                 # K0p <= 512: at the dim9 shapes the fused kernel beats the
                 # hipBLASLt trio (12.29 -> 12.62M); at dim64 (K0p 1696)
                 # hipBLASLt's wide-N tiles win (4.90 vs 4.75M) — measured,
@@ -136,28 +152,23 @@ class _FusedMLP3Fn(torch.autograd.Function):
                                           device=dz1.device)
                     bufs["wscratch"] = scratch
                 ext.mlp3_wgrad(dz1, dz2, dz3, x0, a1, a2, scratch,
-                               w1g, w2g, w3g)
+                               w1g, w2g, w3g,
+                               bscratch if bias_in_wgrad else None)
             else:
                 w1g.addmm_(dz1.t(), x0[:, :K0])
                 w2g.addmm_(dz2.t(), a1)
                 w3g.addmm_(dz3.t(), a2)
             dpart = dout if ctx._has_partial else None
-            if b1g.dtype == torch.bfloat16 and b1g.is_contiguous() \
-                    and b2g.is_contiguous() and b3g.is_contiguous() \
-                    and w4g.is_contiguous():
+            if bias_fused:
                 # bias grads for all 4 layers + the head wgrad in one pass
-                # over the dz mirrors (+ a finisher that folds the fp32
-                # scratch into the bf16 grads and re-zeros it) — replaces
-                # 4 GEMV launches + a reduce + an add + a dout bf16 cast
-                H = a1.shape[1]
-                scratch = bufs.get("bscratch")
-                if scratch is None or scratch.numel() != 4 * H + 1:
-                    scratch = torch.zeros(4 * H + 1, dtype=torch.float32,
-                                          device=a1.device)
-                    bufs["bscratch"] = scratch
+                # (+ a finisher that folds the fp32 scratch into the bf16
+                # grads and re-zeros it) — replaces 4 GEMV launches + a
+                # reduce + an add + a dout bf16 cast. with_dz=False when
+                # the fused wgrad above already summed the dz columns.
                 ext.mlp3_bias_bwd(dout.contiguous(), dz1, dz2, dz3, a3,
-                                  scratch, b1g, b2g, b3g,
-                                  w4g.reshape(-1), b4g)
+                                  bscratch, b1g, b2g, b3g,
+                                  w4g.reshape(-1), b4g,
+                                  not bias_in_wgrad)
             else:
                 d, ones = _d(), _ones()
                 w4g.addmm_(d, a3)

@@ -50,10 +50,10 @@ void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
                       hipStream_t_);
 void emb_mlp3_bias_bwd(const float*, const void*, const void*, const void*,
                        const void*, long, long, float*, void*, void*, void*,
-                       void*, void*, hipStream_t_);
+                       void*, void*, int, hipStream_t_);
 void emb_mlp3_wgrad(const void*, const void*, const void*, const void*,
                     const void*, const void*, long, long, long, long,
-                    float*, void*, void*, void*, hipStream_t_);
+                    float*, void*, void*, void*, float*, hipStream_t_);
 void emb_mlp3_pack(const void*, void*, long, long, long, long, int,
                    const void*, void*, long, long, long, long, int,
                    const void*, void*, long, long, long, long, int,
@@ -548,7 +548,7 @@ void flat_opt(int64_t opt, torch::Tensor param, OptTensor master,
 void mlp3_bias_bwd(torch::Tensor dout, torch::Tensor dz1, torch::Tensor dz2,
                    torch::Tensor dz3, torch::Tensor a3, torch::Tensor scratch,
                    torch::Tensor db1, torch::Tensor db2, torch::Tensor db3,
-                   torch::Tensor dw4, torch::Tensor db4) {
+                   torch::Tensor dw4, torch::Tensor db4, bool with_dz) {
     CHECK_GPU(dout); CHECK_CONT(dout); CHECK_CONT(dz1); CHECK_CONT(dz2);
     CHECK_CONT(dz3); CHECK_CONT(a3); CHECK_CONT(scratch);
     long M = dz1.size(0), H = dz1.size(1);
@@ -567,7 +567,8 @@ void mlp3_bias_bwd(torch::Tensor dout, torch::Tensor dz1, torch::Tensor dz2,
                       dz3.data_ptr(), a3.data_ptr(), M, H,
                       scratch.data_ptr<float>(),
                       db1.data_ptr(), db2.data_ptr(), db3.data_ptr(),
-                      dw4.data_ptr(), db4.data_ptr(), cur_stream());
+                      dw4.data_ptr(), db4.data_ptr(), with_dz ? 1 : 0,
+                      cur_stream());
 }
 
 // ---- CIN implicit-GEMM -------------------------------------------------
@@ -630,7 +631,10 @@ std::tuple<torch::Tensor, torch::Tensor> cin_dx(
 void mlp3_wgrad(torch::Tensor dz1, torch::Tensor dz2, torch::Tensor dz3,
                 torch::Tensor x0, torch::Tensor a1, torch::Tensor a2,
                 torch::Tensor scratch, torch::Tensor dw1, torch::Tensor dw2,
-                torch::Tensor dw3) {
+                torch::Tensor dw3, OptTensor bias_scratch) {
+    // bias_scratch ([4H+1] fp32, the bias pass scratch): when given, the
+    // kernel also accumulates the three dz column sums (bias grads) from
+    // its LDS-staged tiles into segments 0/H/2H
     CHECK_GPU(dz1); CHECK_CONT(dz1); CHECK_CONT(dz2); CHECK_CONT(dz3);
     CHECK_CONT(x0); CHECK_CONT(a1); CHECK_CONT(a2); CHECK_CONT(scratch);
     TORCH_CHECK(dz1.dtype() == torch::kBFloat16
@@ -645,10 +649,18 @@ void mlp3_wgrad(torch::Tensor dz1, torch::Tensor dz2, torch::Tensor dz3,
                 && dw1.is_contiguous() && dw2.is_contiguous()
                 && dw3.is_contiguous(), "mlp3_wgrad shapes");
     const c10::cuda::CUDAGuard guard(dz1.device());
+    float* bptr = nullptr;
+    if (bias_scratch.has_value()) {
+        CHECK_CONT(*bias_scratch);
+        TORCH_CHECK(bias_scratch->numel() >= 3 * H
+                    && bias_scratch->dtype() == torch::kFloat32,
+                    "bias_scratch layout");
+        bptr = bias_scratch->data_ptr<float>();
+    }
     emb_mlp3_wgrad(dz1.data_ptr(), dz2.data_ptr(), dz3.data_ptr(),
                    x0.data_ptr(), a1.data_ptr(), a2.data_ptr(), M, H, K0p,
                    K0, scratch.data_ptr<float>(), dw1.data_ptr(),
-                   dw2.data_ptr(), dw3.data_ptr(), cur_stream());
+                   dw2.data_ptr(), dw3.data_ptr(), bptr, cur_stream());
 }
 
 void mlp3_pack(torch::Tensor s1, torch::Tensor d1, bool t1,
@@ -814,10 +826,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("mlp3_fwd", &mlp3_fwd,
           "fused 3-hidden-layer MLP forward (bf16 MFMA, bias+ReLU fused)");
     m.def("mlp3_wgrad", &mlp3_wgrad,
+          pybind11::arg("dz1"), pybind11::arg("dz2"), pybind11::arg("dz3"),
+          pybind11::arg("x0"), pybind11::arg("a1"), pybind11::arg("a2"),
+          pybind11::arg("scratch"), pybind11::arg("dw1"),
+          pybind11::arg("dw2"), pybind11::arg("dw3"),
+          pybind11::arg("bias_scratch") = pybind11::none(),
           "all three MLP weight grads in one MFMA launch (+= into bf16 "
           "grads via fp32 scratch)");
     m.def("mlp3_bias_bwd", &mlp3_bias_bwd,
-          "MLP bias grads + head wgrad in one pass over the dz mirrors");
+          "MLP bias grads + head wgrad in one pass over the dz mirrors "
+          "(with_dz=false when the fused wgrad carried the dz sums)",
+          pybind11::arg("dout"), pybind11::arg("dz1"), pybind11::arg("dz2"),
+          pybind11::arg("dz3"), pybind11::arg("a3"),
+          pybind11::arg("scratch"), pybind11::arg("db1"),
+          pybind11::arg("db2"), pybind11::arg("db3"), pybind11::arg("dw4"),
+          pybind11::arg("db4"), pybind11::arg("with_dz") = true);
     m.def("cin_fwd", &cin_fwd,
           "CIN layer forward: implicit outer-product MFMA GEMM");
     m.def("cin_dw", &cin_dw,

@@ -295,23 +295,32 @@ extern "C" __global__ void k_mlp3_bias_bwd(
         const mbf16* __restrict__ dz1, const mbf16* __restrict__ dz2,
         const mbf16* __restrict__ dz3, const mbf16* __restrict__ a3,
         long M, long H, long rows_per_blk,
-        float* __restrict__ scratch) {
+        float* __restrict__ scratch, int with_dz) {
     const long r0 = (long)blockIdx.x * rows_per_blk;
     if (r0 >= M) return;
     const long r1 = min(M, r0 + rows_per_blk);
     // column sums: thread t covers columns t, t+blockDim.x, ... of each dz
-    // (+ the head wgrad dw4[c] = sum_r dout[r]*a3[r,c] — same access shape)
+    // (+ the head wgrad dw4[c] = sum_r dout[r]*a3[r,c] — same access shape).
+    // with_dz == 0: the fused wgrad kernel already accumulated the three
+    // dz column sums from its LDS-staged tiles; only the head pass
+    // (dout/a3, a quarter of the traffic) remains here.
     for (long c = threadIdx.x; c < H; c += blockDim.x) {
-        float s1 = 0.f, s2 = 0.f, s3 = 0.f, sw = 0.f;
-        for (long r = r0; r < r1; ++r) {
-            s1 += (float)dz1[r * H + c];
-            s2 += (float)dz2[r * H + c];
-            s3 += (float)dz3[r * H + c];
-            sw += dout[r] * (float)a3[r * H + c];
+        float sw = 0.f;
+        if (with_dz) {
+            float s1 = 0.f, s2 = 0.f, s3 = 0.f;
+            for (long r = r0; r < r1; ++r) {
+                s1 += (float)dz1[r * H + c];
+                s2 += (float)dz2[r * H + c];
+                s3 += (float)dz3[r * H + c];
+                sw += dout[r] * (float)a3[r * H + c];
+            }
+            atomicAdd(scratch + c, s1);
+            atomicAdd(scratch + H + c, s2);
+            atomicAdd(scratch + 2 * H + c, s3);
+        } else {
+            for (long r = r0; r < r1; ++r)
+                sw += dout[r] * (float)a3[r * H + c];
         }
-        atomicAdd(scratch + c, s1);
-        atomicAdd(scratch + H + c, s2);
-        atomicAdd(scratch + 2 * H + c, s3);
         atomicAdd(scratch + 3 * H + c, sw);
     }
     // head bias: sum of dout rows, one atomic per wave
@@ -345,14 +354,14 @@ extern "C" void emb_mlp3_bias_bwd(const float* dout, const void* dz1,
                                   const void* a3,
                                   long M, long H, float* scratch,
                                   void* db1, void* db2, void* db3,
-                                  void* dw4, void* db4,
+                                  void* dw4, void* db4, int with_dz,
                                   hipStream_t stream) {
     if (M == 0) return;
     const long rows_per_blk = 32;
     int ga = (int)((M + rows_per_blk - 1) / rows_per_blk);
     k_mlp3_bias_bwd<<<ga, 256, 0, stream>>>(
         dout, (const mbf16*)dz1, (const mbf16*)dz2, (const mbf16*)dz3,
-        (const mbf16*)a3, M, H, rows_per_blk, scratch);
+        (const mbf16*)a3, M, H, rows_per_blk, scratch, with_dz);
     int gb = (int)((4 * H + 1 + 255) / 256);
     k_mlp3_bias_finish<<<gb, 256, 0, stream>>>(
         scratch, H, (mbf16*)db1, (mbf16*)db2, (mbf16*)db3, (mbf16*)dw4,
@@ -414,22 +423,24 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
                   long M, long H, long K0p,
                   long t1, long tk, long th,       // L1 tiles, K0p/H tiles
                   float* __restrict__ scratch,     // [H*K0p + 2*H*H]
-                  long m_split) {
+                  long m_split,
+                  float* __restrict__ bias_out) {  // optional [3*H] bias
     // block -> (layer, i-tile, j-tile)
     long bt = blockIdx.x;
     const mbf16 *dz, *a;
     float* out;
-    long ti, tj, jdim;
+    long ti, tj, jdim, bseg;
     if (bt < t1) {                    // L1: dz1^T @ x0 -> [H, K0p]
-        dz = dz1; a = x0; out = scratch; jdim = K0p;
+        dz = dz1; a = x0; out = scratch; jdim = K0p; bseg = 0;
         ti = bt / tk; tj = bt - ti * tk;
     } else if (bt < t1 + th * th) {   // L2: dz2^T @ a1 -> [H, H]
         bt -= t1;
-        dz = dz2; a = a1; out = scratch + H * K0p; jdim = H;
+        dz = dz2; a = a1; out = scratch + H * K0p; jdim = H; bseg = H;
         ti = bt / th; tj = bt - ti * th;
     } else {                          // L3: dz3^T @ a2 -> [H, H]
         bt -= t1 + th * th;
         dz = dz3; a = a2; out = scratch + H * K0p + H * H; jdim = H;
+        bseg = 2 * H;
         ti = bt / th; tj = bt - ti * th;
     }
     const long i0 = ti * 64, j0 = tj * 64;
@@ -467,6 +478,23 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
             ldsB[cc * WG_LD + mm] = vb;
         }
         __syncthreads();
+        if (bias_out && tj == 0) {
+            // bias grads ride along: the dz tile this block just staged IS
+            // the operand k_mlp3_bias_bwd used to re-read from HBM — sum
+            // its columns here instead (tj==0 blocks only: each (layer,
+            // i-tile, m-split) contributes once). 8 threads per column,
+            // tree-reduced within the 8-lane group; zeros were staged
+            // beyond mlim/H so no bounds handling is needed in the sum.
+            int cc = (int)threadIdx.x >> 3, sub = (int)threadIdx.x & 7;
+            float s = 0.f;
+            for (int mm = sub; mm < WG_KC; mm += 8)
+                s += (float)ldsA[cc * WG_LD + mm];
+            s += __shfl_down(s, 4);
+            s += __shfl_down(s, 2);
+            s += __shfl_down(s, 1);
+            if (sub == 0 && i0 + cc < H)
+                atomicAdd(&bias_out[bseg + i0 + cc], s);
+        }
         const long koff = (lane >> 4) * 8;
         #pragma unroll
         for (int t = 0; t < 2; ++t) {
@@ -525,6 +553,7 @@ extern "C" void emb_mlp3_wgrad(const void* dz1, const void* dz2,
                                long M, long H, long K0p, long K0,
                                float* scratch,
                                void* dw1, void* dw2, void* dw3,
+                               float* bias_out,
                                hipStream_t stream) {
     if (!M) return;
     const long m_split = 8;
@@ -535,7 +564,7 @@ extern "C" void emb_mlp3_wgrad(const void* dz1, const void* dz2,
     k_mlp3_wgrad<<<grid, 512, 0, stream>>>(
         (const mbf16*)dz1, (const mbf16*)dz2, (const mbf16*)dz3,
         (const mbf16*)x0, (const mbf16*)a1, (const mbf16*)a2, M, H, K0p,
-        t1, tk, th, scratch, m_split);
+        t1, tk, th, scratch, m_split, bias_out);
     long total = H * K0p + 2 * H * H;
     k_mlp3_wgrad_finish<<<(int)((total + 255) / 256), 256, 0, stream>>>(
         scratch, H, K0p, K0, (mbf16*)dw1, (mbf16*)dw2, (mbf16*)dw3);
