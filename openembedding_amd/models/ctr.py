@@ -93,9 +93,26 @@ class _FusedMLP3Fn(torch.autograd.Function):
         # w1tp takes w1p as source so its K0p..K0 tail rows stay zero)
         ext.mlp3_pack(w3, bufs["w3tp"], True, w2, bufs["w2tp"], True,
                       bufs["w1p"], bufs["w1tp"], True)
+        # prebound flat-optimizer grads? decided BEFORE the dgrad launch:
+        # the dz3-assembly loop there carries the head wgrad/bias sums
+        # (it reads dout and a3 anyway), killing the separate head pass
+        params = ctx._params
+        prebound = all(p.grad is not None for p in params)
+        bias_fused = (prebound
+                      and params[1].grad.dtype == torch.bfloat16
+                      and all(params[i].grad.is_contiguous()
+                              for i in (1, 3, 5, 6)))
+        bscratch = None
+        if bias_fused:
+            H0 = a1.shape[1]
+            bscratch = bufs.get("bscratch")
+            if bscratch is None or bscratch.numel() != 4 * H0 + 1:
+                bscratch = torch.zeros(4 * H0 + 1, dtype=torch.float32,
+                                       device=a1.device)
+                bufs["bscratch"] = bscratch
         dx0, dz1, dz2, dz3 = ext.mlp3_bwd(
             dout.contiguous(), a1, a2, a3, w4f,
-            bufs["w3tp"], bufs["w2tp"], bufs["w1tp"])
+            bufs["w3tp"], bufs["w2tp"], bufs["w1tp"], bscratch)
         # bias grads as GEMVs (fallback paths): torch's column-sum of
         # row-major bf16 ran ~16 us each (reduce_kernel); ones@dz is a
         # hipBLASLt GEMV. The default path below does them in the fused
@@ -114,8 +131,7 @@ class _FusedMLP3Fn(torch.autograd.Function):
         # wgrads ACCUMULATE in place via beta=1 addmm_ — one fused GEMM per
         # param instead of GEMM + autograd's separate accumulate-add (8 add
         # kernels/step in the profile) — and autograd gets None.
-        params = ctx._params
-        if all(p.grad is not None for p in params):
+        if prebound:
             w1g, b1g, w2g, b2g, w3g, b3g, w4g, b4g = (p.grad for p in params)
             M, H = dz1.shape
             K0p = x0.shape[1]
@@ -123,20 +139,11 @@ class _FusedMLP3Fn(torch.autograd.Function):
                            and w1g.is_contiguous() and w2g.is_contiguous()
                            and w3g.is_contiguous()
                            and M % 32 == 0 and K0p <= 512)
-            bias_fused = (b1g.dtype == torch.bfloat16
-                          and b1g.is_contiguous() and b2g.is_contiguous()
-                          and b3g.is_contiguous() and w4g.is_contiguous())
             # when BOTH fused paths run, the wgrad kernel carries the three
-            # dz column sums (bias grads) from its LDS-staged tiles and the
-            # bias pass skips its 3x dz re-read (~10 MB/step)
+            # dz column sums (bias grads) from its LDS-staged tiles; the
+            # head sums already rode the dgrad kernel above — the separate
+            # bias pass disappears entirely (only its finisher runs)
             bias_in_wgrad = wgrad_fused and bias_fused
-            bscratch = None
-            if bias_fused:
-                bscratch = bufs.get("bscratch")
-                if bscratch is None or bscratch.numel() != 4 * H + 1:
-                    bscratch = torch.zeros(4 * H + 1, dtype=torch.float32,
-                                           device=a1.device)
-                    bufs["bscratch"] = bscratch
             if wgrad_fused:
                 # K0p <= 512: at the dim9 shapes the fused kernel beats the
                 # hipBLASLt trio (12.29 -> 12.62M); at dim64 (K0p 1696)
@@ -160,15 +167,16 @@ class _FusedMLP3Fn(torch.autograd.Function):
                 w3g.addmm_(dz3.t(), a2)
             dpart = dout if ctx._has_partial else None
             if bias_fused:
-                # bias grads for all 4 layers + the head wgrad in one pass
-                # (+ a finisher that folds the fp32 scratch into the bf16
-                # grads and re-zeros it) — replaces 4 GEMV launches + a
-                # reduce + an add + a dout bf16 cast. with_dz=False when
-                # the fused wgrad above already summed the dz columns.
+                # finisher folds the fp32 scratch (dz sums from the wgrad
+                # kernel or here; head sums from the dgrad kernel) into the
+                # bf16 grads and re-zeros it. with_dz=False when the fused
+                # wgrad already summed the dz columns; with_head always
+                # False (the dgrad kernel carried dw4/db4); both False
+                # launches only the finisher.
                 ext.mlp3_bias_bwd(dout.contiguous(), dz1, dz2, dz3, a3,
                                   bscratch, b1g, b2g, b3g,
                                   w4g.reshape(-1), b4g,
-                                  not bias_in_wgrad)
+                                  not bias_in_wgrad, False)
             else:
                 d, ones = _d(), _ones()
                 w4g.addmm_(d, a3)

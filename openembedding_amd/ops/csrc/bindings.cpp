@@ -50,7 +50,7 @@ void emb_ctr_head_bwd(const float*, const float*, const float*, const void*,
                       hipStream_t_);
 void emb_mlp3_bias_bwd(const float*, const void*, const void*, const void*,
                        const void*, long, long, float*, void*, void*, void*,
-                       void*, void*, int, hipStream_t_);
+                       void*, void*, int, int, hipStream_t_);
 void emb_mlp3_wgrad(const void*, const void*, const void*, const void*,
                     const void*, const void*, long, long, long, long,
                     float*, void*, void*, void*, float*, hipStream_t_);
@@ -98,7 +98,7 @@ void emb_mlp3_fwd(const void*, long, long, const void*, const void*,
 void emb_mlp3_bwd(const float*, long, long, const void*, const void*,
                   const void*, const void*, const void*, const void*,
                   const void*, long, long, void*, void*, void*, void*,
-                  hipStream_t_);
+                  float*, hipStream_t_);
 }
 
 namespace {
@@ -548,7 +548,8 @@ void flat_opt(int64_t opt, torch::Tensor param, OptTensor master,
 void mlp3_bias_bwd(torch::Tensor dout, torch::Tensor dz1, torch::Tensor dz2,
                    torch::Tensor dz3, torch::Tensor a3, torch::Tensor scratch,
                    torch::Tensor db1, torch::Tensor db2, torch::Tensor db3,
-                   torch::Tensor dw4, torch::Tensor db4, bool with_dz) {
+                   torch::Tensor dw4, torch::Tensor db4, bool with_dz,
+                   bool with_head) {
     CHECK_GPU(dout); CHECK_CONT(dout); CHECK_CONT(dz1); CHECK_CONT(dz2);
     CHECK_CONT(dz3); CHECK_CONT(a3); CHECK_CONT(scratch);
     long M = dz1.size(0), H = dz1.size(1);
@@ -568,7 +569,7 @@ void mlp3_bias_bwd(torch::Tensor dout, torch::Tensor dz1, torch::Tensor dz2,
                       scratch.data_ptr<float>(),
                       db1.data_ptr(), db2.data_ptr(), db3.data_ptr(),
                       dw4.data_ptr(), db4.data_ptr(), with_dz ? 1 : 0,
-                      cur_stream());
+                      with_head ? 1 : 0, cur_stream());
 }
 
 // ---- CIN implicit-GEMM -------------------------------------------------
@@ -758,7 +759,9 @@ mlp3_fwd(torch::Tensor x0, torch::Tensor w1, torch::Tensor b1,
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
 mlp3_bwd(torch::Tensor dout, torch::Tensor a1, torch::Tensor a2,
          torch::Tensor a3, torch::Tensor w4, torch::Tensor w3t,
-         torch::Tensor w2t, torch::Tensor w1t) {
+         torch::Tensor w2t, torch::Tensor w1t, OptTensor bias_scratch) {
+    // bias_scratch ([4H+1] fp32): when given, the dz3 assembly loop also
+    // accumulates dw4 (dout.a3 column sums) and db4 into segments 3H/4H
     CHECK_GPU(dout); CHECK_CONT(dout); CHECK_CONT(w3t); CHECK_CONT(w2t);
     CHECK_CONT(w1t);
     const c10::cuda::CUDAGuard guard(dout.device());
@@ -769,10 +772,18 @@ mlp3_bwd(torch::Tensor dout, torch::Tensor a1, torch::Tensor a2,
     auto dz2 = torch::empty_like(a2);
     auto dz3 = torch::empty_like(a3);
     auto dx0 = torch::empty({M, K0p}, a1.options());
+    float* bptr = nullptr;
+    if (bias_scratch.has_value()) {
+        CHECK_CONT(*bias_scratch);
+        TORCH_CHECK(bias_scratch->numel() >= 4 * H + 1
+                    && bias_scratch->dtype() == torch::kFloat32,
+                    "bias_scratch layout");
+        bptr = bias_scratch->data_ptr<float>();
+    }
     emb_mlp3_bwd(dout.data_ptr<float>(), M, K0p, a1.data_ptr(),
                  a2.data_ptr(), a3.data_ptr(), w4.data_ptr(), w3t.data_ptr(),
                  w2t.data_ptr(), w1t.data_ptr(), H, Hp, dz1.data_ptr(),
-                 dz2.data_ptr(), dz3.data_ptr(), dx0.data_ptr(),
+                 dz2.data_ptr(), dz3.data_ptr(), dx0.data_ptr(), bptr,
                  cur_stream());
     return {dx0, dz1, dz2, dz3};
 }
@@ -819,7 +830,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "+ dense linear");
     m.def("ctr_head_bwd", &ctr_head_bwd, "fused CTR head backward");
     m.def("mlp3_bwd", &mlp3_bwd,
-          "fused dgrad chain backward of the 3-layer MLP");
+          "fused dgrad chain backward of the 3-layer MLP (optionally "
+          "carrying the head wgrad/bias sums)",
+          pybind11::arg("dout"), pybind11::arg("a1"), pybind11::arg("a2"),
+          pybind11::arg("a3"), pybind11::arg("w4"), pybind11::arg("w3t"),
+          pybind11::arg("w2t"), pybind11::arg("w1t"),
+          pybind11::arg("bias_scratch") = pybind11::none());
     m.def("mlp3_pack", &mlp3_pack,
           "refresh three padded (or transposed) weight copies in one "
           "launch");
@@ -840,7 +856,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           pybind11::arg("dz3"), pybind11::arg("a3"),
           pybind11::arg("scratch"), pybind11::arg("db1"),
           pybind11::arg("db2"), pybind11::arg("db3"), pybind11::arg("dw4"),
-          pybind11::arg("db4"), pybind11::arg("with_dz") = true);
+          pybind11::arg("db4"), pybind11::arg("with_dz") = true,
+          pybind11::arg("with_head") = true);
     m.def("cin_fwd", &cin_fwd,
           "CIN layer forward: implicit outer-product MFMA GEMM");
     m.def("cin_dw", &cin_dw,

@@ -245,7 +245,8 @@ void k_mlp3_bwd(const float* __restrict__ dout, long M, long K0p,
                 const mbf16* __restrict__ w1t,   // [K0p, Hp]
                 long H, long Hp,
                 mbf16* __restrict__ dz1, mbf16* __restrict__ dz2,
-                mbf16* __restrict__ dz3, mbf16* __restrict__ dx0) {
+                mbf16* __restrict__ dz3, mbf16* __restrict__ dx0,
+                float* __restrict__ bias_out) {  // optional [4H+1] scratch
     __shared__ mbf16 dz[2][MLP_BM * MLP_LD];
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
@@ -258,16 +259,32 @@ void k_mlp3_bwd(const float* __restrict__ dout, long M, long K0p,
     // dz3 = dout ⊗ w4 ⊙ relu'(a3), elementwise; zero the LDS pad columns
     // once (the B-side pads are zero too, but dz tiles are the A side of
     // the NEXT stage whose pads multiply W pads — either side zero is
-    // enough; zeroing here keeps the invariant simple)
-    for (long i = threadIdx.x; i < MLP_BM * MLP_LD; i += blockDim.x) {
-        long row = i / MLP_LD, n = i % MLP_LD;
-        long gm = m0 + row;
-        float v = 0.f;
-        if (n < H && gm < M && (float)a3[gm * H + n] > 0.f)
-            v = dout[gm] * (float)w4[n];
-        mbf16 hv = (mbf16)v;
-        dz[0][i] = hv;
-        if (n < H && gm < M) dz3[gm * H + n] = hv;
+    // enough; zeroing here keeps the invariant simple).
+    // Column-per-thread walk (same writes, same per-row coalescing as the
+    // linear-index version): when bias_out is given the head wgrad
+    // dw4[n] = sum_m dout[m]*a3[m,n] and db4 = sum_m dout[m] ride along —
+    // this loop already touches exactly those operands, so the separate
+    // head pass over a3 (3 MB/step) disappears.
+    for (long n = threadIdx.x; n < MLP_LD; n += blockDim.x) {
+        float sw = 0.f, s4 = 0.f;
+        for (int row = 0; row < MLP_BM; ++row) {
+            long gm = m0 + row;
+            float v = 0.f;
+            if (n < H && gm < M) {
+                float av = (float)a3[gm * H + n];
+                float dv = dout[gm];
+                if (av > 0.f) v = dv * (float)w4[n];
+                sw += dv * av;
+                if (n == 0) s4 += dv;
+            }
+            mbf16 hv = (mbf16)v;
+            dz[0][row * MLP_LD + n] = hv;
+            if (n < H && gm < M) dz3[gm * H + n] = hv;
+        }
+        if (bias_out && n < H) {
+            atomicAdd(&bias_out[3 * H + n], sw);
+            if (n == 0 && s4 != 0.f) atomicAdd(&bias_out[4 * H], s4);
+        }
     }
     __syncthreads();
     // NOTE: dz tiles' pad columns [H, Hp) may hold garbage after a GEMM
@@ -295,15 +312,16 @@ extern "C" __global__ void k_mlp3_bias_bwd(
         const mbf16* __restrict__ dz1, const mbf16* __restrict__ dz2,
         const mbf16* __restrict__ dz3, const mbf16* __restrict__ a3,
         long M, long H, long rows_per_blk,
-        float* __restrict__ scratch, int with_dz) {
+        float* __restrict__ scratch, int with_dz, int with_head) {
     const long r0 = (long)blockIdx.x * rows_per_blk;
     if (r0 >= M) return;
     const long r1 = min(M, r0 + rows_per_blk);
     // column sums: thread t covers columns t, t+blockDim.x, ... of each dz
     // (+ the head wgrad dw4[c] = sum_r dout[r]*a3[r,c] — same access shape).
     // with_dz == 0: the fused wgrad kernel already accumulated the three
-    // dz column sums from its LDS-staged tiles; only the head pass
-    // (dout/a3, a quarter of the traffic) remains here.
+    // dz column sums from its LDS-staged tiles; with_head == 0: the dgrad
+    // kernel carried dw4/db4 (it reads dout/a3 anyway). Both 0: the
+    // launcher never starts this kernel, only the finisher.
     for (long c = threadIdx.x; c < H; c += blockDim.x) {
         float sw = 0.f;
         if (with_dz) {
@@ -312,24 +330,27 @@ extern "C" __global__ void k_mlp3_bias_bwd(
                 s1 += (float)dz1[r * H + c];
                 s2 += (float)dz2[r * H + c];
                 s3 += (float)dz3[r * H + c];
-                sw += dout[r] * (float)a3[r * H + c];
+                if (with_head) sw += dout[r] * (float)a3[r * H + c];
             }
             atomicAdd(scratch + c, s1);
             atomicAdd(scratch + H + c, s2);
             atomicAdd(scratch + 2 * H + c, s3);
-        } else {
+        } else if (with_head) {
             for (long r = r0; r < r1; ++r)
                 sw += dout[r] * (float)a3[r * H + c];
         }
-        atomicAdd(scratch + 3 * H + c, sw);
+        if (with_head) atomicAdd(scratch + 3 * H + c, sw);
     }
-    // head bias: sum of dout rows, one atomic per wave
-    float s4 = 0.f;
-    for (long r = r0 + threadIdx.x; r < r1; r += blockDim.x)
-        s4 += dout[r];
-    #pragma unroll
-    for (int off = 32; off; off >>= 1) s4 += __shfl_down(s4, off, 64);
-    if ((threadIdx.x & 63) == 0 && s4 != 0.f) atomicAdd(scratch + 4 * H, s4);
+    if (with_head) {
+        // head bias: sum of dout rows, one atomic per wave
+        float s4 = 0.f;
+        for (long r = r0 + threadIdx.x; r < r1; r += blockDim.x)
+            s4 += dout[r];
+        #pragma unroll
+        for (int off = 32; off; off >>= 1) s4 += __shfl_down(s4, off, 64);
+        if ((threadIdx.x & 63) == 0 && s4 != 0.f)
+            atomicAdd(scratch + 4 * H, s4);
+    }
 }
 
 extern "C" __global__ void k_mlp3_bias_finish(
@@ -355,13 +376,15 @@ extern "C" void emb_mlp3_bias_bwd(const float* dout, const void* dz1,
                                   long M, long H, float* scratch,
                                   void* db1, void* db2, void* db3,
                                   void* dw4, void* db4, int with_dz,
-                                  hipStream_t stream) {
+                                  int with_head, hipStream_t stream) {
     if (M == 0) return;
     const long rows_per_blk = 32;
     int ga = (int)((M + rows_per_blk - 1) / rows_per_blk);
-    k_mlp3_bias_bwd<<<ga, 256, 0, stream>>>(
-        dout, (const mbf16*)dz1, (const mbf16*)dz2, (const mbf16*)dz3,
-        (const mbf16*)a3, M, H, rows_per_blk, scratch, with_dz);
+    if (with_dz || with_head)
+        k_mlp3_bias_bwd<<<ga, 256, 0, stream>>>(
+            dout, (const mbf16*)dz1, (const mbf16*)dz2, (const mbf16*)dz3,
+            (const mbf16*)a3, M, H, rows_per_blk, scratch, with_dz,
+            with_head);
     int gb = (int)((4 * H + 1 + 255) / 256);
     k_mlp3_bias_finish<<<gb, 256, 0, stream>>>(
         scratch, H, (mbf16*)db1, (mbf16*)db2, (mbf16*)db3, (mbf16*)dw4,
@@ -392,14 +415,14 @@ extern "C" void emb_mlp3_bwd(const float* dout, long M, long K0p,
                              const void* w2t, const void* w1t,
                              long H, long Hp,
                              void* dz1, void* dz2, void* dz3, void* dx0,
-                             hipStream_t stream) {
+                             float* bias_out, hipStream_t stream) {
     if (M == 0) return;
     long grid = (M + MLP_BM - 1) / MLP_BM;
     k_mlp3_bwd<<<(int)grid, 64 * MLP_WAVES, 0, stream>>>(
         dout, M, K0p, (const mbf16*)a1, (const mbf16*)a2, (const mbf16*)a3,
         (const mbf16*)w4, (const mbf16*)w3t, (const mbf16*)w2t,
         (const mbf16*)w1t, H, Hp, (mbf16*)dz1, (mbf16*)dz2, (mbf16*)dz3,
-        (mbf16*)dx0);
+        (mbf16*)dx0, bias_out);
 }
 
 // ---------------------------------------------------------- fused wgrads
