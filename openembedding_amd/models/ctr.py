@@ -351,17 +351,16 @@ class _CINLayerFn(torch.autograd.Function):
     def _vmat(x0p, xkp, B, F, H, d):
         # V [d, B, F*H] = per-slice outer products, built from the
         # d-leading contiguous layouts (one broadcast mul, coalesced)
-        return (x0p.view(d, B, F).unsqueeze(3)
-                * xkp.view(d, B, H).unsqueeze(2)).reshape(d, B, F * H)
+        return (x0p.unsqueeze(3) * xkp.unsqueeze(2)).reshape(d, B, F * H)
 
     @staticmethod
-    def _hip_ok(x0p, W, cdt):
-        if not (x0p.is_cuda and cdt == torch.bfloat16 and hip_available()):
+    def _hip_ok(x0, W, cdt):
+        if not (x0.is_cuda and cdt == torch.bfloat16 and hip_available()):
             return False
         O, K = W.shape
-        F = x0p.shape[1]
+        F = x0.shape[1]
         H = K // F
-        n = x0p.shape[0]                 # columns; dw operands pad to 32
+        n = x0.shape[0] * x0.shape[2]    # columns; dw operands pad to 32
         return (F <= 32 and H <= 128 and 16 <= O <= 128 and O % 16 == 0
                 and n % 32 == 0)
 
@@ -386,16 +385,10 @@ class _CINLayerFn(torch.autograd.Function):
         return bufs
 
     @staticmethod
-    def forward(ctx, x0p, xkp, W, cdt, B):
-        # d-leading 2D interface: x0p [d*B, F], xkp [d*B, H] — CIN keeps
-        # this layout across layers, so the per-layer permute round-trips
-        # ([d,B,*] -> [B,*,d] -> [d,B,*], ~10 us of transform launches
-        # EACH at the benchmark shape) never happen. cin_layer() below
-        # wraps the [B, F, d] interface.
-        N, F = x0p.shape
-        H = xkp.shape[1]
-        d = N // B
-        hip = _CINLayerFn._hip_ok(x0p, W, cdt)
+    def forward(ctx, x0, xk, W, cdt):
+        B, F, d = x0.shape
+        H = xk.shape[1]
+        hip = _CINLayerFn._hip_ok(x0, W, cdt)
         if hip:
             # implicit-GEMM HIP kernels (ops/csrc/cin.hip): V is built in
             # LDS per block and never touches HBM (the torch path below
@@ -403,63 +396,53 @@ class _CINLayerFn(torch.autograd.Function):
             # + re-reads as ~70% of the xDeepFM step)
             from ..ops import require_hip
             ext = require_hip()
-            x0c = x0p.contiguous()
-            xkc = xkp.contiguous()
+            x0p = x0.permute(2, 0, 1).reshape(d * B, F).contiguous()
+            xkp = xk.permute(2, 0, 1).reshape(d * B, H).contiguous()
             bufs = _CINLayerFn._bufs(W, F, H)
-            out = ext.cin_fwd(x0c, xkc, bufs["wp"])       # [d*B, O]
-            ctx.save_for_backward(x0c, xkc, W)
+            out = ext.cin_fwd(x0p, xkp, bufs["wp"])       # [d*B, O]
+            ctx.save_for_backward(x0p, xkp, W)
             ctx.meta = (B, F, H, d, True)
-            return out
-        x0c = x0p.to(cdt).contiguous()
-        xkc = xkp.to(cdt).contiguous()
+            return out.view(d, B, -1).permute(1, 2, 0).float()
+        x0p = x0.permute(2, 0, 1).to(cdt).contiguous()   # [d, B, F]
+        xkp = xk.permute(2, 0, 1).to(cdt).contiguous()   # [d, B, H]
         Wt = W.to(cdt).t().contiguous()                  # [F*H, O]
-        v = _CINLayerFn._vmat(x0c, xkc, B, F, H, d)      # [d, B, F*H]
+        v = _CINLayerFn._vmat(x0p, xkp, B, F, H, d)      # [d, B, F*H]
         out = torch.matmul(v, Wt)                        # [d, B, O]
-        ctx.save_for_backward(x0c, xkc, W)
+        ctx.save_for_backward(x0p, xkp, W)
         ctx.meta = (B, F, H, d, False)
         ctx.cdt = cdt
-        return out.reshape(N, -1).float()                # [d*B, O]
+        return out.permute(1, 2, 0).float()              # [B, O, d]
 
     @staticmethod
-    def backward(ctx, dout):                             # dout [d*B, O]
-        x0c, xkc, W = ctx.saved_tensors
+    def backward(ctx, dout):
+        x0p, xkp, W = ctx.saved_tensors
         B, F, H, d, hip = ctx.meta
         if hip:
             from ..ops import require_hip
             ext = require_hip()
+            N = d * B
             bufs = W._cin_bufs
-            doutp = dout.contiguous()
-            dx0p, dxkp = ext.cin_dx(doutp, bufs["wt"], x0c, xkc)
+            doutp = dout.permute(2, 0, 1).reshape(N, -1).contiguous()
+            dx0p, dxkp = ext.cin_dx(doutp, bufs["wt"], x0p, xkp)
             # transposed bf16 operands for the weight-grad GEMM
             dzt = doutp.t().contiguous().to(torch.bfloat16)
-            x0t = x0c.t().contiguous().to(torch.bfloat16)
-            xkt = xkc.t().contiguous().to(torch.bfloat16)
+            x0t = x0p.t().contiguous().to(torch.bfloat16)
+            xkt = xkp.t().contiguous().to(torch.bfloat16)
             dW = ext.cin_dw(dzt, x0t, xkt, W.shape[0], 8)
-            return dx0p, dxkp, dW.to(W.dtype), None, None
+            return (dx0p.view(d, B, F).permute(1, 2, 0),
+                    dxkp.view(d, B, H).permute(1, 2, 0),
+                    dW.to(W.dtype), None)
         cdt = ctx.cdt
         Wc = W.to(cdt)                                    # [O, F*H]
-        g = dout.to(cdt).view(d, B, -1)                   # [d, B, O]
-        v = _CINLayerFn._vmat(x0c, xkc, B, F, H, d)       # rebuilt, not saved
+        g = dout.permute(2, 0, 1).to(cdt).contiguous()    # [d, B, O]
+        v = _CINLayerFn._vmat(x0p, xkp, B, F, H, d)       # rebuilt, not saved
         # dW = sum_d g_d^T @ v_d  (d-batched GEMM, fp32 accumulate)
         dW = torch.matmul(g.transpose(1, 2), v).float().sum(0)
         p = torch.matmul(g, Wc).view(d, B, F, H)          # [d, B, F, H]
-        dx0 = (p * xkc.view(d, B, 1, H)).sum(3)           # [d, B, F]
-        dxk = (p * x0c.view(d, B, F, 1)).sum(2)           # [d, B, H]
-        return (dx0.reshape(d * B, F).float(),
-                dxk.reshape(d * B, H).float(),
-                dW.to(W.dtype), None, None)
-
-
-def cin_layer(x0: torch.Tensor, xk: torch.Tensor, W: torch.Tensor,
-              cdt) -> torch.Tensor:
-    """[B, F, d] interface over _CINLayerFn ([B,F,d]x[B,H,d] -> [B,O,d]);
-    the CIN module itself stays in the d-leading 2D layout across layers."""
-    B, F, d = x0.shape
-    H = xk.shape[1]
-    x0p = x0.permute(2, 0, 1).reshape(d * B, F)
-    xkp = xk.permute(2, 0, 1).reshape(d * B, H)
-    z = _CINLayerFn.apply(x0p.contiguous(), xkp.contiguous(), W, cdt, B)
-    return z.view(d, B, -1).permute(1, 2, 0)
+        dx0 = (p * xkp.unsqueeze(2)).sum(3)               # [d, B, F]
+        dxk = (p * x0p.unsqueeze(3)).sum(2)               # [d, B, H]
+        return (dx0.permute(1, 2, 0).float(),
+                dxk.permute(1, 2, 0).float(), dW.to(W.dtype), None)
 
 
 class CIN(nn.Module):
@@ -486,17 +469,13 @@ class CIN(nn.Module):
         self.compute_dtype = torch.float32
 
     def forward(self, e: torch.Tensor) -> torch.Tensor:  # e: [B, F, d]
-        B, F, d = e.shape
-        # ONE permute to the d-leading layout; every layer stays in it
-        # (relu and the per-layer sum-over-d work directly on [d*B, Hk]),
-        # so no per-layer [B,*,d] round-trips
-        ep = e.permute(2, 0, 1).contiguous().view(d * B, F)
-        xkp = ep
+        x0 = e
+        xk = e
         outs = []
         for w in self.weights:
-            z = _CINLayerFn.apply(ep, xkp, w, self.compute_dtype, B)
-            xkp = torch.relu(z)                          # [d*B, Hk]
-            outs.append(xkp.view(d, B, -1).sum(0))       # [B, Hk]
+            z = _CINLayerFn.apply(x0, xk, w, self.compute_dtype)
+            xk = torch.relu(z)
+            outs.append(xk.sum(dim=2))                   # [B, Hk]
         return self.fc(torch.cat(outs, dim=1)).squeeze(-1)
 
 

@@ -1,13 +1,13 @@
 """CIN implicit-outer-product layer vs the einsum oracle (CPU).
 
 The reference's CIN (DeepCTR, used by the xDeepFM benchmark row) computes
-xk+1 = relu(conv1d(outer(x0, xk))); cin_layer computes the same
+xk+1 = relu(conv1d(outer(x0, xk))); _CINLayerFn computes the same
 multilinear map without materializing the outer-product tensor. Checked
 here for forward values and all three gradients."""
 
 import torch
 
-from openembedding_amd.models.ctr import CIN, cin_layer
+from openembedding_amd.models.ctr import CIN, _CINLayerFn
 
 
 def _oracle(x0, xk, W):
@@ -24,7 +24,7 @@ def test_cin_layer_matches_einsum():
     xk = torch.randn(B, H, d, generator=g, requires_grad=True)
     W = torch.randn(O, F * H, generator=g, requires_grad=True)
 
-    out = cin_layer(x0, xk, W, torch.float32)
+    out = _CINLayerFn.apply(x0, xk, W, torch.float32)
     ref = _oracle(x0, xk, W)
     torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
 
@@ -60,7 +60,7 @@ def test_cin_x0_equals_xk_first_layer():
     e1 = torch.randn(B, F, d, generator=g, requires_grad=True)
     e2 = e1.detach().clone().requires_grad_(True)
     W = torch.randn(O, F * F, generator=g)
-    out1 = cin_layer(e1, e1, W, torch.float32)
+    out1 = _CINLayerFn.apply(e1, e1, W, torch.float32)
     z = torch.einsum("bfd,bhd->bfhd", e2, e2).reshape(B, F * F, d)
     out2 = torch.einsum("ok,bkd->bod", W, z)
     dout = torch.randn(B, O, d, generator=g)

@@ -22,7 +22,7 @@ def _layer_io(B=512, F=26, H=128, O=128, d=9, seed=0):
 
 
 def _run(x0, xk, W, dout, force_torch):
-    from openembedding_amd.models.ctr import _CINLayerFn, cin_layer
+    from openembedding_amd.models.ctr import _CINLayerFn
     x0 = x0.clone().requires_grad_(True)
     xk = xk.clone().requires_grad_(True)
     W = W.clone().requires_grad_(True)
@@ -30,7 +30,7 @@ def _run(x0, xk, W, dout, force_torch):
         orig = _CINLayerFn._hip_ok
         _CINLayerFn._hip_ok = staticmethod(lambda *a: False)
     try:
-        out = cin_layer(x0, xk, W, torch.bfloat16)
+        out = _CINLayerFn.apply(x0, xk, W, torch.bfloat16)
         out.backward(dout)
     finally:
         if force_torch:
