@@ -259,29 +259,33 @@ void k_mlp3_bwd(const float* __restrict__ dout, long M, long K0p,
     // dz3 = dout ⊗ w4 ⊙ relu'(a3), elementwise; zero the LDS pad columns
     // once (the B-side pads are zero too, but dz tiles are the A side of
     // the NEXT stage whose pads multiply W pads — either side zero is
-    // enough; zeroing here keeps the invariant simple).
-    // Column-per-thread walk (same writes, same per-row coalescing as the
-    // linear-index version): when bias_out is given the head wgrad
-    // dw4[n] = sum_m dout[m]*a3[m,n] and db4 = sum_m dout[m] ride along —
-    // this loop already touches exactly those operands, so the separate
-    // head pass over a3 (3 MB/step) disappears.
-    for (long n = threadIdx.x; n < MLP_LD; n += blockDim.x) {
-        float sw = 0.f, s4 = 0.f;
-        for (int row = 0; row < MLP_BM; ++row) {
-            long gm = m0 + row;
-            float v = 0.f;
-            if (n < H && gm < M) {
-                float av = (float)a3[gm * H + n];
-                float dv = dout[gm];
-                if (av > 0.f) v = dv * (float)w4[n];
-                sw += dv * av;
-                if (n == 0) s4 += dv;
+    // enough; zeroing here keeps the invariant simple)
+    for (long i = threadIdx.x; i < MLP_BM * MLP_LD; i += blockDim.x) {
+        long row = i / MLP_LD, n = i % MLP_LD;
+        long gm = m0 + row;
+        float v = 0.f;
+        if (n < H && gm < M && (float)a3[gm * H + n] > 0.f)
+            v = dout[gm] * (float)w4[n];
+        mbf16 hv = (mbf16)v;
+        dz[0][i] = hv;
+        if (n < H && gm < M) dz3[gm * H + n] = hv;
+    }
+    if (bias_out) {
+        // head wgrad dw4[n] = sum_m dout[m]*a3[m,n] and db4 ride along as
+        // a SECOND walk over the a3 tile the loop above just pulled into
+        // L1/L2 (16 rows x H x 2B = 13 KB — hot): the separate head pass
+        // over a3 (3 MB/step HBM) disappears, while the main loop keeps
+        // its linear-index ILP (a fused column-walk version cost ~4 us)
+        for (long n = threadIdx.x; n < H; n += blockDim.x) {
+            float sw = 0.f, s4 = 0.f;
+            for (int row = 0; row < MLP_BM; ++row) {
+                long gm = m0 + row;
+                if (gm < M) {
+                    float dv = dout[gm];
+                    sw += dv * (float)a3[gm * H + n];
+                    if (n == 0) s4 += dv;
+                }
             }
-            mbf16 hv = (mbf16)v;
-            dz[0][row * MLP_LD + n] = hv;
-            if (n < H && gm < M) dz3[gm * H + n] = hv;
-        }
-        if (bias_out && n < H) {
             atomicAdd(&bias_out[3 * H + n], sw);
             if (n == 0 && s4 != 0.f) atomicAdd(&bias_out[4 * H], s4);
         }
