@@ -476,13 +476,8 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
     const long mend = (mbeg + msz < M) ? (mbeg + msz) : M;
     if (mbeg >= M) return;
 
-    // DOUBLE-BUFFERED LDS tiles (2 x 34.8 KB = 69.6 KB, still 2
-    // blocks/CU): chunk i+1's global loads are issued before chunk i's
-    // MFMAs, so the L2/HBM latency rides under compute. The single-buffer
-    // version measured wait/busy = 31 on SQ_WAIT_ANY — waves parked on
-    // the stage -> barrier -> MFMA -> barrier serialization.
-    __shared__ mbf16 ldsA[2][64 * (WG_KC + 8)];   // dz^T tile  [i][m]
-    __shared__ mbf16 ldsB[2][64 * (WG_KC + 8)];   // a^T tile   [j][m]
+    __shared__ mbf16 ldsA[64 * (WG_KC + 8)];   // dz^T tile  [i][m]
+    __shared__ mbf16 ldsB[64 * (WG_KC + 8)];   // a^T tile   [j][m]
     const int wave = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
 
@@ -491,32 +486,25 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
     acc[0] = f32x4{0.f, 0.f, 0.f, 0.f};
     acc[1] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-#define WG_STAGE(BUF, M0)                                                   \
-    do {                                                                    \
-        const long mlim_ = (mend - (M0) < WG_KC) ? (mend - (M0)) : WG_KC;   \
-        for (int e = (int)threadIdx.x; e < 64 * WG_KC;                      \
-             e += (int)blockDim.x) {                                        \
-            int mm = e >> 6, cc = e & 63;                                   \
-            mbf16 va = (mbf16)0.0f, vb = (mbf16)0.0f;                       \
-            if (mm < mlim_) {                                               \
-                long m = (M0) + mm;                                         \
-                if (i0 + cc < H) va = dz[m * H + i0 + cc];                  \
-                if (j0 + cc < jdim) vb = a[m * jdim + j0 + cc];             \
-            }                                                               \
-            ldsA[BUF][cc * WG_LD + mm] = va;                                \
-            ldsB[BUF][cc * WG_LD + mm] = vb;                                \
-        }                                                                   \
-    } while (0)
-
-    WG_STAGE(0, mbeg);
-    int cur = 0;
     for (long m0 = mbeg; m0 < mend; m0 += WG_KC) {
-        // one barrier per chunk: makes buffer `cur`'s staging visible AND
-        // (lgkmcnt drain) guarantees last chunk's LDS reads of `cur^1`
-        // finished before we overwrite it below
         __syncthreads();
-        const long mn = m0 + WG_KC;
-        if (mn < mend) WG_STAGE(cur ^ 1, mn);   // loads fly under the MFMAs
+        // stage transposed: read rows m (64 consecutive columns
+        // coalesced), write LDS [col][m] — 512 threads x 8 elements
+        const long mlim = (mend - m0 < WG_KC) ? (mend - m0) : WG_KC;
+        for (int e = (int)threadIdx.x; e < 64 * WG_KC;
+             e += (int)blockDim.x) {
+            int mm = e >> 6, cc = e & 63;
+            mbf16 va = (mbf16)0.0f, vb = (mbf16)0.0f;
+            if (mm < mlim) {
+                long m = m0 + mm;
+                if (i0 + cc < H) va = dz[m * H + i0 + cc];
+                if (j0 + cc < jdim)
+                    vb = a[m * jdim + j0 + cc];
+            }
+            ldsA[cc * WG_LD + mm] = va;
+            ldsB[cc * WG_LD + mm] = vb;
+        }
+        __syncthreads();
         if (bias_out && tj == 0) {
             // bias grads ride along: the dz tile this block just staged IS
             // the operand k_mlp3_bias_bwd used to re-read from HBM — sum
@@ -527,7 +515,7 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
             int cc = (int)threadIdx.x >> 3, sub = (int)threadIdx.x & 7;
             float s = 0.f;
             for (int mm = sub; mm < WG_KC; mm += 8)
-                s += (float)ldsA[cur][cc * WG_LD + mm];
+                s += (float)ldsA[cc * WG_LD + mm];
             s += __shfl_down(s, 4);
             s += __shfl_down(s, 2);
             s += __shfl_down(s, 1);
@@ -539,8 +527,8 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
         for (int t = 0; t < 2; ++t) {
             int st = wave + t * 8;            // subtile id 0..15
             int si = (st >> 2) * 16, sj = (st & 3) * 16;
-            const mbf16* pa = ldsA[cur] + (si + (lane & 15)) * WG_LD + koff;
-            const mbf16* pb = ldsB[cur] + (sj + (lane & 15)) * WG_LD + koff;
+            const mbf16* pa = ldsA + (si + (lane & 15)) * WG_LD + koff;
+            const mbf16* pb = ldsB + (sj + (lane & 15)) * WG_LD + koff;
             for (int ks = 0; ks < WG_KC; ks += 32) {
                 bf16x8 av = ld_frag(pa + ks);
                 bf16x8 bv = ld_frag(pb + ks);
@@ -548,9 +536,7 @@ void k_mlp3_wgrad(const mbf16* __restrict__ dz1,   // [M, H]
                     av, bv, acc[t], 0, 0, 0);
             }
         }
-        cur ^= 1;
     }
-#undef WG_STAGE
     #pragma unroll
     for (int t = 0; t < 2; ++t) {
         int st = wave + t * 8;
