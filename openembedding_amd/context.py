@@ -73,7 +73,11 @@ class Context:
         self.world_size = int(os.environ.get("WORLD_SIZE", 1))
         self.local_rank = int(os.environ.get("LOCAL_RANK", self.rank))
         if self.world_size > 1 and not _dist_ready():
-            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            # OEAMD_BACKEND=gloo on a GPU box runs the full engine with the
+            # CPU-staged wire (debug / N-ranks-on-1-GPU rehearsal — RCCL
+            # refuses duplicate devices in one communicator)
+            backend = (os.environ.get("OEAMD_BACKEND")
+                       or ("nccl" if torch.cuda.is_available() else "gloo"))
             if backend == "nccl":
                 torch.cuda.set_device(self.local_rank)
             dist.init_process_group(backend=backend)
@@ -84,8 +88,9 @@ class Context:
             self.rank = dist.get_rank()
             self.world_size = dist.get_world_size()
         if device is None:
-            device = (f"cuda:{self.local_rank}" if torch.cuda.is_available()
-                      else "cpu")
+            device = os.environ.get("OEAMD_DEVICE") or (
+                f"cuda:{self.local_rank}" if torch.cuda.is_available()
+                else "cpu")
         if device.startswith("cuda"):
             torch.cuda.set_device(device)
         self.device = torch.device(device)
