@@ -547,7 +547,8 @@ class DistributedOptimizer:
     UpdateWeights op).
 
     ``flatten_dense=True`` re-bases the dense params onto one flat buffer
-    (single-fill zero_grad, single-tensor allreduce, 3-kernel Adagrad)."""
+    (no-op steady-state zero_grad, single-tensor allreduce, fused flat
+    optimizer kernels)."""
 
     def __init__(self, optimizer: torch.optim.Optimizer,
                  sparse_config: Optional[Dict] = None,
