@@ -159,3 +159,20 @@ def test_top_level_parity_surface():
     assert m.running and m.endpoint == "127.0.0.1:7777"
     s = oe.Server(master_endpoint=m.endpoint)
     assert s.join() is None
+
+
+def test_oeamd_device_env_override(monkeypatch):
+    """OEAMD_DEVICE forces the context device (pairs with OEAMD_BACKEND
+    for the N-ranks-on-1-GPU gloo rehearsal — see
+    profiles/final_validation_r2.md)."""
+    import openembedding_amd.context as cm
+    if cm._context is not None:
+        cm._context.finalize()
+        cm._context = None
+    import openembedding_amd.torch as api
+    api._tracked.clear()
+    monkeypatch.setenv("OEAMD_DEVICE", "cpu")
+    ctx = cm.Context()
+    assert ctx.device.type == "cpu"
+    ctx.finalize()
+    cm._context = None
